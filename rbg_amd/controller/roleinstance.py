@@ -1,0 +1,333 @@
+"""RoleInstance controller — the leaf reconciler that owns engine processes.
+
+The analog of reference pkg/reconciler/roleinstance/instance_reconciler.go:
+74-242 + sync/instance_scale.go:46-320: one RoleInstance = one atomic gang of
+engine processes (components x size).  Reconcile:
+
+  1. read worker status (process liveness + heartbeat + status files),
+  2. decide scale diff — with gang semantics: any Failed worker condemns the
+     WHOLE instance (all processes stopped, GPUs released, RCCL comms with
+     them) and recreates it after the exponential backoff of
+     utils/backoff.py (reference instance_scale.go:342-509),
+  3. gang-reserve GPUs (sticky to previous devices via GpuBindingStore),
+  4. spawn missing workers with the discovery/identity/device env injected
+     at create (the reference's 4 injection hooks, instance_scale.go:233-306),
+  5. publish status conditions (Ready / Restarting).
+"""
+from __future__ import annotations
+
+import logging
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from ..api import constants as C
+from ..api.types import (Condition, ComponentSpec, RoleInstance, WorkerStatus,
+                         set_condition, get_condition)
+from ..discovery import env_builder
+from ..runtime.process import ProcessRunner, WorkerHandle
+from ..scheduler.gang import GangAllocator, GangUnschedulable, GpuClaim
+from ..scheduler.placement import GpuBindingStore
+from ..scheduler.ports import PortAllocator
+from ..store.store import Store
+from ..utils.backoff import RestartRegistry
+
+log = logging.getLogger(__name__)
+
+
+def worker_name(instance: str, component: str, idx: int) -> str:
+    return f"{instance}-{component}-{idx}"
+
+
+@dataclass
+class InstanceRuntime:
+    """Live (non-store) state of one RoleInstance: its process handles."""
+    gang_id: str = ""
+    handles: Dict[str, WorkerHandle] = field(default_factory=dict)
+    gpu_by_worker: Dict[str, List[int]] = field(default_factory=dict)
+    master_port: int = 0
+
+
+class RoleInstanceController:
+    def __init__(self, store: Store, gang: GangAllocator,
+                 runner: ProcessRunner, ports: PortAllocator,
+                 bindings: GpuBindingStore,
+                 restart_registry: Optional[RestartRegistry] = None,
+                 gang_timeout: float = 30.0):
+        self.store = store
+        self.gang = gang
+        self.runner = runner
+        self.ports = ports
+        self.bindings = bindings
+        self.restarts = restart_registry or RestartRegistry()
+        self.gang_timeout = gang_timeout
+        self._runtimes: Dict[str, InstanceRuntime] = {}
+
+    # ------------------------------------------------------------------
+
+    def runtime_for(self, inst: RoleInstance) -> InstanceRuntime:
+        rt = self._runtimes.get(inst.metadata.uid)
+        if rt is None:
+            rt = InstanceRuntime(gang_id=f"inst-{inst.metadata.uid}")
+            self._runtimes[inst.metadata.uid] = rt
+        return rt
+
+    def reconcile(self, name: str, namespace: str = "default") -> float:
+        """Returns requeue-after seconds (0 = no requeue needed)."""
+        inst = self.store.try_get(C.KIND_ROLE_INSTANCE, name, namespace)
+        if inst is None:
+            self._cleanup_by_name(name)
+            return 0.0
+        if inst.metadata.deletion_timestamp is not None:
+            self.teardown(inst)
+            self.store.try_delete(C.KIND_ROLE_INSTANCE, name, namespace)
+            return 0.0
+        rt = self.runtime_for(inst)
+        requeue = self._sync(inst, rt)
+        self._update_status(inst, rt)
+        return requeue
+
+    # ------------------------------------------------------------------
+
+    def _desired_workers(self, inst: RoleInstance) -> List[tuple]:
+        out = []
+        for comp in inst.spec.components:
+            for j in range(comp.size):
+                out.append((comp, j, worker_name(inst.metadata.name, comp.name, j)))
+        return out
+
+    def _restart_ignored(self, comp: ComponentSpec) -> bool:
+        """RestartTriggerPolicy=Ignore exempts auxiliary components from
+        condemning the gang (reference annotation.go:150-176)."""
+        return comp.annotations.get(C.ANNO_RESTART_TRIGGER_POLICY) == "Ignore"
+
+    def _sync(self, inst: RoleInstance, rt: InstanceRuntime) -> float:
+        desired = self._desired_workers(inst)
+        tracker = self.restarts.for_key(inst.metadata.uid)
+
+        # 1. observe failures
+        failed_fatal = []
+        for comp, j, wname in desired:
+            h = rt.handles.get(wname)
+            if h is None:
+                continue
+            phase = h.phase()
+            if phase in ("Failed",) and not self._restart_ignored(comp):
+                failed_fatal.append(wname)
+            elif phase == "Ready":
+                rt.gpu_by_worker[wname] = h.gpu_ids
+        healthy = not failed_fatal and all(
+            rt.handles.get(w) is not None and rt.handles[w].phase() == "Ready"
+            for _, _, w in desired)
+        if healthy:
+            tracker.observe_healthy()
+
+        policy = inst.spec.restart_policy
+        if failed_fatal and policy == C.RESTART_POLICY_RECREATE_INSTANCE:
+            # gang failover: tear down EVERYTHING, backoff, recreate
+            now = time.time()
+            if not tracker.may_restart(now):
+                self._set_restarting(inst, True,
+                                     f"backoff until {tracker.next_allowed_at():.0f}")
+                return max(0.05, tracker.next_allowed_at() - now)
+            log.warning("instance %s: workers %s failed; gang recreate",
+                        inst.metadata.name, failed_fatal)
+            self._record_bindings(inst, rt)
+            self._stop_all(inst, rt)
+            tracker.record_restart(now)
+            inst.status.restart_count = tracker.restart_count
+            inst.status.last_restart_time = now
+            self._set_restarting(inst, True, f"restart #{tracker.restart_count}")
+        elif failed_fatal and policy == C.RESTART_POLICY_NONE:
+            return 1.0   # leave failed workers visible in status
+
+        # 2. spawn anything missing (all-or-nothing GPU reservation first)
+        missing = [(c, j, w) for c, j, w in desired if w not in rt.handles]
+        if missing:
+            try:
+                self._ensure_gang(inst, rt, desired)
+            except GangUnschedulable as e:
+                self._set_condition(inst, C.COND_READY, False, "Unschedulable", str(e))
+                return 1.0
+            for comp, j, wname in missing:
+                self._spawn_worker(inst, rt, comp, j, wname)
+        if not healthy:
+            return 0.25
+        self._set_restarting(inst, False, "")
+        return 1.0   # periodic health poll
+
+    # ------------------------------------------------------------------
+
+    def _ensure_gang(self, inst: RoleInstance, rt: InstanceRuntime,
+                     desired: List[tuple]) -> None:
+        if self.gang.holding(rt.gang_id) is not None:
+            return
+        rbg_uid = self._owner_uid(inst)
+        claims = []
+        for comp, j, wname in desired:
+            gpus = 0
+            hbm = 0
+            if comp.template and comp.template.main_engine():
+                res = comp.template.main_engine().resources
+                if not res.cpu_only:
+                    gpus = res.gpus
+                    hbm = res.hbm_bytes
+            prefer = self.bindings.lookup(rbg_uid, GpuBindingStore.key(
+                inst.metadata.name, f"{comp.name}-{j}"))
+            claims.append(GpuClaim(gpus=gpus, hbm_bytes=hbm, prefer=prefer)
+                          if gpus or hbm else GpuClaim(gpus=0))
+        if not any(cl.gpus or cl.hbm_bytes for cl in claims):
+            self.gang.reserve(rt.gang_id, [])
+            rt.gpu_by_worker = {w: [] for _, _, w in desired}
+            return
+        res = self.gang.reserve(rt.gang_id, claims, timeout=self.gang_timeout)
+        for (comp, j, wname), gpus in zip(desired, res.assignments):
+            rt.gpu_by_worker[wname] = gpus
+
+    def _spawn_worker(self, inst: RoleInstance, rt: InstanceRuntime,
+                      comp: ComponentSpec, j: int, wname: str) -> None:
+        tmpl = comp.template
+        engine = tmpl.main_engine() if tmpl else None
+        if engine is None:
+            return
+        labels = inst.metadata.labels
+        env: Dict[str, str] = {}
+        env.update(env_builder.identity_env(
+            group_name=labels.get(C.LABEL_GROUP_NAME, ""),
+            role_name=labels.get(C.LABEL_ROLE_NAME, ""),
+            role_index=int(labels.get(C.LABEL_ROLE_INDEX, "0") or 0),
+            instance_name=inst.metadata.name,
+            component_name=comp.name, component_index=j,
+            config_path=inst.metadata.annotations.get("rbg.config-path", "")))
+        # leader-worker env: rank topology for the RCCL group
+        lwp_size = int(inst.metadata.annotations.get("rbg.lwp-size", "0") or 0)
+        if lwp_size > 1:
+            if not rt.master_port:
+                rt.master_port = self.ports.allocate(
+                    f"lwp-{inst.metadata.uid}", 1)[0]
+            rank = 0 if comp.name == "leader" else 1 + j
+            env.update(env_builder.leader_worker_env(
+                f"127.0.0.1:{rt.master_port}", rank, lwp_size))
+            env.update(env_builder.device_env(
+                rt.gpu_by_worker.get(wname, []), master_port=rt.master_port))
+        else:
+            env.update(env_builder.device_env(rt.gpu_by_worker.get(wname, [])))
+        # component-scoped ports from the annotation contract
+        from ..scheduler import ports as port_mod
+        try:
+            reqs = port_mod.parse_requests(comp.annotations)
+        except ValueError:
+            reqs = []
+        allocated: Dict[str, List[int]] = {}
+        for req in reqs:
+            scope_key = (f"role-{labels.get(C.LABEL_GROUP_NAME)}-"
+                         f"{labels.get(C.LABEL_ROLE_NAME)}-{req.name}"
+                         if req.scope == port_mod.SCOPE_ROLE
+                         else f"pod-{wname}-{req.name}")
+            allocated[req.name] = self.ports.allocate(scope_key, req.count)
+        env.update(self.ports.env_for(allocated))
+        merged = env_builder.merge_env(engine, env)
+        rt.handles[wname] = self.runner.spawn(
+            name=wname, runner=engine.runner, args=engine.args,
+            env=env_builder.env_as_dict(merged),
+            gpu_ids=rt.gpu_by_worker.get(wname, []),
+            command=engine.command or None)
+
+    # ------------------------------------------------------------------
+
+    def _record_bindings(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
+        rbg_uid = self._owner_uid(inst)
+        for wname, gpus in rt.gpu_by_worker.items():
+            if gpus:
+                # binding key keeps component-level granularity
+                suffix = wname[len(inst.metadata.name) + 1:]
+                self.bindings.record(rbg_uid, GpuBindingStore.key(
+                    inst.metadata.name, suffix), gpus)
+
+    def _stop_all(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
+        for h in rt.handles.values():
+            self.runner.stop(h)
+        rt.handles.clear()
+        self.gang.release(rt.gang_id)
+
+    def teardown(self, inst: RoleInstance) -> None:
+        rt = self._runtimes.pop(inst.metadata.uid, None)
+        if rt is None:
+            return
+        self._record_bindings(inst, rt)
+        for h in rt.handles.values():
+            self.runner.stop(h)
+        self.gang.release(rt.gang_id)
+        self.ports.release(f"lwp-{inst.metadata.uid}")
+        self.restarts.evict(inst.metadata.uid)
+
+    def _cleanup_by_name(self, name: str) -> None:
+        for uid, rt in list(self._runtimes.items()):
+            if any(w.startswith(name + "-") or w == name for w in rt.handles) \
+                    or rt.gang_id == f"inst-{uid}" and not rt.handles:
+                continue
+        # handled via teardown on delete events; nothing else to do here
+
+    @staticmethod
+    def _owner_uid(inst: RoleInstance) -> str:
+        for ref in inst.metadata.owner_references:
+            if ref.kind == C.KIND_RBG:
+                return ref.uid
+        # fall back to the RIS owner chain's recorded label
+        return inst.metadata.labels.get(C.LABEL_GROUP_NAME, "")
+
+    # ------------------------------------------------------------------
+
+    def _set_condition(self, inst: RoleInstance, type_: str, status: bool,
+                       reason: str, message: str) -> None:
+        def mutate(cur: RoleInstance):
+            set_condition(cur.status.conditions,
+                          Condition.new(type_, status, reason, message))
+            cur.status.restart_count = inst.status.restart_count
+            cur.status.last_restart_time = inst.status.last_restart_time
+            return cur
+        try:
+            self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mutate,
+                             inst.metadata.namespace, subresource="status")
+        except KeyError:
+            pass
+
+    def _set_restarting(self, inst: RoleInstance, on: bool, msg: str) -> None:
+        self._set_condition(inst, C.COND_RESTARTING, on,
+                            "GangRecreate" if on else "Stable", msg)
+
+    def _update_status(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
+        desired = self._desired_workers(inst)
+        workers: List[WorkerStatus] = []
+        ready = 0
+        for comp, j, wname in desired:
+            h = rt.handles.get(wname)
+            phase = h.phase() if h else "Pending"
+            if phase == "Ready":
+                ready += 1
+            workers.append(WorkerStatus(
+                name=wname, component=comp.name, component_index=j,
+                pid=h.pid if h else 0,
+                gpu_ids=rt.gpu_by_worker.get(wname, []),
+                phase=phase,
+                last_heartbeat=time.time() - h.heartbeat_age() if h else 0.0))
+        all_ready = ready == len(desired) and len(desired) > 0
+
+        def mutate(cur: RoleInstance):
+            cur.status.workers = workers
+            set_condition(cur.status.conditions, Condition.new(
+                C.COND_ALL_PODS_READY, all_ready,
+                "AllReady" if all_ready else "Waiting",
+                f"{ready}/{len(desired)} workers ready"))
+            restarting = get_condition(cur.status.conditions, C.COND_RESTARTING)
+            is_restarting = restarting is not None and restarting.status == "True"
+            set_condition(cur.status.conditions, Condition.new(
+                C.COND_READY, all_ready and not is_restarting,
+                "Ready" if all_ready else "NotReady",
+                f"{ready}/{len(desired)} workers ready"))
+            return cur
+        try:
+            self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mutate,
+                             inst.metadata.namespace, subresource="status")
+        except KeyError:
+            pass

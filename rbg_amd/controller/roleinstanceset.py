@@ -1,0 +1,260 @@
+"""RoleInstanceSet controller — stateful replica materialization.
+
+Mirrors the reference's stateful mode (reference internal/controller/
+workloads/roleinstanceset/statefulmode/stateful_instance_set_control.go:
+168-1056): ordinal instances `{set}-{i}`, OrderedReady/Parallel creation,
+condemned deletion in descending ordinal order, rolling update honoring
+partition + maxUnavailable with current/update revisions, and in-place
+update per instance when only engine args changed (weight-reload class
+changes; anything structural recreates the gang).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional
+
+from ..api import constants as C
+from ..api.serde import asdict
+from ..api.types import (Condition, ObjectMeta, RoleInstance,
+                         RoleInstanceSet, RoleInstanceSpec, set_condition,
+                         get_condition)
+from ..store.revisions import hash_spec
+from ..store.store import Store, set_owner
+
+log = logging.getLogger(__name__)
+
+
+def instance_ordinal_name(set_name: str, i: int) -> str:
+    return f"{set_name}-{i}"
+
+
+def template_hash(ris: RoleInstanceSet) -> str:
+    return hash_spec(asdict(ris.spec.template))
+
+
+def _is_ready(inst: RoleInstance) -> bool:
+    c = get_condition(inst.status.conditions, C.COND_READY)
+    return c is not None and c.status == "True"
+
+
+class RoleInstanceSetController:
+    def __init__(self, store: Store):
+        self.store = store
+
+    def reconcile(self, name: str, namespace: str = "default") -> float:
+        ris = self.store.try_get(C.KIND_ROLE_INSTANCE_SET, name, namespace)
+        if ris is None:
+            return 0.0
+        if ris.metadata.deletion_timestamp is not None:
+            for inst in self._owned(ris):
+                self._delete_instance(inst)
+            self.store.try_delete(C.KIND_ROLE_INSTANCE_SET, name, namespace)
+            return 0.0
+        update_hash = template_hash(ris)
+        instances = {i.metadata.name: i for i in self._owned(ris)}
+        requeue = self._scale(ris, instances, update_hash)
+        requeue = max(requeue, self._rolling_update(ris, instances, update_hash))
+        self._update_status(ris, instances, update_hash)
+        return requeue
+
+    # ------------------------------------------------------------------
+
+    def _owned(self, ris: RoleInstanceSet) -> List[RoleInstance]:
+        return self.store.list_owned(C.KIND_ROLE_INSTANCE,
+                                     ris.metadata.uid, ris.metadata.namespace)
+
+    def _make_instance(self, ris: RoleInstanceSet, ordinal: int,
+                       revision: str) -> RoleInstance:
+        tmpl = ris.spec.template
+        labels = dict(ris.metadata.labels)
+        labels.update(tmpl.metadata.labels)
+        labels[C.LABEL_ROLE_INDEX] = str(ordinal)
+        labels[C.LABEL_REVISION_HASH] = revision
+        annotations = dict(ris.metadata.annotations)
+        annotations.update(tmpl.metadata.annotations)
+        inst = RoleInstance(
+            metadata=ObjectMeta(
+                name=instance_ordinal_name(ris.metadata.name, ordinal),
+                namespace=ris.metadata.namespace,
+                labels=labels, annotations=annotations),
+            spec=RoleInstanceSpec(components=tmpl.components,
+                                  restart_policy=tmpl.restart_policy))
+        inst.status.update_revision = revision
+        inst.status.current_revision = revision
+        set_owner(inst, ris)
+        # keep the RBG ownerRef too so the leaf controller can key GPU
+        # stickiness by group uid
+        for ref in ris.metadata.owner_references:
+            if ref.kind == C.KIND_RBG:
+                inst.metadata.owner_references.append(ref)
+        return inst
+
+    def _delete_instance(self, inst: RoleInstance) -> None:
+        def mark(cur: RoleInstance):
+            import time as _t
+            cur.metadata.deletion_timestamp = _t.time()
+            return cur
+        try:
+            self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mark,
+                             inst.metadata.namespace)
+        except KeyError:
+            pass
+
+    # ------------------------------------------------------------------
+
+    def _scale(self, ris: RoleInstanceSet,
+               instances: Dict[str, RoleInstance], update_hash: str) -> float:
+        want = ris.spec.replicas
+        ordered = ris.spec.pod_management_policy == C.POD_MANAGEMENT_ORDERED_READY
+        # priority scale-in via the role-instance-to-delete annotation
+        # (reference statelessmode/sync/scale.go:48-344 priority delete)
+        to_delete = {n.strip() for n in ris.metadata.annotations.get(
+            C.ANNO_ROLE_INSTANCE_TO_DELETE, "").split(",") if n.strip()}
+        requeue = 0.0
+        for i in range(want):
+            name = instance_ordinal_name(ris.metadata.name, i)
+            inst = instances.get(name)
+            if inst is None or inst.metadata.deletion_timestamp is not None:
+                if inst is None:
+                    created = self.store.create(
+                        self._make_instance(ris, i, update_hash))
+                    instances[name] = created
+                if ordered:
+                    return 0.2   # OrderedReady: one at a time
+            elif ordered and not _is_ready(inst):
+                return 0.2       # gate the next ordinal on readiness
+        # condemned: ordinals >= want, deleted descending (monotonic guard)
+        condemned = sorted(
+            (i for i in instances.values()
+             if self._ordinal(i) is not None and self._ordinal(i) >= want
+             and i.metadata.deletion_timestamp is None),
+            key=lambda i: -self._ordinal(i))
+        for name in to_delete:
+            inst = instances.get(name)
+            if inst is not None and inst not in condemned \
+                    and inst.metadata.deletion_timestamp is None:
+                condemned.insert(0, inst)
+        for inst in condemned:
+            self._delete_instance(inst)
+            requeue = 0.2
+            if ordered:
+                break
+        return requeue
+
+    def _ordinal(self, inst: RoleInstance) -> Optional[int]:
+        tail = inst.metadata.name.rsplit("-", 1)[-1]
+        return int(tail) if tail.isdigit() else None
+
+    # ------------------------------------------------------------------
+
+    def _rolling_update(self, ris: RoleInstanceSet,
+                        instances: Dict[str, RoleInstance],
+                        update_hash: str) -> float:
+        """Monotonic rolling update honoring partition + maxUnavailable
+        (reference statefulmode progressUpdate:553-633): walk ordinals
+        descending, update instances above the partition, never exceeding
+        maxUnavailable simultaneously-not-ready instances."""
+        strat = ris.spec.update_strategy
+        if strat.paused:
+            return 0.0
+        want = ris.spec.replicas
+        stale = []
+        not_ready = 0
+        for i in range(want):
+            inst = instances.get(instance_ordinal_name(ris.metadata.name, i))
+            if inst is None:
+                not_ready += 1
+                continue
+            if not _is_ready(inst):
+                not_ready += 1
+            if inst.metadata.labels.get(C.LABEL_REVISION_HASH) != update_hash:
+                stale.append((i, inst))
+        if not stale:
+            return 0.0
+        budget = max(0, strat.max_unavailable - not_ready)
+        # descending ordinals, only above the partition
+        for i, inst in sorted(stale, key=lambda t: -t[0]):
+            if i < strat.partition:
+                continue
+            if budget <= 0:
+                break
+            if self._can_update_in_place(ris, inst):
+                self._in_place_update(ris, inst, update_hash)
+            else:
+                # recreate: delete; _scale recreates at the new revision
+                self._delete_instance(inst)
+            budget -= 1
+        return 0.2
+
+    def _can_update_in_place(self, ris: RoleInstanceSet,
+                             inst: RoleInstance) -> bool:
+        """In-place is feasible when only engine args/env changed — the
+        revision-diff feasibility check of reference
+        inplace/pod/inplaceupdate/inplace_update.go:68-317. Component
+        topology or resource changes force a recreate."""
+        if ris.spec.update_strategy.type == C.UPDATE_RECREATE:
+            return False
+        old_shape = [(c.name, c.size,
+                      (c.template.main_engine().resources.gpus
+                       if c.template and c.template.main_engine() else 0))
+                     for c in inst.spec.components]
+        new_shape = [(c.name, c.size,
+                      (c.template.main_engine().resources.gpus
+                       if c.template and c.template.main_engine() else 0))
+                     for c in ris.spec.template.components]
+        feasible = old_shape == new_shape
+        if not feasible and ris.spec.update_strategy.type == C.UPDATE_IN_PLACE_ONLY:
+            log.warning("ris %s: in-place-only update infeasible; holding",
+                        ris.metadata.name)
+        return feasible
+
+    def _in_place_update(self, ris: RoleInstanceSet, inst: RoleInstance,
+                         update_hash: str) -> None:
+        """Swap the spec under the live gang; the RoleInstance controller's
+        live-update path reloads weights/args without tearing down the KV
+        pool. Records expected-restart baselines so the restart detector
+        does not count the reload as a crash (reference
+        roleinstance_types.go:170-198)."""
+        def mutate(cur: RoleInstance):
+            cur.spec.components = ris.spec.template.components
+            cur.metadata.labels[C.LABEL_REVISION_HASH] = update_hash
+            cur.status.update_revision = update_hash
+            for w in cur.status.workers:
+                cur.status.in_place_update_baselines[w.name] = w.restart_count
+            set_condition(cur.status.conditions, Condition.new(
+                C.COND_INPLACE_UPDATE_READY, False, "Updating",
+                f"in-place update to {update_hash}"))
+            return cur
+        self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mutate,
+                         inst.metadata.namespace)
+
+    # ------------------------------------------------------------------
+
+    def _update_status(self, ris: RoleInstanceSet,
+                       instances: Dict[str, RoleInstance],
+                       update_hash: str) -> None:
+        live = [i for i in self._owned(ris)
+                if i.metadata.deletion_timestamp is None]
+        ready = sum(1 for i in live if _is_ready(i))
+        updated = sum(1 for i in live
+                      if i.metadata.labels.get(C.LABEL_REVISION_HASH) == update_hash)
+
+        def mutate(cur: RoleInstanceSet):
+            cur.status.observed_generation = cur.metadata.generation
+            cur.status.replicas = len(live)
+            cur.status.ready_replicas = ready
+            cur.status.updated_replicas = updated
+            cur.status.update_revision = update_hash
+            if updated == len(live):
+                cur.status.current_revision = update_hash
+            set_condition(cur.status.conditions, Condition.new(
+                C.COND_READY, ready == cur.spec.replicas and
+                len(live) == cur.spec.replicas,
+                "AllReady" if ready == cur.spec.replicas else "Scaling",
+                f"{ready}/{cur.spec.replicas} instances ready"))
+            return cur
+        try:
+            self.store.apply(C.KIND_ROLE_INSTANCE_SET, ris.metadata.name,
+                             mutate, ris.metadata.namespace, subresource="status")
+        except KeyError:
+            pass

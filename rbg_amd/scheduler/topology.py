@@ -1,0 +1,112 @@
+"""GPU topology discovery — xGMI link map of one MI355X node.
+
+The placement analog of the reference's exclusive-topology / node-affinity
+machinery (reference pkg/reconciler/pod_reconciler.go:160-241): instead of
+zone labels we rank GPU pairs by xGMI connectivity.  On an MI355X OAM board
+every GPU exposes 7 point-to-point xGMI links (~153 GB/s each), one to each
+peer — fully connected — so any pair is "adjacent"; the ranking still matters
+for partial-node visibility and future multi-hop parts, and the measured
+per-pair bandwidth feeds KV-transfer chunk sizing.
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from ..api import constants as C
+
+LINK_XGMI = "XGMI"
+LINK_PCIE = "PCIE"
+LINK_NONE = "NONE"
+
+_RANK = {LINK_XGMI: 2, LINK_PCIE: 1, LINK_NONE: 0}
+
+
+@dataclass
+class NodeTopology:
+    num_gpus: int
+    link_type: Dict[Tuple[int, int], str] = field(default_factory=dict)
+    link_gbps: Dict[Tuple[int, int], float] = field(default_factory=dict)
+    hbm_bytes: int = C.MI355X_HBM_BYTES_PER_GPU
+
+    def link(self, a: int, b: int) -> str:
+        if a == b:
+            return LINK_XGMI
+        return self.link_type.get((min(a, b), max(a, b)), LINK_NONE)
+
+    def bandwidth(self, a: int, b: int) -> float:
+        if a == b:
+            return float("inf")
+        return self.link_gbps.get((min(a, b), max(a, b)), 0.0)
+
+    def peers_ranked(self, gpu: int) -> List[int]:
+        """Peers of ``gpu`` best-first: xGMI > PCIe > none, then bandwidth."""
+        others = [g for g in range(self.num_gpus) if g != gpu]
+        return sorted(others, key=lambda g: (-_RANK[self.link(gpu, g)],
+                                             -self.bandwidth(gpu, g), g))
+
+    def adjacency_score(self, gpus: List[int]) -> float:
+        """Sum of pairwise link ranks — higher = tighter gang placement."""
+        score = 0.0
+        for i, a in enumerate(gpus):
+            for b in gpus[i + 1:]:
+                score += _RANK[self.link(a, b)] + self.bandwidth(a, b) / 1000.0
+        return score
+
+
+def fully_connected(num_gpus: int = C.MI355X_GPUS_PER_NODE,
+                    gbps: float = C.MI355X_XGMI_GBPS_PER_LINK) -> NodeTopology:
+    """The MI355X OAM default: all-pairs single-hop xGMI."""
+    topo = NodeTopology(num_gpus=num_gpus)
+    for a in range(num_gpus):
+        for b in range(a + 1, num_gpus):
+            topo.link_type[(a, b)] = LINK_XGMI
+            topo.link_gbps[(a, b)] = gbps
+    return topo
+
+
+def _detect_num_gpus() -> int:
+    try:
+        import torch
+        if torch.cuda.is_available():
+            return torch.cuda.device_count()
+    except Exception:
+        pass
+    return 0
+
+
+def discover(timeout: float = 10.0) -> NodeTopology:
+    """Best-effort live discovery: GPU count from HIP, link map from
+    `rocm-smi --showtopotype`; falls back to the fully-connected default
+    shape for whatever GPU count is visible (0 GPUs -> 8-GPU fake for CPU
+    development, matching the KWOK-style simulated backend of SURVEY §4)."""
+    n = _detect_num_gpus()
+    if n == 0:
+        return fully_connected()
+    topo = fully_connected(n)
+    try:
+        out = subprocess.run(["rocm-smi", "--showtopotype"],
+                             capture_output=True, text=True, timeout=timeout).stdout
+        parse_topo_type(out, topo)
+    except Exception:
+        pass
+    return topo
+
+
+def parse_topo_type(text: str, topo: NodeTopology) -> None:
+    """Parse rocm-smi topology-type output lines of the form
+    `Link type between DRM devices 0 and 1: XGMI` (format tolerant)."""
+    import re
+    pat = re.compile(r"(\d+)\s+and\s+(\d+).*?:\s*([A-Z]+)", re.IGNORECASE)
+    for line in text.splitlines():
+        m = pat.search(line)
+        if not m:
+            continue
+        a, b = int(m.group(1)), int(m.group(2))
+        lt = m.group(3).upper()
+        if a < topo.num_gpus and b < topo.num_gpus and a != b:
+            key = (min(a, b), max(a, b))
+            topo.link_type[key] = LINK_XGMI if "XGMI" in lt else (
+                LINK_PCIE if "PCIE" in lt else LINK_NONE)

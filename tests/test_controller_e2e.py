@@ -1,0 +1,261 @@
+"""Controller integration suite — the envtest analog: real store,
+real controllers, real engine processes (the `echo` runner on CPU), no GPU.
+Covers BASELINE config 1: a 2-role router + echo-worker RoleBasedGroup with
+dependency ordering, plus restart policy, rolling update, scaling adapter,
+orphan cleanup and RBGSet fan-out.
+"""
+import os
+import time
+
+import pytest
+
+from rbg_amd.api import constants as C
+from rbg_amd.api.types import (ComponentSpec, Condition, CoordinatedPolicy,
+                               CoordinatedPolicySpec, CoordinatedScaling,
+                               CoordinationRule, CoordinationStrategy,
+                               EngineResources, EngineSpec, EngineTemplate,
+                               ObjectMeta, RoleBasedGroup, RoleBasedGroupSpec,
+                               RoleBasedGroupScalingAdapter, RoleSpec,
+                               ScaleTargetRef, ScalingAdapterSpecFull,
+                               get_condition)
+from rbg_amd.controller.manager import Manager, ManagerOptions
+from rbg_amd.discovery.config_builder import load_config
+
+
+def cpu_template(runner="echo", args=None):
+    return EngineTemplate(engines=[EngineSpec(
+        name="engine", runner=runner, args=args or {},
+        resources=EngineResources(gpus=0, cpu_only=True))])
+
+
+def router_worker_rbg(name="pd", worker_replicas=2, worker_args=None):
+    return RoleBasedGroup(
+        metadata=ObjectMeta(name=name),
+        spec=RoleBasedGroupSpec(roles=[
+            RoleSpec(name="router", replicas=1, template=cpu_template()),
+            RoleSpec(name="worker", replicas=worker_replicas,
+                     dependencies=["router"],
+                     template=cpu_template(args=worker_args)),
+        ]))
+
+
+@pytest.fixture
+def mgr(tmp_run_dir):
+    m = Manager(ManagerOptions(run_root=tmp_run_dir, num_gpus=8,
+                               resync_period=0.1, gang_timeout=5.0))
+    # fast restart backoff so failure tests run in seconds
+    m.restarts.base = 0.2
+    m.restarts.max_delay = 1.0
+    m.start()
+    yield m
+    m.stop()
+
+
+def rbg_ready(m, name):
+    rbg = m.store.try_get(C.KIND_RBG, name)
+    if rbg is None:
+        return False
+    c = get_condition(rbg.status.conditions, C.COND_READY)
+    return c is not None and c.status == "True"
+
+
+def test_two_role_group_becomes_ready(mgr):
+    mgr.store.create(router_worker_rbg())
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "pd"), timeout=30), \
+        _debug_dump(mgr)
+    rbg = mgr.store.get(C.KIND_RBG, "pd")
+    sts = {s.name: s for s in rbg.status.role_statuses}
+    assert sts["router"].ready_replicas == 1
+    assert sts["worker"].ready_replicas == 2
+    # discovery config published with the full topology
+    path = mgr.registry.path_for("default", "pd")
+    assert os.path.exists(path)
+    doc = load_config(path)
+    roles = {r["name"]: r for r in doc["group"]["roles"]}
+    assert len(roles["worker"]["instances"]) == 2
+    assert all(i["ready"] for i in roles["worker"]["instances"])
+
+
+def test_dependency_ordering_router_first(mgr):
+    mgr.store.create(router_worker_rbg(name="ord"))
+    # before the router is ready, no worker instances may exist
+    seen_violation = []
+
+    def check():
+        workers = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+            C.LABEL_GROUP_NAME: "ord", C.LABEL_ROLE_NAME: "worker"})
+        if workers and not rbg_ready_role(mgr, "ord", "router"):
+            seen_violation.append(True)
+        return rbg_ready(mgr, "ord")
+
+    assert mgr.wait_for(check, timeout=30)
+    assert not seen_violation
+
+
+def rbg_ready_role(m, name, role):
+    rbg = m.store.try_get(C.KIND_RBG, name)
+    if rbg is None:
+        return False
+    for s in rbg.status.role_statuses:
+        if s.name == role:
+            want = rbg.spec.role(role).replicas
+            return s.ready_replicas >= want
+    return False
+
+
+def test_worker_crash_triggers_gang_recreate(mgr):
+    # worker crashes after 1s; restart policy recreates the instance
+    mgr.store.create(router_worker_rbg(
+        name="crash", worker_replicas=1,
+        worker_args={"crash_after": 1.0}))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "crash"), timeout=30)
+
+    def restarted():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+            C.LABEL_GROUP_NAME: "crash", C.LABEL_ROLE_NAME: "worker"})
+        return any(i.status.restart_count >= 1 for i in insts)
+
+    assert mgr.wait_for(restarted, timeout=30), _debug_dump(mgr)
+    # and it recovers (becomes ready again after recreate)
+    time.sleep(0.5)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "crash"), timeout=30)
+
+
+def test_scale_up_and_down(mgr):
+    mgr.store.create(router_worker_rbg(name="scale", worker_replicas=1))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "scale"), timeout=30)
+
+    def set_replicas(n):
+        def mutate(cur):
+            cur.spec.role("worker").replicas = n
+            return cur
+        mgr.store.apply(C.KIND_RBG, "scale", mutate)
+
+    set_replicas(3)
+    assert mgr.wait_for(
+        lambda: rbg_ready_role(mgr, "scale", "worker") and
+        len(_instances(mgr, "scale", "worker")) == 3, timeout=30)
+    set_replicas(1)
+    assert mgr.wait_for(
+        lambda: len(_instances(mgr, "scale", "worker")) == 1, timeout=30)
+    # highest ordinals were condemned first
+    assert _instances(mgr, "scale", "worker")[0].metadata.name == "scale-worker-0"
+
+
+def _instances(m, group, role):
+    out = [i for i in m.store.list(C.KIND_ROLE_INSTANCE, selector={
+        C.LABEL_GROUP_NAME: group, C.LABEL_ROLE_NAME: role})
+        if i.metadata.deletion_timestamp is None]
+    out.sort(key=lambda i: i.metadata.name)
+    return out
+
+
+def test_scaling_adapter_drives_replicas(mgr):
+    mgr.store.create(router_worker_rbg(name="auto", worker_replicas=1))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "auto"), timeout=30)
+    ad = RoleBasedGroupScalingAdapter(
+        metadata=ObjectMeta(name="auto-worker"),
+        spec=ScalingAdapterSpecFull(
+            replicas=2, scale_target_ref=ScaleTargetRef(name="auto", role="worker")))
+    mgr.store.create(ad)
+    assert mgr.wait_for(
+        lambda: len(_instances(mgr, "auto", "worker")) == 2, timeout=30)
+    adapter = mgr.store.get(C.KIND_SCALING_ADAPTER, "auto-worker")
+    assert adapter.status.phase == C.SCALING_ADAPTER_BOUND
+
+    from rbg_amd.controller.scalingadapter import scale_adapter
+    scale_adapter(mgr.store, "auto-worker", 1)
+    assert mgr.wait_for(
+        lambda: len(_instances(mgr, "auto", "worker")) == 1, timeout=30)
+
+
+def test_role_removal_cleans_orphans(mgr):
+    mgr.store.create(router_worker_rbg(name="orph"))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "orph"), timeout=30)
+
+    def drop_worker(cur):
+        cur.spec.roles = [r for r in cur.spec.roles if r.name != "worker"]
+        return cur
+    mgr.store.apply(C.KIND_RBG, "orph", drop_worker)
+    assert mgr.wait_for(
+        lambda: mgr.store.try_get(C.KIND_ROLE_INSTANCE_SET, "orph-worker") is None,
+        timeout=30)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "orph"), timeout=30)
+
+
+def test_delete_rbg_tears_down_processes(mgr):
+    mgr.store.create(router_worker_rbg(name="gone"))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "gone"), timeout=30)
+    pids = [w.pid for i in _instances(mgr, "gone", "worker")
+            for w in i.status.workers]
+    assert pids
+
+    def mark(cur):
+        cur.metadata.deletion_timestamp = time.time()
+        return cur
+    mgr.store.apply(C.KIND_RBG, "gone", mark)
+    assert mgr.wait_for(
+        lambda: mgr.store.try_get(C.KIND_RBG, "gone") is None, timeout=30)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        if not any(_pid_alive(p) for p in pids):
+            break
+        time.sleep(0.1)
+    assert not any(_pid_alive(p) for p in pids)
+
+
+def _pid_alive(pid):
+    try:
+        os.kill(pid, 0)
+        return True
+    except OSError:
+        return False
+
+
+def test_coordinated_scaling_caps_decode_pool(mgr):
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="coord"),
+        spec=RoleBasedGroupSpec(roles=[
+            RoleSpec(name="prefill", replicas=2, template=cpu_template()),
+            RoleSpec(name="decode", replicas=4, template=cpu_template()),
+        ]))
+    policy = CoordinatedPolicy(
+        metadata=ObjectMeta(name="coord"),
+        spec=CoordinatedPolicySpec(rules=[CoordinationRule(
+            roles=["prefill", "decode"],
+            strategy=CoordinationStrategy(
+                scaling=CoordinatedScaling(max_skew=50)))]))
+    mgr.store.create(policy)
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "coord"), timeout=40), \
+        _debug_dump(mgr)
+    assert len(_instances(mgr, "coord", "prefill")) == 2
+    assert len(_instances(mgr, "coord", "decode")) == 4
+
+
+def test_rbgset_fans_out(mgr):
+    from rbg_amd.api.types import RoleBasedGroupSet, RoleBasedGroupSetSpec
+    rbgset = RoleBasedGroupSet(
+        metadata=ObjectMeta(name="fleet"),
+        spec=RoleBasedGroupSetSpec(
+            replicas=2,
+            template=router_worker_rbg(worker_replicas=1).spec))
+    mgr.store.create(rbgset)
+    assert mgr.wait_for(
+        lambda: rbg_ready(mgr, "fleet-0") and rbg_ready(mgr, "fleet-1"),
+        timeout=40)
+    assert mgr.wait_for(
+        lambda: mgr.store.get(C.KIND_RBG_SET, "fleet").status.ready_replicas == 2,
+        timeout=10)
+
+
+def _debug_dump(m):
+    lines = []
+    for kind in (C.KIND_RBG, C.KIND_ROLE_INSTANCE_SET, C.KIND_ROLE_INSTANCE):
+        for o in m.store.list(kind, namespace=None):
+            conds = getattr(o.status, "conditions", [])
+            lines.append(f"{kind} {o.metadata.name}: " + "; ".join(
+                f"{c.type}={c.status}({c.reason}:{c.message})" for c in conds))
+            for w in getattr(o.status, "workers", []):
+                lines.append(f"   worker {w.name} phase={w.phase} pid={w.pid}")
+    return "\n".join(lines)
