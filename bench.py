@@ -1,0 +1,156 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark — Llama-3-8B continuous-batching decode.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs the
+serving step on N GPUs of one node (torchrun, one rank per GPU, RCCL).  Each
+rank serves its own engine replica (weak scaling, "dpN"): batch PROMPTS of
+2048 synthetic tokens are prefilled during setup, then K timed decode steps
+run over the fixed running batch.  Rank 0 prints ONE JSON line; `value` is
+whole-job output tok/s (all GPUs); p50 TTFT of the prefill phase rides in
+`config`.  Baseline: the reference's only published serving number
+(1300.41 tok/s total throughput, Qwen3-32B on NVIDIA).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=64)
+    p.add_argument("--warmup", type=int, default=16)
+    p.add_argument("--batch", type=int, default=64,
+                   help="decode batch per GPU")
+    p.add_argument("--seq-len", type=int, default=2048,
+                   help="synthetic prompt length")
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--eager", action="store_true",
+                   help="disable hipGraph capture")
+    return p.parse_args()
+
+
+def main() -> int:
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(args.gpus, world)
+    distributed = world > 1
+    if distributed:
+        import torch.distributed as dist
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+    device = args.device
+    if device == "cuda" and not torch.cuda.is_available():
+        print(json.dumps({"error": "no GPU visible"}), flush=True)
+        return 1
+
+    from rbg_amd.engine.config import EngineConfig, ModelConfig
+    from rbg_amd.engine.engine import LLMEngine
+    from rbg_amd.engine.sequence import SamplingParams
+
+    model_cfg = ModelConfig.preset(args.model)
+    cfg = EngineConfig(
+        model=model_cfg, device=device,
+        max_batch_size=max(args.batch, 8),
+        max_seq_len=args.seq_len + args.steps + args.warmup + 64,
+        max_prefill_tokens=8192,
+        enforce_eager=args.eager or device != "cuda",
+        kv_pool_tokens=(args.batch *
+                        (args.seq_len + args.steps + args.warmup + 64) + 4096),
+    )
+    eng = LLMEngine(cfg)
+
+    torch.manual_seed(123 + rank)
+    prompts = [torch.randint(0, model_cfg.vocab_size, (args.seq_len,)).tolist()
+               for _ in range(args.batch)]
+    sampling = SamplingParams(
+        max_new_tokens=args.steps + args.warmup + 32, ignore_eos=True)
+    for prompt in prompts:
+        eng.add_request(prompt, sampling)
+
+    # ---- setup phase: prefill everything (TTFT measured here) -------------
+    t0 = time.monotonic()
+    while eng.scheduler.waiting:
+        mode = eng.step()
+        if mode == "idle":
+            break
+    if device == "cuda":
+        torch.cuda.synchronize()
+    prefill_wall = time.monotonic() - t0
+    assert len(eng.scheduler.running) == args.batch, \
+        f"only {len(eng.scheduler.running)} running"
+
+    # ---- warmup decode steps (captures hipGraphs) -------------------------
+    for _ in range(args.warmup):
+        m = eng.step()
+        assert m == "decode", m
+    if device == "cuda":
+        torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+
+    # ---- timed region: exactly K decode steps -----------------------------
+    t_start = time.monotonic()
+    for _ in range(args.steps):
+        m = eng.step()
+        assert m == "decode", m
+    if device == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.monotonic() - t_start
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if device == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_tok_s = args.batch * n_gpus * args.steps / elapsed
+    ttfts = sorted(eng.stats.ttfts)
+    p50_ttft_ms = (ttfts[len(ttfts) // 2] * 1000.0) if ttfts else 0.0
+
+    if rank == 0:
+        baseline = 1300.41
+        print(json.dumps({
+            "metric": "output tok/s (Llama-3-8B serving, continuous batching)",
+            "value": round(total_tok_s, 2),
+            "unit": "tok/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(total_tok_s / baseline, 3),
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{n_gpus}",
+                "p50_ttft_ms": round(p50_ttft_ms, 1),
+                "prefill_wall_s": round(prefill_wall, 3),
+                "prefill_tok_s": round(
+                    args.batch * args.seq_len * n_gpus / prefill_wall, 1),
+            },
+        }), flush=True)
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

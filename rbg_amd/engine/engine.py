@@ -1,0 +1,113 @@
+"""LLM serving engine — scheduler + model runner + metrics.
+
+The GPU-resident realization of one "role instance" engine process
+(SURVEY §2.3): add_request() enqueues, step() runs one prefill or decode
+iteration, and the stats feed the Prometheus-style gauges the controller
+scrapes (tok/s, TTFT, pool occupancy).
+"""
+from __future__ import annotations
+
+import logging
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from ..models.llama import TPContext
+from .config import EngineConfig
+from .model_runner import ModelRunner
+from .scheduler import Scheduler
+from .sequence import FINISHED, SamplingParams, Sequence
+
+log = logging.getLogger(__name__)
+
+
+@dataclass
+class EngineStats:
+    prefill_tokens: int = 0
+    decode_tokens: int = 0
+    prefill_steps: int = 0
+    decode_steps: int = 0
+    finished: int = 0
+    ttfts: List[float] = field(default_factory=list)
+    started: float = field(default_factory=time.monotonic)
+
+    def snapshot(self) -> Dict[str, float]:
+        wall = max(1e-9, time.monotonic() - self.started)
+        ttfts = sorted(self.ttfts)
+        p50 = ttfts[len(ttfts) // 2] if ttfts else 0.0
+        return {
+            "prefill_tokens": self.prefill_tokens,
+            "decode_tokens": self.decode_tokens,
+            "output_tok_per_s": self.decode_tokens / wall,
+            "finished": self.finished,
+            "p50_ttft_s": p50,
+        }
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None):
+        self.cfg = cfg
+        self.runner = ModelRunner(cfg, tp)
+        self.scheduler = Scheduler(cfg, self.runner.cache)
+        self.stats = EngineStats()
+        self._finished: List[Sequence] = []
+
+    # ------------------------------------------------------------------
+
+    def add_request(self, prompt_tokens: List[int],
+                    sampling: Optional[SamplingParams] = None) -> Sequence:
+        seq = Sequence(prompt_tokens, sampling)
+        self.scheduler.add(seq)
+        return seq
+
+    def step(self) -> str:
+        """Run one engine iteration; returns the mode executed."""
+        mode, seqs = self.scheduler.schedule()
+        if mode == "prefill":
+            self.runner.prefill(seqs)
+            self.stats.prefill_tokens += sum(s.num_prompt_tokens for s in seqs)
+            self.stats.prefill_steps += 1
+            for s in seqs:
+                t = s.ttft()
+                if t is not None:
+                    self.stats.ttfts.append(t)
+            self.scheduler.finish_prefill(seqs)
+        elif mode == "decode":
+            self.runner.decode(seqs)
+            self.stats.decode_tokens += len(seqs)
+            self.stats.decode_steps += 1
+            self.scheduler.finish_decode()
+        self._collect_finished()
+        return mode
+
+    def _collect_finished(self) -> None:
+        # sequences leave scheduler.running when finished; track them here
+        pass
+
+    def run_until_done(self, max_steps: int = 1_000_000) -> List[Sequence]:
+        done: List[Sequence] = []
+        seen = set()
+        all_seqs: List[Sequence] = []
+        for _ in range(max_steps):
+            if not self.scheduler.has_work():
+                break
+            self.step()
+        return done
+
+    def generate(self, prompts: List[List[int]],
+                 sampling: Optional[SamplingParams] = None,
+                 max_steps: int = 1_000_000) -> List[Sequence]:
+        seqs = [self.add_request(p, sampling) for p in prompts]
+        steps = 0
+        while self.scheduler.has_work() and steps < max_steps:
+            self.step()
+            steps += 1
+        self.stats.finished += sum(1 for s in seqs if s.status == FINISHED)
+        return seqs
+
+    # -- live update (weight reload keeping the KV pool) ------
+
+    def reload_weights(self, seed: int) -> None:
+        self.runner.model.reload_weights(seed)

@@ -1,0 +1,194 @@
+"""Model runner — builds kernel batches and executes model steps.
+
+Owns the device, model, KV pool and (optionally) the hipGraph-captured
+decode path: decode steps at a given batch size are launch-bound (dozens of
+small kernels), so the runner captures one graph per padded batch-size
+bucket and replays it with fresh inputs copied into static buffers
+(guide: "capture launch-bound inner loops in hipGraphs").
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..models.llama import ForwardBatch, LlamaForCausalLM, TPContext
+from .config import EngineConfig
+from .kv_cache import PagedKVCache
+from .sequence import Sequence
+
+log = logging.getLogger(__name__)
+
+GRAPH_BATCH_SIZES = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256)
+
+
+class ModelRunner:
+    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None):
+        self.cfg = cfg
+        self.device = torch.device(cfg.device)
+        torch.manual_seed(cfg.seed)
+        self.model = LlamaForCausalLM(cfg.model, self.device, tp)
+        self.cache = PagedKVCache(cfg, self.device)
+        self.max_pages_per_seq = (cfg.max_seq_len + cfg.page_size - 1) // \
+            cfg.page_size
+        self._graphs: Dict[int, Tuple] = {}
+        self._graph_pool = None
+        log.info("model %s on %s: %d KV pages (%.1f GB pool)",
+                 cfg.model.name, self.device, self.cache.num_pages,
+                 self.cache.kv.numel() * 2 / 1e9)
+
+    # ------------------------------------------------------------------
+
+    def _i32(self, data) -> torch.Tensor:
+        return torch.tensor(data, dtype=torch.int32, device=self.device)
+
+    @torch.inference_mode()
+    def prefill(self, seqs: List[Sequence]) -> List[int]:
+        tokens: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        cu = [0]
+        last_idx = []
+        for seq in seqs:
+            n = seq.num_prompt_tokens
+            tokens.extend(seq.prompt_tokens)
+            positions.extend(range(n))
+            slots.extend(seq.block_table.slots_for(0, n))
+            cu.append(cu[-1] + n)
+            last_idx.append(cu[-1] - 1)
+        batch = ForwardBatch(
+            mode="prefill",
+            positions=self._i32(positions),
+            slot_mapping=self._i32(slots),
+            cu_seqlens=self._i32(cu))
+        hidden = self.model.forward(
+            torch.tensor(tokens, dtype=torch.int64, device=self.device),
+            batch, self.cache)
+        logits = self.model.logits(hidden, self._i32(last_idx))
+        next_tokens = self.sample(logits, seqs)
+        for seq, tok in zip(seqs, next_tokens):
+            seq.append_token(tok)
+        return next_tokens
+
+    # ------------------------------------------------------------------
+
+    def _decode_inputs(self, seqs: List[Sequence]):
+        tokens = [seq.last_token for seq in seqs]
+        positions = [seq.num_tokens - 1 for seq in seqs]
+        slots = [seq.block_table.slots_for(seq.num_tokens - 1, 1)[0]
+                 for seq in seqs]
+        ctx = [seq.num_tokens for seq in seqs]
+        max_pages = max(len(seq.block_table.pages) for seq in seqs)
+        bt = torch.zeros(len(seqs), max_pages, dtype=torch.int32)
+        for i, seq in enumerate(seqs):
+            bt[i, :len(seq.block_table.pages)] = torch.tensor(
+                seq.block_table.pages, dtype=torch.int32)
+        return tokens, positions, slots, ctx, bt
+
+    @torch.inference_mode()
+    def decode(self, seqs: List[Sequence]) -> List[int]:
+        tokens, positions, slots, ctx, bt = self._decode_inputs(seqs)
+        use_graph = (not self.cfg.enforce_eager and
+                     self.device.type == "cuda" and
+                     len(seqs) <= GRAPH_BATCH_SIZES[-1])
+        if use_graph:
+            logits = self._decode_graph(tokens, positions, slots, ctx, bt)
+        else:
+            batch = ForwardBatch(
+                mode="decode",
+                positions=self._i32(positions),
+                slot_mapping=self._i32(slots),
+                block_tables=bt.to(self.device),
+                context_lens=self._i32(ctx),
+                decode_num_splits=self._splits_for(len(seqs)))
+            hidden = self.model.forward(
+                torch.tensor(tokens, dtype=torch.int64, device=self.device),
+                batch, self.cache)
+            logits = self.model.logits(hidden)
+        next_tokens = self.sample(logits[:len(seqs)], seqs)
+        for seq, tok in zip(seqs, next_tokens):
+            seq.append_token(tok)
+        return next_tokens
+
+    # ------------------------------------------------------------------
+
+    def _splits_for(self, batch_size: int) -> int:
+        """Sync-free split-KV policy (graph-safe: depends only on the padded
+        batch size and the config's max context)."""
+        from .. import ops as _ops
+        m = self.cfg.model
+        tp = max(1, getattr(self.model.tp, "size", 1))
+        return _ops.pick_decode_splits(batch_size, m.num_kv_heads // tp,
+                                       self.cfg.max_seq_len)
+
+    def _graph_bucket(self, n: int) -> int:
+        for b in GRAPH_BATCH_SIZES:
+            if n <= b:
+                return b
+        return GRAPH_BATCH_SIZES[-1]
+
+    def _build_graph(self, bs: int):
+        """Capture one decode step at batch size ``bs`` into a hipGraph."""
+        m = self.cfg.model
+        static = {
+            "tokens": torch.zeros(bs, dtype=torch.int64, device=self.device),
+            "positions": torch.zeros(bs, dtype=torch.int32, device=self.device),
+            "slots": torch.zeros(bs, dtype=torch.int32, device=self.device),
+            "ctx": torch.ones(bs, dtype=torch.int32, device=self.device),
+            "bt": torch.zeros(bs, self.max_pages_per_seq, dtype=torch.int32,
+                              device=self.device),
+        }
+        batch = ForwardBatch(mode="decode", positions=static["positions"],
+                             slot_mapping=static["slots"],
+                             block_tables=static["bt"],
+                             context_lens=static["ctx"],
+                             decode_num_splits=self._splits_for(bs))
+        # warm up allocations outside capture
+        torch.cuda.synchronize()
+        hidden = self.model.forward(static["tokens"], batch, self.cache)
+        logits = self.model.logits(hidden)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph, pool=self._graph_pool):
+            hidden = self.model.forward(static["tokens"], batch, self.cache)
+            logits = self.model.logits(hidden)
+        if self._graph_pool is None:
+            self._graph_pool = graph.pool()
+        self._graphs[bs] = (graph, static, logits)
+        log.info("captured decode hipGraph for batch %d", bs)
+        return self._graphs[bs]
+
+    def _decode_graph(self, tokens, positions, slots, ctx, bt) -> torch.Tensor:
+        n = len(tokens)
+        bs = self._graph_bucket(n)
+        if bs not in self._graphs:
+            self._build_graph(bs)
+        graph, static, logits = self._graphs[bs]
+        static["tokens"][:n] = torch.tensor(tokens, dtype=torch.int64,
+                                            device=self.device)
+        static["positions"][:n] = self._i32(positions)
+        static["slots"][:n] = self._i32(slots)
+        static["ctx"].fill_(1)
+        static["ctx"][:n] = self._i32(ctx)
+        static["bt"].zero_()
+        static["bt"][:n, :bt.shape[1]] = bt.to(self.device)
+        if n < bs:
+            # padding rows decode against page 0 with ctx=1: harmless work,
+            # their slot writes go to slot 0 of page 0 — reserve it
+            static["slots"][n:] = 0
+            static["positions"][n:] = 0
+        graph.replay()
+        return logits
+
+    # ------------------------------------------------------------------
+
+    @torch.inference_mode()
+    def sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> List[int]:
+        temps = [seq.sampling.temperature for seq in seqs]
+        if all(t == 0.0 for t in temps):
+            return logits.argmax(dim=-1).tolist()
+        t = torch.tensor([max(tt, 1e-5) for tt in temps],
+                         device=logits.device).unsqueeze(1)
+        probs = torch.softmax(logits / t, dim=-1)
+        return torch.multinomial(probs, 1).squeeze(1).tolist()

@@ -1,0 +1,211 @@
+"""Llama-family decoder built directly on the rbg_amd CDNA4 ops.
+
+MI355X-first structure: the four hot non-GEMM ops (RMSNorm with fused
+residual add, RoPE fused with the paged-KV write, prefill/decode paged
+attention, SiLU*up) are the hand-written HIP kernels of rbg_amd.ops; plain
+projections go through hipBLASLt via torch.nn.functional.linear with QKV and
+gate/up fused into single GEMMs so each layer runs 4 GEMMs total.
+
+Tensor parallelism is Megatron-style over RCCL/xGMI: QKV and gate_up are
+column-parallel (head-aligned shards), o_proj and down_proj row-parallel
+with one all-reduce each per layer (2 per layer), sized for the 7-link xGMI
+fabric by keeping the TP degree within one node.
+
+Capability analog: the model executor of the SGLang engines the reference
+orchestrates.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..engine.config import ModelConfig
+from ..engine.kv_cache import PagedKVCache
+
+
+@dataclass
+class ForwardBatch:
+    """Everything the kernels need for one step; built by the model runner."""
+    mode: str                       # "prefill" | "decode"
+    positions: torch.Tensor         # [T] int32
+    slot_mapping: torch.Tensor      # [T] int32 (KV write slots)
+    # prefill
+    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
+    # decode
+    block_tables: Optional[torch.Tensor] = None  # [S, max_pages] int32
+    context_lens: Optional[torch.Tensor] = None  # [S] int32
+    decode_num_splits: int = 0      # 0 = pick host-side (sync-free in graphs)
+
+
+def _linear_weight(out_f: int, in_f: int, device, dtype, gen) -> nn.Parameter:
+    w = torch.empty(out_f, in_f, device=device, dtype=dtype)
+    w.normal_(0.0, 0.02, generator=gen)
+    return nn.Parameter(w, requires_grad=False)
+
+
+class TPContext:
+    """Holds the tensor-parallel process group (RCCL over xGMI)."""
+
+    def __init__(self, size: int = 1, rank: int = 0, group=None):
+        self.size = size
+        self.rank = rank
+        self.group = group
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.size > 1:
+            torch.distributed.all_reduce(t, group=self.group)
+        return t
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int, tp: TPContext,
+                 device, dtype, gen):
+        super().__init__()
+        self.cfg = cfg
+        self.layer_idx = layer_idx
+        self.tp = tp
+        assert cfg.num_heads % tp.size == 0 and cfg.num_kv_heads % tp.size == 0, \
+            "TP degree must divide head counts"
+        self.heads = cfg.num_heads // tp.size
+        self.kv_heads = cfg.num_kv_heads // tp.size
+        self.head_dim = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        q_out = self.heads * cfg.head_dim
+        kv_out = self.kv_heads * cfg.head_dim
+        # column-parallel fused QKV; row-parallel O
+        self.wqkv = _linear_weight(q_out + 2 * kv_out, cfg.hidden_size,
+                                   device, dtype, gen)
+        self.wo = _linear_weight(cfg.hidden_size, q_out, device, dtype, gen)
+        self.q_out, self.kv_out = q_out, kv_out
+
+    def forward(self, x: torch.Tensor, batch: ForwardBatch,
+                kv: PagedKVCache, cos_sin: torch.Tensor) -> torch.Tensor:
+        T = x.shape[0]
+        qkv = torch.nn.functional.linear(x, self.wqkv)
+        q = qkv[:, :self.q_out].contiguous()
+        k = qkv[:, self.q_out:self.q_out + self.kv_out].contiguous()
+        v = qkv[:, self.q_out + self.kv_out:].contiguous()
+        key_cache = kv.key_cache(self.layer_idx)
+        value_cache = kv.value_cache(self.layer_idx)
+        # fused RoPE + paged KV write (q,k rotated in place)
+        ops.rope_store_kv(q, k, v, key_cache, value_cache, cos_sin,
+                          batch.positions, batch.slot_mapping)
+        q3 = q.view(T, self.heads, self.head_dim)
+        if batch.mode == "prefill":
+            o = ops.prefill_attention(
+                q3, k.view(T, self.kv_heads, self.head_dim),
+                v.view(T, self.kv_heads, self.head_dim),
+                batch.cu_seqlens, self.scale)
+        else:
+            o = ops.decode_attention(q3, key_cache, value_cache,
+                                     batch.block_tables, batch.context_lens,
+                                     self.scale,
+                                     num_splits=batch.decode_num_splits)
+        out = torch.nn.functional.linear(o.view(T, self.q_out), self.wo)
+        return self.tp.all_reduce(out)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, tp: TPContext, device, dtype, gen):
+        super().__init__()
+        self.tp = tp
+        assert cfg.intermediate_size % tp.size == 0
+        inter = cfg.intermediate_size // tp.size
+        self.w_gate_up = _linear_weight(2 * inter, cfg.hidden_size,
+                                        device, dtype, gen)
+        self.w_down = _linear_weight(cfg.hidden_size, inter, device, dtype, gen)
+        self.inter = inter
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        gu = torch.nn.functional.linear(x, self.w_gate_up)
+        h = ops.silu_mul(gu)
+        return self.tp.all_reduce(torch.nn.functional.linear(h, self.w_down))
+
+
+class LlamaLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int, tp: TPContext,
+                 device, dtype, gen):
+        super().__init__()
+        self.attn = LlamaAttention(cfg, layer_idx, tp, device, dtype, gen)
+        self.mlp = LlamaMLP(cfg, tp, device, dtype, gen)
+        self.input_norm = nn.Parameter(
+            torch.ones(cfg.hidden_size, device=device, dtype=dtype),
+            requires_grad=False)
+        self.post_norm = nn.Parameter(
+            torch.ones(cfg.hidden_size, device=device, dtype=dtype),
+            requires_grad=False)
+        self.eps = cfg.rms_eps
+
+    def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor],
+                batch: ForwardBatch, kv: PagedKVCache,
+                cos_sin: torch.Tensor):
+        if residual is None:
+            residual = x.clone()
+            h = ops.rmsnorm(x, self.input_norm, self.eps)
+        else:
+            # h := rmsnorm(x + residual); residual := x + residual  (fused)
+            ops.fused_add_rmsnorm(x, residual, self.input_norm, self.eps)
+            h = x
+        h = self.attn(h, batch, kv, cos_sin)
+        ops.fused_add_rmsnorm(h, residual, self.post_norm, self.eps)
+        h = self.mlp(h)
+        return h, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: ModelConfig, device: torch.device,
+                 tp: Optional[TPContext] = None):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp or TPContext()
+        dtype = torch.bfloat16
+        gen = torch.Generator(device=device)
+        gen.manual_seed(1234 + self.tp.rank)
+        self.embed = nn.Parameter(
+            torch.empty(cfg.vocab_size, cfg.hidden_size, device=device,
+                        dtype=dtype).normal_(0, 0.02, generator=gen),
+            requires_grad=False)
+        self.layers = nn.ModuleList([
+            LlamaLayer(cfg, i, self.tp, device, dtype, gen)
+            for i in range(cfg.num_layers)])
+        self.final_norm = nn.Parameter(
+            torch.ones(cfg.hidden_size, device=device, dtype=dtype),
+            requires_grad=False)
+        self.lm_head = _linear_weight(cfg.vocab_size, cfg.hidden_size,
+                                      device, dtype, gen)
+        self.cos_sin = ops.build_cos_sin_table(
+            cfg.head_dim, cfg.max_position, cfg.rope_theta,
+            device=device)
+
+    @torch.inference_mode()
+    def forward(self, tokens: torch.Tensor, batch: ForwardBatch,
+                kv: PagedKVCache) -> torch.Tensor:
+        x = self.embed[tokens.long()]
+        residual = None
+        for layer in self.layers:
+            x, residual = layer(x, residual, batch, kv, self.cos_sin)
+        x = x + residual
+        x = ops.rmsnorm(x, self.final_norm, self.cfg.rms_eps)
+        return x
+
+    @torch.inference_mode()
+    def logits(self, hidden: torch.Tensor,
+               gather_idx: Optional[torch.Tensor] = None) -> torch.Tensor:
+        if gather_idx is not None:
+            hidden = hidden[gather_idx.long()]
+        return torch.nn.functional.linear(hidden, self.lm_head).float()
+
+    def reload_weights(self, seed: int) -> None:
+        """In-place weight refresh — the live-update path (SURVEY §2.3
+        "Live engine update": new weights without tearing down the KV pool)."""
+        gen = torch.Generator(device=self.embed.device)
+        gen.manual_seed(seed + self.tp.rank)
+        with torch.no_grad():
+            for p in self.parameters():
+                if p.dim() == 2:
+                    p.normal_(0, 0.02, generator=gen)

@@ -1,0 +1,102 @@
+"""rbg_amd.ops — dispatch layer over the CDNA4 HIP kernels.
+
+On a GPU the compiled in-tree extension (_hip_ops, built by setup.py for
+gfx950) is MANDATORY: a missing extension raises instead of silently running
+an eager fallback, so GPU tests always exercise the native path.  On CPU
+(unit tests, GPU-less CI) the fp32 torch references in ops.reference serve
+as the implementation.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import reference
+
+_hip = None
+_hip_err: Exception | None = None
+try:
+    from . import _hip_ops as _hip   # type: ignore[attr-defined]
+except Exception as e:  # noqa: BLE001
+    _hip_err = e
+
+HAVE_HIP = _hip is not None
+
+
+def _require_hip():
+    if _hip is None:
+        raise RuntimeError(
+            "rbg_amd HIP extension is not built but a GPU tensor was passed. "
+            "Build it in-tree: PYTORCH_ROCM_ARCH=gfx950 python setup.py "
+            f"build_ext --inplace (import error: {_hip_err!r})")
+    return _hip
+
+
+def _on_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if _on_gpu(x):
+        return _require_hip().rmsnorm(x, weight, eps)
+    return reference.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float) -> None:
+    if _on_gpu(x):
+        _require_hip().fused_add_rmsnorm(x, residual, weight, eps)
+        return
+    reference.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def silu_mul(x: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(x):
+        return _require_hip().silu_mul(x)
+    return reference.silu_mul(x)
+
+
+def rope_store_kv(q, k, v, key_cache, value_cache, cos_sin, positions,
+                  slot_mapping) -> None:
+    if _on_gpu(q):
+        _require_hip().rope_store_kv(q, k, v, key_cache, value_cache,
+                                     cos_sin, positions, slot_mapping)
+        return
+    reference.rope_store_kv(q, k, v, key_cache, value_cache, cos_sin,
+                            positions, slot_mapping)
+
+
+def pick_decode_splits(num_seqs: int, num_kv_heads: int,
+                       max_context: int) -> int:
+    """Fill the chip: MI355X has 256 CUs; aim for >=512 workgroups unless the
+    context is too short to split into 256-key chunks."""
+    base = num_seqs * num_kv_heads
+    if base >= 512:
+        return 1
+    want = max(1, 512 // max(base, 1))
+    by_ctx = max(1, max_context // 256)
+    return int(min(want, by_ctx, 16))
+
+
+def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
+                     scale: float, num_splits: int = 0) -> torch.Tensor:
+    if _on_gpu(q):
+        if num_splits <= 0:
+            max_ctx = int(context_lens.max().item()) if context_lens.numel() else 1
+            num_splits = pick_decode_splits(q.shape[0], key_cache.shape[1],
+                                            max_ctx)
+        return _require_hip().decode_attention(
+            q, key_cache, value_cache, block_tables, context_lens, scale,
+            num_splits)
+    return reference.decode_attention(q, key_cache, value_cache, block_tables,
+                                      context_lens, scale)
+
+
+def prefill_attention(q, k, v, cu_seqlens, scale: float) -> torch.Tensor:
+    if _on_gpu(q):
+        block_info, seq_lens = reference.prefill_block_info(cu_seqlens.cpu())
+        return _require_hip().prefill_attention(
+            q, k, v, block_info.to(q.device), seq_lens.to(q.device), scale)
+    return reference.prefill_attention(q, k, v, cu_seqlens, scale)
+
+
+build_cos_sin_table = reference.build_cos_sin_table

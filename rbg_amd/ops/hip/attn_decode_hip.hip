@@ -1,0 +1,244 @@
+#include "hip/hip_runtime.h"
+// Paged flash-decode attention (gfx950) — single query token per sequence.
+//
+// HBM-bound: the cost is streaming the sequence's paged K/V once.  Layout
+// and access pattern are designed for that: K/V rows are head_dim=128
+// contiguous bf16 (256 B), read as 16-byte lane vectors by 16-lane groups
+// (4 keys per wave-instruction), GQA handled by computing all q-heads of a
+// kv-head in one pass so K/V bytes are read exactly once.  Split-KV
+// partitions ("flash-decoding") keep >=256 workgroups resident at small
+// batch; partials merge in a combine kernel via the (m, l) log-sum-exp
+// algebra.
+//
+// Capability analog: the decode-side paged attention of the engines the
+// reference orchestrates (SURVEY §2.3 decode engine row).
+#include "common.h"
+
+namespace {
+
+constexpr int HEAD_DIM = 128;
+constexpr float NEG_INF = -1e30f;
+
+// QPG = q heads per kv head (GQA group). One block = (seq, kv_head, split).
+// 4 waves; each wave covers 4 keys per iteration (16-lane groups, 16 B/lane).
+template <int QPG>
+__global__ __launch_bounds__(256) void decode_attn_kernel(
+    float* __restrict__ partial_o,        // [splits, seqs, QH, D]
+    float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
+    __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
+    const __hip_bfloat16* __restrict__ q, // [seqs, QH, D]
+    const __hip_bfloat16* __restrict__ key_cache,  // [pages, KVH, page, D]
+    const __hip_bfloat16* __restrict__ val_cache,
+    const int* __restrict__ block_tables, // [seqs, max_pages]
+    const int* __restrict__ context_lens, // [seqs]
+    const float scale, const int num_kv_heads, const int page_size,
+    const int max_pages, const int num_splits) {
+  const int kvh = blockIdx.x;
+  const int seq = blockIdx.y;
+  const int split = blockIdx.z;
+  const int num_q_heads = num_kv_heads * QPG;
+  const int ctx = context_lens[seq];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int group = lane >> 4;        // which of the wave's 4 keys
+  const int gl = lane & 15;           // lane within the 16-lane group
+  const int dbase = gl * 8;           // this lane's 8 dims
+
+  // split bounds (rounded to 16-key chunks)
+  const int chunk = 16;
+  const int nchunks = (ctx + chunk - 1) / chunk;
+  const int per_split = (nchunks + num_splits - 1) / num_splits;
+  const int key_begin = split * per_split * chunk;
+  const int key_end = min(ctx, (split + 1) * per_split * chunk);
+
+  // q fragment: this lane's 8 dims for each of the QPG heads
+  float qf[QPG][8];
+  {
+    const __hip_bfloat16* qrow =
+        q + ((size_t)seq * num_q_heads + kvh * QPG) * HEAD_DIM;
+#pragma unroll
+    for (int h = 0; h < QPG; ++h) {
+      Bf16x8U qv;
+      qv.u = *reinterpret_cast<const uint4*>(qrow + h * HEAD_DIM + dbase);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qf[h][j] = bf2f(qv.e[j]) * scale;
+    }
+  }
+
+  float m[QPG], l[QPG], acc[QPG][8];
+#pragma unroll
+  for (int h = 0; h < QPG; ++h) {
+    m[h] = NEG_INF;
+    l[h] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[h][j] = 0.f;
+  }
+
+  const int* btab = block_tables + (size_t)seq * max_pages;
+  // wave stride: 4 waves x 4 keys
+  for (int base = key_begin + wave * 4; base < key_end; base += 16) {
+    const int key = base + group;
+    const bool valid = key < key_end;
+    const int kslot = valid ? key : key_begin;   // clamp for safe address
+    const int page = btab[kslot / page_size];
+    const size_t row_off =
+        (((size_t)page * num_kv_heads + kvh) * page_size +
+         (kslot % page_size)) * HEAD_DIM;
+    Bf16x8U kv;
+    kv.u = *reinterpret_cast<const uint4*>(key_cache + row_off + dbase);
+    // scores for this key, all QPG heads
+    float p[QPG];
+    float tile_max[QPG];
+#pragma unroll
+    for (int h = 0; h < QPG; ++h) {
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s += qf[h][j] * bf2f(kv.e[j]);
+      s = group16_sum(s);             // full dot across the 16-lane group
+      if (!valid) s = NEG_INF;
+      // tile max across the wave's 4 keys
+      float tm = s;
+      tm = fmaxf(tm, __shfl_xor(tm, 16, 64));
+      tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
+      tile_max[h] = tm;
+      p[h] = s;
+    }
+    Bf16x8U vv;
+    vv.u = *reinterpret_cast<const uint4*>(val_cache + row_off + dbase);
+#pragma unroll
+    for (int h = 0; h < QPG; ++h) {
+      const float m_new = fmaxf(m[h], tile_max[h]);
+      const float alpha = __expf(m[h] - m_new);
+      const float pv = valid ? __expf(p[h] - m_new) : 0.f;
+      // sum of p over the 4 keys of this wave-iteration
+      float psum = pv;
+      psum += __shfl_xor(psum, 16, 64);
+      psum += __shfl_xor(psum, 32, 64);
+      l[h] = l[h] * alpha + psum;
+      m[h] = m_new;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc[h][j] = acc[h][j] * alpha + pv * bf2f(vv.e[j]);
+    }
+  }
+
+  // fold the 4 key-groups of the wave (same dims, disjoint keys, same m/l)
+#pragma unroll
+  for (int h = 0; h < QPG; ++h)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[h][j] += __shfl_xor(acc[h][j], 16, 64);
+      acc[h][j] += __shfl_xor(acc[h][j], 32, 64);
+    }
+
+  // cross-wave combine through LDS
+  __shared__ float lds_acc[4][QPG][HEAD_DIM];
+  __shared__ float lds_ml[4][QPG][2];
+  if (lane < 16) {
+#pragma unroll
+    for (int h = 0; h < QPG; ++h) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_acc[wave][h][dbase + j] = acc[h][j];
+      if (gl == 0) {
+        lds_ml[wave][h][0] = m[h];
+        lds_ml[wave][h][1] = l[h];
+      }
+    }
+  }
+  __syncthreads();
+
+  // threads cover (head, dim): QPG*128 outputs
+  for (int i = tid; i < QPG * HEAD_DIM; i += blockDim.x) {
+    const int h = i / HEAD_DIM, d = i % HEAD_DIM;
+    float m_star = NEG_INF;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) m_star = fmaxf(m_star, lds_ml[w][h][0]);
+    float o = 0.f, lsum = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      const float f = __expf(lds_ml[w][h][0] - m_star);
+      o += lds_acc[w][h][d] * f;
+      lsum += lds_ml[w][h][1] * f;
+    }
+    const int qh = kvh * QPG + h;
+    if (num_splits == 1) {
+      out[((size_t)seq * num_q_heads + qh) * HEAD_DIM + d] =
+          f2bf(lsum > 0.f ? o / lsum : 0.f);
+    } else {
+      partial_o[(((size_t)split * gridDim.y + seq) * num_q_heads + qh) *
+                    HEAD_DIM + d] = o;
+      if (d == 0) {
+        float* ml = partial_ml +
+            (((size_t)split * gridDim.y + seq) * num_q_heads + qh) * 2;
+        ml[0] = m_star;
+        ml[1] = lsum;
+      }
+    }
+  }
+}
+
+// Merge split partials: one block per (seq, q_head).
+__global__ void decode_combine_kernel(
+    __hip_bfloat16* __restrict__ out,       // [seqs, QH, D]
+    const float* __restrict__ partial_o,    // [splits, seqs, QH, D]
+    const float* __restrict__ partial_ml,   // [splits, seqs, QH, 2]
+    const int num_splits, const int num_q_heads) {
+  const int seq = blockIdx.x, qh = blockIdx.y, d = threadIdx.x;
+  const int seqs = gridDim.x;
+  float m_star = NEG_INF;
+  for (int s = 0; s < num_splits; ++s)
+    m_star = fmaxf(m_star,
+                   partial_ml[(((size_t)s * seqs + seq) * num_q_heads + qh) * 2]);
+  float o = 0.f, lsum = 0.f;
+  for (int s = 0; s < num_splits; ++s) {
+    const float* ml =
+        partial_ml + (((size_t)s * seqs + seq) * num_q_heads + qh) * 2;
+    const float f = __expf(ml[0] - m_star);
+    o += partial_o[(((size_t)s * seqs + seq) * num_q_heads + qh) * HEAD_DIM +
+                   d] * f;
+    lsum += ml[1] * f;
+  }
+  out[((size_t)seq * num_q_heads + qh) * HEAD_DIM + d] =
+      f2bf(lsum > 0.f ? o / lsum : 0.f);
+}
+
+}  // namespace
+
+extern "C" {
+
+void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
+                             const void* q, const void* key_cache,
+                             const void* val_cache, const void* block_tables,
+                             const void* context_lens, float scale,
+                             int num_seqs, int num_q_heads, int num_kv_heads,
+                             int page_size, int max_pages, int num_splits,
+                             hipStream_t stream) {
+  const int qpg = num_q_heads / num_kv_heads;
+  dim3 grid(num_kv_heads, num_seqs, num_splits), block(256);
+#define LAUNCH_QPG(QPG)                                                       \
+  hipLaunchKernelGGL(decode_attn_kernel<QPG>, grid, block, 0, stream,         \
+                     (float*)partial_o, (float*)partial_ml,                   \
+                     (__hip_bfloat16*)out, (const __hip_bfloat16*)q,          \
+                     (const __hip_bfloat16*)key_cache,                        \
+                     (const __hip_bfloat16*)val_cache,                        \
+                     (const int*)block_tables, (const int*)context_lens,      \
+                     scale, num_kv_heads, page_size, max_pages, num_splits)
+  switch (qpg) {
+    case 1: LAUNCH_QPG(1); break;
+    case 2: LAUNCH_QPG(2); break;
+    case 4: LAUNCH_QPG(4); break;
+    case 8: LAUNCH_QPG(8); break;
+    default: return;   // validated host-side
+  }
+#undef LAUNCH_QPG
+  if (num_splits > 1) {
+    dim3 cgrid(num_seqs, num_q_heads), cblock(HEAD_DIM);
+    hipLaunchKernelGGL(decode_combine_kernel, cgrid, cblock, 0, stream,
+                       (__hip_bfloat16*)out, (const float*)partial_o,
+                       (const float*)partial_ml, num_splits, num_q_heads);
+  }
+}
+
+}  // extern "C"

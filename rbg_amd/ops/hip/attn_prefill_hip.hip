@@ -1,0 +1,220 @@
+#include "hip/hip_runtime.h"
+// Varlen causal flash-attention prefill (gfx950), MFMA-tiled.
+//
+// Structure (guide §5/Appendix B "fused attention prefill"): one workgroup =
+// 64 query rows x one q-head; 4 waves, 16 q-rows each; K/V tiles of 64 keys
+// staged in LDS (K row-major, V transposed for the PV B-operand), rows
+// padded by 8 bf16 (16 B) so the 16-lane ds_read_b128 groups are
+// bank-conflict-free; QK^T and PV on v_mfma_f32_16x16x32_bf16; online
+// softmax with the running (m, l) per q-row kept in registers, row
+// reductions via 16-lane xor shuffles (no serial-lane softmax —
+// guide common-mistake #6).
+//
+// v1 is the correctness-first instance of the guide's ladder; the staged
+// upgrades (XOR-swizzle + tr_read V, 8-wave 32x32 structure, async-STAGE)
+// are applied in later rounds against rocprof evidence.
+//
+// Capability analog: prefill attention of the engines the reference
+// orchestrates (SURVEY §2.3 prefill engine row).
+#include "common.h"
+
+namespace {
+
+constexpr int HEAD_DIM = 128;
+constexpr int QTILE = 64;      // q rows per block (16 per wave)
+constexpr int KTILE = 64;      // keys per LDS tile
+constexpr int PAD = 8;         // bf16 row padding (16 B) — bank-conflict fix
+constexpr float NEG_INF = -1e30f;
+
+typedef __attribute__((ext_vector_type(8))) __bf16 mfma_bf8;
+typedef __attribute__((ext_vector_type(4))) float mfma_f4;
+
+DEV_INLINE mfma_bf8 load_bf8(const __hip_bfloat16* p) {
+  union { uint4 u; mfma_bf8 v; } cvt;
+  cvt.u = *reinterpret_cast<const uint4*>(p);
+  return cvt.v;
+}
+
+DEV_INLINE mfma_bf8 lds_bf8(const __hip_bfloat16* p) {
+  union { uint4 u; mfma_bf8 v; } cvt;
+  cvt.u = *reinterpret_cast<const uint4*>(p);
+  return cvt.v;
+}
+
+// grid.x = total q-blocks (host-computed map), grid.y = num q heads.
+__global__ __launch_bounds__(256) void prefill_attn_kernel(
+    __hip_bfloat16* __restrict__ out,        // [T, QH, D]
+    const __hip_bfloat16* __restrict__ q,    // [T, QH, D]
+    const __hip_bfloat16* __restrict__ k,    // [T, KVH, D]
+    const __hip_bfloat16* __restrict__ v,    // [T, KVH, D]
+    const int* __restrict__ block_info,      // [nblocks, 2] = (seq_start_row, q_block)
+    const int* __restrict__ seq_lens,        // [nblocks] length of this block's seq
+    const float scale, const int num_q_heads, const int num_kv_heads) {
+  const int qh = blockIdx.y;
+  const int kvh = qh / (num_q_heads / num_kv_heads);
+  const int seq_start = block_info[blockIdx.x * 2];
+  const int qblock = block_info[blockIdx.x * 2 + 1];
+  const int seq_len = seq_lens[blockIdx.x];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int gl = lane & 15;          // fragment row/col lane
+  const int gslice = lane >> 4;      // fragment k-slice (x8 elements)
+
+  __shared__ __hip_bfloat16 k_lds[KTILE][HEAD_DIM + PAD];
+  __shared__ __hip_bfloat16 v_lds[HEAD_DIM][KTILE + PAD];   // transposed
+  __shared__ __hip_bfloat16 p_lds[4][16][KTILE + PAD];
+
+  // ---- Q fragments: wave's 16 rows, 4 k-steps of 32 dims ---------------
+  const int q_row_in_seq = qblock * QTILE + wave * 16 + gl;
+  const bool q_valid = q_row_in_seq < seq_len;
+  const int q_row = seq_start + min(q_row_in_seq, seq_len - 1);
+  mfma_bf8 q_frag[4];
+#pragma unroll
+  for (int ks = 0; ks < 4; ++ks)
+    q_frag[ks] = load_bf8(q + ((size_t)q_row * num_q_heads + qh) * HEAD_DIM +
+                          ks * 32 + gslice * 8);
+
+  // per-lane softmax state: 4 q-rows (rows 4*gslice + r of the wave tile)
+  float m_run[4], l_run[4];
+  mfma_f4 acc_o[8];    // O[16 x 128]: 8 col-tiles of 16 dims
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = NEG_INF; l_run[r] = 0.f; }
+#pragma unroll
+  for (int dt = 0; dt < 8; ++dt) acc_o[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  // causal: this block needs keys up to its last valid q row
+  const int block_q_max = min(qblock * QTILE + QTILE - 1, seq_len - 1);
+  const int wave_q_max = min(qblock * QTILE + wave * 16 + 15, seq_len - 1);
+  const int nktiles = block_q_max / KTILE + 1;
+
+  for (int kt = 0; kt < nktiles; ++kt) {
+    const int k_base = kt * KTILE;
+    // ---- stage K tile: 64 rows x 128 dims, 16 B per thread-iteration ----
+    __syncthreads();
+    for (int i = tid; i < KTILE * (HEAD_DIM / 8); i += 256) {
+      const int key = i >> 4;          // 16 chunks of 8 dims per key
+      const int chunk = (i & 15) * 8;
+      const int krow = k_base + key;
+      uint4 val = {0, 0, 0, 0};
+      if (krow < seq_len)
+        val = *reinterpret_cast<const uint4*>(
+            k + ((size_t)(seq_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
+            chunk);
+      *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = val;
+      // V transposed: scatter the 8 dims as single bf16 stores
+      uint4 vv = {0, 0, 0, 0};
+      if (krow < seq_len)
+        vv = *reinterpret_cast<const uint4*>(
+            v + ((size_t)(seq_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
+            chunk);
+      const __hip_bfloat16* ve = reinterpret_cast<const __hip_bfloat16*>(&vv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v_lds[chunk + j][key] = ve[j];
+    }
+    __syncthreads();
+
+    if (k_base > wave_q_max) continue;   // fully-masked for this wave
+
+    // ---- S = Q K^T : 4 col-tiles x 4 k-steps of MFMA --------------------
+    mfma_f4 s[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+      s[ct] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        mfma_bf8 kf = lds_bf8(&k_lds[ct * 16 + gl][ks * 32 + gslice * 8]);
+        s[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, s[ct],
+                                                        0, 0, 0);
+      }
+    }
+
+    // ---- online softmax (rows 4*gslice+r, key col = ct*16+gl) -----------
+    float p[4][4];
+    float m_new[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q_pos = qblock * QTILE + wave * 16 + 4 * gslice + r;
+      float row_max = NEG_INF;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const int k_pos = k_base + ct * 16 + gl;
+        float sv = s[ct][r] * scale;
+        if (k_pos > q_pos || k_pos >= seq_len || q_pos >= seq_len)
+          sv = NEG_INF;
+        p[ct][r] = sv;
+        row_max = fmaxf(row_max, sv);
+      }
+      row_max = group16_max(row_max);
+      m_new[r] = fmaxf(m_run[r], row_max);
+      const float alpha = __expf(m_run[r] - m_new[r]);
+      float psum = 0.f;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        p[ct][r] = (p[ct][r] <= NEG_INF) ? 0.f : __expf(p[ct][r] - m_new[r]);
+        psum += p[ct][r];
+      }
+      l_run[r] = l_run[r] * alpha + psum;
+      m_run[r] = m_new[r];
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) acc_o[dt][r] *= alpha;
+    }
+
+    // ---- P -> LDS (transpose to the A-fragment layout) ------------------
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct)
+        p_lds[wave][4 * gslice + r][ct * 16 + gl] = f2bf(p[ct][r]);
+    // same-wave LDS read-after-write across lanes: drain the DS queue
+    // explicitly (no cross-wave sharing, so no barrier needed)
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- O += P V : 8 col-tiles x 2 key-steps ---------------------------
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      mfma_bf8 pa = lds_bf8(&p_lds[wave][gl][ks * 32 + gslice * 8]);
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) {
+        mfma_bf8 vf = lds_bf8(&v_lds[dt * 16 + gl][ks * 32 + gslice * 8]);
+        acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf, acc_o[dt],
+                                                            0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: normalize and store ------------------------------------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) l_run[r] = group16_sum(l_run[r]);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int q_pos = qblock * QTILE + wave * 16 + 4 * gslice + r;
+    if (q_pos >= seq_len) continue;
+    const float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+    __hip_bfloat16* orow =
+        out + ((size_t)(seq_start + q_pos) * num_q_heads + qh) * HEAD_DIM;
+#pragma unroll
+    for (int dt = 0; dt < 8; ++dt)
+      orow[dt * 16 + gl] = f2bf(acc_o[dt][r] * inv_l);
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void launch_prefill_attention(void* out, const void* q, const void* k,
+                              const void* v, const void* block_info,
+                              const void* seq_lens, float scale, int nblocks,
+                              int num_q_heads, int num_kv_heads,
+                              hipStream_t stream) {
+  dim3 grid(nblocks, num_q_heads), block(256);
+  hipLaunchKernelGGL(prefill_attn_kernel, grid, block, 0, stream,
+                     (__hip_bfloat16*)out, (const __hip_bfloat16*)q,
+                     (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,
+                     (const int*)block_info, (const int*)seq_lens, scale,
+                     num_q_heads, num_kv_heads);
+}
+
+}  // extern "C"

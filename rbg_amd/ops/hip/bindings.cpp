@@ -1,0 +1,170 @@
+// PyTorch bindings for the rbg_amd CDNA4 HIP kernels.
+// Thin shape/dtype checks here; all math lives in the .hip translation units.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void launch_rmsnorm(void*, void*, const void*, float, int, int, hipStream_t);
+void launch_fused_add_rmsnorm(void*, void*, const void*, float, int, int,
+                              hipStream_t);
+void launch_silu_mul(void*, const void*, int, int, hipStream_t);
+void launch_rope_store_kv(void*, void*, const void*, void*, void*,
+                          const void*, const void*, const void*, int, int,
+                          int, int, int, hipStream_t);
+void launch_decode_attention(void*, void*, void*, const void*, const void*,
+                             const void*, const void*, const void*, float,
+                             int, int, int, int, int, int, hipStream_t);
+void launch_prefill_attention(void*, const void*, const void*, const void*,
+                              const void*, const void*, float, int, int, int,
+                              hipStream_t);
+void launch_mfma_probe(void*, const void*, const void*, int, int, hipStream_t);
+}
+
+namespace {
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor weight, double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(weight, "weight");
+  const int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  const int tokens = x.numel() / hidden;
+  auto out = torch::empty_like(x);
+  launch_rmsnorm(out.data_ptr(), x.data_ptr(), weight.data_ptr(), (float)eps,
+                 tokens, hidden, current_stream());
+  return out;
+}
+
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor weight, double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(residual, "residual");
+  const int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  const int tokens = x.numel() / hidden;
+  launch_fused_add_rmsnorm(x.data_ptr(), residual.data_ptr(),
+                           weight.data_ptr(), (float)eps, tokens, hidden,
+                           current_stream());
+}
+
+torch::Tensor silu_mul(torch::Tensor x) {
+  check_bf16_contig(x, "x");
+  const int inter2 = x.size(-1);
+  TORCH_CHECK(inter2 % 16 == 0, "2*inter must be a multiple of 16");
+  const int inter = inter2 / 2;
+  const int tokens = x.numel() / inter2;
+  auto out = torch::empty({x.size(0), inter}, x.options());
+  launch_silu_mul(out.data_ptr(), x.data_ptr(), tokens, inter,
+                  current_stream());
+  return out;
+}
+
+void rope_store_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                   torch::Tensor key_cache, torch::Tensor value_cache,
+                   torch::Tensor cos_sin, torch::Tensor positions,
+                   torch::Tensor slot_mapping) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32, "cos_sin fp32");
+  TORCH_CHECK(positions.scalar_type() == torch::kInt32, "positions int32");
+  TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt32, "slots int32");
+  const int tokens = q.size(0);
+  const int head_dim = key_cache.size(3);
+  const int num_kv_heads = key_cache.size(1);
+  const int page_size = key_cache.size(2);
+  const int num_q_heads = q.numel() / tokens / head_dim;
+  launch_rope_store_kv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       key_cache.data_ptr(), value_cache.data_ptr(),
+                       cos_sin.data_ptr(), positions.data_ptr(),
+                       slot_mapping.data_ptr(), tokens, num_q_heads,
+                       num_kv_heads, head_dim, page_size, current_stream());
+}
+
+torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
+                               torch::Tensor value_cache,
+                               torch::Tensor block_tables,
+                               torch::Tensor context_lens, double scale,
+                               int64_t num_splits) {
+  check_bf16_contig(q, "q");
+  const int num_seqs = q.size(0);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  TORCH_CHECK(head_dim == 128, "head_dim must be 128");
+  const int num_kv_heads = key_cache.size(1);
+  const int page_size = key_cache.size(2);
+  const int max_pages = block_tables.size(1);
+  const int qpg = num_q_heads / num_kv_heads;
+  TORCH_CHECK(qpg == 1 || qpg == 2 || qpg == 4 || qpg == 8,
+              "GQA group must be 1/2/4/8");
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32, "tables int32");
+  TORCH_CHECK(context_lens.scalar_type() == torch::kInt32, "lens int32");
+  auto out = torch::empty_like(q);
+  torch::Tensor partial_o, partial_ml;
+  void *po = nullptr, *pml = nullptr;
+  if (num_splits > 1) {
+    auto opts = q.options().dtype(torch::kFloat32);
+    partial_o = torch::empty({num_splits, num_seqs, num_q_heads, head_dim}, opts);
+    partial_ml = torch::empty({num_splits, num_seqs, num_q_heads, 2}, opts);
+    po = partial_o.data_ptr();
+    pml = partial_ml.data_ptr();
+  }
+  launch_decode_attention(out.data_ptr(), po, pml, q.data_ptr(),
+                          key_cache.data_ptr(), value_cache.data_ptr(),
+                          block_tables.data_ptr(), context_lens.data_ptr(),
+                          (float)scale, num_seqs, num_q_heads, num_kv_heads,
+                          page_size, max_pages, (int)num_splits,
+                          current_stream());
+  return out;
+}
+
+torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
+                                torch::Tensor v, torch::Tensor block_info,
+                                torch::Tensor seq_lens, double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  TORCH_CHECK(q.size(2) == 128, "head_dim must be 128");
+  TORCH_CHECK(block_info.scalar_type() == torch::kInt32, "block_info int32");
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32, "seq_lens int32");
+  const int nblocks = block_info.size(0);
+  const int num_q_heads = q.size(1);
+  const int num_kv_heads = k.size(1);
+  auto out = torch::empty_like(q);
+  launch_prefill_attention(out.data_ptr(), q.data_ptr(), k.data_ptr(),
+                           v.data_ptr(), block_info.data_ptr(),
+                           seq_lens.data_ptr(), (float)scale, nblocks,
+                           num_q_heads, num_kv_heads, current_stream());
+  return out;
+}
+
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b, int64_t a_split,
+                         int64_t b_split) {
+  auto d = torch::zeros({16, 16}, a.options().dtype(torch::kFloat32));
+  launch_mfma_probe(d.data_ptr(), a.data_ptr(), b.data_ptr(), (int)a_split,
+                    (int)b_split, current_stream());
+  return d;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "x,residual += ; rmsnorm");
+  m.def("silu_mul", &silu_mul, "silu(gate)*up");
+  m.def("rope_store_kv", &rope_store_kv, "RoPE + paged KV write");
+  m.def("decode_attention", &decode_attention, "paged flash-decode");
+  m.def("prefill_attention", &prefill_attention, "varlen causal flash prefill");
+  m.def("mfma_probe", &mfma_probe, "MFMA layout probe");
+}

@@ -1,0 +1,95 @@
+"""Engine-on-CPU suite: tiny model, reference ops — continuous batching,
+paged KV bookkeeping, determinism (batched == sequential), live reload."""
+import pytest
+import torch
+
+from rbg_amd.engine.config import EngineConfig, ModelConfig
+from rbg_amd.engine.engine import LLMEngine
+from rbg_amd.engine.kv_cache import OutOfPages, PagedKVCache
+from rbg_amd.engine.sequence import SamplingParams
+
+
+def tiny_engine(**kw):
+    cfg = EngineConfig(model=ModelConfig.preset("tiny"), device="cpu",
+                       kv_pool_tokens=kw.pop("kv_pool_tokens", 4096),
+                       max_prefill_tokens=kw.pop("max_prefill_tokens", 512),
+                       enforce_eager=True, **kw)
+    return LLMEngine(cfg)
+
+
+def test_generate_greedy_batch():
+    eng = tiny_engine()
+    torch.manual_seed(0)
+    prompts = [torch.randint(0, 500, (n,)).tolist() for n in (7, 19, 33)]
+    seqs = eng.generate(prompts, SamplingParams(max_new_tokens=6))
+    for s in seqs:
+        assert len(s.output_tokens) == 6
+        assert s.status == "finished"
+    assert eng.stats.decode_tokens > 0
+    assert eng.runner.cache.free_pages == eng.runner.cache.num_pages - 1
+
+
+def test_batched_equals_sequential():
+    """Continuous batching must not change greedy outputs — validates paged
+    bookkeeping end to end."""
+    torch.manual_seed(0)
+    prompts = [torch.randint(0, 500, (n,)).tolist() for n in (11, 23)]
+    eng1 = tiny_engine()
+    batched = eng1.generate(prompts, SamplingParams(max_new_tokens=5))
+    outs_b = [s.output_tokens for s in batched]
+    eng2 = tiny_engine()
+    outs_s = []
+    for p in prompts:
+        (s,) = eng2.generate([p], SamplingParams(max_new_tokens=5))
+        outs_s.append(s.output_tokens)
+    assert outs_b == outs_s
+
+
+def test_prefill_budget_splits_batches():
+    eng = tiny_engine(max_prefill_tokens=32)
+    prompts = [[1] * 20, [2] * 20, [3] * 20]
+    seqs = eng.generate(prompts, SamplingParams(max_new_tokens=2))
+    assert all(len(s.output_tokens) == 2 for s in seqs)
+    assert eng.stats.prefill_steps >= 2  # 60 tokens vs budget 32
+
+
+def test_kv_pool_admission_gate():
+    # pool of 256 tokens (16 pages); a prompt needing more must wait forever
+    eng = tiny_engine(kv_pool_tokens=256)
+    big = [1] * 200
+    small = [2] * 10
+    s_big = eng.add_request(big, SamplingParams(max_new_tokens=100))
+    s_small = eng.add_request(small, SamplingParams(max_new_tokens=4))
+    for _ in range(50):
+        if not eng.scheduler.has_work():
+            break
+        eng.step()
+    # big cannot be admitted (200+100 > 240 free tokens); small is behind it
+    assert s_big.status == "waiting"
+
+
+def test_out_of_pages_raises():
+    cfg = EngineConfig(model=ModelConfig.preset("tiny"), device="cpu",
+                       kv_pool_tokens=64)
+    cache = PagedKVCache(cfg, torch.device("cpu"))
+    with pytest.raises(OutOfPages):
+        cache.alloc(cache.num_pages)   # page 0 reserved
+
+
+def test_live_weight_reload_changes_outputs():
+    eng = tiny_engine()
+    p = [list(range(12))]
+    (a,) = eng.generate(p, SamplingParams(max_new_tokens=4))
+    eng.reload_weights(seed=999)
+    (b,) = eng.generate(p, SamplingParams(max_new_tokens=4))
+    assert a.output_tokens != b.output_tokens  # weights actually changed
+    # KV pool survived the reload
+    assert eng.runner.cache.num_pages > 0
+
+
+def test_stats_snapshot():
+    eng = tiny_engine()
+    eng.generate([[1, 2, 3]], SamplingParams(max_new_tokens=3))
+    snap = eng.stats.snapshot()
+    assert snap["decode_tokens"] >= 2
+    assert snap["p50_ttft_s"] >= 0.0

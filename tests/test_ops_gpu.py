@@ -1,0 +1,155 @@
+"""GPU numerics: every HIP kernel vs its plain-PyTorch fp32 reference
+(tests run on a real MI355X via gpurun; marked gpu)."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+import rbg_amd.ops as ops
+from rbg_amd.ops import reference as ref
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    torch.manual_seed(1234)
+    return torch.device("cuda:0")
+
+
+def test_hip_extension_loaded(dev):
+    # on a GPU box the native extension must be the path that runs
+    assert ops.HAVE_HIP, f"HIP extension missing: {ops._hip_err!r}"
+
+
+def test_mfma_fragment_layout(dev):
+    """Empirically identify the v_mfma_f32_16x16x32_bf16 operand layout
+    (guide §3: asymmetric operands, transpose-detecting)."""
+    a = torch.randn(16, 32, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(32, 16, dtype=torch.bfloat16, device=dev)
+    want = a.float() @ b.float()
+    results = {}
+    for a_s in (0, 1):
+        for b_s in (0, 1):
+            d = ops._hip.mfma_probe(a.contiguous(), b.contiguous(), a_s, b_s)
+            results[(a_s, b_s)] = (d - want).abs().max().item()
+    best = min(results, key=results.get)
+    assert results[best] < 0.1, f"no layout matched: {results}"
+    # the kernels assume contiguous-8 per lane (0,0); fail loudly otherwise
+    assert best == (0, 0), (
+        f"MFMA layout is {best}, kernels assume (0,0); errors={results}")
+
+
+def test_rmsnorm(dev):
+    for T, H in ((1, 4096), (17, 4096), (256, 8192), (64, 1024)):
+        x = torch.randn(T, H, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(H, dtype=torch.bfloat16, device=dev)
+        got = ops.rmsnorm(x, w, 1e-5).float()
+        want = ref.rmsnorm(x, w, 1e-5).float()
+        assert torch.allclose(got, want, atol=2e-2, rtol=2e-2), \
+            (T, H, (got - want).abs().max().item())
+
+
+def test_fused_add_rmsnorm(dev):
+    T, H = 33, 4096
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev)
+    res = torch.randn(T, H, dtype=torch.bfloat16, device=dev)
+    w = torch.randn(H, dtype=torch.bfloat16, device=dev)
+    x2, res2 = x.clone(), res.clone()
+    ops.fused_add_rmsnorm(x, res, w, 1e-5)
+    ref.fused_add_rmsnorm(x2, res2, w, 1e-5)
+    assert torch.allclose(res.float(), res2.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(x.float(), x2.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_silu_mul(dev):
+    T, I = 129, 14336
+    x = torch.randn(T, 2 * I, dtype=torch.bfloat16, device=dev)
+    got = ops.silu_mul(x).float()
+    want = ref.silu_mul(x).float()
+    assert torch.allclose(got, want, atol=2e-2, rtol=2e-2)
+
+
+def _mk_cache(dev, pages=64, kvh=8, page=16, d=128):
+    kc = torch.zeros(pages, kvh, page, d, dtype=torch.bfloat16, device=dev)
+    vc = torch.zeros_like(kc)
+    return kc, vc
+
+
+def test_rope_store_kv(dev):
+    T, QH, KVH, D, page = 37, 32, 8, 128, 16
+    q = torch.randn(T, QH, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev)
+    cos_sin = ref.build_cos_sin_table(D, 4096, device=dev)
+    positions = torch.randint(0, 4096, (T,), dtype=torch.int32, device=dev)
+    slots = torch.randperm(64 * page, device=dev)[:T].to(torch.int32)
+    kc, vc = _mk_cache(dev)
+    kc2, vc2 = _mk_cache(dev)
+    q2, k2, v2 = q.clone(), k.clone(), v.clone()
+    ops.rope_store_kv(q, k, v, kc, vc, cos_sin, positions, slots)
+    ref.rope_store_kv(q2, k2, v2, kc2, vc2, cos_sin, positions, slots)
+    assert torch.allclose(q.float(), q2.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(k.float(), k2.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(kc.float(), kc2.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(vc.float(), vc2.float(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("qpg,splits", [(4, 1), (4, 4), (8, 1), (1, 2)])
+def test_decode_attention(dev, qpg, splits):
+    KVH, D, page = 8, 128, 16
+    QH = KVH * qpg
+    S = 5
+    ctx_lens = torch.tensor([1, 16, 57, 300, 777], dtype=torch.int32,
+                            device=dev)
+    max_pages = (int(ctx_lens.max()) + page - 1) // page
+    total_pages = int((torch.div(ctx_lens + page - 1, page,
+                                 rounding_mode="floor")).sum())
+    kc = torch.randn(total_pages + 1, KVH, page, D, dtype=torch.bfloat16,
+                     device=dev)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(S, max_pages, dtype=torch.int32, device=dev)
+    nxt = 0
+    for s in range(S):
+        n = (int(ctx_lens[s]) + page - 1) // page
+        bt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32, device=dev)
+        nxt += n
+    q = torch.randn(S, QH, D, dtype=torch.bfloat16, device=dev)
+    scale = 1.0 / math.sqrt(D)
+    got = ops._hip.decode_attention(q, kc, vc, bt, ctx_lens, scale, splits)
+    want = ref.decode_attention(q, kc, vc, bt, ctx_lens, scale)
+    err = (got.float() - want.float()).abs().max().item()
+    assert err < 3e-2, f"qpg={qpg} splits={splits} err={err}"
+
+
+@pytest.mark.parametrize("lens", [[128], [64, 200, 1], [2048], [33, 129]])
+def test_prefill_attention(dev, lens):
+    QH, KVH, D = 32, 8, 128
+    T = sum(lens)
+    q = torch.randn(T, QH, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=dev)
+    scale = 1.0 / math.sqrt(D)
+    got = ops.prefill_attention(q, k, v, cu, scale)
+    want = ref.prefill_attention(q, k, v, cu, scale)
+    err = (got.float() - want.float()).abs().max().item()
+    assert err < 3e-2, f"lens={lens} err={err}"
+
+
+def test_prefill_attention_spiked_scores(dev):
+    """Force large per-tile max jumps (rule 26: exercise the rescale path)."""
+    QH, KVH, D = 4, 4, 128
+    L = 512
+    q = torch.randn(L, QH, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(L, KVH, D, dtype=torch.bfloat16, device=dev)
+    # spike one late key against everything
+    k[400] *= 30.0
+    v = torch.randn(L, KVH, D, dtype=torch.bfloat16, device=dev)
+    cu = torch.tensor([0, L], dtype=torch.int32, device=dev)
+    got = ops.prefill_attention(q, k, v, cu, 1.0 / math.sqrt(D))
+    want = ref.prefill_attention(q, k, v, cu, 1.0 / math.sqrt(D))
+    err = (got.float() - want.float()).abs().max().item()
+    assert err < 5e-2, f"spiked err={err}"
