@@ -302,6 +302,7 @@ class WorkerStatus:
     component_index: int = 0
     pid: int = 0
     gpu_ids: List[int] = field(default_factory=list)
+    ports: List[int] = field(default_factory=list)   # worker-reported (rpc/http)
     phase: str = "Pending"        # Pending | Running | Ready | Failed | Succeeded
     restart_count: int = 0
     last_heartbeat: float = 0.0

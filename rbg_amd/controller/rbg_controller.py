@@ -302,9 +302,10 @@ class RoleBasedGroupController:
                 ready = c is not None and c.status == "True"
                 gpu_ids = sorted({g for w in inst.status.workers
                                   for g in w.gpu_ids})
+                ports = [p for w in inst.status.workers for p in w.ports]
                 items.append({"name": inst.metadata.name,
                               "address": "127.0.0.1",
-                              "ports": role.service_ports,
+                              "ports": ports or role.service_ports,
                               "gpu_ids": gpu_ids, "ready": ready})
             items.sort(key=lambda i: i["name"])
             instances[role.name] = items

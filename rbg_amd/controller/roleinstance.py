@@ -305,11 +305,17 @@ class RoleInstanceController:
             phase = h.phase() if h else "Pending"
             if phase == "Ready":
                 ready += 1
+            ports: List[int] = []
+            if h is not None:
+                st = h.read_status()
+                for key in ("rpc_port", "http_port"):
+                    if st.get(key):
+                        ports.append(int(st[key]))
             workers.append(WorkerStatus(
                 name=wname, component=comp.name, component_index=j,
                 pid=h.pid if h else 0,
                 gpu_ids=rt.gpu_by_worker.get(wname, []),
-                phase=phase,
+                ports=ports, phase=phase,
                 last_heartbeat=time.time() - h.heartbeat_age() if h else 0.0))
         all_ready = ready == len(desired) and len(desired) > 0
 

@@ -150,8 +150,12 @@ def main(argv: Optional[List[str]] = None) -> int:
     hb.start()
     write_status(run_dir, name, "Running", pid=os.getpid())
 
-    _load_entry_runners()
-    runner = RUNNERS.get(opts.runner)
+    # when executed as `python -m rbg_amd.runtime.worker` this file is the
+    # __main__ module; runners register on the canonical import, so consult
+    # that one (not this module's RUNNERS copy)
+    import rbg_amd.runtime.worker as canonical
+    canonical._load_entry_runners()
+    runner = canonical.RUNNERS.get(opts.runner)
     if runner is None:
         write_status(run_dir, name, "Failed",
                      error=f"unknown runner {opts.runner!r}")

@@ -1,0 +1,205 @@
+"""Router role — HTTP front-end + prefill/decode dispatcher.
+
+The in-repo realization of the reference's router (SMG) role
+(reference examples/inference/pd-disagg-standalone.yaml:55-66): terminates
+an OpenAI-style HTTP API, picks a prefill and a decode instance from the
+discovery topology, drives the migration handshake and streams the decode
+side's tokens back.  In colocated mode it simply load-balances generate
+calls round-robin across worker instances.
+
+No external web framework: stdlib ThreadingHTTPServer keeps the worker
+process dependency-free and fork-fast.
+"""
+from __future__ import annotations
+
+import itertools
+import json
+import logging
+import os
+import threading
+import time
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from typing import Any, Dict, List, Optional
+
+from ..runtime.worker import WorkerContext, register_runner
+from .rpc import RpcClient
+
+log = logging.getLogger(__name__)
+
+
+def simple_tokenize(text: str, vocab_size: int = 128000) -> List[int]:
+    """Deterministic byte-pair-free stand-in tokenizer (no network for real
+    tokenizer files; weights are random-init anyway)."""
+    return [(b * 7 + i) % (vocab_size - 10) + 1
+            for i, b in enumerate(text.encode())]
+
+
+class Router:
+    def __init__(self, ctx: WorkerContext):
+        self.ctx = ctx
+        args = ctx.args
+        self.mode = args.get("dispatch", "colocated")   # colocated | pd
+        self.vocab_size = int(args.get("vocab_size", 128000))
+        self.worker_roles = args.get("worker_roles", ["worker"])
+        self.prefill_roles = args.get("prefill_roles", ["prefill"])
+        self.decode_roles = args.get("decode_roles", ["decode"])
+        self._clients: Dict[str, RpcClient] = {}
+        self._rr = itertools.count()
+        self._lock = threading.Lock()
+
+    # -- topology ------------------------------------------------------------
+
+    def _instances(self, roles: List[str]) -> List[Dict[str, Any]]:
+        topo = self.ctx.load_topology().get("group", {})
+        out = []
+        for role in topo.get("roles", []):
+            if role.get("name") in roles:
+                for inst in role.get("instances", []):
+                    if inst.get("ready") and inst.get("ports"):
+                        out.append(inst)
+        return out
+
+    def _client(self, inst: Dict[str, Any]) -> RpcClient:
+        key = inst["name"]
+        with self._lock:
+            c = self._clients.get(key)
+            if c is None:
+                c = RpcClient(inst.get("address", "127.0.0.1"),
+                              int(inst["ports"][0]))
+                self._clients[key] = c
+            return c
+
+    def _pick_role(self, roles: List[str], timeout: float = 15.0
+                   ) -> Dict[str, Any]:
+        """Round-robin over ready instances; brief retry covers the window
+        between role readiness and the next discovery publish."""
+        deadline = time.monotonic() + timeout
+        while True:
+            insts = self._instances(roles)
+            if insts:
+                return insts[next(self._rr) % len(insts)]
+            if time.monotonic() >= deadline:
+                raise RuntimeError(f"no ready instances for roles {roles}")
+            time.sleep(0.1)
+
+    # -- request paths -------------------------------------------------------
+
+    def generate(self, tokens: List[int], max_new_tokens: int,
+                 temperature: float = 0.0) -> Dict[str, Any]:
+        t0 = time.monotonic()
+        if self.mode == "pd":
+            return self._generate_pd(tokens, max_new_tokens, temperature, t0)
+        inst = self._pick_role(self.worker_roles)
+        res = self._client(inst).call("generate", tokens=tokens,
+                                      max_new_tokens=max_new_tokens,
+                                      temperature=temperature)
+        res["wall_s"] = time.monotonic() - t0
+        res["instance"] = inst["name"]
+        return res
+
+    def _generate_pd(self, tokens: List[int], max_new_tokens: int,
+                     temperature: float, t0: float) -> Dict[str, Any]:
+        prefill = self._pick_role(self.prefill_roles)
+        decode = self._pick_role(self.decode_roles)
+        pres = self._client(prefill).call(
+            "prefill", tokens=tokens, max_new_tokens=max_new_tokens,
+            decode_instance=decode["name"], temperature=temperature)
+        ttft = time.monotonic() - t0
+        seq_id = pres["decode_seq_id"]
+        dclient = self._client(decode)
+        while True:
+            res = dclient.call("poll", seq_id=seq_id)
+            if res["finished"]:
+                res["ttft_s"] = ttft
+                res["wall_s"] = time.monotonic() - t0
+                res["prefill_instance"] = prefill["name"]
+                res["decode_instance"] = decode["name"]
+                return res
+            time.sleep(0.005)
+
+    def stats(self) -> Dict[str, Any]:
+        out: Dict[str, Any] = {"mode": self.mode, "instances": {}}
+        for role_list in (self.worker_roles, self.prefill_roles,
+                          self.decode_roles):
+            for inst in self._instances(role_list):
+                try:
+                    out["instances"][inst["name"]] = \
+                        self._client(inst).call("stats")
+                except Exception as e:  # noqa: BLE001
+                    out["instances"][inst["name"]] = {"error": repr(e)}
+        return out
+
+
+class _Handler(BaseHTTPRequestHandler):
+    router: Router = None   # set by serve()
+
+    def log_message(self, *a):  # quiet
+        pass
+
+    def _reply(self, code: int, obj: Any) -> None:
+        data = json.dumps(obj).encode()
+        self.send_response(code)
+        self.send_header("Content-Type", "application/json")
+        self.send_header("Content-Length", str(len(data)))
+        self.end_headers()
+        self.wfile.write(data)
+
+    def do_GET(self):
+        if self.path in ("/health", "/healthz"):
+            self._reply(200, {"status": "ok"})
+        elif self.path == "/metrics":
+            self._reply(200, self.router.stats())
+        else:
+            self._reply(404, {"error": "not found"})
+
+    def do_POST(self):
+        try:
+            n = int(self.headers.get("Content-Length", 0))
+            body = json.loads(self.rfile.read(n) or b"{}")
+            if self.path in ("/generate", "/v1/completions"):
+                tokens = body.get("prompt_tokens")
+                if tokens is None:
+                    tokens = simple_tokenize(str(body.get("prompt", "")),
+                                             self.router.vocab_size)
+                res = self.router.generate(
+                    tokens,
+                    int(body.get("max_tokens", body.get("max_new_tokens", 64))),
+                    float(body.get("temperature", 0.0)))
+                if self.path == "/v1/completions":
+                    res = {
+                        "id": f"cmpl-{int(time.time()*1000)}",
+                        "object": "text_completion",
+                        "model": "rbg-mi355x",
+                        "choices": [{"index": 0,
+                                     "tokens": res["tokens"],
+                                     "finish_reason": "length"}],
+                        "usage": {"prompt_tokens": len(tokens),
+                                  "completion_tokens": len(res["tokens"])},
+                        "timing": {"ttft_s": res.get("ttft_s"),
+                                   "wall_s": res.get("wall_s")},
+                    }
+                self._reply(200, res)
+            else:
+                self._reply(404, {"error": "not found"})
+        except Exception as e:  # noqa: BLE001
+            self._reply(500, {"error": repr(e)})
+
+
+@register_runner("router")
+def router_runner(ctx: WorkerContext) -> None:
+    router = Router(ctx)
+    port = 0
+    raw = os.environ.get("PORT_HTTP", "")
+    if raw:
+        port = int(raw.split(",")[0])
+    handler = type("BoundHandler", (_Handler,), {"router": router})
+    httpd = ThreadingHTTPServer(("127.0.0.1", port), handler)
+    port = httpd.server_address[1]
+    thread = threading.Thread(target=httpd.serve_forever, daemon=True)
+    thread.start()
+    ctx.set_ready(http_port=port, mode=router.mode)
+    try:
+        while not ctx.should_stop():
+            ctx.wait(0.2)
+    finally:
+        httpd.shutdown()
