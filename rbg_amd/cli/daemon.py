@@ -1,0 +1,112 @@
+"""rbgd — the node daemon (manager entrypoint analog).
+
+Mirrors reference cmd/rbgs/main.go: parses flags, starts the controller
+manager over the in-memory store, and exposes the store verbs + health over
+the RPC socket so rbgctl and external tools can drive it.
+
+Run: python -m rbg_amd.cli.daemon [--port 7471] [--run-root DIR] [--gpus N]
+"""
+from __future__ import annotations
+
+import argparse
+import logging
+import signal
+import sys
+import threading
+from typing import Any, Dict, List, Optional
+
+from ..api import constants as C
+from ..api.serde import asdict
+from ..api.types import load_object
+from ..controller.manager import Manager, ManagerOptions
+from ..server.rpc import RpcServer
+
+log = logging.getLogger(__name__)
+
+DEFAULT_PORT = 7471
+
+
+class Daemon:
+    def __init__(self, opts: ManagerOptions, port: int = DEFAULT_PORT):
+        self.manager = Manager(opts)
+        self.rpc = RpcServer(port=port)
+        self.port = self.rpc.port
+        self._register()
+
+    def _register(self) -> None:
+        store = self.manager.store
+        reg = self.rpc.register
+        reg("ping", lambda: "pong")
+        reg("healthz", lambda: {"status": "ok",
+                                "gpus": self.manager.topo.num_gpus,
+                                "free_gpus": self.manager.gang.free_gpus()})
+        reg("store_create", lambda obj: asdict(store.create(load_object(obj))))
+
+        def store_get(kind: str, name: str, namespace: str = "default"):
+            got = store.try_get(kind, name, namespace)
+            return asdict(got) if got else None
+        reg("store_get", store_get)
+
+        def store_list(kind: str, namespace: Optional[str] = "default",
+                       selector: Optional[Dict[str, str]] = None):
+            return [asdict(o) for o in store.list(kind, namespace, selector)]
+        reg("store_list", store_list)
+
+        def store_update(obj: Dict[str, Any]):
+            typed = load_object(obj)
+            cur = store.try_get(typed.kind, typed.metadata.name,
+                                typed.metadata.namespace)
+            if cur is not None:
+                typed.metadata.resource_version = cur.metadata.resource_version
+            return asdict(store.update(typed))
+        reg("store_update", store_update)
+
+        def store_delete(kind: str, name: str, namespace: str = "default"):
+            # graceful: mark for deletion so controllers tear down processes
+            obj = store.try_get(kind, name, namespace)
+            if obj is None:
+                return False
+            import time as _t
+            obj.metadata.deletion_timestamp = _t.time()
+            store.update(obj)
+            return True
+        reg("store_delete", store_delete)
+
+    def run_forever(self) -> None:
+        self.manager.start()
+        self.rpc.start()
+        log.info("rbgd listening on 127.0.0.1:%d (%d GPUs)",
+                 self.port, self.manager.topo.num_gpus)
+        stop = threading.Event()
+
+        def on_sig(sig, frame):
+            stop.set()
+        signal.signal(signal.SIGTERM, on_sig)
+        signal.signal(signal.SIGINT, on_sig)
+        stop.wait()
+        self.rpc.stop()
+        self.manager.stop()
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    ap = argparse.ArgumentParser(prog="rbgd")
+    ap.add_argument("--port", type=int, default=DEFAULT_PORT)
+    ap.add_argument("--run-root", default="/tmp/rbg-run")
+    ap.add_argument("--gpus", type=int, default=0, help="0 = discover")
+    ap.add_argument("--gang-timeout", type=float, default=30.0)
+    ap.add_argument("--history-limit", type=int, default=10)
+    ap.add_argument("--log-level", default="info")
+    args = ap.parse_args(argv)
+    logging.basicConfig(
+        level=getattr(logging, args.log_level.upper(), logging.INFO),
+        format="%(asctime)s %(name)s %(levelname)s %(message)s")
+    daemon = Daemon(ManagerOptions(
+        run_root=args.run_root, num_gpus=args.gpus,
+        gang_timeout=args.gang_timeout, history_limit=args.history_limit),
+        port=args.port)
+    daemon.run_forever()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,225 @@
+"""rbgctl — the kubectl-rbg plugin analog.
+
+Commands (reference cmd/cli/cmd/{root,status,rollout}):
+  apply -f FILE              create or update objects from YAML
+  get KIND [NAME]            list / show objects
+  status RBG                 role/worker table with readiness
+  scale ADAPTER --replicas N drive a scaling adapter
+  delete KIND NAME
+  rollout history RBG        ControllerRevision list
+  rollout diff RBG [--to N]  diff current spec vs a revision
+  rollout undo RBG [--to N]  restore a previous revision's spec
+
+Talks to a running rbgd daemon (--port) via the RemoteClient.
+"""
+from __future__ import annotations
+
+import argparse
+import difflib
+import json
+import sys
+from typing import List, Optional
+
+import yaml
+
+from ..api import constants as C
+from ..api.serde import asdict, fromdict
+from ..api.types import RoleBasedGroupSpec, load_object
+from ..client.client import BaseClient, RemoteClient
+
+
+def _print_table(rows: List[List[str]], header: List[str]) -> None:
+    widths = [max(len(str(r[i])) for r in [header] + rows)
+              for i in range(len(header))]
+    fmt = "  ".join("{:<%d}" % w for w in widths)
+    print(fmt.format(*header))
+    for r in rows:
+        print(fmt.format(*[str(c) for c in r]))
+
+
+def cmd_apply(client: BaseClient, args) -> int:
+    with open(args.file) as f:
+        docs = [d for d in yaml.safe_load_all(f) if d]
+    for doc in docs:
+        obj = load_object(doc)
+        existing = client.get(obj.kind, obj.metadata.name,
+                              obj.metadata.namespace)
+        if existing is None:
+            client.create(obj)
+            print(f"{obj.kind}/{obj.metadata.name} created")
+        else:
+            existing.spec = obj.spec
+            existing.metadata.labels = obj.metadata.labels
+            existing.metadata.annotations = obj.metadata.annotations
+            client.update(existing)
+            print(f"{obj.kind}/{obj.metadata.name} configured")
+    return 0
+
+
+def cmd_get(client: BaseClient, args) -> int:
+    kind = _resolve_kind(args.kind)
+    if args.name:
+        obj = client.get(kind, args.name, args.namespace)
+        if obj is None:
+            print(f"{kind}/{args.name} not found", file=sys.stderr)
+            return 1
+        print(yaml.safe_dump(asdict(obj), sort_keys=False))
+        return 0
+    rows = []
+    for obj in client.list(kind, args.namespace):
+        ready = ""
+        for cond in getattr(obj.status, "conditions", []):
+            if cond.type == C.COND_READY:
+                ready = cond.status
+        rows.append([obj.metadata.name, ready,
+                     obj.metadata.resource_version])
+    _print_table(rows, ["NAME", "READY", "RV"])
+    return 0
+
+
+def cmd_status(client: BaseClient, args) -> int:
+    rbg = client.get(C.KIND_RBG, args.name, args.namespace)
+    if rbg is None:
+        print(f"rolebasedgroup {args.name} not found", file=sys.stderr)
+        return 1
+    print(f"RoleBasedGroup: {rbg.metadata.name}")
+    for cond in rbg.status.conditions:
+        print(f"  {cond.type}: {cond.status} ({cond.reason}) {cond.message}")
+    rows = []
+    for rs in rbg.status.role_statuses:
+        role = rbg.spec.role(rs.name)
+        rows.append([rs.name, f"{rs.ready_replicas}/{role.replicas if role else '?'}",
+                     rs.updated_replicas])
+    print()
+    _print_table(rows, ["ROLE", "READY", "UPDATED"])
+    insts = client.list(C.KIND_ROLE_INSTANCE, args.namespace,
+                        selector={C.LABEL_GROUP_NAME: args.name})
+    rows = []
+    for inst in sorted(insts, key=lambda i: i.metadata.name):
+        for w in inst.status.workers:
+            rows.append([inst.metadata.name, w.name, w.phase, w.pid,
+                         ",".join(map(str, w.gpu_ids)) or "-",
+                         inst.status.restart_count])
+    print()
+    _print_table(rows, ["INSTANCE", "WORKER", "PHASE", "PID", "GPUS",
+                        "RESTARTS"])
+    return 0
+
+
+def cmd_scale(client: BaseClient, args) -> int:
+    client.scale(args.name, args.replicas, args.namespace)
+    print(f"scalingadapter/{args.name} scaled to {args.replicas}")
+    return 0
+
+
+def cmd_delete(client: BaseClient, args) -> int:
+    ok = client.delete(_resolve_kind(args.kind), args.name, args.namespace)
+    print(f"{args.kind}/{args.name} " + ("deleted" if ok else "not found"))
+    return 0 if ok else 1
+
+
+def _spec_yaml(data: dict) -> str:
+    return yaml.safe_dump(data, sort_keys=True)
+
+
+def cmd_rollout(client: BaseClient, args) -> int:
+    revs = client.revisions(args.name, args.namespace)
+    if args.action == "history":
+        _print_table([[r.revision, r.metadata.name,
+                       r.metadata.labels.get(C.LABEL_REVISION_HASH, "")]
+                      for r in revs], ["REVISION", "NAME", "HASH"])
+        return 0
+    rbg = client.get(C.KIND_RBG, args.name, args.namespace)
+    if rbg is None:
+        print(f"rolebasedgroup {args.name} not found", file=sys.stderr)
+        return 1
+    if not revs:
+        print("no revisions recorded", file=sys.stderr)
+        return 1
+    target = None
+    if args.to:
+        target = next((r for r in revs if r.revision == args.to), None)
+    else:
+        # default: previous revision (reference rollout_undo.go semantics)
+        target = revs[-2] if len(revs) >= 2 else revs[-1]
+    if target is None:
+        print(f"revision {args.to} not found", file=sys.stderr)
+        return 1
+    if args.action == "diff":
+        cur = _spec_yaml(asdict(rbg.spec)).splitlines(keepends=True)
+        old = _spec_yaml(target.data).splitlines(keepends=True)
+        sys.stdout.writelines(difflib.unified_diff(
+            old, cur, fromfile=f"revision-{target.revision}",
+            tofile="current"))
+        return 0
+    if args.action == "undo":
+        rbg.spec = fromdict(RoleBasedGroupSpec, target.data)
+        client.update(rbg)
+        print(f"rolebasedgroup/{args.name} rolled back to revision "
+              f"{target.revision}")
+        return 0
+    print(f"unknown rollout action {args.action}", file=sys.stderr)
+    return 1
+
+
+_KIND_ALIASES = {
+    "rbg": C.KIND_RBG, "rolebasedgroup": C.KIND_RBG,
+    "rbgset": C.KIND_RBG_SET, "ris": C.KIND_ROLE_INSTANCE_SET,
+    "roleinstanceset": C.KIND_ROLE_INSTANCE_SET,
+    "roleinstance": C.KIND_ROLE_INSTANCE, "ri": C.KIND_ROLE_INSTANCE,
+    "scalingadapter": C.KIND_SCALING_ADAPTER,
+    "rbgsa": C.KIND_SCALING_ADAPTER,
+    "coordinatedpolicy": C.KIND_COORDINATED_POLICY,
+    "warmup": C.KIND_WARMUP,
+    "revision": C.KIND_CONTROLLER_REVISION,
+}
+
+
+def _resolve_kind(kind: str) -> str:
+    k = _KIND_ALIASES.get(kind.lower())
+    if k is None and kind in C.ALL_KINDS:
+        return kind
+    if k is None:
+        raise SystemExit(f"unknown kind {kind!r}")
+    return k
+
+
+def build_parser() -> argparse.ArgumentParser:
+    ap = argparse.ArgumentParser(prog="rbgctl")
+    ap.add_argument("--port", type=int, default=7471)
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("-n", "--namespace", default="default")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+    p = sub.add_parser("apply")
+    p.add_argument("-f", "--file", required=True)
+    p = sub.add_parser("get")
+    p.add_argument("kind")
+    p.add_argument("name", nargs="?")
+    p = sub.add_parser("status")
+    p.add_argument("name")
+    p = sub.add_parser("scale")
+    p.add_argument("name")
+    p.add_argument("--replicas", type=int, required=True)
+    p = sub.add_parser("delete")
+    p.add_argument("kind")
+    p.add_argument("name")
+    p = sub.add_parser("rollout")
+    p.add_argument("action", choices=["history", "diff", "undo"])
+    p.add_argument("name")
+    p.add_argument("--to", type=int, default=0)
+    return ap
+
+
+def main(argv: Optional[List[str]] = None,
+         client: Optional[BaseClient] = None) -> int:
+    args = build_parser().parse_args(argv)
+    if client is None:
+        client = RemoteClient(args.host, args.port)
+    return {
+        "apply": cmd_apply, "get": cmd_get, "status": cmd_status,
+        "scale": cmd_scale, "delete": cmd_delete, "rollout": cmd_rollout,
+    }[args.cmd](client, args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,130 @@
+"""CLI + client SDK + daemon suite (reference analog: cmd/cli tests +
+client-go fakes)."""
+import io
+import sys
+import textwrap
+
+import pytest
+
+from rbg_amd.api import constants as C
+from rbg_amd.api.serde import asdict
+from rbg_amd.cli.daemon import Daemon
+from rbg_amd.cli.main import main as cli_main
+from rbg_amd.client.client import InProcessClient, RemoteClient
+from rbg_amd.controller.manager import ManagerOptions
+from rbg_amd.store.revisions import RevisionManager
+from rbg_amd.store.store import Store
+from tests.test_api_types import make_rbg
+
+
+@pytest.fixture
+def daemon(tmp_run_dir):
+    d = Daemon(ManagerOptions(run_root=tmp_run_dir, num_gpus=8,
+                              resync_period=0.1), port=0)
+    d.manager.start()
+    d.rpc.start()
+    yield d
+    d.rpc.stop()
+    d.manager.stop()
+
+
+def run_cli(daemon, *argv):
+    out = io.StringIO()
+    old = sys.stdout
+    sys.stdout = out
+    try:
+        rc = cli_main(["--port", str(daemon.port), *argv])
+    finally:
+        sys.stdout = old
+    return rc, out.getvalue()
+
+
+def test_inprocess_client_crud():
+    store = Store()
+    client = InProcessClient(store)
+    created = client.create(make_rbg())
+    assert created.metadata.uid
+    got = client.get(C.KIND_RBG, "demo")
+    got.spec.roles[0].replicas = 3
+    client.update(got)
+    assert client.get(C.KIND_RBG, "demo").spec.roles[0].replicas == 3
+    assert [o.metadata.name for o in client.list(C.KIND_RBG)] == ["demo"]
+    assert client.delete(C.KIND_RBG, "demo")
+
+
+def test_cli_apply_status_rollout(daemon, tmp_path):
+    yaml_file = tmp_path / "rbg.yaml"
+    yaml_file.write_text(textwrap.dedent("""
+        apiVersion: workloads.x-k8s.io/v1alpha2
+        kind: RoleBasedGroup
+        metadata: {name: demo}
+        spec:
+          roles:
+          - name: router
+            replicas: 1
+            template:
+              engines:
+              - {name: engine, runner: echo, resources: {cpuOnly: true}}
+          - name: worker
+            replicas: 2
+            dependencies: [router]
+            template:
+              engines:
+              - {name: engine, runner: echo, resources: {cpuOnly: true}}
+    """))
+    rc, out = run_cli(daemon, "apply", "-f", str(yaml_file))
+    assert rc == 0 and "created" in out
+    # wait until ready
+    from tests.test_controller_e2e import rbg_ready
+    assert daemon.manager.wait_for(
+        lambda: rbg_ready(daemon.manager, "demo"), timeout=60)
+    rc, out = run_cli(daemon, "status", "demo")
+    assert rc == 0
+    assert "router" in out and "worker" in out and "Ready: True" in out
+    rc, out = run_cli(daemon, "get", "rbg")
+    assert rc == 0 and "demo" in out
+
+    # mutate spec -> new revision
+    yaml_file.write_text(yaml_file.read_text().replace("replicas: 2",
+                                                       "replicas: 3"))
+    rc, out = run_cli(daemon, "apply", "-f", str(yaml_file))
+    assert rc == 0 and "configured" in out
+    assert daemon.manager.wait_for(
+        lambda: len(RevisionManager(daemon.manager.store).list_for(
+            daemon.manager.store.get(C.KIND_RBG, "demo"))) >= 2, timeout=30)
+    rc, out = run_cli(daemon, "rollout", "history", "demo")
+    assert rc == 0 and "REVISION" in out
+    rc, out = run_cli(daemon, "rollout", "diff", "demo")
+    assert rc == 0 and ("replicas" in out)
+    rc, out = run_cli(daemon, "rollout", "undo", "demo")
+    assert rc == 0 and "rolled back" in out
+    assert daemon.manager.wait_for(
+        lambda: daemon.manager.store.get(
+            C.KIND_RBG, "demo").spec.role("worker").replicas == 2,
+        timeout=30)
+    rc, out = run_cli(daemon, "delete", "rbg", "demo")
+    assert rc == 0
+
+
+def test_cli_scale(daemon):
+    from rbg_amd.api.types import (ObjectMeta, RoleBasedGroupScalingAdapter,
+                                   ScaleTargetRef, ScalingAdapterSpecFull)
+    client = RemoteClient(port=daemon.port)
+    client.create(make_rbg("auto"))
+    client.create(RoleBasedGroupScalingAdapter(
+        metadata=ObjectMeta(name="auto-worker"),
+        spec=ScalingAdapterSpecFull(
+            replicas=1,
+            scale_target_ref=ScaleTargetRef(name="auto", role="worker"))))
+    rc, out = run_cli(daemon, "scale", "auto-worker", "--replicas", "4")
+    assert rc == 0
+    assert daemon.manager.wait_for(
+        lambda: daemon.manager.store.get(
+            C.KIND_RBG, "auto").spec.role("worker").replicas == 4, timeout=30)
+
+
+def test_daemon_healthz(daemon):
+    from rbg_amd.server.rpc import RpcClient
+    c = RpcClient("127.0.0.1", daemon.port)
+    h = c.call("healthz")
+    assert h["status"] == "ok" and h["gpus"] == 8
