@@ -33,9 +33,9 @@ class PagedKVCache:
         if num_pages is None:
             num_pages = self._size_pool(cfg, device)
         self.num_pages = num_pages
+        kvh = m.num_kv_heads // max(1, cfg.tp_size)   # TP shards the heads
         self.kv = torch.zeros(
-            (m.num_layers, 2, num_pages, m.num_kv_heads, cfg.page_size,
-             m.head_dim),
+            (m.num_layers, 2, num_pages, kvh, cfg.page_size, m.head_dim),
             dtype=torch.bfloat16, device=device)
         # page 0 is reserved scratch: hipGraph decode padding rows write
         # their (dead) KV slot there (model_runner._decode_graph)
@@ -44,7 +44,8 @@ class PagedKVCache:
     @staticmethod
     def _size_pool(cfg: EngineConfig, device: torch.device) -> int:
         m = cfg.model
-        page_bytes = (m.num_layers * 2 * m.num_kv_heads * cfg.page_size *
+        kvh = m.num_kv_heads // max(1, cfg.tp_size)
+        page_bytes = (m.num_layers * 2 * kvh * cfg.page_size *
                       m.head_dim * 2)
         if device.type == "cuda" and torch.cuda.is_available():
             free, _total = torch.cuda.mem_get_info(device)

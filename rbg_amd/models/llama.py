@@ -42,9 +42,16 @@ class ForwardBatch:
     decode_num_splits: int = 0      # 0 = pick host-side (sync-free in graphs)
 
 
-def _linear_weight(out_f: int, in_f: int, device, dtype, gen) -> nn.Parameter:
+def _linear_weight(out_f: int, in_f: int, device, dtype, gen,
+                   shard_dim: int = -1, tp_size: int = 1,
+                   tp_rank: int = 0) -> nn.Parameter:
+    """Generate the FULL weight deterministically, then slice this rank's
+    shard — the model is bit-identical across TP degrees (modulo all-reduce
+    summation order), so TP=N serving is directly comparable to TP=1."""
     w = torch.empty(out_f, in_f, device=device, dtype=dtype)
     w.normal_(0.0, 0.02, generator=gen)
+    if tp_size > 1 and shard_dim >= 0:
+        w = w.chunk(tp_size, dim=shard_dim)[tp_rank].contiguous()
     return nn.Parameter(w, requires_grad=False)
 
 
@@ -77,10 +84,22 @@ class LlamaAttention(nn.Module):
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
         q_out = self.heads * cfg.head_dim
         kv_out = self.kv_heads * cfg.head_dim
-        # column-parallel fused QKV; row-parallel O
-        self.wqkv = _linear_weight(q_out + 2 * kv_out, cfg.hidden_size,
-                                   device, dtype, gen)
-        self.wo = _linear_weight(cfg.hidden_size, q_out, device, dtype, gen)
+        # column-parallel fused QKV (head-aligned shards); row-parallel O.
+        # Full weights are generated then sliced so every TP degree sees
+        # the same model (see _linear_weight).
+        full_q = cfg.num_heads * cfg.head_dim
+        full_kv = cfg.num_kv_heads * cfg.head_dim
+        wq = _linear_weight(full_q, cfg.hidden_size, device, dtype, gen,
+                            shard_dim=0, tp_size=tp.size, tp_rank=tp.rank)
+        wk = _linear_weight(full_kv, cfg.hidden_size, device, dtype, gen,
+                            shard_dim=0, tp_size=tp.size, tp_rank=tp.rank)
+        wv = _linear_weight(full_kv, cfg.hidden_size, device, dtype, gen,
+                            shard_dim=0, tp_size=tp.size, tp_rank=tp.rank)
+        self.wqkv = nn.Parameter(torch.cat([wq, wk, wv], dim=0),
+                                 requires_grad=False)
+        self.wo = _linear_weight(cfg.hidden_size, full_q, device, dtype, gen,
+                                 shard_dim=1, tp_size=tp.size,
+                                 tp_rank=tp.rank)
         self.q_out, self.kv_out = q_out, kv_out
 
     def forward(self, x: torch.Tensor, batch: ForwardBatch,
@@ -116,9 +135,17 @@ class LlamaMLP(nn.Module):
         self.tp = tp
         assert cfg.intermediate_size % tp.size == 0
         inter = cfg.intermediate_size // tp.size
-        self.w_gate_up = _linear_weight(2 * inter, cfg.hidden_size,
-                                        device, dtype, gen)
-        self.w_down = _linear_weight(cfg.hidden_size, inter, device, dtype, gen)
+        wg = _linear_weight(cfg.intermediate_size, cfg.hidden_size, device,
+                            dtype, gen, shard_dim=0, tp_size=tp.size,
+                            tp_rank=tp.rank)
+        wu = _linear_weight(cfg.intermediate_size, cfg.hidden_size, device,
+                            dtype, gen, shard_dim=0, tp_size=tp.size,
+                            tp_rank=tp.rank)
+        self.w_gate_up = nn.Parameter(torch.cat([wg, wu], dim=0),
+                                      requires_grad=False)
+        self.w_down = _linear_weight(cfg.hidden_size, cfg.intermediate_size,
+                                     device, dtype, gen, shard_dim=1,
+                                     tp_size=tp.size, tp_rank=tp.rank)
         self.inter = inter
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -159,13 +186,15 @@ class LlamaLayer(nn.Module):
 
 class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: ModelConfig, device: torch.device,
-                 tp: Optional[TPContext] = None):
+                 tp: Optional[TPContext] = None, base_seed: int = 1234):
         super().__init__()
         self.cfg = cfg
         self.tp = tp or TPContext()
+        self.base_seed = base_seed
+        self.device = device
         dtype = torch.bfloat16
         gen = torch.Generator(device=device)
-        gen.manual_seed(1234 + self.tp.rank)
+        gen.manual_seed(base_seed)
         self.embed = nn.Parameter(
             torch.empty(cfg.vocab_size, cfg.hidden_size, device=device,
                         dtype=dtype).normal_(0, 0.02, generator=gen),
@@ -202,10 +231,17 @@ class LlamaForCausalLM(nn.Module):
 
     def reload_weights(self, seed: int) -> None:
         """In-place weight refresh — the live-update path (SURVEY §2.3
-        "Live engine update": new weights without tearing down the KV pool)."""
-        gen = torch.Generator(device=self.embed.device)
-        gen.manual_seed(seed + self.tp.rank)
+        "Live engine update": new weights without tearing down the KV pool).
+        Replays construction with the new seed so TP shards stay consistent
+        across ranks, then copies into the live parameters (the KV pool and
+        any captured hipGraphs keep their addresses)."""
+        fresh = LlamaForCausalLM(self.cfg, self.device, self.tp,
+                                 base_seed=seed)
         with torch.no_grad():
-            for p in self.parameters():
-                if p.dim() == 2:
-                    p.normal_(0, 0.02, generator=gen)
+            for p, q in zip(self.parameters(), fresh.parameters()):
+                p.copy_(q)
+        del fresh
+        if self.device.type == "cuda":
+            torch.cuda.empty_cache()
+        if self.tp.size > 1:
+            torch.distributed.barrier(group=self.tp.group)

@@ -77,17 +77,42 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
   }
 
   const int* btab = block_tables + (size_t)seq * max_pages;
-  // wave stride: 4 waves x 4 keys
-  for (int base = key_begin + wave * 4; base < key_end; base += 16) {
+  // page_size is a power of two (engine config); shift/mask beats idiv
+  const int ps_shift = 31 - __clz(page_size);
+  const int ps_mask = page_size - 1;
+
+  auto row_offset = [&](int base) -> size_t {
     const int key = base + group;
-    const bool valid = key < key_end;
-    const int kslot = valid ? key : key_begin;   // clamp for safe address
-    const int page = btab[kslot / page_size];
-    const size_t row_off =
-        (((size_t)page * num_kv_heads + kvh) * page_size +
-         (kslot % page_size)) * HEAD_DIM;
-    Bf16x8U kv;
-    kv.u = *reinterpret_cast<const uint4*>(key_cache + row_off + dbase);
+    const int kslot = key < key_end ? key : key_begin;  // clamp: safe addr
+    const int page = btab[kslot >> ps_shift];
+    return (((size_t)page * num_kv_heads + kvh) * page_size +
+            (kslot & ps_mask)) * HEAD_DIM;
+  };
+
+  // wave stride: 4 waves x 4 keys; 2-deep software prefetch keeps the next
+  // iteration's K AND V in flight under the current iteration's VALU work
+  int base = key_begin + wave * 4;
+  Bf16x8U k_pref, v_pref;
+  if (base < key_end) {
+    const size_t off = row_offset(base);
+    k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
+    v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
+  }
+  for (; base < key_end; base += 16) {
+    const bool valid = base + group < key_end;
+    Bf16x8U kv = k_pref, vv = v_pref;
+    const int nxt = base + 16;
+    if (nxt < key_end) {
+      const size_t off = row_offset(nxt);
+      k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
+      v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
+    }
+    float kf[8], vf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      kf[j] = bf2f(kv.e[j]);
+      vf[j] = bf2f(vv.e[j]);
+    }
     // scores for this key, all QPG heads
     float p[QPG];
     float tile_max[QPG];
@@ -95,7 +120,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     for (int h = 0; h < QPG; ++h) {
       float s = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) s += qf[h][j] * bf2f(kv.e[j]);
+      for (int j = 0; j < 8; ++j) s += qf[h][j] * kf[j];
       s = group16_sum(s);             // full dot across the 16-lane group
       if (!valid) s = NEG_INF;
       // tile max across the wave's 4 keys
@@ -105,8 +130,6 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
       tile_max[h] = tm;
       p[h] = s;
     }
-    Bf16x8U vv;
-    vv.u = *reinterpret_cast<const uint4*>(val_cache + row_off + dbase);
 #pragma unroll
     for (int h = 0; h < QPG; ++h) {
       const float m_new = fmaxf(m[h], tile_max[h]);
@@ -120,7 +143,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
       m[h] = m_new;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        acc[h][j] = acc[h][j] * alpha + pv * bf2f(vv.e[j]);
+        acc[h][j] = acc[h][j] * alpha + pv * vf[j];
     }
   }
 

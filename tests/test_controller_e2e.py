@@ -78,13 +78,18 @@ def test_two_role_group_becomes_ready(mgr):
 
 def test_dependency_ordering_router_first(mgr):
     mgr.store.create(router_worker_rbg(name="ord"))
-    # before the router is ready, no worker instances may exist
+    # before the router is ready, no worker instances may exist; probe the
+    # RoleInstanceSet status (the same source the readiness gate reads)
     seen_violation = []
+
+    def router_ris_ready():
+        ris = mgr.store.try_get(C.KIND_ROLE_INSTANCE_SET, "ord-router")
+        return ris is not None and ris.status.ready_replicas >= 1
 
     def check():
         workers = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
             C.LABEL_GROUP_NAME: "ord", C.LABEL_ROLE_NAME: "worker"})
-        if workers and not rbg_ready_role(mgr, "ord", "router"):
+        if workers and not router_ris_ready():
             seen_violation.append(True)
         return rbg_ready(mgr, "ord")
 
