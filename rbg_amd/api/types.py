@@ -320,6 +320,9 @@ class RoleInstanceStatus:
     workers: List[WorkerStatus] = field(default_factory=list)
     restart_count: int = 0
     last_restart_time: float = 0.0
+    # group-recovery time: failure detection -> gang Ready again (seconds);
+    # one of the BASELINE.json headline metrics
+    last_recovery_duration: float = 0.0
     # in-place-update baselines: expected restarts vs crashes
     # (reference roleinstance_types.go:170-198)
     in_place_update_baselines: Dict[str, int] = field(default_factory=dict)

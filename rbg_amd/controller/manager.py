@@ -39,15 +39,22 @@ Key = Tuple[str, str]   # (name, namespace)
 
 
 class WorkQueue:
-    """Deduplicating delayed workqueue (controller-runtime analog)."""
+    """Deduplicating delayed workqueue with in-flight tracking so concurrent
+    workers never reconcile the same key simultaneously (controller-runtime
+    semantics: an event for an in-flight key re-queues it after done)."""
 
     def __init__(self) -> None:
         self._cond = threading.Condition()
         self._ready: Dict[Key, None] = {}
         self._delayed: Dict[Key, float] = {}
+        self._inflight: set = set()
+        self._dirty: set = set()
 
     def add(self, key: Key, after: float = 0.0) -> None:
         with self._cond:
+            if key in self._inflight:
+                self._dirty.add(key)
+                return
             if after <= 0:
                 self._delayed.pop(key, None)
                 self._ready[key] = None
@@ -66,10 +73,14 @@ class WorkQueue:
                 for key, due in list(self._delayed.items()):
                     if due <= now:
                         del self._delayed[key]
-                        self._ready[key] = None
+                        if key in self._inflight:
+                            self._dirty.add(key)
+                        else:
+                            self._ready[key] = None
                 if self._ready:
                     key = next(iter(self._ready))
                     del self._ready[key]
+                    self._inflight.add(key)
                     return key
                 wait = deadline - now
                 if self._delayed:
@@ -78,9 +89,20 @@ class WorkQueue:
                     return None
                 self._cond.wait(timeout=wait)
 
+    def done(self, key: Key, requeue_after: float = 0.0) -> None:
+        with self._cond:
+            self._inflight.discard(key)
+            if key in self._dirty:
+                self._dirty.discard(key)
+                self._ready[key] = None
+                self._cond.notify_all()
+                return
+        if requeue_after > 0:
+            self.add(key, after=requeue_after)
+
     def empty(self) -> bool:
         with self._cond:
-            return not self._ready and not self._delayed
+            return not self._ready and not self._delayed and not self._inflight
 
 
 @dataclass
@@ -91,6 +113,7 @@ class ManagerOptions:
     resync_period: float = 0.5
     port_range: Tuple[int, int] = (30000, 40000)
     history_limit: int = 10
+    concurrency: int = 4          # reconcile workers per controller
 
 
 class Manager:
@@ -204,14 +227,14 @@ class Manager:
                 except Exception:
                     log.exception("reconcile %s %s failed", kind, key)
                     requeue = 1.0
-                if requeue and requeue > 0:
-                    q.add(key, after=requeue)
+                q.done(key, requeue_after=requeue if requeue else 0.0)
 
         self._threads = [threading.Thread(target=watch_loop, daemon=True),
                          threading.Thread(target=resync_loop, daemon=True)]
         for kind in self._queues:
-            self._threads.append(
-                threading.Thread(target=worker_loop, args=(kind,), daemon=True))
+            for _ in range(self.opts.concurrency):
+                self._threads.append(threading.Thread(
+                    target=worker_loop, args=(kind,), daemon=True))
         for t in self._threads:
             t.start()
 

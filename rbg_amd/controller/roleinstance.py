@@ -46,6 +46,8 @@ class InstanceRuntime:
     handles: Dict[str, WorkerHandle] = field(default_factory=dict)
     gpu_by_worker: Dict[str, List[int]] = field(default_factory=dict)
     master_port: int = 0
+    recovery_started: float = 0.0        # failure detected, not Ready yet
+    applied_args: Dict[str, str] = field(default_factory=dict)  # in-place
 
 
 class RoleInstanceController:
@@ -121,6 +123,15 @@ class RoleInstanceController:
             for _, _, w in desired)
         if healthy:
             tracker.observe_healthy()
+            if rt.recovery_started:
+                # group-recovery time: first failure -> whole gang Ready
+                duration = time.time() - rt.recovery_started
+                rt.recovery_started = 0.0
+                inst.status.last_recovery_duration = duration
+                self._record_recovery(inst, duration)
+            self._apply_in_place_updates(inst, rt, desired)
+        elif failed_fatal and not rt.recovery_started:
+            rt.recovery_started = time.time()
 
         policy = inst.spec.restart_policy
         if failed_fatal and policy == C.RESTART_POLICY_RECREATE_INSTANCE:
@@ -157,6 +168,58 @@ class RoleInstanceController:
         return 1.0   # periodic health poll
 
     # ------------------------------------------------------------------
+
+    def _record_recovery(self, inst: RoleInstance, duration: float) -> None:
+        def mutate(cur: RoleInstance):
+            cur.status.last_recovery_duration = duration
+            return cur
+        try:
+            self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mutate,
+                             inst.metadata.namespace, subresource="status")
+        except KeyError:
+            pass
+
+    def _apply_in_place_updates(self, inst: RoleInstance, rt: InstanceRuntime,
+                                desired: List[tuple]) -> None:
+        """Live engine update: when only engine args changed
+        (the in-place-feasible diff, roleinstanceset._can_update_in_place),
+        push the new args to the running worker over its RPC port — weights
+        reload in place, the KV pool and RCCL comms survive."""
+        import json as _json
+        changed_any = False
+        all_applied = True
+        for comp, j, wname in desired:
+            engine = comp.template.main_engine() if comp.template else None
+            if engine is None:
+                continue
+            want = _json.dumps(engine.args, sort_keys=True)
+            have = rt.applied_args.get(wname)
+            if have is None:
+                rt.applied_args[wname] = want    # initial spawn args
+                continue
+            if have == want:
+                continue
+            changed_any = True
+            h = rt.handles.get(wname)
+            st = h.read_status() if h else {}
+            port = st.get("rpc_port")
+            if not port:
+                all_applied = False
+                continue
+            try:
+                from ..server.rpc import RpcClient
+                client = RpcClient("127.0.0.1", int(port), timeout=30.0)
+                client.call("apply_update", args=engine.args)
+                client.close()
+                rt.applied_args[wname] = want
+            except Exception as e:  # noqa: BLE001
+                log.warning("in-place update of %s failed: %r", wname, e)
+                all_applied = False
+        if changed_any:
+            self._set_condition(inst, C.COND_INPLACE_UPDATE_READY,
+                                all_applied,
+                                "Applied" if all_applied else "Applying",
+                                "live engine update")
 
     def _ensure_gang(self, inst: RoleInstance, rt: InstanceRuntime,
                      desired: List[tuple]) -> None:

@@ -73,40 +73,64 @@ class ModelRunner:
 
     # ------------------------------------------------------------------
 
-    def _decode_inputs(self, seqs: List[Sequence]):
-        tokens = [seq.last_token for seq in seqs]
-        positions = [seq.num_tokens - 1 for seq in seqs]
-        slots = [seq.block_table.slots_for(seq.num_tokens - 1, 1)[0]
-                 for seq in seqs]
-        ctx = [seq.num_tokens for seq in seqs]
+    def _build_decode_state(self, seqs: List[Sequence]) -> dict:
+        """Device-resident decode inputs.  Valid across steps for the same
+        batch composition: positions/ctx advance on-device, the block tables
+        never change mid-decode (pages are reserved for prompt+max_new at
+        admission — engine/scheduler.py), so the only host->device traffic
+        per step is nothing at all; sampled tokens stay on-device as the
+        next step's input."""
+        dev = self.device
+        positions = self._i32([seq.num_tokens - 1 for seq in seqs])
+        ctx = self._i32([seq.num_tokens for seq in seqs])
+        tokens = torch.tensor([seq.last_token for seq in seqs],
+                              dtype=torch.int64, device=dev)
         max_pages = max(len(seq.block_table.pages) for seq in seqs)
         bt = torch.zeros(len(seqs), max_pages, dtype=torch.int32)
         for i, seq in enumerate(seqs):
             bt[i, :len(seq.block_table.pages)] = torch.tensor(
                 seq.block_table.pages, dtype=torch.int32)
-        return tokens, positions, slots, ctx, bt
+        return {"ids": tuple(s.seq_id for s in seqs),
+                "positions": positions, "ctx": ctx, "tokens": tokens,
+                "bt": bt.to(dev), "fresh": True}
+
+    def _decode_slots(self, state: dict) -> torch.Tensor:
+        ps = self.cfg.page_size
+        pos = state["positions"].long()
+        page = torch.gather(state["bt"], 1,
+                            (pos // ps).unsqueeze(1)).squeeze(1)
+        return (page * ps + (pos % ps).to(torch.int32)).to(torch.int32)
 
     @torch.inference_mode()
     def decode(self, seqs: List[Sequence]) -> List[int]:
-        tokens, positions, slots, ctx, bt = self._decode_inputs(seqs)
+        n = len(seqs)
+        ids = tuple(s.seq_id for s in seqs)
+        state = self._dstate if getattr(self, "_dstate", None) else None
+        if state is None or state["ids"] != ids:
+            state = self._build_decode_state(seqs)
+            self._dstate = state
+        elif not state["fresh"]:
+            state["positions"] += 1
+            state["ctx"] += 1
+        state["fresh"] = False
+        slots = self._decode_slots(state)
         use_graph = (not self.cfg.enforce_eager and
                      self.device.type == "cuda" and
-                     len(seqs) <= GRAPH_BATCH_SIZES[-1])
+                     n <= GRAPH_BATCH_SIZES[-1])
         if use_graph:
-            logits = self._decode_graph(tokens, positions, slots, ctx, bt)
+            logits = self._decode_graph(state, slots, n)
         else:
             batch = ForwardBatch(
                 mode="decode",
-                positions=self._i32(positions),
-                slot_mapping=self._i32(slots),
-                block_tables=bt.to(self.device),
-                context_lens=self._i32(ctx),
-                decode_num_splits=self._splits_for(len(seqs)))
-            hidden = self.model.forward(
-                torch.tensor(tokens, dtype=torch.int64, device=self.device),
-                batch, self.cache)
+                positions=state["positions"],
+                slot_mapping=slots,
+                block_tables=state["bt"],
+                context_lens=state["ctx"],
+                decode_num_splits=self._splits_for(n))
+            hidden = self.model.forward(state["tokens"], batch, self.cache)
             logits = self.model.logits(hidden)
-        next_tokens = self.sample(logits[:len(seqs)], seqs)
+        next_dev, next_tokens = self.sample_device(logits[:n], seqs)
+        state["tokens"] = next_dev
         for seq, tok in zip(seqs, next_tokens):
             seq.append_token(tok)
         return next_tokens
@@ -159,25 +183,28 @@ class ModelRunner:
         log.info("captured decode hipGraph for batch %d", bs)
         return self._graphs[bs]
 
-    def _decode_graph(self, tokens, positions, slots, ctx, bt) -> torch.Tensor:
-        n = len(tokens)
+    def _decode_graph(self, state: dict, slots: torch.Tensor,
+                      n: int) -> torch.Tensor:
         bs = self._graph_bucket(n)
         if bs not in self._graphs:
             self._build_graph(bs)
         graph, static, logits = self._graphs[bs]
-        static["tokens"][:n] = torch.tensor(tokens, dtype=torch.int64,
-                                            device=self.device)
-        static["positions"][:n] = self._i32(positions)
-        static["slots"][:n] = self._i32(slots)
-        static["ctx"].fill_(1)
-        static["ctx"][:n] = self._i32(ctx)
-        static["bt"].zero_()
-        static["bt"][:n, :bt.shape[1]] = bt.to(self.device)
-        if n < bs:
-            # padding rows decode against page 0 with ctx=1: harmless work,
-            # their slot writes go to slot 0 of page 0 — reserve it
-            static["slots"][n:] = 0
-            static["positions"][n:] = 0
+        static["tokens"][:n].copy_(state["tokens"][:n])
+        static["positions"][:n].copy_(state["positions"])
+        static["slots"][:n].copy_(slots)
+        static["ctx"][:n].copy_(state["ctx"])
+        bt = state["bt"]
+        if state.get("graph_bt_loaded") != (state["ids"], bs):
+            # block tables are stable for this batch: upload once per batch
+            static["bt"].zero_()
+            static["bt"][:n, :bt.shape[1]].copy_(bt)
+            if n < bs:
+                # padding rows decode against page 0 with ctx=1: harmless
+                # work; their slot writes land in reserved scratch page 0
+                static["slots"][n:] = 0
+                static["positions"][n:] = 0
+                static["ctx"][n:] = 1
+            state["graph_bt_loaded"] = (state["ids"], bs)
         graph.replay()
         return logits
 
@@ -185,10 +212,18 @@ class ModelRunner:
 
     @torch.inference_mode()
     def sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> List[int]:
+        return self.sample_device(logits, seqs)[1]
+
+    @torch.inference_mode()
+    def sample_device(self, logits: torch.Tensor, seqs: List[Sequence]):
+        """Returns (device int64 tensor, host list) of next tokens — the
+        device tensor feeds the next decode step without a host round trip."""
         temps = [seq.sampling.temperature for seq in seqs]
         if all(t == 0.0 for t in temps):
-            return logits.argmax(dim=-1).tolist()
-        t = torch.tensor([max(tt, 1e-5) for tt in temps],
-                         device=logits.device).unsqueeze(1)
-        probs = torch.softmax(logits / t, dim=-1)
-        return torch.multinomial(probs, 1).squeeze(1).tolist()
+            dev = logits.argmax(dim=-1)
+        else:
+            t = torch.tensor([max(tt, 1e-5) for tt in temps],
+                             device=logits.device).unsqueeze(1)
+            probs = torch.softmax(logits / t, dim=-1)
+            dev = torch.multinomial(probs, 1).squeeze(1)
+        return dev, dev.tolist()

@@ -147,10 +147,27 @@ class ServeWorker:
         self.rpc.register("poll", self._rpc_poll)
         self.rpc.register("reload_weights",
                           lambda seed: self.engine.reload_weights(int(seed)))
+        self.rpc.register("apply_update", self._rpc_apply_update)
         if self.mode == "prefill":
             self.rpc.register("prefill", self._rpc_prefill)
         if self.mode == "decode":
             self.rpc.register("import_seq", self._rpc_import_seq)
+
+    def _rpc_apply_update(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        """Live in-place update: apply the arg diffs an engine can absorb
+        without restart.  weights_seed -> weight reload keeping the KV pool
+        (reference pkg/inplace semantics on a live engine)."""
+        applied = []
+        new_seed = args.get("weights_seed")
+        if new_seed is not None and \
+                int(new_seed) != int(self.ctx.args.get("weights_seed", -1)):
+            with self.lock:
+                self.engine.reload_weights(int(new_seed))
+            self.ctx.args["weights_seed"] = int(new_seed)
+            applied.append("weights_seed")
+        self.ctx.args.update({k: v for k, v in args.items()
+                              if k not in ("weights_seed",)})
+        return {"applied": applied}
 
     # -- colocated / generic ------------------------------------------------
 

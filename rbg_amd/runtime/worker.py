@@ -154,8 +154,13 @@ def main(argv: Optional[List[str]] = None) -> int:
     # __main__ module; runners register on the canonical import, so consult
     # that one (not this module's RUNNERS copy)
     import rbg_amd.runtime.worker as canonical
-    canonical._load_entry_runners()
     runner = canonical.RUNNERS.get(opts.runner)
+    if runner is None:
+        # engine runners import torch (~seconds); load lazily so built-in
+        # runners (echo) start fast — the controller-latency path the stress
+        # harness measures
+        canonical._load_entry_runners()
+        runner = canonical.RUNNERS.get(opts.runner)
     if runner is None:
         write_status(run_dir, name, "Failed",
                      error=f"unknown runner {opts.runner!r}")

@@ -1,0 +1,93 @@
+"""In-place live update + group-recovery metric (reference analog:
+pkg/inplace tests, failure-handling doc semantics, restart_policy envtest)."""
+import time
+
+import pytest
+import torch
+
+from rbg_amd.api import constants as C
+from rbg_amd.api.types import get_condition
+from rbg_amd.controller.manager import Manager, ManagerOptions
+from tests.test_controller_e2e import rbg_ready
+from tests.test_serving_e2e import engine_role, router_role, _router_http_port, _http_post
+
+
+@pytest.fixture
+def mgr(tmp_run_dir):
+    m = Manager(ManagerOptions(run_root=tmp_run_dir, num_gpus=8,
+                               resync_period=0.1))
+    m.restarts.base = 0.2
+    m.restarts.max_delay = 1.0
+    m.start()
+    yield m
+    m.stop()
+
+
+@pytest.mark.timeout(300)
+def test_inplace_weight_reload_keeps_processes(mgr):
+    from rbg_amd.api.types import ObjectMeta, RoleBasedGroup, RoleBasedGroupSpec
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="live"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("colocated", {"worker_roles": ["worker"],
+                                      "vocab_size": 500}),
+            engine_role("worker", "colocated", {"weights_seed": 1}),
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "live"), timeout=120)
+    port = _router_http_port(mgr, "live")
+    torch.manual_seed(3)
+    prompt = torch.randint(0, 500, (10,)).tolist()
+    before = _http_post(port, "/generate",
+                        {"prompt_tokens": prompt, "max_new_tokens": 4})
+    pid_before = [w.pid for i in mgr.store.list(
+        C.KIND_ROLE_INSTANCE,
+        selector={C.LABEL_GROUP_NAME: "live", C.LABEL_ROLE_NAME: "worker"})
+        for w in i.status.workers]
+
+    # change ONLY engine args -> in-place feasible -> live reload
+    def mutate(cur):
+        role = cur.spec.role("worker")
+        role.template.engines[0].args["weights_seed"] = 999
+        return cur
+    mgr.store.apply(C.KIND_RBG, "live", mutate)
+
+    def updated():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+            C.LABEL_GROUP_NAME: "live", C.LABEL_ROLE_NAME: "worker"})
+        for i in insts:
+            cond = get_condition(i.status.conditions,
+                                 C.COND_INPLACE_UPDATE_READY)
+            if cond is not None and cond.status == "True":
+                return True
+        return False
+    assert mgr.wait_for(updated, timeout=60)
+    # same processes (no recreate)
+    pid_after = [w.pid for i in mgr.store.list(
+        C.KIND_ROLE_INSTANCE,
+        selector={C.LABEL_GROUP_NAME: "live", C.LABEL_ROLE_NAME: "worker"})
+        for w in i.status.workers]
+    assert pid_before == pid_after
+    # outputs actually changed (new weights)
+    after = _http_post(port, "/generate",
+                       {"prompt_tokens": prompt, "max_new_tokens": 4})
+    assert after["tokens"] != before["tokens"]
+
+
+@pytest.mark.timeout(300)
+def test_group_recovery_time_recorded(mgr):
+    from tests.test_controller_e2e import router_worker_rbg
+    mgr.store.create(router_worker_rbg(
+        name="rec", worker_replicas=1, worker_args={"crash_after": 1.0}))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "rec"), timeout=60)
+
+    def recovered():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+            C.LABEL_GROUP_NAME: "rec", C.LABEL_ROLE_NAME: "worker"})
+        return any(i.status.last_recovery_duration > 0 for i in insts)
+    assert mgr.wait_for(recovered, timeout=90)
+    insts = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+        C.LABEL_GROUP_NAME: "rec", C.LABEL_ROLE_NAME: "worker"})
+    dur = max(i.status.last_recovery_duration for i in insts)
+    # crash -> backoff(0.2s) -> respawn -> ready; should be seconds, not minutes
+    assert 0.0 < dur < 60.0

@@ -1,0 +1,164 @@
+#!/usr/bin/env python3
+"""Controller stress harness — the KWOK load driver analog.
+
+Mirrors reference test/stress/{main,client,scenario,timing,report}.go:
+drives create / update / delete phases of N RoleBasedGroups at a target QPS
+against an in-process Manager whose engines are the CPU `echo` stub (the
+fake-engine stand-in for KWOK fake pods), measures watch-based
+submit->Ready / submit->Gone latency, and reports P50/P90/P99 + achieved
+QPS as JSON.
+
+Usage: python tools/stress.py [--groups 10] [--qps 5] [--roles 2]
+                              [--replicas 2] [--out stress.json]
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import tempfile
+import time
+from typing import Dict, List
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from rbg_amd.api import constants as C
+from rbg_amd.api.types import (EngineResources, EngineSpec, EngineTemplate,
+                               ObjectMeta, RoleBasedGroup, RoleBasedGroupSpec,
+                               RoleSpec, get_condition)
+from rbg_amd.controller.manager import Manager, ManagerOptions
+
+
+def pct(sorted_vals: List[float], p: float) -> float:
+    if not sorted_vals:
+        return 0.0
+    idx = min(len(sorted_vals) - 1, int(p / 100.0 * len(sorted_vals)))
+    return sorted_vals[idx]
+
+
+def echo_rbg(name: str, roles: int, replicas: int) -> RoleBasedGroup:
+    tmpl = EngineTemplate(engines=[EngineSpec(
+        name="engine", runner="echo",
+        resources=EngineResources(cpu_only=True))])
+    role_specs = []
+    for r in range(roles):
+        deps = [f"role{r-1}"] if r > 0 else []
+        role_specs.append(RoleSpec(name=f"role{r}", replicas=replicas,
+                                   dependencies=deps, template=tmpl))
+    return RoleBasedGroup(metadata=ObjectMeta(name=name),
+                          spec=RoleBasedGroupSpec(roles=role_specs))
+
+
+def ready(mgr: Manager, name: str) -> bool:
+    rbg = mgr.store.try_get(C.KIND_RBG, name)
+    if rbg is None:
+        return False
+    c = get_condition(rbg.status.conditions, C.COND_READY)
+    return c is not None and c.status == "True"
+
+
+def run_phase(mgr: Manager, names: List[str], qps: float, submit, settled,
+              timeout: float) -> Dict:
+    """Submit at qps, then wait for every name to settle; per-item latency is
+    submit->settled (watch-equivalent: condition poll at 10 ms)."""
+    submit_times: Dict[str, float] = {}
+    t0 = time.monotonic()
+    for i, name in enumerate(names):
+        target = t0 + i / qps
+        now = time.monotonic()
+        if target > now:
+            time.sleep(target - now)
+        submit(name)
+        submit_times[name] = time.monotonic()
+    latencies: Dict[str, float] = {}
+    deadline = time.monotonic() + timeout
+    pending = set(names)
+    while pending and time.monotonic() < deadline:
+        for name in list(pending):
+            if settled(name):
+                latencies[name] = time.monotonic() - submit_times[name]
+                pending.discard(name)
+        time.sleep(0.01)
+    vals = sorted(latencies.values())
+    return {
+        "submitted": len(names),
+        "settled": len(latencies),
+        "timed_out": len(pending),
+        "achieved_qps": round(len(names) / max(1e-9,
+                              max(submit_times.values()) - t0 + 1 / qps), 2),
+        "p50_s": round(pct(vals, 50), 3),
+        "p90_s": round(pct(vals, 90), 3),
+        "p99_s": round(pct(vals, 99), 3),
+        "max_s": round(vals[-1], 3) if vals else 0.0,
+    }
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--groups", type=int, default=10)
+    ap.add_argument("--qps", type=float, default=5.0)
+    ap.add_argument("--roles", type=int, default=2)
+    ap.add_argument("--replicas", type=int, default=2)
+    ap.add_argument("--timeout", type=float, default=120.0)
+    ap.add_argument("--out", default="")
+    args = ap.parse_args()
+
+    run_root = tempfile.mkdtemp(prefix="rbg-stress-")
+    mgr = Manager(ManagerOptions(run_root=run_root, num_gpus=8,
+                                 resync_period=0.1))
+    mgr.start()
+    names = [f"stress-{i}" for i in range(args.groups)]
+    report = {"config": vars(args)}
+    try:
+        # ---- create phase -------------------------------------------------
+        report["create"] = run_phase(
+            mgr, names, args.qps,
+            submit=lambda n: mgr.store.create(
+                echo_rbg(n, args.roles, args.replicas)),
+            settled=lambda n: ready(mgr, n),
+            timeout=args.timeout)
+
+        # ---- update phase (scale each group's last role +1) ---------------
+        def scale_up(n):
+            def mutate(cur):
+                cur.spec.roles[-1].replicas += 1
+                return cur
+            mgr.store.apply(C.KIND_RBG, n, mutate)
+
+        def scaled(n):
+            rbg = mgr.store.try_get(C.KIND_RBG, n)
+            if rbg is None or not ready(mgr, n):
+                return False
+            want = rbg.spec.roles[-1].replicas
+            for rs in rbg.status.role_statuses:
+                if rs.name == rbg.spec.roles[-1].name:
+                    return rs.ready_replicas >= want
+            return False
+        report["update"] = run_phase(mgr, names, args.qps, scale_up, scaled,
+                                     args.timeout)
+
+        # ---- delete phase -------------------------------------------------
+        def delete(n):
+            def mark(cur):
+                cur.metadata.deletion_timestamp = time.time()
+                return cur
+            mgr.store.apply(C.KIND_RBG, n, mark)
+        report["delete"] = run_phase(
+            mgr, names, args.qps, delete,
+            settled=lambda n: mgr.store.try_get(C.KIND_RBG, n) is None,
+            timeout=args.timeout)
+    finally:
+        mgr.stop()
+    out = json.dumps(report, indent=1)
+    print(out)
+    if args.out:
+        with open(args.out, "w") as f:
+            f.write(out)
+    ok = all(report[ph]["timed_out"] == 0
+             for ph in ("create", "update", "delete"))
+    return 0 if ok else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
