@@ -152,16 +152,34 @@ class RoleInstanceController:
         elif failed_fatal and policy == C.RESTART_POLICY_NONE:
             return 1.0   # leave failed workers visible in status
 
-        # 2. spawn anything missing (all-or-nothing GPU reservation first)
+        # 2. spawn anything missing (all-or-nothing GPU reservation first);
+        #    component start-ordering gates (discovery/component.py)
         missing = [(c, j, w) for c, j, w in desired if w not in rt.handles]
+        gated = False
         if missing:
             try:
                 self._ensure_gang(inst, rt, desired)
             except GangUnschedulable as e:
                 self._set_condition(inst, C.COND_READY, False, "Unschedulable", str(e))
                 return 1.0
+            from ..discovery import component as comp_disc
+            deps = comp_disc.parse_depends_on(inst.metadata.annotations)
+            for comp in inst.spec.components:
+                deps.update(comp_disc.parse_depends_on(comp.annotations))
+            comp_ready = {
+                comp.name: all(
+                    rt.handles.get(w) is not None and
+                    rt.handles[w].phase() == "Ready"
+                    for c2, j2, w in desired if c2.name == comp.name)
+                and any(c2.name == comp.name for c2, _, _ in desired)
+                for comp in inst.spec.components}
             for comp, j, wname in missing:
+                if not comp_disc.start_gate(comp.name, deps, comp_ready):
+                    gated = True
+                    continue
                 self._spawn_worker(inst, rt, comp, j, wname)
+        if gated:
+            return 0.2
         if not healthy:
             return 0.25
         self._set_restarting(inst, False, "")
@@ -289,6 +307,22 @@ class RoleInstanceController:
                          else f"pod-{wname}-{req.name}")
             allocated[req.name] = self.ports.allocate(scope_key, req.count)
         env.update(self.ports.env_for(allocated))
+        # sibling-component discovery env (discovery/component.py)
+        from ..discovery import component as comp_disc
+        disc = comp_disc.parse_discovery(comp.annotations) or \
+            comp_disc.parse_discovery(inst.metadata.annotations)
+        if disc:
+            sibling_ports: Dict[str, List[int]] = {}
+            for other in inst.spec.components:
+                ports_list: List[int] = []
+                for w2, h2 in rt.handles.items():
+                    if w2.startswith(f"{inst.metadata.name}-{other.name}-"):
+                        st2 = h2.read_status()
+                        for key in ("rpc_port", "http_port"):
+                            if st2.get(key):
+                                ports_list.append(int(st2[key]))
+                sibling_ports[other.name] = ports_list
+            env.update(comp_disc.sibling_env(comp.name, disc, sibling_ports))
         merged = env_builder.merge_env(engine, env)
         rt.handles[wname] = self.runner.spawn(
             name=wname, runner=engine.runner, args=engine.args,

@@ -52,8 +52,16 @@ class RoleInstanceSetController:
             return 0.0
         update_hash = template_hash(ris)
         instances = {i.metadata.name: i for i in self._owned(ris)}
-        requeue = self._scale(ris, instances, update_hash)
-        requeue = max(requeue, self._rolling_update(ris, instances, update_hash))
+        pattern = ris.metadata.annotations.get(C.ANNO_INSTANCE_PATTERN,
+                                               "Stateful")
+        if pattern == "Stateless":
+            requeue = self._scale_stateless(ris, instances, update_hash)
+            requeue = max(requeue, self._rolling_update_stateless(
+                ris, instances, update_hash))
+        else:
+            requeue = self._scale(ris, instances, update_hash)
+            requeue = max(requeue,
+                          self._rolling_update(ris, instances, update_hash))
         self._update_status(ris, instances, update_hash)
         return requeue
 
@@ -140,6 +148,67 @@ class RoleInstanceSetController:
             if ordered:
                 break
         return requeue
+
+    # ------------------------------------------------------------------
+    # Stateless mode (reference statelessmode/: CloneSet-style control —
+    # random instance ids, priority delete via the role-instance-to-delete
+    # annotation, unready-then-youngest scale-in order)
+
+    def _scale_stateless(self, ris: RoleInstanceSet,
+                         instances: Dict[str, RoleInstance],
+                         update_hash: str) -> float:
+        import uuid
+        live = [i for i in instances.values()
+                if i.metadata.deletion_timestamp is None]
+        want = ris.spec.replicas
+        requeue = 0.0
+        for _ in range(want - len(live)):
+            name = f"{ris.metadata.name}-{uuid.uuid4().hex[:5]}"
+            inst = self._make_instance_named(ris, name, update_hash)
+            self.store.create(inst)
+            requeue = 0.2
+        if len(live) > want:
+            to_delete = {n.strip() for n in ris.metadata.annotations.get(
+                C.ANNO_ROLE_INSTANCE_TO_DELETE, "").split(",") if n.strip()}
+
+            def order(i: RoleInstance):
+                prio = 0 if i.metadata.name in to_delete else 1
+                unready = 0 if not _is_ready(i) else 1
+                return (prio, unready, -i.metadata.creation_timestamp)
+            for inst in sorted(live, key=order)[:len(live) - want]:
+                self._delete_instance(inst)
+                requeue = 0.2
+        return requeue
+
+    def _rolling_update_stateless(self, ris: RoleInstanceSet,
+                                  instances: Dict[str, RoleInstance],
+                                  update_hash: str) -> float:
+        strat = ris.spec.update_strategy
+        if strat.paused:
+            return 0.0
+        live = [i for i in instances.values()
+                if i.metadata.deletion_timestamp is None]
+        stale = [i for i in live
+                 if i.metadata.labels.get(C.LABEL_REVISION_HASH) != update_hash]
+        if not stale:
+            return 0.0
+        not_ready = sum(1 for i in live if not _is_ready(i))
+        budget = max(0, strat.max_unavailable - not_ready)
+        # unready stale first (free progress), then oldest
+        stale.sort(key=lambda i: (_is_ready(i), i.metadata.creation_timestamp))
+        for inst in stale[:budget]:
+            if self._can_update_in_place(ris, inst):
+                self._in_place_update(ris, inst, update_hash)
+            else:
+                self._delete_instance(inst)
+        return 0.2
+
+    def _make_instance_named(self, ris: RoleInstanceSet, name: str,
+                             revision: str) -> RoleInstance:
+        inst = self._make_instance(ris, 0, revision)
+        inst.metadata.name = name
+        inst.metadata.labels[C.LABEL_ROLE_INDEX] = "0"
+        return inst
 
     def _ordinal(self, inst: RoleInstance) -> Optional[int]:
         tail = inst.metadata.name.rsplit("-", 1)[-1]

@@ -1,0 +1,118 @@
+"""Component discovery/ordering + stateless RoleInstanceSet mode
+(reference analogs: pkg/component-discovery tests, statelessmode suite)."""
+import json
+
+import pytest
+
+from rbg_amd.api import constants as C
+from rbg_amd.api.types import (ComponentSpec, CustomComponentsPattern,
+                               EngineResources, EngineSpec, EngineTemplate,
+                               ObjectMeta, RoleBasedGroup, RoleBasedGroupSpec,
+                               RoleSpec)
+from rbg_amd.controller.manager import Manager, ManagerOptions
+from rbg_amd.discovery import component as comp_disc
+from tests.test_controller_e2e import rbg_ready
+
+
+def test_depends_on_parse_and_cycle():
+    deps = comp_disc.parse_depends_on(
+        {C.ANNO_COMPONENT_DEPENDS_ON: json.dumps({"worker": ["leader"]})})
+    assert deps == {"worker": ["leader"]}
+    assert not comp_disc.has_cycle(deps)
+    assert comp_disc.has_cycle({"a": ["b"], "b": ["a"]})
+    # cycle -> parallel fallback: everything may start
+    assert comp_disc.start_gate("a", {"a": ["b"], "b": ["a"]}, {})
+
+
+def test_start_gate():
+    deps = {"worker": ["leader"]}
+    assert comp_disc.start_gate("leader", deps, {})
+    assert not comp_disc.start_gate("worker", deps, {"leader": False})
+    assert comp_disc.start_gate("worker", deps, {"leader": True})
+
+
+def test_sibling_env():
+    env = comp_disc.sibling_env(
+        "worker", [{"name": "LEADER_ADDR", "component": "leader"}],
+        {"leader": [1234]})
+    assert env == {"LEADER_ADDR": "127.0.0.1:1234"}
+
+
+@pytest.fixture
+def mgr(tmp_run_dir):
+    m = Manager(ManagerOptions(run_root=tmp_run_dir, num_gpus=8,
+                               resync_period=0.1))
+    m.start()
+    yield m
+    m.stop()
+
+
+def cpu_engine(runner="echo", args=None):
+    return EngineTemplate(engines=[EngineSpec(
+        name="engine", runner=runner, args=args or {},
+        resources=EngineResources(cpu_only=True))])
+
+
+@pytest.mark.timeout(180)
+def test_component_ordering_e2e(mgr):
+    """customComponents role: 'second' must not start before 'first' is
+    Ready (reference component_ordering e2e case)."""
+    role = RoleSpec(
+        name="combo", replicas=1, pattern=C.PATTERN_CUSTOM_COMPONENTS,
+        custom_components_pattern=CustomComponentsPattern(components=[
+            ComponentSpec(name="first", size=1,
+                          template=cpu_engine(args={"ready_delay": 1.5})),
+            ComponentSpec(name="second", size=1, template=cpu_engine(),
+                          annotations={C.ANNO_COMPONENT_DEPENDS_ON:
+                                       json.dumps({"second": ["first"]})}),
+        ]))
+    mgr.store.create(RoleBasedGroup(metadata=ObjectMeta(name="ord"),
+                                    spec=RoleBasedGroupSpec(roles=[role])))
+    violations = []
+
+    def check():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                               selector={C.LABEL_GROUP_NAME: "ord"})
+        for inst in insts:
+            phases = {w.component: w.phase for w in inst.status.workers}
+            if phases.get("second") not in (None, "Pending") and \
+                    phases.get("first") != "Ready":
+                violations.append(dict(phases))
+        return rbg_ready(mgr, "ord")
+
+    assert mgr.wait_for(check, timeout=60)
+    assert not violations, violations
+
+
+@pytest.mark.timeout(180)
+def test_stateless_mode_scale_and_names(mgr):
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(
+            name="free",
+            annotations={C.ANNO_INSTANCE_PATTERN: "Stateless"}),
+        spec=RoleBasedGroupSpec(roles=[
+            RoleSpec(name="w", replicas=3, template=cpu_engine())]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "free"), timeout=90)
+    insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                           selector={C.LABEL_GROUP_NAME: "free"})
+    assert len(insts) == 3
+    # stateless: random-suffix names, not ordinals
+    suffixes = [i.metadata.name.rsplit("-", 1)[-1] for i in insts]
+    assert not all(s.isdigit() and int(s) < 3 for s in suffixes)
+
+    # priority scale-in via the role-instance-to-delete annotation
+    victim = insts[0].metadata.name
+
+    def scale_down(cur):
+        cur.spec.role("w").replicas = 2
+        cur.metadata.annotations[C.ANNO_ROLE_INSTANCE_TO_DELETE] = victim
+        return cur
+    mgr.store.apply(C.KIND_RBG, "free", scale_down)
+
+    def settled():
+        live = [i for i in mgr.store.list(
+            C.KIND_ROLE_INSTANCE, selector={C.LABEL_GROUP_NAME: "free"})
+            if i.metadata.deletion_timestamp is None]
+        return len(live) == 2 and victim not in [i.metadata.name for i in live]
+    assert mgr.wait_for(settled, timeout=60)
