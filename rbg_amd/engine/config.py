@@ -49,6 +49,11 @@ class ModelConfig:
                                intermediate_size=512, num_layers=2,
                                num_heads=2, num_kv_heads=1, head_dim=128,
                                vocab_size=512, max_position=2048)
+        if name == "tiny-tp":   # TP-divisible CPU test shape
+            return ModelConfig(name="tiny-tp", hidden_size=512,
+                               intermediate_size=1024, num_layers=2,
+                               num_heads=4, num_kv_heads=2, head_dim=128,
+                               vocab_size=512, max_position=2048)
         raise ValueError(f"unknown model preset {name!r}")
 
 

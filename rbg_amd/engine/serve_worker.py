@@ -96,8 +96,28 @@ class ServeWorker:
         self.ctx = ctx
         self.mode = mode
         self.cfg = _engine_cfg(ctx)
-        self.engine = LLMEngine(self.cfg)
-        self.rpc = RpcServer(port=_rpc_port(ctx))
+        # TP rank group (leaderWorker pattern): join the RCCL/xGMI
+        # communicator from the injected RBG_LWP_* env; every rank runs the
+        # engine in lockstep (leader broadcasts request ops each step —
+        # greedy sampling is deterministic across ranks because activations
+        # are all-reduced, so no token broadcast is needed)
+        self.tp = None
+        self.comm = None
+        if ctx.args.get("tp_from_env") and \
+                int(os.environ.get(C.ENV_LWP_GROUP_SIZE, "1") or 1) > 1:
+            from ..models.llama import TPContext
+            from ..parallel import comm
+            self.comm = comm.init_from_env(
+                backend=ctx.args.get("tp_backend"))
+            comm.warmup_collectives(self.comm)
+            self.cfg.tp_size = self.comm.world_size
+            self.cfg.tp_rank = self.comm.rank
+            self.tp = TPContext(size=self.comm.world_size,
+                                rank=self.comm.rank)
+        self.engine = LLMEngine(self.cfg, self.tp)
+        self.is_leader = self.tp is None or self.tp.rank == 0
+        self._pending_ops: List[Dict[str, Any]] = []
+        self.rpc = RpcServer(port=_rpc_port(ctx)) if self.is_leader else None
         self.lock = threading.Lock()
         self.results: Dict[int, Sequence] = {}
         self._finished_pages: Dict[int, List[int]] = {}
@@ -137,21 +157,98 @@ class ServeWorker:
                 backend=self.ctx.args.get("transfer_backend"))
         return self._transfer
 
+    # -- TP lockstep --------------------------------------------------------
+
+    def _tp_submit_op(self, op: Dict[str, Any]) -> Sequence:
+        """Queue an op for broadcast and wait until the lockstep loop has
+        applied it on this (leader) rank; returns the created sequence."""
+        ev = threading.Event()
+        op["_ev"] = ev
+        op["_result"] = None
+        with self.lock:
+            self._pending_ops.append(op)
+        ev.wait(timeout=120)
+        return op["_result"]
+
+    def _tp_apply_op(self, op: Dict[str, Any]) -> None:
+        kind = op.get("kind")
+        if kind == "submit":
+            seq = self.engine.add_request(
+                list(op["tokens"]),
+                SamplingParams(max_new_tokens=int(op["max_new_tokens"]),
+                               temperature=float(op.get("temperature", 0.0))))
+            self.results[seq.seq_id] = seq
+            op["_result"] = seq
+        elif kind == "reload":
+            self.engine.reload_weights(int(op["seed"]))
+        ev = op.get("_ev")
+        if ev is not None:
+            ev.set()
+
+    def _run_tp_lockstep(self) -> None:
+        """Every TP rank executes the identical schedule: the leader
+        broadcasts queued ops each iteration (broadcast_object_list over the
+        group), all ranks apply them and step the engine in lockstep.
+        Sampling needs no cross-rank exchange: activations are all-reduced,
+        so logits (and greedy tokens) are identical on every rank."""
+        import torch.distributed as dist
+        self.ctx.set_ready(rpc_port=self.rpc.port if self.rpc else 0,
+                           mode=self.mode, tp_rank=self.tp.rank,
+                           tp_size=self.tp.size, device=self.cfg.device,
+                           model=self.cfg.model.name)
+        try:
+            while True:
+                if self.is_leader:
+                    with self.lock:
+                        ops, self._pending_ops = self._pending_ops, []
+                    wire = [{k: v for k, v in op.items()
+                             if not k.startswith("_")} for op in ops]
+                    if self.ctx.should_stop():
+                        wire = [{"kind": "stop"}]
+                    payload = [wire]
+                else:
+                    ops = []
+                    payload = [None]
+                dist.broadcast_object_list(payload, src=0)
+                wire = payload[0]
+                if any(op.get("kind") == "stop" for op in wire):
+                    return
+                if self.is_leader:
+                    for op in ops:
+                        self._tp_apply_op(op)
+                else:
+                    for op in wire:
+                        self._tp_apply_op(op)
+                mode = self.engine.step()
+                if mode == "idle" and not wire:
+                    time.sleep(0.002)
+        finally:
+            if self.rpc:
+                self.rpc.stop()
+
     # ------------------------------------------------------------------
 
     def _register_rpc(self) -> None:
+        if self.rpc is None:
+            return     # TP follower ranks take orders via broadcast only
         self.rpc.register("ping", lambda: "pong")
         self.rpc.register("stats", lambda: self.engine.stats.snapshot())
         self.rpc.register("generate", self._rpc_generate)
         self.rpc.register("submit", self._rpc_submit)
         self.rpc.register("poll", self._rpc_poll)
-        self.rpc.register("reload_weights",
-                          lambda seed: self.engine.reload_weights(int(seed)))
+        self.rpc.register("reload_weights", self._rpc_reload)
         self.rpc.register("apply_update", self._rpc_apply_update)
         if self.mode == "prefill":
             self.rpc.register("prefill", self._rpc_prefill)
         if self.mode == "decode":
             self.rpc.register("import_seq", self._rpc_import_seq)
+
+    def _rpc_reload(self, seed) -> None:
+        if self.tp is not None:
+            self._tp_submit_op({"kind": "reload", "seed": int(seed)})
+        else:
+            with self.lock:
+                self.engine.reload_weights(int(seed))
 
     def _rpc_apply_update(self, args: Dict[str, Any]) -> Dict[str, Any]:
         """Live in-place update: apply the arg diffs an engine can absorb
@@ -161,8 +258,7 @@ class ServeWorker:
         new_seed = args.get("weights_seed")
         if new_seed is not None and \
                 int(new_seed) != int(self.ctx.args.get("weights_seed", -1)):
-            with self.lock:
-                self.engine.reload_weights(int(new_seed))
+            self._rpc_reload(int(new_seed))
             self.ctx.args["weights_seed"] = int(new_seed)
             applied.append("weights_seed")
         self.ctx.args.update({k: v for k, v in args.items()
@@ -177,6 +273,13 @@ class ServeWorker:
         if not tokens or min(tokens) < 0 or max(tokens) >= vocab:
             raise ValueError(
                 f"prompt tokens out of range [0,{vocab}) or empty")
+        if self.tp is not None:
+            seq = self._tp_submit_op({"kind": "submit", "tokens": tokens,
+                                      "max_new_tokens": max_new_tokens,
+                                      "temperature": temperature})
+            if seq is None:
+                raise RuntimeError("TP lockstep loop did not apply the op")
+            return seq.seq_id
         with self.lock:
             seq = self.engine.add_request(
                 tokens, SamplingParams(max_new_tokens=max_new_tokens,
@@ -280,7 +383,11 @@ class ServeWorker:
     # -- engine loop ---------------------------------------------------------
 
     def run(self) -> None:
-        self.rpc.start()
+        if self.rpc:
+            self.rpc.start()
+        if self.tp is not None:
+            self._run_tp_lockstep()
+            return
         if self.mode == "prefill":
             # prefill keeps finished sequences' pages for migration: patch
             # the scheduler's release with a park list

@@ -81,11 +81,11 @@ def _http_post(port, path, payload, timeout=120):
         raise AssertionError(f"HTTP {e.code} on {path}: {e.read().decode()}")
 
 
-def _local_reference_tokens(prompt, max_new):
+def _local_reference_tokens(prompt, max_new, model="tiny"):
     from rbg_amd.engine.config import EngineConfig, ModelConfig
     from rbg_amd.engine.engine import LLMEngine
     from rbg_amd.engine.sequence import SamplingParams
-    eng = LLMEngine(EngineConfig(model=ModelConfig.preset("tiny"),
+    eng = LLMEngine(EngineConfig(model=ModelConfig.preset(model),
                                  device="cpu", kv_pool_tokens=4096,
                                  enforce_eager=True))
     (s,) = eng.generate([prompt], SamplingParams(max_new_tokens=max_new))
@@ -153,3 +153,40 @@ def test_pd_disaggregated_serving(mgr):
                       {"prompt_tokens": prompt2, "max_new_tokens": 4},
                       timeout=240)
     assert res2["tokens"] == _local_reference_tokens(prompt2, 4)
+
+
+@pytest.mark.timeout(420)
+def test_tp2_leader_worker_serving(mgr):
+    """leaderWorker pattern (TP=2 rank group over gloo on CPU): the
+    controller injects RBG_LWP_* rank env; the engine forms the group and
+    serves in lockstep.  Degree-invariant weights make the output exactly
+    equal the TP=1 colocated reference."""
+    from rbg_amd.api.types import LeaderWorkerPattern
+    args = dict(ENGINE_ARGS, mode="colocated", model="tiny-tp",
+                cpu_model="tiny-tp", tp_from_env=True, tp_backend="gloo")
+    role = RoleSpec(
+        name="worker", replicas=1, dependencies=["router"],
+        pattern=C.PATTERN_LEADER_WORKER,
+        leader_worker_pattern=LeaderWorkerPattern(size=2),
+        template=EngineTemplate(engines=[EngineSpec(
+            name="engine", runner="llm-engine", args=args,
+            resources=EngineResources(cpu_only=True))]))
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="tp"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("colocated", {"worker_roles": ["worker"],
+                                      "vocab_size": 500}),
+            role,
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "tp"), timeout=180)
+    assert mgr.wait_for(
+        lambda: (_router_http_port(mgr, "tp") is not None), timeout=30)
+    port = _router_http_port(mgr, "tp")
+    torch.manual_seed(21)
+    prompt = torch.randint(0, 500, (14,)).tolist()
+    res = _http_post(port, "/generate",
+                     {"prompt_tokens": prompt, "max_new_tokens": 5},
+                     timeout=240)
+    assert len(res["tokens"]) == 5
+    assert res["tokens"] == _local_reference_tokens(prompt, 5, model="tiny-tp")
