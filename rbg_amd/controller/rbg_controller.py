@@ -228,6 +228,7 @@ class RoleBasedGroupController:
                                partitions: Dict[str, int]) -> None:
         name = ris_name(rbg.metadata.name, role.name)
         components = expand_pattern(rbg, role)
+        components = self._apply_engine_runtimes(rbg, role, components)
         replicas = role.replicas
         if role.name in scale_caps:
             replicas = min(role.replicas, scale_caps[role.name])
@@ -284,6 +285,36 @@ class RoleBasedGroupController:
                 self.store.apply(C.KIND_ROLE_INSTANCE_SET, name,
                                  lambda cur: fill(cur),
                                  rbg.metadata.namespace)
+
+    def _apply_engine_runtimes(self, rbg: RoleBasedGroup, role: RoleSpec,
+                               components: List[ComponentSpec]
+                               ) -> List[ComponentSpec]:
+        """Inject ClusterEngineRuntimeProfile bundles named by the role
+        (reference pkg/discovery/sidecar_builder.go:47-158): profile env is
+        prepended (user env wins), profile args are defaults under the
+        engine's own args."""
+        if not role.engine_runtimes:
+            return components
+        from ..api.types import EnvVar
+        components = [clone(c) for c in components]
+        for prof_name in role.engine_runtimes:
+            prof = self.store.try_get(C.KIND_ENGINE_RUNTIME_PROFILE,
+                                      prof_name, rbg.metadata.namespace)
+            if prof is None:
+                log.warning("engine runtime profile %r not found", prof_name)
+                continue
+            for comp in components:
+                if comp.template is None:
+                    continue
+                for eng in comp.template.engines:
+                    have = {e.name for e in eng.env}
+                    eng.env = [EnvVar(name=e.name, value=e.value)
+                               for e in prof.spec.env
+                               if e.name not in have] + eng.env
+                    merged = dict(prof.spec.args)
+                    merged.update(eng.args)
+                    eng.args = merged
+        return components
 
     # ------------------------------------------------------------------
 

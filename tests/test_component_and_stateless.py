@@ -116,3 +116,24 @@ def test_stateless_mode_scale_and_names(mgr):
             if i.metadata.deletion_timestamp is None]
         return len(live) == 2 and victim not in [i.metadata.name for i in live]
     assert mgr.wait_for(settled, timeout=60)
+
+
+@pytest.mark.timeout(180)
+def test_engine_runtime_profile_injection(mgr):
+    from rbg_amd.api.types import (ClusterEngineRuntimeProfile,
+                                   EngineRuntimeProfileSpec, EnvVar)
+    prof = ClusterEngineRuntimeProfile(
+        metadata=ObjectMeta(name="base-runtime"),
+        spec=EngineRuntimeProfileSpec(
+            env=[EnvVar(name="RUNTIME_FLAG", value="on")],
+            args={"ready_delay": 0.1, "crash_after": 0}))
+    mgr.store.create(prof)
+    role = RoleSpec(name="w", replicas=1, template=cpu_engine(),
+                    engine_runtimes=["base-runtime"])
+    mgr.store.create(RoleBasedGroup(metadata=ObjectMeta(name="prof"),
+                                    spec=RoleBasedGroupSpec(roles=[role])))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "prof"), timeout=60)
+    ris = mgr.store.get(C.KIND_ROLE_INSTANCE_SET, "prof-w")
+    eng = ris.spec.template.components[0].template.engines[0]
+    assert {"RUNTIME_FLAG": "on"} == {e.name: e.value for e in eng.env}
+    assert eng.args["ready_delay"] == 0.1
