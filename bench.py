@@ -26,7 +26,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=64)
     p.add_argument("--warmup", type=int, default=16)
-    p.add_argument("--batch", type=int, default=64,
+    p.add_argument("--batch", type=int, default=128,
                    help="decode batch per GPU")
     p.add_argument("--seq-len", type=int, default=2048,
                    help="synthetic prompt length")

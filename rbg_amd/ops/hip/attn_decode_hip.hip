@@ -89,31 +89,27 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
             (kslot & ps_mask)) * HEAD_DIM;
   };
 
-  // wave stride: 4 waves x 4 keys; a 2-ahead software-prefetch ring keeps
-  // ~8 KB per wave in flight (latency-bound regime: the shuffles/VALU of
-  // one iteration cover only part of the ~800-cycle HBM latency)
+  // wave stride: 4 waves x 4 keys; 1-ahead software prefetch keeps the next
+  // iteration's K AND V in flight under the current iteration's VALU work.
+  // (A 2-ahead ring measured WORSE: 3.45 -> 2.68 TB/s at batch 64 —
+  // profiles/decode_breakdown.md; the ring's loop-carried conditionals
+  // defeat the schedule.)
   int base = key_begin + wave * 4;
-  Bf16x8U k_ring[2], v_ring[2];
-#pragma unroll
-  for (int s = 0; s < 2; ++s) {
-    const int pre = base + s * 16;
-    if (pre < key_end) {
-      const size_t off = row_offset(pre);
-      k_ring[s].u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
-      v_ring[s].u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
-    }
+  Bf16x8U k_pref, v_pref;
+  if (base < key_end) {
+    const size_t off = row_offset(base);
+    k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
+    v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
   }
-  int ring = 0;
   for (; base < key_end; base += 16) {
     const bool valid = base + group < key_end;
-    Bf16x8U kv = k_ring[ring], vv = v_ring[ring];
-    const int nxt = base + 32;
+    Bf16x8U kv = k_pref, vv = v_pref;
+    const int nxt = base + 16;
     if (nxt < key_end) {
       const size_t off = row_offset(nxt);
-      k_ring[ring].u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
-      v_ring[ring].u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
+      k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
+      v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
     }
-    ring ^= 1;
     float kf[8], vf[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {

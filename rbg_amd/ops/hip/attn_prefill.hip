@@ -40,6 +40,19 @@ DEV_INLINE mfma_bf8 lds_bf8(const __hip_bfloat16* p) {
   return cvt.v;
 }
 
+// V image element offset for the ds_read_b64_tr_b16 PV path.  Per 32-key
+// step (ks) and 16-dim column tile (dt), a 16-lane group's [4 key][16 dim]
+// block lives at elems {q*64 + j*16 + d16} with the tr-read redistributing
+// it so lane l ends with keys q*8+half*4+j at its dim column (l&15) —
+// exactly the MFMA B-fragment (guide §2 tr-read layout; conflict-free
+// subtiling).  Writes land as contiguous 8-element (16 B) runs per
+// (key, 8-dim chunk), so staging stays fully vectorized.
+DEV_INLINE int v_img_off(int key, int dim) {
+  const int ks = key >> 5, kk = key & 31, q = kk >> 3, jj = kk & 7;
+  return ks * 4096 + (dim >> 4) * 512 + (jj >> 2) * 256 + q * 64 +
+         (jj & 3) * 16 + (dim & 15);
+}
+
 // grid.x = total q-blocks (host-computed map), grid.y = num q heads.
 __global__ __launch_bounds__(256) void prefill_attn_kernel(
     __hip_bfloat16* __restrict__ out,        // [T, QH, D]
@@ -62,7 +75,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   const int gslice = lane >> 4;      // fragment k-slice (x8 elements)
 
   __shared__ __hip_bfloat16 k_lds[KTILE][HEAD_DIM + PAD];
-  __shared__ __hip_bfloat16 v_lds[HEAD_DIM][KTILE + PAD];   // transposed
+  __shared__ __hip_bfloat16 v_img[KTILE * HEAD_DIM];   // tr-read image
   __shared__ __hip_bfloat16 p_lds[4][16][KTILE + PAD];
 
   // ---- Q fragments: wave's 16 rows, 4 k-steps of 32 dims ---------------
@@ -108,9 +121,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
         vv = *reinterpret_cast<const uint4*>(
             v + ((size_t)(seq_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
             chunk);
-      const __hip_bfloat16* ve = reinterpret_cast<const __hip_bfloat16*>(&vv);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) v_lds[chunk + j][key] = ve[j];
+      *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) = vv;
     }
     __syncthreads();
 
@@ -170,15 +181,46 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     // explicitly (no cross-wave sharing, so no barrier needed)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    // ---- O += P V : 8 col-tiles x 2 key-steps ---------------------------
+    // ---- O += P V : 8 col-tiles x 2 key-steps; V arrives by batched
+    // hardware-transpose reads (ds_read_b64_tr_b16) ------------------------
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       mfma_bf8 pa = lds_bf8(&p_lds[wave][gl][ks * 32 + gslice * 8]);
+      const unsigned vaddr = (unsigned)(unsigned long long)(
+          &v_img[ks * 4096 + gslice * 64 + gl * 4]);
+      unsigned long long vlo[8], vhi[8];
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %16\n\t"
+          "ds_read_b64_tr_b16 %1, %16 offset:512\n\t"
+          "ds_read_b64_tr_b16 %2, %16 offset:1024\n\t"
+          "ds_read_b64_tr_b16 %3, %16 offset:1536\n\t"
+          "ds_read_b64_tr_b16 %4, %16 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %5, %16 offset:2560\n\t"
+          "ds_read_b64_tr_b16 %6, %16 offset:3072\n\t"
+          "ds_read_b64_tr_b16 %7, %16 offset:3584\n\t"
+          "ds_read_b64_tr_b16 %8, %16 offset:4096\n\t"
+          "ds_read_b64_tr_b16 %9, %16 offset:4608\n\t"
+          "ds_read_b64_tr_b16 %10, %16 offset:5120\n\t"
+          "ds_read_b64_tr_b16 %11, %16 offset:5632\n\t"
+          "ds_read_b64_tr_b16 %12, %16 offset:6144\n\t"
+          "ds_read_b64_tr_b16 %13, %16 offset:6656\n\t"
+          "ds_read_b64_tr_b16 %14, %16 offset:7168\n\t"
+          "ds_read_b64_tr_b16 %15, %16 offset:7680\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(vlo[0]), "=&v"(vhi[0]), "=&v"(vlo[1]), "=&v"(vhi[1]),
+            "=&v"(vlo[2]), "=&v"(vhi[2]), "=&v"(vlo[3]), "=&v"(vhi[3]),
+            "=&v"(vlo[4]), "=&v"(vhi[4]), "=&v"(vlo[5]), "=&v"(vhi[5]),
+            "=&v"(vlo[6]), "=&v"(vhi[6]), "=&v"(vlo[7]), "=&v"(vhi[7])
+          : "v"(vaddr)
+          : "memory");
+      __builtin_amdgcn_sched_barrier(0);   // rule 18: keep MFMAs below
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
-        mfma_bf8 vf = lds_bf8(&v_lds[dt * 16 + gl][ks * 32 + gslice * 8]);
-        acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf, acc_o[dt],
-                                                            0, 0, 0);
+        union { struct { unsigned long long lo, hi; } u; mfma_bf8 v; } vf;
+        vf.u.lo = vlo[dt];
+        vf.u.hi = vhi[dt];
+        acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf.v,
+                                                            acc_o[dt], 0, 0, 0);
       }
     }
   }
