@@ -61,6 +61,36 @@ class Daemon:
             return asdict(store.update(typed))
         reg("store_update", store_update)
 
+        def metrics() -> str:
+            """Prometheus text-format gauges (reference: controller-runtime
+            metrics server + status conditions as the public state machine)."""
+            from ..api.types import get_condition
+            lines = [
+                "# TYPE rbg_groups gauge",
+                f"rbg_groups {len(store.list(C.KIND_RBG, namespace=None))}",
+                "# TYPE rbg_free_gpus gauge",
+                f"rbg_free_gpus {len(self.manager.gang.free_gpus())}",
+            ]
+            lines.append("# TYPE rbg_group_ready gauge")
+            for rbg in store.list(C.KIND_RBG, namespace=None):
+                c = get_condition(rbg.status.conditions, C.COND_READY)
+                val = 1 if (c is not None and c.status == "True") else 0
+                lines.append(
+                    f'rbg_group_ready{{group="{rbg.metadata.name}"}} {val}')
+            lines.append("# TYPE rbg_instance_restarts counter")
+            lines.append("# TYPE rbg_instance_recovery_seconds gauge")
+            for inst in store.list(C.KIND_ROLE_INSTANCE, namespace=None):
+                n = inst.metadata.name
+                lines.append(
+                    f'rbg_instance_restarts{{instance="{n}"}} '
+                    f"{inst.status.restart_count}")
+                if inst.status.last_recovery_duration:
+                    lines.append(
+                        f'rbg_instance_recovery_seconds{{instance="{n}"}} '
+                        f"{inst.status.last_recovery_duration:.3f}")
+            return "\n".join(lines) + "\n"
+        reg("metrics", metrics)
+
         def store_delete(kind: str, name: str, namespace: str = "default"):
             # graceful: mark for deletion so controllers tear down processes
             obj = store.try_get(kind, name, namespace)

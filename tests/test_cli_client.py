@@ -128,3 +128,13 @@ def test_daemon_healthz(daemon):
     c = RpcClient("127.0.0.1", daemon.port)
     h = c.call("healthz")
     assert h["status"] == "ok" and h["gpus"] == 8
+
+
+def test_daemon_metrics(daemon):
+    from rbg_amd.server.rpc import RpcClient
+    daemon.manager.store.create(make_rbg("metrics-demo"))
+    c = RpcClient("127.0.0.1", daemon.port)
+    text = c.call("metrics")
+    assert "rbg_groups 1" in text
+    assert 'rbg_group_ready{group="metrics-demo"}' in text
+    assert "rbg_free_gpus" in text
