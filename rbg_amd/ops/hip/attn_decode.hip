@@ -20,7 +20,9 @@ constexpr float NEG_INF = -1e30f;
 
 // QPG = q heads per kv head (GQA group). One block = (seq, kv_head, split).
 // 4 waves; each wave covers 4 keys per iteration (16-lane groups, 16 B/lane).
-template <int QPG>
+// WIDE=1: 8 keys per iteration with branch-free clamped prefetch (2x the
+// loads in flight; A/B-selected at runtime via tools/profile_decode.py).
+template <int QPG, int WIDE>
 __global__ __launch_bounds__(256) void decode_attn_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
@@ -88,34 +90,15 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
             (kslot & ps_mask)) * HEAD_DIM;
   };
 
-  // wave stride: 4 waves x 4 keys; 1-ahead software prefetch keeps the next
-  // iteration's K AND V in flight under the current iteration's VALU work.
-  // (A 2-ahead ring measured WORSE: 3.45 -> 2.68 TB/s at batch 64 —
-  // profiles/decode_breakdown.md; the ring's loop-carried conditionals
-  // defeat the schedule.)
-  int base = key_begin + wave * 4;
-  Bf16x8U k_pref, v_pref;
-  if (base < key_end) {
-    const size_t off = row_offset(base);
-    k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
-    v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
-  }
-  for (; base < key_end; base += 16) {
-    const bool valid = base + group < key_end;
-    Bf16x8U kv = k_pref, vv = v_pref;
-    const int nxt = base + 16;
-    if (nxt < key_end) {
-      const size_t off = row_offset(nxt);
-      k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
-      v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
-    }
+  // One 4-key quad of online-softmax work (scores via 16-lane-group dot
+  // reduction, tile max/psum across the wave's quad, V accumulate).
+  auto process4 = [&](const Bf16x8U& kv, const Bf16x8U& vv, bool valid) {
     float kf[8], vf[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       kf[j] = bf2f(kv.e[j]);
       vf[j] = bf2f(vv.e[j]);
     }
-    // scores for this key, all QPG heads
     float p[QPG];
     float tile_max[QPG];
 #pragma unroll
@@ -125,7 +108,6 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
       for (int j = 0; j < 8; ++j) s += qf[h][j] * kf[j];
       s = group16_sum(s);             // full dot across the 16-lane group
       if (!valid) s = NEG_INF;
-      // tile max across the wave's 4 keys
       float tm = s;
       tm = fmaxf(tm, __shfl_xor(tm, 16, 64));
       tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
@@ -137,7 +119,6 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
       const float m_new = fmaxf(m[h], tile_max[h]);
       const float alpha = __expf(m[h] - m_new);
       const float pv = valid ? __expf(p[h] - m_new) : 0.f;
-      // sum of p over the 4 keys of this wave-iteration
       float psum = pv;
       psum += __shfl_xor(psum, 16, 64);
       psum += __shfl_xor(psum, 32, 64);
@@ -146,6 +127,46 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         acc[h][j] = acc[h][j] * alpha + pv * vf[j];
+    }
+  };
+
+  auto load_pair = [&](int b, Bf16x8U& kd, Bf16x8U& vd) {
+    const size_t off = row_offset(b);
+    kd.u = *reinterpret_cast<const uint4*>(key_cache + off + dbase);
+    vd.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
+  };
+
+  if (WIDE) {
+    // 8 keys per wave-iteration, branch-free clamped prefetch: 8 loads in
+    // flight per wave without loop-carried conditionals.
+    int base = key_begin + wave * 8;
+    Bf16x8U kA, vA, kB, vB, kC, vC, kD, vD;
+    if (base < key_end) {
+      load_pair(base, kA, vA);
+      load_pair(base + 4, kB, vB);
+    }
+    for (; base < key_end; base += 32) {
+      load_pair(base + 32 < key_end ? base + 32 : base, kC, vC);
+      load_pair(base + 36 < key_end ? base + 36 : base, kD, vD);
+      process4(kA, vA, base + group < key_end);
+      process4(kB, vB, base + 4 + group < key_end);
+      kA = kC; vA = vC; kB = kD; vB = vD;
+    }
+  } else {
+    // 4 keys per wave-iteration with 1-ahead prefetch.  (A 2-ahead ring
+    // measured WORSE: 3.45 -> 2.68 TB/s at batch 64 — its loop-carried
+    // conditional loads de-pipeline the schedule; profiles/decode_breakdown.md.)
+    int base = key_begin + wave * 4;
+    Bf16x8U k_pref, v_pref;
+    if (base < key_end)
+      load_pair(base, k_pref, v_pref);
+    for (; base < key_end; base += 16) {
+      const bool valid = base + group < key_end;
+      Bf16x8U kv = k_pref, vv = v_pref;
+      const int nxt = base + 16;
+      if (nxt < key_end)
+        load_pair(nxt, k_pref, v_pref);
+      process4(kv, vv, valid);
     }
   }
 
@@ -239,22 +260,26 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                              const void* context_lens, float scale,
                              int num_seqs, int num_q_heads, int num_kv_heads,
                              int page_size, int max_pages, int num_splits,
-                             hipStream_t stream) {
+                             int wide, hipStream_t stream) {
   const int qpg = num_q_heads / num_kv_heads;
   dim3 grid(num_kv_heads, num_seqs, num_splits), block(256);
-#define LAUNCH_QPG(QPG)                                                       \
-  hipLaunchKernelGGL(decode_attn_kernel<QPG>, grid, block, 0, stream,         \
+#define LAUNCH_QPG(QPG, WIDE)                                                 \
+  hipLaunchKernelGGL((decode_attn_kernel<QPG, WIDE>), grid, block, 0, stream, \
                      (float*)partial_o, (float*)partial_ml,                   \
                      (__hip_bfloat16*)out, (const __hip_bfloat16*)q,          \
                      (const __hip_bfloat16*)key_cache,                        \
                      (const __hip_bfloat16*)val_cache,                        \
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits)
-  switch (qpg) {
-    case 1: LAUNCH_QPG(1); break;
-    case 2: LAUNCH_QPG(2); break;
-    case 4: LAUNCH_QPG(4); break;
-    case 8: LAUNCH_QPG(8); break;
+  switch (qpg * 2 + (wide ? 1 : 0)) {
+    case 2: LAUNCH_QPG(1, 0); break;
+    case 3: LAUNCH_QPG(1, 1); break;
+    case 4: LAUNCH_QPG(2, 0); break;
+    case 5: LAUNCH_QPG(2, 1); break;
+    case 8: LAUNCH_QPG(4, 0); break;
+    case 9: LAUNCH_QPG(4, 1); break;
+    case 16: LAUNCH_QPG(8, 0); break;
+    case 17: LAUNCH_QPG(8, 1); break;
     default: return;   // validated host-side
   }
 #undef LAUNCH_QPG

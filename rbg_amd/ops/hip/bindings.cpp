@@ -15,7 +15,7 @@ void launch_rope_store_kv(void*, void*, const void*, void*, void*,
                           int, int, int, hipStream_t);
 void launch_decode_attention(void*, void*, void*, const void*, const void*,
                              const void*, const void*, const void*, float,
-                             int, int, int, int, int, int, hipStream_t);
+                             int, int, int, int, int, int, int, hipStream_t);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const void*, const void*, float, int, int, int,
                               hipStream_t);
@@ -96,7 +96,7 @@ torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
                                torch::Tensor value_cache,
                                torch::Tensor block_tables,
                                torch::Tensor context_lens, double scale,
-                               int64_t num_splits) {
+                               int64_t num_splits, int64_t wide) {
   check_bf16_contig(q, "q");
   const int num_seqs = q.size(0);
   const int num_q_heads = q.size(1);
@@ -124,7 +124,7 @@ torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
                           key_cache.data_ptr(), value_cache.data_ptr(),
                           block_tables.data_ptr(), context_lens.data_ptr(),
                           (float)scale, num_seqs, num_q_heads, num_kv_heads,
-                          page_size, max_pages, (int)num_splits,
+                          page_size, max_pages, (int)num_splits, (int)wide,
                           current_stream());
   return out;
 }

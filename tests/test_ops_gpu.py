@@ -117,10 +117,13 @@ def test_decode_attention(dev, qpg, splits):
         nxt += n
     q = torch.randn(S, QH, D, dtype=torch.bfloat16, device=dev)
     scale = 1.0 / math.sqrt(D)
-    got = ops._hip.decode_attention(q, kc, vc, bt, ctx_lens, scale, splits)
+    got = ops._hip.decode_attention(q, kc, vc, bt, ctx_lens, scale, splits, 0)
+    got_w = ops._hip.decode_attention(q, kc, vc, bt, ctx_lens, scale, splits, 1)
     want = ref.decode_attention(q, kc, vc, bt, ctx_lens, scale)
     err = (got.float() - want.float()).abs().max().item()
     assert err < 3e-2, f"qpg={qpg} splits={splits} err={err}"
+    err_w = (got_w.float() - want.float()).abs().max().item()
+    assert err_w < 3e-2, f"WIDE qpg={qpg} splits={splits} err={err_w}"
 
 
 @pytest.mark.parametrize("lens", [[128], [64, 200, 1], [2048], [33, 129]])

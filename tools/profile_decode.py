@@ -55,12 +55,13 @@ def main():
     ctx_lens = torch.full((B,), CTX, dtype=torch.int32, device=dev)
     q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev)
     kv_bytes = B * CTX * KVH * D * 2 * 2
-    for splits in (1, 2, 4, 8):
-        ms = timeit(lambda: ops._hip.decode_attention(
-            q, kc, vc, bt, ctx_lens, 0.088, splits))
-        report[f"attn_decode_split{splits}_ms"] = round(ms, 4)
-        report[f"attn_decode_split{splits}_GBps"] = round(
-            kv_bytes / ms / 1e6, 1)
+    for wide in (0, 1):
+        for splits in (1, 2, 4, 8):
+            ms = timeit(lambda: ops._hip.decode_attention(
+                q, kc, vc, bt, ctx_lens, 0.088, splits, wide))
+            key = f"attn_decode_w{wide}_split{splits}"
+            report[key + "_ms"] = round(ms, 4)
+            report[key + "_GBps"] = round(kv_bytes / ms / 1e6, 1)
     report["attn_kv_bytes_per_layer_MB"] = kv_bytes // (1 << 20)
 
     # ---- GEMMs at decode shapes ------------------------------------------
