@@ -91,3 +91,28 @@ def test_group_recovery_time_recorded(mgr):
     dur = max(i.status.last_recovery_duration for i in insts)
     # crash -> backoff(0.2s) -> respawn -> ready; should be seconds, not minutes
     assert 0.0 < dur < 60.0
+
+
+@pytest.mark.timeout(300)
+def test_external_kill_recovers(mgr):
+    """Fault injection (reference e2e stability.go analog): SIGKILL a worker
+    process from outside; the gang recreates and the group returns Ready."""
+    import os
+    import signal
+    from tests.test_controller_e2e import router_worker_rbg
+    mgr.store.create(router_worker_rbg(name="kill", worker_replicas=2))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "kill"), timeout=60)
+    insts = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+        C.LABEL_GROUP_NAME: "kill", C.LABEL_ROLE_NAME: "worker"})
+    victim_pid = insts[0].status.workers[0].pid
+    os.kill(victim_pid, signal.SIGKILL)
+
+    def recovered():
+        if not rbg_ready(mgr, "kill"):
+            return False
+        insts2 = mgr.store.list(C.KIND_ROLE_INSTANCE, selector={
+            C.LABEL_GROUP_NAME: "kill", C.LABEL_ROLE_NAME: "worker"})
+        pids = [w.pid for i in insts2 for w in i.status.workers]
+        return victim_pid not in pids and all(
+            w.phase == "Ready" for i in insts2 for w in i.status.workers)
+    assert mgr.wait_for(recovered, timeout=90)
