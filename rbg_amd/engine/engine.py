@@ -83,18 +83,9 @@ class LLMEngine:
         return mode
 
     def _collect_finished(self) -> None:
-        # sequences leave scheduler.running when finished; track them here
+        # sequences leave scheduler.running when finished; callers keep their
+        # own Sequence handles (status/output live on the object)
         pass
-
-    def run_until_done(self, max_steps: int = 1_000_000) -> List[Sequence]:
-        done: List[Sequence] = []
-        seen = set()
-        all_seqs: List[Sequence] = []
-        for _ in range(max_steps):
-            if not self.scheduler.has_work():
-                break
-            self.step()
-        return done
 
     def generate(self, prompts: List[List[int]],
                  sampling: Optional[SamplingParams] = None,
