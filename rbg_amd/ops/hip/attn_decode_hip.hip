@@ -24,7 +24,7 @@ constexpr float NEG_INF = -1e30f;
 // WIDE=1: 8 keys per iteration with branch-free clamped prefetch (2x the
 // loads in flight; A/B-selected at runtime via tools/profile_decode.py).
 template <int QPG, int WIDE>
-__global__ __launch_bounds__(256) void decode_attn_kernel(
+__global__ __launch_bounds__(256, QPG <= 4 ? 5 : 3) void decode_attn_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
     __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
@@ -55,8 +55,11 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
   const int key_begin = split * per_split * chunk;
   const int key_end = min(ctx, (split + 1) * per_split * chunk);
 
-  // q fragment: this lane's 8 dims for each of the QPG heads
-  float qf[QPG][8];
+  // q fragment: this lane's 8 dims per head, kept PACKED bf16 (v_dot2
+  // consumes pairs directly — halves the q registers and drops the per-key
+  // f32 conversions; scale is applied to the f32 dot result instead)
+  typedef __attribute__((ext_vector_type(2))) __bf16 bfpair;
+  bfpair qp[QPG][4];
   {
     const __hip_bfloat16* qrow =
         q + ((size_t)seq * num_q_heads + kvh * QPG) * HEAD_DIM;
@@ -65,7 +68,8 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
       Bf16x8U qv;
       qv.u = *reinterpret_cast<const uint4*>(qrow + h * HEAD_DIM + dbase);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) qf[h][j] = bf2f(qv.e[j]) * scale;
+      for (int j = 0; j < 4; ++j)
+        qp[h][j] = reinterpret_cast<const bfpair*>(&qv)[j];
     }
   }
 
@@ -94,19 +98,16 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
   // One 4-key quad of online-softmax work (scores via 16-lane-group dot
   // reduction, tile max/psum across the wave's quad, V accumulate).
   auto process4 = [&](const Bf16x8U& kv, const Bf16x8U& vv, bool valid) {
-    float kf[8], vf[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      kf[j] = bf2f(kv.e[j]);
-      vf[j] = bf2f(vv.e[j]);
-    }
+    const bfpair* kp = reinterpret_cast<const bfpair*>(&kv);
     float p[QPG];
     float tile_max[QPG];
 #pragma unroll
     for (int h = 0; h < QPG; ++h) {
       float s = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) s += qf[h][j] * kf[j];
+      for (int j = 0; j < 4; ++j)
+        s = __builtin_amdgcn_fdot2_f32_bf16(qp[h][j], kp[j], s, false);
+      s *= scale;
       s = group16_sum(s);             // full dot across the 16-lane group
       if (!valid) s = NEG_INF;
       float tm = s;
@@ -127,7 +128,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
       m[h] = m_new;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        acc[h][j] = acc[h][j] * alpha + pv * vf[j];
+        acc[h][j] = acc[h][j] * alpha + pv * bf2f(vv.e[j]);
     }
   };
 
