@@ -46,8 +46,11 @@ def main() -> int:
     if distributed:
         import torch.distributed as dist
         local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+            dist.init_process_group("nccl")   # = RCCL over xGMI on ROCm
+        else:
+            dist.init_process_group("gloo")   # CPU test path
     device = args.device
     if device == "cuda" and not torch.cuda.is_available():
         print(json.dumps({"error": "no GPU visible"}), flush=True)
@@ -111,7 +114,9 @@ def main() -> int:
         import torch.distributed as dist
         dist.barrier()
         t = torch.tensor([elapsed], dtype=torch.float64,
-                         device="cuda" if device == "cuda" else "cpu")
+                         device="cuda" if (device == "cuda" and
+                                           torch.cuda.is_available())
+                         else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
@@ -123,7 +128,8 @@ def main() -> int:
     if rank == 0:
         baseline = 1300.41
         print(json.dumps({
-            "metric": "output tok/s (Llama-3-8B serving, continuous batching)",
+            "metric": f"output tok/s ({model_cfg.name} serving, "
+                      "continuous batching)",
             "value": round(total_tok_s, 2),
             "unit": "tok/s",
             "n_gpus": n_gpus,
