@@ -79,22 +79,22 @@ def pick_decode_splits(num_seqs: int, num_kv_heads: int,
     return int(min(want, by_ctx, 16))
 
 
-DECODE_WIDE = int(__import__("os").environ.get("RBG_DECODE_WIDE", "0"))
+DECODE_VARIANT = int(__import__("os").environ.get("RBG_DECODE_VARIANT", "1"))
 
 
 def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
                      scale: float, num_splits: int = 0,
-                     wide: int = -1) -> torch.Tensor:
+                     variant: int = -1) -> torch.Tensor:
     if _on_gpu(q):
         if num_splits <= 0:
             max_ctx = int(context_lens.max().item()) if context_lens.numel() else 1
             num_splits = pick_decode_splits(q.shape[0], key_cache.shape[1],
                                             max_ctx)
-        if wide < 0:
-            wide = DECODE_WIDE
+        if variant < 0:
+            variant = DECODE_VARIANT
         return _require_hip().decode_attention(
             q, key_cache, value_cache, block_tables, context_lens, scale,
-            num_splits, wide)
+            num_splits, variant)
     return reference.decode_attention(q, key_cache, value_cache, block_tables,
                                       context_lens, scale)
 

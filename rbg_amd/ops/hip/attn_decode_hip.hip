@@ -20,11 +20,14 @@ constexpr int HEAD_DIM = 128;
 constexpr float NEG_INF = -1e30f;
 
 // QPG = q heads per kv head (GQA group). One block = (seq, kv_head, split).
-// 4 waves; each wave covers 4 keys per iteration (16-lane groups, 16 B/lane).
-// WIDE=1: 8 keys per iteration with branch-free clamped prefetch (2x the
-// loads in flight; A/B-selected at runtime via tools/profile_decode.py).
-template <int QPG, int WIDE>
-__global__ __launch_bounds__(256, QPG <= 4 ? 5 : 3) void decode_attn_kernel(
+// 4 waves; each wave covers 4 keys per iteration (16-lane groups, 16 B/lane),
+// 1-ahead K/V prefetch.
+// VARIANT 0: f32 FMA scores (more VGPRs, 3-4 waves/SIMD);
+// VARIANT 1: packed-bf16 v_dot2 scores (96 VGPRs, 5 waves/SIMD for QPG<=4).
+// Runtime-selected (RBG_DECODE_VARIANT) for within-probe A/B.
+template <int QPG, int VARIANT>
+__global__ __launch_bounds__(256, (VARIANT == 1 && QPG <= 4) ? 5 : 2)
+void decode_attn_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
     __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
@@ -55,11 +58,11 @@ __global__ __launch_bounds__(256, QPG <= 4 ? 5 : 3) void decode_attn_kernel(
   const int key_begin = split * per_split * chunk;
   const int key_end = min(ctx, (split + 1) * per_split * chunk);
 
-  // q fragment: this lane's 8 dims per head, kept PACKED bf16 (v_dot2
-  // consumes pairs directly — halves the q registers and drops the per-key
-  // f32 conversions; scale is applied to the f32 dot result instead)
+  // q fragment: VARIANT 1 keeps it packed bf16 (v_dot2 pairs, half the
+  // registers); VARIANT 0 pre-converts to f32 with the scale folded in
   typedef __attribute__((ext_vector_type(2))) __bf16 bfpair;
-  bfpair qp[QPG][4];
+  bfpair qp[VARIANT == 1 ? QPG : 1][4];
+  float qf[VARIANT == 0 ? QPG : 1][8];
   {
     const __hip_bfloat16* qrow =
         q + ((size_t)seq * num_q_heads + kvh * QPG) * HEAD_DIM;
@@ -67,9 +70,14 @@ __global__ __launch_bounds__(256, QPG <= 4 ? 5 : 3) void decode_attn_kernel(
     for (int h = 0; h < QPG; ++h) {
       Bf16x8U qv;
       qv.u = *reinterpret_cast<const uint4*>(qrow + h * HEAD_DIM + dbase);
+      if (VARIANT == 1) {
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        qp[h][j] = reinterpret_cast<const bfpair*>(&qv)[j];
+        for (int j = 0; j < 4; ++j)
+          qp[h][j] = reinterpret_cast<const bfpair*>(&qv)[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) qf[h][j] = bf2f(qv.e[j]) * scale;
+      }
     }
   }
 
@@ -99,15 +107,25 @@ __global__ __launch_bounds__(256, QPG <= 4 ? 5 : 3) void decode_attn_kernel(
   // reduction, tile max/psum across the wave's quad, V accumulate).
   auto process4 = [&](const Bf16x8U& kv, const Bf16x8U& vv, bool valid) {
     const bfpair* kp = reinterpret_cast<const bfpair*>(&kv);
+    float kf[VARIANT == 0 ? 8 : 1];
+    if (VARIANT == 0) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kf[j] = bf2f(kv.e[j]);
+    }
     float p[QPG];
     float tile_max[QPG];
 #pragma unroll
     for (int h = 0; h < QPG; ++h) {
       float s = 0.f;
+      if (VARIANT == 1) {
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        s = __builtin_amdgcn_fdot2_f32_bf16(qp[h][j], kp[j], s, false);
-      s *= scale;
+        for (int j = 0; j < 4; ++j)
+          s = __builtin_amdgcn_fdot2_f32_bf16(qp[h][j], kp[j], s, false);
+        s *= scale;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s += qf[h][j] * kf[j];
+      }
       s = group16_sum(s);             // full dot across the 16-lane group
       if (!valid) s = NEG_INF;
       float tm = s;
@@ -138,26 +156,10 @@ __global__ __launch_bounds__(256, QPG <= 4 ? 5 : 3) void decode_attn_kernel(
     vd.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
   };
 
-  if (WIDE) {
-    // 8 keys per wave-iteration, branch-free clamped prefetch: 8 loads in
-    // flight per wave without loop-carried conditionals.
-    int base = key_begin + wave * 8;
-    Bf16x8U kA, vA, kB, vB, kC, vC, kD, vD;
-    if (base < key_end) {
-      load_pair(base, kA, vA);
-      load_pair(base + 4, kB, vB);
-    }
-    for (; base < key_end; base += 32) {
-      load_pair(base + 32 < key_end ? base + 32 : base, kC, vC);
-      load_pair(base + 36 < key_end ? base + 36 : base, kD, vD);
-      process4(kA, vA, base + group < key_end);
-      process4(kB, vB, base + 4 + group < key_end);
-      kA = kC; vA = vC; kB = kD; vB = vD;
-    }
-  } else {
-    // 4 keys per wave-iteration with 1-ahead prefetch.  (A 2-ahead ring
-    // measured WORSE: 3.45 -> 2.68 TB/s at batch 64 — its loop-carried
-    // conditional loads de-pipeline the schedule; profiles/decode_breakdown.md.)
+  // 4 keys per wave-iteration with 1-ahead prefetch.  (Wider 8-key and
+  // 2-ahead-ring variants both measured null-to-negative —
+  // profiles/decode_breakdown.md.)
+  {
     int base = key_begin + wave * 4;
     Bf16x8U k_pref, v_pref;
     if (base < key_end)
@@ -262,18 +264,18 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                              const void* context_lens, float scale,
                              int num_seqs, int num_q_heads, int num_kv_heads,
                              int page_size, int max_pages, int num_splits,
-                             int wide, hipStream_t stream) {
+                             int variant, hipStream_t stream) {
   const int qpg = num_q_heads / num_kv_heads;
   dim3 grid(num_kv_heads, num_seqs, num_splits), block(256);
-#define LAUNCH_QPG(QPG, WIDE)                                                 \
-  hipLaunchKernelGGL((decode_attn_kernel<QPG, WIDE>), grid, block, 0, stream, \
+#define LAUNCH_QPG(QPG, VAR)                                                  \
+  hipLaunchKernelGGL((decode_attn_kernel<QPG, VAR>), grid, block, 0, stream,  \
                      (float*)partial_o, (float*)partial_ml,                   \
                      (__hip_bfloat16*)out, (const __hip_bfloat16*)q,          \
                      (const __hip_bfloat16*)key_cache,                        \
                      (const __hip_bfloat16*)val_cache,                        \
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits)
-  switch (qpg * 2 + (wide ? 1 : 0)) {
+  switch (qpg * 2 + (variant ? 1 : 0)) {
     case 2: LAUNCH_QPG(1, 0); break;
     case 3: LAUNCH_QPG(1, 1); break;
     case 4: LAUNCH_QPG(2, 0); break;

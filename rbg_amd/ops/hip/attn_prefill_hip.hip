@@ -60,14 +60,19 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     const __hip_bfloat16* __restrict__ q,    // [T, QH, D]
     const __hip_bfloat16* __restrict__ k,    // [T, KVH, D]
     const __hip_bfloat16* __restrict__ v,    // [T, KVH, D]
-    const int* __restrict__ block_info,      // [nblocks, 2] = (seq_start_row, q_block)
-    const int* __restrict__ seq_lens,        // [nblocks] length of this block's seq
+    const int* __restrict__ block_info,      // [nblocks, 4] =
+    //   (q_start_row, q_block, kv_start_row, q_offset) — q_offset is the
+    //   position of q row 0 within the sequence (prefix caching / chunked
+    //   prefill attend over [past; new] K/V)
+    const int* __restrict__ seq_lens,        // [nblocks] KV length of the seq
     const float scale, const int num_q_heads, const int num_kv_heads) {
   const int qh = blockIdx.y;
   const int kvh = qh / (num_q_heads / num_kv_heads);
-  const int seq_start = block_info[blockIdx.x * 2];
-  const int qblock = block_info[blockIdx.x * 2 + 1];
-  const int seq_len = seq_lens[blockIdx.x];
+  const int q_start = block_info[blockIdx.x * 4];
+  const int qblock = block_info[blockIdx.x * 4 + 1];
+  const int kv_start = block_info[blockIdx.x * 4 + 2];
+  const int q_offset = block_info[blockIdx.x * 4 + 3];
+  const int seq_len = seq_lens[blockIdx.x];     // total KV tokens
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -80,9 +85,9 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   __shared__ __hip_bfloat16 p_lds[4][16][KTILE + PAD];
 
   // ---- Q fragments: wave's 16 rows, 4 k-steps of 32 dims ---------------
-  const int q_row_in_seq = qblock * QTILE + wave * 16 + gl;
-  const bool q_valid = q_row_in_seq < seq_len;
-  const int q_row = seq_start + min(q_row_in_seq, seq_len - 1);
+  const int num_q_rows = seq_len - q_offset;    // rows in the q tensor
+  const int q_row_in_chunk = qblock * QTILE + wave * 16 + gl;
+  const int q_row = q_start + min(q_row_in_chunk, num_q_rows - 1);
   mfma_bf8 q_frag[4];
 #pragma unroll
   for (int ks = 0; ks < 4; ++ks)
@@ -97,9 +102,11 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
 #pragma unroll
   for (int dt = 0; dt < 8; ++dt) acc_o[dt] = {0.f, 0.f, 0.f, 0.f};
 
-  // causal: this block needs keys up to its last valid q row
-  const int block_q_max = min(qblock * QTILE + QTILE - 1, seq_len - 1);
-  const int wave_q_max = min(qblock * QTILE + wave * 16 + 15, seq_len - 1);
+  // causal: this block needs keys up to its last valid q POSITION
+  const int block_q_max = min(q_offset + qblock * QTILE + QTILE - 1,
+                              seq_len - 1);
+  const int wave_q_max = min(q_offset + qblock * QTILE + wave * 16 + 15,
+                             seq_len - 1);
   const int nktiles = block_q_max / KTILE + 1;
 
   for (int kt = 0; kt < nktiles; ++kt) {
@@ -113,14 +120,13 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
       uint4 val = {0, 0, 0, 0};
       if (krow < seq_len)
         val = *reinterpret_cast<const uint4*>(
-            k + ((size_t)(seq_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
+            k + ((size_t)(kv_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
             chunk);
       *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = val;
-      // V transposed: scatter the 8 dims as single bf16 stores
       uint4 vv = {0, 0, 0, 0};
       if (krow < seq_len)
         vv = *reinterpret_cast<const uint4*>(
-            v + ((size_t)(seq_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
+            v + ((size_t)(kv_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
             chunk);
       *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) = vv;
     }
@@ -146,7 +152,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     float m_new[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int q_pos = qblock * QTILE + wave * 16 + 4 * gslice + r;
+      const int q_pos = q_offset + qblock * QTILE + wave * 16 + 4 * gslice + r;
       float row_max = NEG_INF;
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
@@ -231,11 +237,11 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   for (int r = 0; r < 4; ++r) l_run[r] = group16_sum(l_run[r]);
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const int q_pos = qblock * QTILE + wave * 16 + 4 * gslice + r;
-    if (q_pos >= seq_len) continue;
+    const int q_row_out = qblock * QTILE + wave * 16 + 4 * gslice + r;
+    if (q_row_out >= num_q_rows) continue;
     const float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
     __hip_bfloat16* orow =
-        out + ((size_t)(seq_start + q_pos) * num_q_heads + qh) * HEAD_DIM;
+        out + ((size_t)(q_start + q_row_out) * num_q_heads + qh) * HEAD_DIM;
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt)
       orow[dt * 16 + gl] = f2bf(acc_o[dt][r] * inv_l);

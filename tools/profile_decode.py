@@ -55,7 +55,7 @@ def main():
     ctx_lens = torch.full((B,), CTX, dtype=torch.int32, device=dev)
     q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev)
     kv_bytes = B * CTX * KVH * D * 2 * 2
-    for wide in (0, 1):
+    for wide in (0, 1):  # variant
         for splits in (1, 2, 4, 8):
             ms = timeit(lambda: ops._hip.decode_attention(
                 q, kc, vc, bt, ctx_lens, 0.088, splits, wide))
