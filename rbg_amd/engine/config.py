@@ -70,5 +70,6 @@ class EngineConfig:
     kv_pool_tokens: int = 0            # explicit override (tests)
     tp_size: int = 1
     tp_rank: int = 0
+    enable_prefix_cache: bool = True   # full-page KV reuse (Mooncake analog)
     enforce_eager: bool = False        # False: capture decode in hipGraphs
     seed: int = 0

@@ -67,12 +67,19 @@ class LLMEngine:
         mode, seqs = self.scheduler.schedule()
         if mode == "prefill":
             self.runner.prefill(seqs)
-            self.stats.prefill_tokens += sum(s.num_prompt_tokens for s in seqs)
+            self.stats.prefill_tokens += sum(
+                s.num_prompt_tokens - s.cached_prefix_len for s in seqs)
             self.stats.prefill_steps += 1
+            prefix = self.runner.cache.prefix
             for s in seqs:
                 t = s.ttft()
                 if t is not None:
                     self.stats.ttfts.append(t)
+                if prefix is not None and s.block_table is not None:
+                    # publish this prompt's full pages for reuse
+                    s.block_table.num_shared = prefix.register(
+                        s.prompt_tokens, s.block_table.pages,
+                        s.block_table.num_shared)
             self.scheduler.finish_prefill(seqs)
         elif mode == "decode":
             self.runner.decode(seqs)

@@ -40,6 +40,8 @@ class PagedKVCache:
         # page 0 is reserved scratch: hipGraph decode padding rows write
         # their (dead) KV slot there (model_runner._decode_graph)
         self._free: List[int] = list(range(num_pages - 1, 0, -1))
+        self.prefix: Optional["PrefixCache"] = (
+            PrefixCache(self) if cfg.enable_prefix_cache else None)
 
     @staticmethod
     def _size_pool(cfg: EngineConfig, device: torch.device) -> int:
@@ -62,7 +64,15 @@ class PagedKVCache:
     def free_pages(self) -> int:
         return len(self._free)
 
+    @property
+    def free_pages_evictable(self) -> int:
+        extra = len(self.prefix.lru) if self.prefix is not None else 0
+        return len(self._free) + extra
+
     def alloc(self, n: int) -> List[int]:
+        if n > len(self._free) and self.prefix is not None:
+            # reclaim idle cached-prefix pages (LRU) under pressure
+            self._free.extend(self.prefix.evict(n - len(self._free)))
         if n > len(self._free):
             raise OutOfPages(f"need {n} pages, {len(self._free)} free")
         out = [self._free.pop() for _ in range(n)]
@@ -84,17 +94,26 @@ class PagedKVCache:
 
 
 class BlockTable:
-    """Per-sequence page list + slot mapping helpers."""
+    """Per-sequence page list + slot mapping helpers.  The first
+    ``num_shared`` pages are owned by the prefix cache (full, immutable,
+    possibly shared across sequences) and are released back to it rather
+    than to the free list."""
 
     def __init__(self, cache: PagedKVCache):
         self.cache = cache
         self.pages: List[int] = []
+        self.num_shared = 0
         self.num_tokens = 0
 
     def ensure(self, num_tokens: int) -> None:
         need = (num_tokens + self.cache.page_size - 1) // self.cache.page_size
         if need > len(self.pages):
             self.pages.extend(self.cache.alloc(need - len(self.pages)))
+
+    def adopt_shared(self, pages: List[int]) -> None:
+        assert not self.pages, "adopt before any allocation"
+        self.pages = list(pages)
+        self.num_shared = len(pages)
 
     def slots_for(self, start: int, count: int) -> List[int]:
         """Global slot ids (page*page_size + offset) for token positions
@@ -105,6 +124,94 @@ class BlockTable:
                 for i in range(count)]
 
     def release(self) -> None:
-        self.cache.free(self.pages)
+        if self.num_shared and self.cache.prefix is not None:
+            for pg in self.pages[:self.num_shared]:
+                self.cache.prefix.release_page(pg)
+            self.cache.free(self.pages[self.num_shared:])
+        else:
+            self.cache.free(self.pages)
         self.pages = []
+        self.num_shared = 0
         self.num_tokens = 0
+
+
+class PrefixCache:
+    """Hash-based full-page prefix reuse — the in-engine realization of the
+    reference's Mooncake KV-reuse role (reference keps/74-mooncake-integration;
+    the multi-turn TTFT win).  Each FULL page of a prompt is keyed by the
+    hash chain of every token up to its end; matched pages are adopted
+    instead of re-prefilled.  Pages with refcount 0 park in an LRU pool the
+    allocator evicts from under memory pressure, so a hot 288 GB HBM pool
+    doubles as the reuse store."""
+
+    def __init__(self, cache: "PagedKVCache"):
+        self.cache = cache
+        self.by_hash: Dict[int, int] = {}           # chain hash -> page id
+        self.page_info: Dict[int, Tuple[int, int]] = {}  # page -> (hash, refs)
+        from collections import OrderedDict
+        self.lru: "OrderedDict[int, None]" = OrderedDict()
+        self.hits = 0
+        self.misses = 0
+
+    def _chain(self, tokens: List[int]):
+        ps = self.cache.page_size
+        h = 0x9e3779b9
+        for i in range(0, (len(tokens) // ps) * ps, ps):
+            h = hash((h, tuple(tokens[i:i + ps])))
+            yield h
+
+    def match(self, prompt: List[int]) -> List[int]:
+        """Longest cached page chain for prompt[:-1] (at least one token is
+        always left to prefill so logits exist); acquires the pages."""
+        got: List[int] = []
+        for h in self._chain(prompt[:-1]):
+            pg = self.by_hash.get(h)
+            if pg is None:
+                break
+            got.append(pg)
+        for pg in got:
+            h, refs = self.page_info[pg]
+            self.page_info[pg] = (h, refs + 1)
+            self.lru.pop(pg, None)
+        self.hits += len(got)
+        self.misses += max(0, len(prompt[:-1]) // self.cache.page_size
+                           - len(got))
+        return got
+
+    def register(self, prompt: List[int], pages: List[int],
+                 already_shared: int) -> int:
+        """After prefill: publish the prompt's full pages.  Returns the new
+        shared-page count (callers update BlockTable.num_shared)."""
+        shared = already_shared
+        for idx, h in enumerate(self._chain(prompt)):
+            if idx < already_shared:
+                continue
+            if h in self.by_hash:
+                break     # someone registered concurrently; keep ours private
+            pg = pages[idx]
+            self.by_hash[h] = pg
+            self.page_info[pg] = (h, 1)
+            shared = idx + 1
+        return shared
+
+    def release_page(self, pg: int) -> None:
+        h, refs = self.page_info[pg]
+        if refs <= 1:
+            self.page_info[pg] = (h, 0)
+            self.lru[pg] = None
+        else:
+            self.page_info[pg] = (h, refs - 1)
+
+    def evict(self, n: int) -> List[int]:
+        out = []
+        while self.lru and len(out) < n:
+            pg, _ = self.lru.popitem(last=False)
+            h, _refs = self.page_info.pop(pg)
+            self.by_hash.pop(h, None)
+            out.append(pg)
+        return out
+
+    def stats(self) -> Dict[str, int]:
+        return {"hits": self.hits, "misses": self.misses,
+                "cached_pages": len(self.page_info),
+                "evictable": len(self.lru)}

@@ -48,20 +48,29 @@ class ModelRunner:
         tokens: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
+        gather: List[int] = []
         cu = [0]
+        cu_k = [0]
         last_idx = []
+        any_cached = any(seq.cached_prefix_len for seq in seqs)
         for seq in seqs:
             n = seq.num_prompt_tokens
-            tokens.extend(seq.prompt_tokens)
-            positions.extend(range(n))
-            slots.extend(seq.block_table.slots_for(0, n))
-            cu.append(cu[-1] + n)
+            c = seq.cached_prefix_len     # suffix-only prefill (prefix cache)
+            tokens.extend(seq.prompt_tokens[c:])
+            positions.extend(range(c, n))
+            slots.extend(seq.block_table.slots_for(c, n - c))
+            cu.append(cu[-1] + (n - c))
+            cu_k.append(cu_k[-1] + n)
             last_idx.append(cu[-1] - 1)
+            if any_cached:
+                gather.extend(seq.block_table.slots_for(0, n))
         batch = ForwardBatch(
             mode="prefill",
             positions=self._i32(positions),
             slot_mapping=self._i32(slots),
-            cu_seqlens=self._i32(cu))
+            cu_seqlens=self._i32(cu),
+            cu_seqlens_k=self._i32(cu_k) if any_cached else None,
+            kv_gather_slots=self._i32(gather) if any_cached else None)
         hidden = self.model.forward(
             torch.tensor(tokens, dtype=torch.int64, device=self.device),
             batch, self.cache)

@@ -45,16 +45,21 @@ class Scheduler:
             seq = self.waiting[0]
             if seq.num_prompt_tokens > budget and batch:
                 break
-            if self._pages_needed(seq) > self.cache.free_pages:
+            if self._pages_needed(seq) > self.cache.free_pages_evictable:
                 break
             self.waiting.popleft()
             seq.block_table = BlockTable(self.cache)
+            if self.cache.prefix is not None and not seq.imported_kv:
+                cached = self.cache.prefix.match(seq.prompt_tokens)
+                if cached:
+                    seq.block_table.adopt_shared(cached)
+                    seq.cached_prefix_len = len(cached) * self.cfg.page_size
             seq.block_table.ensure(min(
                 seq.num_prompt_tokens + seq.sampling.max_new_tokens,
                 self.cfg.max_seq_len))
             seq.status = RUNNING
             batch.append(seq)
-            budget -= seq.num_prompt_tokens
+            budget -= seq.num_prompt_tokens - seq.cached_prefix_len
             if budget <= 0:
                 break
         if batch:

@@ -40,6 +40,9 @@ class Sequence:
         # P/D disaggregation: sequences migrated from a prefill engine carry
         # their KV pages + first generated token instead of re-prefilling
         self.imported_kv = False
+        # prefix cache: leading tokens whose KV pages were adopted from the
+        # cache (prefill runs only on the suffix)
+        self.cached_prefix_len = 0
 
     @property
     def num_prompt_tokens(self) -> int:

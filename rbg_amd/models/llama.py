@@ -35,7 +35,11 @@ class ForwardBatch:
     positions: torch.Tensor         # [T] int32
     slot_mapping: torch.Tensor      # [T] int32 (KV write slots)
     # prefill
-    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
+    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32 (q rows)
+    cu_seqlens_k: Optional[torch.Tensor] = None  # [B+1] int32 (total KV)
+    kv_gather_slots: Optional[torch.Tensor] = None  # [Tkv] int32 — prefix
+    #   caching / chunked prefill: gather the FULL [past; new] K/V from the
+    #   paged pool (the new tokens' rows were just written by rope_store_kv)
     # decode
     block_tables: Optional[torch.Tensor] = None  # [S, max_pages] int32
     context_lens: Optional[torch.Tensor] = None  # [S] int32
@@ -116,10 +120,20 @@ class LlamaAttention(nn.Module):
                           batch.positions, batch.slot_mapping)
         q3 = q.view(T, self.heads, self.head_dim)
         if batch.mode == "prefill":
-            o = ops.prefill_attention(
-                q3, k.view(T, self.kv_heads, self.head_dim),
-                v.view(T, self.kv_heads, self.head_dim),
-                batch.cu_seqlens, self.scale)
+            if batch.kv_gather_slots is not None:
+                ps = key_cache.shape[2]
+                slots = batch.kv_gather_slots.long()
+                pages, offs = slots // ps, slots % ps
+                k_full = key_cache[pages, :, offs]
+                v_full = value_cache[pages, :, offs]
+                o = ops.prefill_attention(q3, k_full, v_full,
+                                          batch.cu_seqlens, self.scale,
+                                          batch.cu_seqlens_k)
+            else:
+                o = ops.prefill_attention(
+                    q3, k.view(T, self.kv_heads, self.head_dim),
+                    v.view(T, self.kv_heads, self.head_dim),
+                    batch.cu_seqlens, self.scale)
         else:
             o = ops.decode_attention(q3, key_cache, value_cache,
                                      batch.block_tables, batch.context_lens,

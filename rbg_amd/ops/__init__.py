@@ -99,12 +99,16 @@ def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
                                       context_lens, scale)
 
 
-def prefill_attention(q, k, v, cu_seqlens, scale: float) -> torch.Tensor:
+def prefill_attention(q, k, v, cu_seqlens, scale: float,
+                      cu_seqlens_k=None) -> torch.Tensor:
     if _on_gpu(q):
-        block_info, seq_lens = reference.prefill_block_info(cu_seqlens.cpu())
+        block_info, seq_lens = reference.prefill_block_info(
+            cu_seqlens.cpu(),
+            cu_seqlens_k=None if cu_seqlens_k is None else cu_seqlens_k.cpu())
         return _require_hip().prefill_attention(
             q, k, v, block_info.to(q.device), seq_lens.to(q.device), scale)
-    return reference.prefill_attention(q, k, v, cu_seqlens, scale)
+    return reference.prefill_attention(q, k, v, cu_seqlens, scale,
+                                       cu_seqlens_k)
 
 
 build_cos_sin_table = reference.build_cos_sin_table

@@ -107,42 +107,55 @@ def decode_attention(q: torch.Tensor, key_cache: torch.Tensor,
 
 
 def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                      cu_seqlens: torch.Tensor, scale: float) -> torch.Tensor:
-    """Varlen causal attention; q [T,QH,D], k/v [T,KVH,D]."""
-    T, QH, D = q.shape
+                      cu_seqlens: torch.Tensor, scale: float,
+                      cu_seqlens_k: Optional[torch.Tensor] = None
+                      ) -> torch.Tensor:
+    """Varlen causal attention; q [Tq,QH,D], k/v [Tkv,KVH,D].  With
+    cu_seqlens_k, each sequence's q rows are its LAST (Lq) positions of the
+    Lk-token KV (prefix caching / chunked prefill)."""
+    Tq, QH, D = q.shape
     kvh = k.shape[1]
     qpg = QH // kvh
     out = torch.empty_like(q)
     cu = cu_seqlens.long().tolist()
+    cuk = (cu_seqlens_k.long().tolist()
+           if cu_seqlens_k is not None else cu)
     for b in range(len(cu) - 1):
         s, e = cu[b], cu[b + 1]
-        L = e - s
-        qs = q[s:e].float().view(L, kvh, qpg, D)
-        ks = k[s:e].float()
-        vs = v[s:e].float()
+        sk, ek = cuk[b], cuk[b + 1]
+        Lq, Lk = e - s, ek - sk
+        off = Lk - Lq
+        qs = q[s:e].float().view(Lq, kvh, qpg, D)
+        ks = k[sk:ek].float()
+        vs = v[sk:ek].float()
         attn = torch.einsum("qhgd,khd->hgqk", qs, ks) * scale
-        mask = torch.triu(torch.ones(L, L, dtype=torch.bool,
-                                     device=q.device), diagonal=1)
-        attn.masked_fill_(mask, float("-inf"))
+        qpos = torch.arange(off, Lk, device=q.device).unsqueeze(1)
+        kpos = torch.arange(Lk, device=q.device).unsqueeze(0)
+        attn.masked_fill_(kpos > qpos, float("-inf"))
         p = torch.softmax(attn, dim=-1)
         o = torch.einsum("hgqk,khd->qhgd", p, vs)
-        out[s:e] = o.reshape(L, QH, D).to(q.dtype)
+        out[s:e] = o.reshape(Lq, QH, D).to(q.dtype)
     return out
 
 
-def prefill_block_info(cu_seqlens: torch.Tensor,
-                       qtile: int = 64) -> Tuple[torch.Tensor, torch.Tensor]:
+def prefill_block_info(cu_seqlens: torch.Tensor, qtile: int = 64,
+                       cu_seqlens_k: Optional[torch.Tensor] = None
+                       ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Host-side block map for the prefill kernel: per block
-    (seq_start_row, q_block_index) plus that block's sequence length."""
+    (q_start_row, q_block_index, kv_start_row, q_offset) plus the block's
+    total KV length."""
     infos = []
     lens = []
     cu = cu_seqlens.tolist()
+    cuk = cu_seqlens_k.tolist() if cu_seqlens_k is not None else cu
     for b in range(len(cu) - 1):
-        start, end = cu[b], cu[b + 1]
-        L = end - start
-        for qb in range((L + qtile - 1) // qtile):
-            infos.append((start, qb))
-            lens.append(L)
+        qs, qe = cu[b], cu[b + 1]
+        ks, ke = cuk[b], cuk[b + 1]
+        Lq, Lk = qe - qs, ke - ks
+        off = Lk - Lq
+        for qb in range((Lq + qtile - 1) // qtile):
+            infos.append((qs, qb, ks, off))
+            lens.append(Lk)
     device = cu_seqlens.device
-    return (torch.tensor(infos, dtype=torch.int32, device=device).view(-1, 2),
+    return (torch.tensor(infos, dtype=torch.int32, device=device).view(-1, 4),
             torch.tensor(lens, dtype=torch.int32, device=device))

@@ -156,3 +156,27 @@ def test_prefill_attention_spiked_scores(dev):
     want = ref.prefill_attention(q, k, v, cu, 1.0 / math.sqrt(D))
     err = (got.float() - want.float()).abs().max().item()
     assert err < 5e-2, f"spiked err={err}"
+
+
+@pytest.mark.parametrize("lens", [[(96, 32)], [(200, 80), (64, 64), (40, 1)]])
+def test_prefill_attention_with_prefix(dev, lens):
+    """Extended kernel: q rows are the LAST Lq positions of an Lk-token KV
+    (prefix caching / chunked prefill)."""
+    QH, KVH, D = 32, 8, 128
+    Tk = sum(lk for lk, _ in lens)
+    Tq = sum(lq for _, lq in lens)
+    k = torch.randn(Tk, KVH, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(Tk, KVH, D, dtype=torch.bfloat16, device=dev)
+    q = torch.randn(Tq, QH, D, dtype=torch.bfloat16, device=dev)
+    cu_q = [0]
+    cu_k = [0]
+    for lk, lq in lens:
+        cu_q.append(cu_q[-1] + lq)
+        cu_k.append(cu_k[-1] + lk)
+    cu_q = torch.tensor(cu_q, dtype=torch.int32, device=dev)
+    cu_k = torch.tensor(cu_k, dtype=torch.int32, device=dev)
+    scale = 1.0 / math.sqrt(D)
+    got = ops.prefill_attention(q, k, v, cu_q, scale, cu_k)
+    want = ref.prefill_attention(q, k, v, cu_q, scale, cu_k)
+    err = (got.float() - want.float()).abs().max().item()
+    assert err < 3e-2, f"lens={lens} err={err}"
