@@ -167,9 +167,9 @@ void decode_attn_kernel(
     for (; base < key_end; base += 16) {
       const bool valid = base + group < key_end;
       Bf16x8U kv = k_pref, vv = v_pref;
-      const int nxt = base + 16;
-      if (nxt < key_end)
-        load_pair(nxt, k_pref, v_pref);
+      // unconditional clamped prefetch (row_offset clamps): a branch
+      // around the loads forces a vmcnt drain per iteration (guide trap 4c)
+      load_pair(base + 16, k_pref, v_pref);
       process4(kv, vv, valid);
     }
   }

@@ -112,22 +112,21 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     const int k_base = kt * KTILE;
     // ---- stage K tile: 64 rows x 128 dims, 16 B per thread-iteration ----
     __syncthreads();
+    // UNCONDITIONAL clamped loads: a per-iteration `if (krow < seq_len)`
+    // guard makes hipcc branch around each load and drain vmcnt(0) per
+    // element (guide §5 trap 4c — dependent HBM round trips).  Out-of-range
+    // rows carry garbage that the causal/seq-len mask already zeroes in
+    // softmax, so clamping the address is sufficient.
     for (int i = tid; i < KTILE * (HEAD_DIM / 8); i += 256) {
       const int key = i >> 4;          // 16 chunks of 8 dims per key
       const int chunk = (i & 15) * 8;
-      const int krow = k_base + key;
-      uint4 val = {0, 0, 0, 0};
-      if (krow < seq_len)
-        val = *reinterpret_cast<const uint4*>(
-            k + ((size_t)(kv_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
-            chunk);
-      *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = val;
-      uint4 vv = {0, 0, 0, 0};
-      if (krow < seq_len)
-        vv = *reinterpret_cast<const uint4*>(
-            v + ((size_t)(kv_start + krow) * num_kv_heads + kvh) * HEAD_DIM +
-            chunk);
-      *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) = vv;
+      const int krow = min(k_base + key, seq_len - 1);
+      const size_t row = ((size_t)(kv_start + krow) * num_kv_heads + kvh) *
+                         HEAD_DIM + chunk;
+      *reinterpret_cast<uint4*>(&k_lds[key][chunk]) =
+          *reinterpret_cast<const uint4*>(k + row);
+      *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) =
+          *reinterpret_cast<const uint4*>(v + row);
     }
     __syncthreads();
 
