@@ -80,6 +80,7 @@ def pick_decode_splits(num_seqs: int, num_kv_heads: int,
 
 
 DECODE_VARIANT = int(__import__("os").environ.get("RBG_DECODE_VARIANT", "1"))
+PREFILL_SWZ = int(__import__("os").environ.get("RBG_PREFILL_SWZ", "1"))
 
 
 def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
@@ -106,7 +107,8 @@ def prefill_attention(q, k, v, cu_seqlens, scale: float,
             cu_seqlens.cpu(),
             cu_seqlens_k=None if cu_seqlens_k is None else cu_seqlens_k.cpu())
         return _require_hip().prefill_attention(
-            q, k, v, block_info.to(q.device), seq_lens.to(q.device), scale)
+            q, k, v, block_info.to(q.device), seq_lens.to(q.device), scale,
+            PREFILL_SWZ)
     return reference.prefill_attention(q, k, v, cu_seqlens, scale,
                                        cu_seqlens_k)
 

@@ -54,7 +54,12 @@ DEV_INLINE int v_img_off(int key, int dim) {
          (jj & 3) * 16 + (dim & 15);
 }
 
-// grid.x = total q-blocks (host-computed map), grid.y = num q heads.
+// grid: 1D, nblocks * num_q_heads.  SWZ=1 remaps block ids so the QPG
+// q-heads sharing one (q-block, kv-head) K/V tile land on the SAME XCD
+// (dispatcher places block b on XCD b%8 — guide T1) temporally adjacent:
+// the sharers then hit that XCD's L2 instead of re-pulling HBM (the GQA
+// staging re-read is the dominant prefill cost, profiles/).
+template <int SWZ>
 __global__ __launch_bounds__(256) void prefill_attn_kernel(
     __hip_bfloat16* __restrict__ out,        // [T, QH, D]
     const __hip_bfloat16* __restrict__ q,    // [T, QH, D]
@@ -65,14 +70,38 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     //   position of q row 0 within the sequence (prefix caching / chunked
     //   prefill attend over [past; new] K/V)
     const int* __restrict__ seq_lens,        // [nblocks] KV length of the seq
-    const float scale, const int num_q_heads, const int num_kv_heads) {
-  const int qh = blockIdx.y;
-  const int kvh = qh / (num_q_heads / num_kv_heads);
-  const int q_start = block_info[blockIdx.x * 4];
-  const int qblock = block_info[blockIdx.x * 4 + 1];
-  const int kv_start = block_info[blockIdx.x * 4 + 2];
-  const int q_offset = block_info[blockIdx.x * 4 + 3];
-  const int seq_len = seq_lens[blockIdx.x];     // total KV tokens
+    const float scale, const int num_q_heads, const int num_kv_heads,
+    const int nblocks) {
+  const int qpg_n = num_q_heads / num_kv_heads;
+  int blk, qh;
+  if (SWZ && qpg_n > 1) {
+    // group g = (block, kvh); its qpg_n sharers get ids c*8*qpg + q*8 + g%8
+    const int lin = blockIdx.x;
+    const int G = nblocks * num_kv_heads;
+    const int chunk = 8 * qpg_n;
+    const int full = (G / 8) * chunk;
+    int g, qpg_idx;
+    if (lin < full) {
+      const int c = lin / chunk, r = lin % chunk;
+      qpg_idx = r >> 3;
+      g = c * 8 + (r & 7);
+    } else {
+      const int idx = lin - full;
+      g = (G / 8) * 8 + idx / qpg_n;
+      qpg_idx = idx % qpg_n;
+    }
+    blk = g / num_kv_heads;
+    qh = (g % num_kv_heads) * qpg_n + qpg_idx;
+  } else {
+    blk = blockIdx.x / num_q_heads;
+    qh = blockIdx.x % num_q_heads;
+  }
+  const int kvh = qh / qpg_n;
+  const int q_start = block_info[blk * 4];
+  const int qblock = block_info[blk * 4 + 1];
+  const int kv_start = block_info[blk * 4 + 2];
+  const int q_offset = block_info[blk * 4 + 3];
+  const int seq_len = seq_lens[blk];     // total KV tokens
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -254,14 +283,21 @@ extern "C" {
 void launch_prefill_attention(void* out, const void* q, const void* k,
                               const void* v, const void* block_info,
                               const void* seq_lens, float scale, int nblocks,
-                              int num_q_heads, int num_kv_heads,
+                              int num_q_heads, int num_kv_heads, int swz,
                               hipStream_t stream) {
-  dim3 grid(nblocks, num_q_heads), block(256);
-  hipLaunchKernelGGL(prefill_attn_kernel, grid, block, 0, stream,
-                     (__hip_bfloat16*)out, (const __hip_bfloat16*)q,
-                     (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,
-                     (const int*)block_info, (const int*)seq_lens, scale,
-                     num_q_heads, num_kv_heads);
+  dim3 grid(nblocks * num_q_heads), block(256);
+  if (swz)
+    hipLaunchKernelGGL(prefill_attn_kernel<1>, grid, block, 0, stream,
+                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,
+                       (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,
+                       (const int*)block_info, (const int*)seq_lens, scale,
+                       num_q_heads, num_kv_heads, nblocks);
+  else
+    hipLaunchKernelGGL(prefill_attn_kernel<0>, grid, block, 0, stream,
+                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,
+                       (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,
+                       (const int*)block_info, (const int*)seq_lens, scale,
+                       num_q_heads, num_kv_heads, nblocks);
 }
 
 }  // extern "C"

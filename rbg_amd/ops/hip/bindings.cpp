@@ -18,7 +18,7 @@ void launch_decode_attention(void*, void*, void*, const void*, const void*,
                              int, int, int, int, int, int, int, hipStream_t);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const void*, const void*, float, int, int, int,
-                              hipStream_t);
+                              int, hipStream_t);
 void launch_mfma_probe(void*, const void*, const void*, int, int, hipStream_t);
 }
 
@@ -131,7 +131,8 @@ torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
 
 torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
                                 torch::Tensor v, torch::Tensor block_info,
-                                torch::Tensor seq_lens, double scale) {
+                                torch::Tensor seq_lens, double scale,
+                                int64_t swz) {
   check_bf16_contig(q, "q");
   check_bf16_contig(k, "k");
   check_bf16_contig(v, "v");
@@ -145,7 +146,8 @@ torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
   launch_prefill_attention(out.data_ptr(), q.data_ptr(), k.data_ptr(),
                            v.data_ptr(), block_info.data_ptr(),
                            seq_lens.data_ptr(), (float)scale, nblocks,
-                           num_q_heads, num_kv_heads, current_stream());
+                           num_q_heads, num_kv_heads, (int)swz,
+                           current_stream());
   return out;
 }
 

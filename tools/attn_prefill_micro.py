@@ -16,6 +16,7 @@ ap.add_argument("--len", type=int, default=2048, dest="slen")
 ap.add_argument("--qh", type=int, default=32)
 ap.add_argument("--kvh", type=int, default=8)
 ap.add_argument("--iters", type=int, default=20)
+ap.add_argument("--rounds", type=int, default=5)
 a = ap.parse_args()
 dev = torch.device("cuda:0")
 D = 128
@@ -25,15 +26,27 @@ q = torch.randn(T, a.qh, D, dtype=torch.bfloat16, device=dev)
 k = torch.randn(T, a.kvh, D, dtype=torch.bfloat16, device=dev)
 v = torch.randn_like(k)
 cu = torch.arange(0, T + 1, a.slen, dtype=torch.int32, device=dev)
-for _ in range(3):
-    out = ops.prefill_attention(q, k, v, cu, 0.088)
-torch.cuda.synchronize()
-t0 = time.monotonic()
-for _ in range(a.iters):
-    out = ops.prefill_attention(q, k, v, cu, 0.088)
-torch.cuda.synchronize()
-dt = (time.monotonic() - t0) / a.iters
-# causal: 2 matmuls x S^2/2 x D x heads per seq
+from rbg_amd.ops import reference as refmod
+block_info, seq_lens = refmod.prefill_block_info(cu.cpu())
+block_info, seq_lens = block_info.to(dev), seq_lens.to(dev)
+
+def run(swz):
+    torch.cuda.synchronize()
+    t0 = time.monotonic()
+    for _ in range(a.iters):
+        ops._hip.prefill_attention(q, k, v, block_info, seq_lens, 0.088, swz)
+    torch.cuda.synchronize()
+    return (time.monotonic() - t0) / a.iters
+
 flops = a.seqs * 2 * 2 * (a.slen * a.slen / 2) * D * a.qh
-print(f"prefill attn B{a.seqs}xS{a.slen} H{a.qh}/{a.kvh}: "
-      f"{dt*1e3:.2f} ms, {flops/dt/1e12:.1f} TF/s")
+for swz in (0, 1):
+    run(swz)
+import statistics
+res = {0: [], 1: []}
+for _ in range(a.rounds):
+    for swz in (0, 1):
+        res[swz].append(run(swz))
+for swz in (0, 1):
+    dt = statistics.median(res[swz])
+    print(f"prefill attn B{a.seqs}xS{a.slen} H{a.qh}/{a.kvh} swz={swz}: "
+          f"{dt*1e3:.2f} ms, {flops/dt/1e12:.1f} TF/s")
