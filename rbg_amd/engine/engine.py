@@ -67,11 +67,12 @@ class LLMEngine:
         mode, seqs = self.scheduler.schedule()
         if mode == "prefill":
             self.runner.prefill(seqs)
-            self.stats.prefill_tokens += sum(
-                s.num_prompt_tokens - s.cached_prefix_len for s in seqs)
+            self.stats.prefill_tokens += sum(s.chunk_len for s in seqs)
             self.stats.prefill_steps += 1
             prefix = self.runner.cache.prefix
-            for s in seqs:
+            completed = [s for s in seqs
+                         if s.num_prefilled == s.num_prompt_tokens]
+            for s in completed:
                 t = s.ttft()
                 if t is not None:
                     self.stats.ttfts.append(t)
@@ -80,7 +81,7 @@ class LLMEngine:
                     s.block_table.num_shared = prefix.register(
                         s.prompt_tokens, s.block_table.pages,
                         s.block_table.num_shared)
-            self.scheduler.finish_prefill(seqs)
+            self.scheduler.finish_prefill(completed)
         elif mode == "decode":
             self.runner.decode(seqs)
             self.stats.decode_tokens += len(seqs)

@@ -45,38 +45,49 @@ class ModelRunner:
 
     @torch.inference_mode()
     def prefill(self, seqs: List[Sequence]) -> List[int]:
+        """One prefill step: each sequence contributes its current chunk
+        [num_prefilled, num_prefilled + chunk_len) — a full prompt, a
+        cached-prefix suffix, or one piece of a chunked long prompt.  Only
+        sequences whose chunk reaches the prompt end get a token sampled."""
         tokens: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
         gather: List[int] = []
         cu = [0]
         cu_k = [0]
-        last_idx = []
-        any_cached = any(seq.cached_prefix_len for seq in seqs)
+        completing: List[Sequence] = []
+        last_idx: List[int] = []
+        any_past = any(seq.num_prefilled > 0 for seq in seqs)
         for seq in seqs:
-            n = seq.num_prompt_tokens
-            c = seq.cached_prefix_len     # suffix-only prefill (prefix cache)
-            tokens.extend(seq.prompt_tokens[c:])
-            positions.extend(range(c, n))
-            slots.extend(seq.block_table.slots_for(c, n - c))
-            cu.append(cu[-1] + (n - c))
-            cu_k.append(cu_k[-1] + n)
-            last_idx.append(cu[-1] - 1)
-            if any_cached:
-                gather.extend(seq.block_table.slots_for(0, n))
+            start = seq.num_prefilled
+            end = start + (seq.chunk_len or
+                           (seq.num_prompt_tokens - start))
+            tokens.extend(seq.prompt_tokens[start:end])
+            positions.extend(range(start, end))
+            slots.extend(seq.block_table.slots_for(start, end - start))
+            cu.append(cu[-1] + (end - start))
+            cu_k.append(cu_k[-1] + end)
+            if any_past:
+                gather.extend(seq.block_table.slots_for(0, end))
+            if end == seq.num_prompt_tokens:
+                completing.append(seq)
+                last_idx.append(cu[-1] - 1)
+            seq.num_prefilled = end
         batch = ForwardBatch(
             mode="prefill",
             positions=self._i32(positions),
             slot_mapping=self._i32(slots),
             cu_seqlens=self._i32(cu),
-            cu_seqlens_k=self._i32(cu_k) if any_cached else None,
-            kv_gather_slots=self._i32(gather) if any_cached else None)
+            cu_seqlens_k=self._i32(cu_k) if any_past else None,
+            kv_gather_slots=self._i32(gather) if any_past else None)
         hidden = self.model.forward(
             torch.tensor(tokens, dtype=torch.int64, device=self.device),
             batch, self.cache)
+        if not completing:
+            return []
         logits = self.model.logits(hidden, self._i32(last_idx))
-        next_tokens = self.sample(logits, seqs)
-        for seq, tok in zip(seqs, next_tokens):
+        next_tokens = self.sample(logits, completing)
+        for seq, tok in zip(completing, next_tokens):
             seq.append_token(tok)
         return next_tokens
 

@@ -37,31 +37,37 @@ class Scheduler:
 
     def schedule(self) -> Tuple[str, List[Sequence]]:
         """Returns ("prefill"|"decode"|"idle", sequences)."""
-        # prefill first: new prompts keep the decode batch full
+        # prefill first: new prompts keep the decode batch full.  A prompt
+        # whose remaining suffix exceeds the step budget is CHUNKED: it stays
+        # at the head of the queue and prefills budget-sized pieces per step
+        # (the attend-over-[past; new] path of the prefill kernel).
         batch: List[Sequence] = []
         budget = self.cfg.max_prefill_tokens
         while self.waiting and len(self.running) + len(batch) < \
-                self.cfg.max_batch_size:
+                self.cfg.max_batch_size and budget > 0:
             seq = self.waiting[0]
-            if seq.num_prompt_tokens > budget and batch:
-                break
-            if self._pages_needed(seq) > self.cache.free_pages_evictable:
-                break
-            self.waiting.popleft()
-            seq.block_table = BlockTable(self.cache)
-            if self.cache.prefix is not None and not seq.imported_kv:
-                cached = self.cache.prefix.match(seq.prompt_tokens)
-                if cached:
-                    seq.block_table.adopt_shared(cached)
-                    seq.cached_prefix_len = len(cached) * self.cfg.page_size
-            seq.block_table.ensure(min(
-                seq.num_prompt_tokens + seq.sampling.max_new_tokens,
-                self.cfg.max_seq_len))
-            seq.status = RUNNING
+            if seq.block_table is None:
+                if self._pages_needed(seq) > self.cache.free_pages_evictable:
+                    break
+                seq.block_table = BlockTable(self.cache)
+                if self.cache.prefix is not None and not seq.imported_kv:
+                    cached = self.cache.prefix.match(seq.prompt_tokens)
+                    if cached:
+                        seq.block_table.adopt_shared(cached)
+                        seq.cached_prefix_len = \
+                            len(cached) * self.cfg.page_size
+                seq.num_prefilled = seq.cached_prefix_len
+                seq.block_table.ensure(min(
+                    seq.num_prompt_tokens + seq.sampling.max_new_tokens,
+                    self.cfg.max_seq_len))
+            remaining = seq.num_prompt_tokens - seq.num_prefilled
+            seq.chunk_len = min(remaining, budget)
+            budget -= seq.chunk_len
             batch.append(seq)
-            budget -= seq.num_prompt_tokens - seq.cached_prefix_len
-            if budget <= 0:
-                break
+            if seq.num_prefilled + seq.chunk_len < seq.num_prompt_tokens:
+                break     # partial chunk: seq stays queued, batch is full
+            self.waiting.popleft()
+            seq.status = RUNNING
         if batch:
             return "prefill", batch
         if self.running:

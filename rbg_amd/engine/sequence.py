@@ -43,6 +43,10 @@ class Sequence:
         # prefix cache: leading tokens whose KV pages were adopted from the
         # cache (prefill runs only on the suffix)
         self.cached_prefix_len = 0
+        # chunked prefill: prompt tokens whose KV is already computed; the
+        # scheduler feeds [num_prefilled, num_prefilled+chunk) per step
+        self.num_prefilled = 0
+        self.chunk_len = 0
 
     @property
     def num_prompt_tokens(self) -> int:

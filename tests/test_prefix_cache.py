@@ -92,3 +92,50 @@ def test_shared_pages_not_freed_while_running():
     assert a.output_tokens[:2] == seq_c.output_tokens
     # refcounts drained back to LRU after both finished
     assert eng.runner.cache.prefix.stats()["evictable"] >= 3
+
+
+def test_chunked_prefill_exact():
+    """A prompt longer than max_prefill_tokens prefills in chunks across
+    steps (attend-over-past path); greedy output must exactly match an
+    engine with an unbounded budget."""
+    torch.manual_seed(5)
+    prompt = torch.randint(0, 500, (150,)).tolist()
+    small = engine(prefix=False, max_prefill_tokens=48)
+    (a,) = small.generate([prompt], SamplingParams(max_new_tokens=5))
+    assert small.stats.prefill_steps >= 4     # 150 tokens / 48 budget
+    big = engine(prefix=False, max_prefill_tokens=4096)
+    (b,) = big.generate([prompt], SamplingParams(max_new_tokens=5))
+    assert big.stats.prefill_steps == 1
+    assert a.output_tokens == b.output_tokens
+
+
+def test_chunked_prefill_with_prefix_cache():
+    torch.manual_seed(6)
+    shared = torch.randint(0, 500, (64,)).tolist()
+    p1 = shared + torch.randint(0, 500, (90,)).tolist()
+    p2 = shared + torch.randint(0, 500, (70,)).tolist()
+    eng = engine(prefix=True, max_prefill_tokens=40)
+    (a1,) = eng.generate([p1], SamplingParams(max_new_tokens=3))
+    (a2,) = eng.generate([p2], SamplingParams(max_new_tokens=3))
+    assert a2.cached_prefix_len == 64        # 4 pages reused
+    ref = engine(prefix=False, max_prefill_tokens=4096)
+    (b1,) = ref.generate([p1], SamplingParams(max_new_tokens=3))
+    (b2,) = ref.generate([p2], SamplingParams(max_new_tokens=3))
+    assert a1.output_tokens == b1.output_tokens
+    assert a2.output_tokens == b2.output_tokens
+
+
+def test_mixed_batch_chunked_and_short():
+    torch.manual_seed(7)
+    long_p = torch.randint(0, 500, (120,)).tolist()
+    short_p = torch.randint(0, 500, (10,)).tolist()
+    eng = engine(prefix=False, max_prefill_tokens=64)
+    seq_l = eng.add_request(long_p, SamplingParams(max_new_tokens=3))
+    seq_s = eng.add_request(short_p, SamplingParams(max_new_tokens=3))
+    while eng.scheduler.has_work():
+        eng.step()
+    ref = engine(prefix=False, max_prefill_tokens=4096)
+    (bl,) = ref.generate([long_p], SamplingParams(max_new_tokens=3))
+    (bs,) = ref.generate([short_p], SamplingParams(max_new_tokens=3))
+    assert seq_l.output_tokens == bl.output_tokens
+    assert seq_s.output_tokens == bs.output_tokens
