@@ -32,6 +32,9 @@ def parse_args():
                    help="synthetic prompt length")
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--device", default="cuda")
+    p.add_argument("--parallel", choices=["dp", "tp"], default="dp",
+                   help="dp: one engine replica per rank (weak scaling); "
+                        "tp: all ranks form one tensor-parallel engine")
     p.add_argument("--eager", action="store_true",
                    help="disable hipGraph capture")
     return p.parse_args()
@@ -59,20 +62,28 @@ def main() -> int:
     from rbg_amd.engine.config import EngineConfig, ModelConfig
     from rbg_amd.engine.engine import LLMEngine
     from rbg_amd.engine.sequence import SamplingParams
+    from rbg_amd.models.llama import TPContext
 
+    tp_mode = args.parallel == "tp" and world > 1
     model_cfg = ModelConfig.preset(args.model)
     cfg = EngineConfig(
         model=model_cfg, device=device,
         max_batch_size=max(args.batch, 8),
         max_seq_len=args.seq_len + args.steps + args.warmup + 64,
         max_prefill_tokens=8192,
-        enforce_eager=args.eager or device != "cuda",
+        # TP: collectives inside hipGraph capture are deferred to round 2
+        enforce_eager=args.eager or device != "cuda" or tp_mode,
         kv_pool_tokens=(args.batch *
                         (args.seq_len + args.steps + args.warmup + 64) + 4096),
+        tp_size=world if tp_mode else 1,
+        tp_rank=rank if tp_mode else 0,
     )
-    eng = LLMEngine(cfg)
+    tp = TPContext(size=world, rank=rank) if tp_mode else None
+    eng = LLMEngine(cfg, tp)
 
-    torch.manual_seed(123 + rank)
+    # TP lockstep: every rank must build the IDENTICAL schedule, so the
+    # prompt seed must not depend on rank
+    torch.manual_seed(123 if tp_mode else 123 + rank)
     prompts = [torch.randint(0, model_cfg.vocab_size, (args.seq_len,)).tolist()
                for _ in range(args.batch)]
     sampling = SamplingParams(
@@ -121,7 +132,8 @@ def main() -> int:
         elapsed = float(t.item())
 
     ms_per_step = elapsed / args.steps * 1000.0
-    total_tok_s = args.batch * n_gpus * args.steps / elapsed
+    replicas = 1 if tp_mode else n_gpus
+    total_tok_s = args.batch * replicas * args.steps / elapsed
     ttfts = sorted(eng.stats.ttfts)
     p50_ttft_ms = (ttfts[len(ttfts) // 2] * 1000.0) if ttfts else 0.0
 
@@ -143,9 +155,10 @@ def main() -> int:
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch * n_gpus,
+                "global_batch": args.batch * replicas,
                 "seq_len": args.seq_len,
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": (f"tp{n_gpus}" if tp_mode else
+                                f"dp{n_gpus}"),
                 "p50_ttft_ms": round(p50_ttft_ms, 1),
                 "prefill_wall_s": round(prefill_wall, 3),
                 "prefill_tok_s": round(

@@ -359,11 +359,9 @@ class RoleInstanceController:
         self.restarts.evict(inst.metadata.uid)
 
     def _cleanup_by_name(self, name: str) -> None:
-        for uid, rt in list(self._runtimes.items()):
-            if any(w.startswith(name + "-") or w == name for w in rt.handles) \
-                    or rt.gang_id == f"inst-{uid}" and not rt.handles:
-                continue
-        # handled via teardown on delete events; nothing else to do here
+        # deletion runs through teardown() on the delete path; a reconcile of
+        # a vanished key has nothing left to clean
+        return
 
     @staticmethod
     def _owner_uid(inst: RoleInstance) -> str:

@@ -23,7 +23,6 @@ from __future__ import annotations
 
 import logging
 import os
-import queue
 import threading
 import time
 from typing import Any, Dict, List, Optional
