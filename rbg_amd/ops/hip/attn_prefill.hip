@@ -58,7 +58,10 @@ DEV_INLINE int v_img_off(int key, int dim) {
 // (dispatcher places block b on XCD b%8 — guide T1) temporally adjacent:
 // the sharers then hit that XCD's L2 instead of re-pulling HBM (the GQA
 // staging re-read is the dominant prefill cost, profiles/).
-template <int SWZ>
+// DB=1: T14 register staging (issue tile t+1's global loads BEFORE tile
+// t's compute, write them to LDS after the barrier) — HBM latency hides
+// under the MFMAs at the cost of ~32 staging VGPRs (guide G15/T14).
+template <int SWZ, int DB>
 __global__ __launch_bounds__(256) void prefill_attn_kernel(
     __hip_bfloat16* __restrict__ out,        // [T, QH, D]
     const __hip_bfloat16* __restrict__ q,    // [T, QH, D]
@@ -137,29 +140,55 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
                              seq_len - 1);
   const int nktiles = block_q_max / KTILE + 1;
 
-  for (int kt = 0; kt < nktiles; ++kt) {
-    const int k_base = kt * KTILE;
-    // ---- stage K tile: 64 rows x 128 dims, 16 B per thread-iteration ----
-    __syncthreads();
-    // UNCONDITIONAL clamped loads: a per-iteration `if (krow < seq_len)`
-    // guard makes hipcc branch around each load and drain vmcnt(0) per
-    // element (guide §5 trap 4c — dependent HBM round trips).  Out-of-range
-    // rows carry garbage that the causal/seq-len mask already zeroes in
-    // softmax, so clamping the address is sufficient.
-    for (int i = tid; i < KTILE * (HEAD_DIM / 8); i += 256) {
-      const int key = i >> 4;          // 16 chunks of 8 dims per key
+  // staging helpers: each thread owns 4 (key, chunk) pieces of a tile.
+  // UNCONDITIONAL clamped loads: a per-piece `if (krow < seq_len)` guard
+  // makes hipcc branch around each load and drain vmcnt(0) per element
+  // (guide §5 trap 4c); out-of-range rows are masked in softmax anyway.
+  uint4 kreg[4], vreg[4];
+  auto issue_loads = [&](int kt2) {
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int i = tid + s * 256;
+      const int key = i >> 4;
       const int chunk = (i & 15) * 8;
-      const int krow = min(k_base + key, seq_len - 1);
+      const int krow = min(kt2 * KTILE + key, seq_len - 1);
       const size_t row = ((size_t)(kv_start + krow) * num_kv_heads + kvh) *
                          HEAD_DIM + chunk;
-      *reinterpret_cast<uint4*>(&k_lds[key][chunk]) =
-          *reinterpret_cast<const uint4*>(k + row);
-      *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) =
-          *reinterpret_cast<const uint4*>(v + row);
+      kreg[s] = *reinterpret_cast<const uint4*>(k + row);
+      vreg[s] = *reinterpret_cast<const uint4*>(v + row);
     }
-    __syncthreads();
+  };
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int i = tid + s * 256;
+      const int key = i >> 4;
+      const int chunk = (i & 15) * 8;
+      *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = kreg[s];
+      *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) = vreg[s];
+    }
+  };
 
-    if (k_base > wave_q_max) continue;   // fully-masked for this wave
+  if (DB) {   // prologue: tile 0 resident before the loop
+    issue_loads(0);
+    write_tile();
+  }
+
+  for (int kt = 0; kt < nktiles; ++kt) {
+    const int k_base = kt * KTILE;
+    if (DB) {
+      __syncthreads();                 // tile kt visible to every wave
+      if (kt + 1 < nktiles)
+        issue_loads(kt + 1);           // in flight under the MFMAs below
+    } else {
+      __syncthreads();
+      issue_loads(kt);
+      write_tile();
+      __syncthreads();
+    }
+
+    const bool compute_this = k_base <= wave_q_max;
+    if (compute_this) {
 
     // ---- S = Q K^T : 4 col-tiles x 4 k-steps of MFMA --------------------
     mfma_f4 s[4];
@@ -250,12 +279,18 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
       __builtin_amdgcn_sched_barrier(0);   // rule 18: keep MFMAs below
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
-        union { struct { unsigned long long lo, hi; } u; mfma_bf8 v; } vf;
+        union { struct { unsigned long long lo, hi; } u; mfma_bf8 vf2; } vf;
         vf.u.lo = vlo[dt];
         vf.u.hi = vhi[dt];
-        acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf.v,
+        acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf.vf2,
                                                             acc_o[dt], 0, 0, 0);
       }
+    }
+    }  // compute_this
+
+    if (DB && kt + 1 < nktiles) {
+      __syncthreads();                 // every wave done READING tile kt
+      write_tile();                    // land tile kt+1 (write-after-barrier)
     }
   }
 
@@ -285,18 +320,20 @@ void launch_prefill_attention(void* out, const void* q, const void* k,
                               int num_q_heads, int num_kv_heads, int swz,
                               hipStream_t stream) {
   dim3 grid(nblocks * num_q_heads), block(256);
-  if (swz)
-    hipLaunchKernelGGL(prefill_attn_kernel<1>, grid, block, 0, stream,
-                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,
-                       (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,
-                       (const int*)block_info, (const int*)seq_lens, scale,
-                       num_q_heads, num_kv_heads, nblocks);
-  else
-    hipLaunchKernelGGL(prefill_attn_kernel<0>, grid, block, 0, stream,
-                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,
-                       (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,
-                       (const int*)block_info, (const int*)seq_lens, scale,
-                       num_q_heads, num_kv_heads, nblocks);
+  // swz bit 0: XCD-affine GQA swizzle; bit 1: T14 double-buffer staging
+#define PF_LAUNCH(S, D)                                                     \
+  hipLaunchKernelGGL((prefill_attn_kernel<S, D>), grid, block, 0, stream,   \
+                     (__hip_bfloat16*)out, (const __hip_bfloat16*)q,        \
+                     (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,    \
+                     (const int*)block_info, (const int*)seq_lens, scale,   \
+                     num_q_heads, num_kv_heads, nblocks)
+  switch (swz & 3) {
+    case 0: PF_LAUNCH(0, 0); break;
+    case 1: PF_LAUNCH(1, 0); break;
+    case 2: PF_LAUNCH(0, 1); break;
+    case 3: PF_LAUNCH(1, 1); break;
+  }
+#undef PF_LAUNCH
 }
 
 }  // extern "C"

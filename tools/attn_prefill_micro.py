@@ -39,14 +39,15 @@ def run(swz):
     return (time.monotonic() - t0) / a.iters
 
 flops = a.seqs * 2 * 2 * (a.slen * a.slen / 2) * D * a.qh
-for swz in (0, 1):
+for swz in (0, 2):
     run(swz)
 import statistics
-res = {0: [], 1: []}
+variants = (0, 2)
+res = {v: [] for v in variants}
 for _ in range(a.rounds):
-    for swz in (0, 1):
+    for swz in variants:
         res[swz].append(run(swz))
-for swz in (0, 1):
+for swz in variants:
     dt = statistics.median(res[swz])
     print(f"prefill attn B{a.seqs}xS{a.slen} H{a.qh}/{a.kvh} swz={swz}: "
           f"{dt*1e3:.2f} ms, {flops/dt/1e12:.1f} TF/s")
