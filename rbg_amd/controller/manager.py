@@ -136,7 +136,8 @@ class Manager:
         self.restarts = RestartRegistry()
 
         self.rbg = RoleBasedGroupController(self.store, self.registry,
-                                            self.opts.history_limit)
+                                            self.opts.history_limit,
+                                            ports=self.ports)
         self.ris = RoleInstanceSetController(self.store)
         self.instance = RoleInstanceController(
             self.store, self.gang, self.runner, self.ports, self.bindings,

@@ -76,10 +76,63 @@ def _merge_template(base: EngineTemplate, patch: Dict) -> EngineTemplate:
 
 class RoleBasedGroupController:
     def __init__(self, store: Store, registry: Optional[TopologyRegistry] = None,
-                 history_limit: int = 10):
+                 history_limit: int = 10, ports=None):
         self.store = store
         self.registry = registry
         self.revisions = RevisionManager(store, history_limit)
+        self.ports = ports        # PortAllocator (comm rendezvous ports)
+
+    # ------------------------------------------------------------------
+
+    def _comm_plan(self, rbg: RoleBasedGroup) -> Dict[str, str]:
+        """Assign a GLOBAL rank to every llm-engine worker of the group's
+        communicating roles (leaderWorker TP groups, prefill/decode
+        migration peers) — the gang-scheduling-as-communicator-formation
+        realization.  Deterministic from the spec: roles in
+        spec order, instances by ordinal, components in template order.
+        Returns the annotation set injected into every instance; empty when
+        no role communicates.  Membership is fixed at the current spec:
+        scaling a communicating role recreates its engines (the gang is
+        atomic)."""
+        import json as _json
+        rank = 0
+        groups: List[List[int]] = []
+        members: Dict[str, List[int]] = {}
+        rank_map: Dict[str, int] = {}
+        for role in rbg.spec.roles:
+            comps = self._apply_engine_runtimes(
+                rbg, role, expand_pattern(rbg, role))
+            engines = [(c, e) for c in comps
+                       for e in (c.template.engines if c.template else [])
+                       if e.runner == "llm-engine"]
+            communicates = (role.pattern == C.PATTERN_LEADER_WORKER or any(
+                e.args.get("mode") in ("prefill", "decode")
+                for _, e in engines))
+            if not engines or not communicates:
+                continue
+            for i in range(role.replicas):
+                inst = instance_name(rbg.metadata.name, role.name, i)
+                inst_ranks = []
+                for comp, eng in engines:
+                    for j in range(comp.size):
+                        wname = f"{inst}-{comp.name}-{j}"
+                        rank_map[wname] = rank
+                        inst_ranks.append(rank)
+                        rank += 1
+                groups.append(inst_ranks)
+                members[inst] = inst_ranks
+        if rank <= 1 and len(groups) <= 1:
+            return {}
+        port = 29500
+        if self.ports is not None:
+            port = self.ports.allocate(f"comm-{rbg.metadata.uid}", 1)[0]
+        return {
+            "rbg.comm-world": str(rank),
+            "rbg.comm-port": str(port),
+            "rbg.comm-groups": _json.dumps(groups),
+            "rbg.comm-members": _json.dumps(members),
+            "rbg.comm-rank-map": _json.dumps(rank_map),
+        }
 
     # ------------------------------------------------------------------
 
@@ -104,8 +157,9 @@ class RoleBasedGroupController:
                                     rbg.metadata.name, rbg.metadata.namespace)
         scale_caps = self._coordination_scale_caps(rbg, policy, statuses)
         partitions = self._coordination_partitions(rbg, policy, statuses)
+        comm_annos = self._comm_plan(rbg)
         requeue = self._reconcile_roles(rbg, revision.metadata.labels.get(
-            C.LABEL_REVISION_HASH, ""), scale_caps, partitions)
+            C.LABEL_REVISION_HASH, ""), scale_caps, partitions, comm_annos)
         self._publish_discovery(rbg)
         self._update_group_status(rbg, statuses)
         self._cleanup_orphans(rbg)
@@ -207,7 +261,8 @@ class RoleBasedGroupController:
 
     def _reconcile_roles(self, rbg: RoleBasedGroup, revision_hash: str,
                          scale_caps: Dict[str, int],
-                         partitions: Dict[str, int]) -> float:
+                         partitions: Dict[str, int],
+                         comm_annos: Dict[str, str]) -> float:
         """Dependency-ordered waves with readiness gates
         (reference :458-567 + dependency.go:94-117)."""
         statuses = self._construct_role_statuses(rbg)
@@ -215,7 +270,8 @@ class RoleBasedGroupController:
         for wave in sort_roles(rbg.spec.roles):
             for role in wave:
                 self._reconcile_single_role(rbg, role, revision_hash,
-                                            scale_caps, partitions)
+                                            scale_caps, partitions,
+                                            comm_annos)
             if not all(self._role_ready(rbg, r, self._construct_role_statuses(rbg))
                        for r in wave):
                 requeue = 0.3   # downstream waves wait for this one
@@ -225,7 +281,8 @@ class RoleBasedGroupController:
     def _reconcile_single_role(self, rbg: RoleBasedGroup, role: RoleSpec,
                                revision_hash: str,
                                scale_caps: Dict[str, int],
-                               partitions: Dict[str, int]) -> None:
+                               partitions: Dict[str, int],
+                               comm_annos: Dict[str, str] = {}) -> None:
         name = ris_name(rbg.metadata.name, role.name)
         components = expand_pattern(rbg, role)
         components = self._apply_engine_runtimes(rbg, role, components)
@@ -239,6 +296,8 @@ class RoleBasedGroupController:
                     if role.pattern == C.PATTERN_LEADER_WORKER
                     and role.leader_worker_pattern else 0)
         tmpl_annotations = {"rbg.lwp-size": str(lwp_size)} if lwp_size > 1 else {}
+        if comm_annos:
+            tmpl_annotations.update(comm_annos)
         if self.registry is not None:
             tmpl_annotations["rbg.config-path"] = self.registry.path_for(
                 rbg.metadata.namespace, rbg.metadata.name)

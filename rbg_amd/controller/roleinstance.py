@@ -307,6 +307,21 @@ class RoleInstanceController:
                          else f"pod-{wname}-{req.name}")
             allocated[req.name] = self.ports.allocate(scope_key, req.count)
         env.update(self.ports.env_for(allocated))
+        # RBG-wide communicator world (controller comm plan): global rank +
+        # rendezvous + subgroup layout — injected LAST so the group-wide
+        # rendezvous port wins over the legacy per-instance LWP master port
+        annos = inst.metadata.annotations
+        rank_map_raw = annos.get("rbg.comm-rank-map", "")
+        if rank_map_raw:
+            import json as _json
+            rank_map = _json.loads(rank_map_raw)
+            if wname in rank_map:
+                env["RBG_GLOBAL_RANK"] = str(rank_map[wname])
+                env["RBG_GLOBAL_WORLD"] = annos.get("rbg.comm-world", "1")
+                env[C.ENV_MASTER_ADDR] = "127.0.0.1"
+                env[C.ENV_MASTER_PORT] = annos.get("rbg.comm-port", "29500")
+                env["RBG_COMM_GROUPS"] = annos.get("rbg.comm-groups", "[]")
+                env["RBG_COMM_MEMBERS"] = annos.get("rbg.comm-members", "{}")
         # sibling-component discovery env (discovery/component.py)
         from ..discovery import component as comp_disc
         disc = comp_disc.parse_discovery(comp.annotations) or \

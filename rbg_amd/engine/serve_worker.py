@@ -95,25 +95,33 @@ class ServeWorker:
         self.ctx = ctx
         self.mode = mode
         self.cfg = _engine_cfg(ctx)
-        # TP rank group (leaderWorker pattern): join the RCCL/xGMI
-        # communicator from the injected RBG_LWP_* env; every rank runs the
-        # engine in lockstep (leader broadcasts request ops each step —
-        # greedy sampling is deterministic across ranks because activations
-        # are all-reduced, so no token broadcast is needed)
+        # RBG-wide communicator world (controller comm plan): blocking join
+        # is the gang barrier — the engine is not Ready until every member
+        # of the group is up.  The TP communicator is this world's subgroup
+        # for my instance; KV migration is point-to-point on the default
+        # group.  Lockstep semantics: leader broadcasts request ops; greedy
+        # sampling is deterministic across ranks because activations are
+        # all-reduced, so no token exchange is needed.
         self.tp = None
-        self.comm = None
-        if ctx.args.get("tp_from_env") and \
-                int(os.environ.get(C.ENV_LWP_GROUP_SIZE, "1") or 1) > 1:
+        self.gcomm = None
+        if int(os.environ.get("RBG_GLOBAL_WORLD", "1") or 1) > 1 and \
+                ctx.args.get("comm_from_env", True):
+            from ..parallel import comm as commmod
+            backend = (ctx.args.get("comm_backend") or
+                       ctx.args.get("tp_backend") or
+                       ctx.args.get("transfer_backend"))
+            self.gcomm = commmod.init_global_from_env(backend=backend)
+        if self.gcomm is not None and self.gcomm.tp_size > 1:
             from ..models.llama import TPContext
-            from ..parallel import comm
-            self.comm = comm.init_from_env(
-                backend=ctx.args.get("tp_backend"))
-            comm.warmup_collectives(self.comm)
-            self.cfg.tp_size = self.comm.world_size
-            self.cfg.tp_rank = self.comm.rank
-            self.tp = TPContext(size=self.comm.world_size,
-                                rank=self.comm.rank)
+            self.cfg.tp_size = self.gcomm.tp_size
+            self.cfg.tp_rank = self.gcomm.tp_rank
+            self.tp = TPContext(size=self.gcomm.tp_size,
+                                rank=self.gcomm.tp_rank,
+                                group=self.gcomm.tp_group)
+        self.my_instance = os.environ.get(C.ENV_ROLE_INSTANCE_NAME, "")
         self.engine = LLMEngine(self.cfg, self.tp)
+        self._xfer_lock = threading.Lock()
+        self._tickets: Dict[str, int] = {}
         self.is_leader = self.tp is None or self.tp.rank == 0
         self._pending_ops: List[Dict[str, Any]] = []
         self.rpc = RpcServer(port=_rpc_port(ctx)) if self.is_leader else None
@@ -143,10 +151,13 @@ class ServeWorker:
 
     @property
     def transfer(self):
+        """Legacy standalone transfer group (explicit transfer_port args);
+        the controller comm plan supersedes it — with a gcomm, page moves
+        ride the default group (_send_pages/_recv_pages)."""
         if self._transfer is None:
             from ..parallel.kv_transfer import TransferEngine
             plan = self.plan()
-            inst = os.environ.get(C.ENV_ROLE_INSTANCE_NAME, "")
+            inst = self.my_instance
             rank = plan.rank_of(inst) if inst in plan.members else 0
             self._transfer = TransferEngine(
                 rank=rank, world_size=plan.world, master_addr="127.0.0.1",
@@ -155,6 +166,52 @@ class ServeWorker:
                 if self.cfg.device == "cuda" else None,
                 backend=self.ctx.args.get("transfer_backend"))
         return self._transfer
+
+    def _xfer_buf_device(self):
+        if self.gcomm is not None and self.gcomm.backend == "gloo":
+            return torch.device("cpu")
+        return self.engine.runner.cache.kv.device
+
+    def _send_pages(self, pages: List[int], dst_rank: int) -> None:
+        import torch.distributed as dist
+        cache = self.engine.runner.cache
+        with self._xfer_lock:
+            idx = torch.tensor(pages, dtype=torch.int64, device=cache.kv.device)
+            buf = cache.kv.index_select(2, idx).contiguous()
+            if buf.device != self._xfer_buf_device():
+                buf = buf.to(self._xfer_buf_device())
+            dist.send(buf, dst_rank)
+
+    def _recv_pages(self, pages: List[int], src_rank: int) -> None:
+        import torch.distributed as dist
+        cache = self.engine.runner.cache
+        with self._xfer_lock:
+            m = cache.kv.shape
+            buf = torch.empty((m[0], m[1], len(pages), m[3], m[4], m[5]),
+                              dtype=cache.kv.dtype,
+                              device=self._xfer_buf_device())
+            dist.recv(buf, src_rank)
+            idx = torch.tensor(pages, dtype=torch.int64, device=cache.kv.device)
+            cache.kv.index_copy_(2, idx, buf.to(cache.kv.device))
+
+    def _peer_rank(self, peer_instance: str) -> int:
+        """My point-to-point counterpart in `peer_instance`: same position
+        within the instance's rank list (TP shard i talks to shard i)."""
+        mine = self.gcomm.ranks_of(self.my_instance)
+        pos = mine.index(self.gcomm.rank) if self.gcomm.rank in mine else 0
+        peers = self.gcomm.ranks_of(peer_instance)
+        return peers[pos]
+
+    def _decode_port(self, decode_instance: str, timeout: float = 30.0) -> int:
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            topo = self.ctx.load_topology().get("group", {})
+            for role in topo.get("roles", []):
+                for inst in role.get("instances", []):
+                    if inst.get("name") == decode_instance and inst.get("ports"):
+                        return int(inst["ports"][0])
+            time.sleep(0.2)
+        raise TimeoutError(f"no RPC port published for {decode_instance}")
 
     # -- TP lockstep --------------------------------------------------------
 
@@ -180,6 +237,21 @@ class ServeWorker:
             op["_result"] = seq
         elif kind == "reload":
             self.engine.reload_weights(int(op["seed"]))
+        elif kind == "migrate":
+            # prefill TP: each rank ships its own KV shard to its decode
+            # counterpart; page ids are rank-identical (lockstep allocators)
+            parked = self._finished_pages.pop(int(op["seq_id"]))
+            self._send_pages(parked[0], self._peer_rank(op["dst_instance"]))
+            self._release_parked(parked)
+        elif kind == "import":
+            seq = self._import_alloc(op["tokens"], int(op["num_pages"]),
+                                     int(op["max_new_tokens"]),
+                                     float(op.get("temperature", 0.0)))
+            self._import_finish(seq, int(op["first_token"]),
+                                int(op["max_new_tokens"]),
+                                float(op.get("arrival_time", 0.0)),
+                                self._peer_rank(op["src_instance"]))
+            self._tickets[op["ticket"]] = seq.seq_id
         ev = op.get("_ev")
         if ev is not None:
             ev.set()
@@ -195,6 +267,13 @@ class ServeWorker:
                            mode=self.mode, tp_rank=self.tp.rank,
                            tp_size=self.tp.size, device=self.cfg.device,
                            model=self.cfg.model.name)
+        # broadcast within THIS instance's subgroup only: the world may hold
+        # several lockstep groups (e.g. prefill TP + decode TP) whose ops
+        # must not interleave on the default group
+        my_ranks = (self.gcomm.ranks_of(self.my_instance)
+                    if self.gcomm is not None else [0])
+        bc_src = my_ranks[0] if my_ranks else 0
+        bc_group = self.tp.group
         try:
             while True:
                 if self.is_leader:
@@ -208,7 +287,8 @@ class ServeWorker:
                 else:
                     ops = []
                     payload = [None]
-                dist.broadcast_object_list(payload, src=0)
+                dist.broadcast_object_list(payload, src=bc_src,
+                                           group=bc_group)
                 wire = payload[0]
                 if any(op.get("kind") == "stop" for op in wire):
                     return
@@ -241,6 +321,7 @@ class ServeWorker:
             self.rpc.register("prefill", self._rpc_prefill)
         if self.mode == "decode":
             self.rpc.register("import_seq", self._rpc_import_seq)
+            self.rpc.register("resolve_ticket", self._rpc_resolve_ticket)
 
     def _rpc_reload(self, seed) -> None:
         if self.tp is not None:
@@ -305,47 +386,84 @@ class ServeWorker:
 
     # -- prefill role --------------------------------------------------------
 
+    def _release_parked(self, parked) -> None:
+        pages, nshared = parked
+        cache = self.engine.runner.cache
+        if nshared and cache.prefix is not None:
+            for pg in pages[:nshared]:
+                cache.prefix.release_page(pg)
+            cache.free(pages[nshared:])
+        else:
+            cache.free(pages)
+
     def _rpc_prefill(self, tokens: List[int], max_new_tokens: int,
                      decode_instance: str, temperature: float = 0.0
                      ) -> Dict[str, Any]:
         """Prefill + first token, then migrate KV to the decode peer.
-        Returns the decode-side seq id."""
-        with self.lock:
-            seq = self.engine.add_request(
-                tokens, SamplingParams(max_new_tokens=1))
-            self.results[seq.seq_id] = seq
+        Returns the decode-side seq id (or a ticket resolving to it)."""
+        if self.tp is not None:
+            seq = self._tp_submit_op({"kind": "submit", "tokens": tokens,
+                                      "max_new_tokens": 1,
+                                      "temperature": 0.0})
+        else:
+            with self.lock:
+                seq = self.engine.add_request(
+                    tokens, SamplingParams(max_new_tokens=1))
+                self.results[seq.seq_id] = seq
         while self._rpc_poll(seq.seq_id)["finished"] is False:
             time.sleep(0.002)
-        # keep pages alive past finish for the transfer
-        pages = list(self._finished_pages.pop(seq.seq_id))
+        if self.tp is not None:
+            # peek only: the migrate op pops on EVERY rank (leader included)
+            parked = self._finished_pages[seq.seq_id]
+        else:
+            parked = self._finished_pages.pop(seq.seq_id)
+        pages, nshared = parked
         first_token = seq.output_tokens[0]
-        # handshake with decode: it allocates pages and posts the recv
+        meta = dict(tokens=tokens, first_token=first_token,
+                    num_pages=len(pages), max_new_tokens=max_new_tokens,
+                    temperature=temperature, arrival_time=seq.arrival_time)
+        if self.gcomm is not None:
+            client = RpcClient("127.0.0.1",
+                               self._decode_port(decode_instance))
+            try:
+                res = client.call("import_seq",
+                                  src_instance=self.my_instance, **meta)
+                if self.tp is not None:
+                    # every rank sends its KV shard to its counterpart;
+                    # followers park identical page ids (lockstep)
+                    self._tp_submit_op({"kind": "migrate",
+                                        "seq_id": seq.seq_id,
+                                        "dst_instance": decode_instance})
+                else:
+                    self._send_pages(pages,
+                                     self._peer_rank(decode_instance))
+                    self._release_parked(parked)
+            finally:
+                client.close()
+            return {"decode_seq_id": res.get("seq_id"),
+                    "ticket": res.get("ticket"),
+                    "first_token": first_token, "ttft_s": seq.ttft()}
+        # legacy standalone transfer group (explicit transfer_port args)
         plan = self.plan()
         dinst = next(i for i in plan.decode_instances
                      if i["name"] == decode_instance)
         client = RpcClient("127.0.0.1", int(dinst["ports"][0]))
         try:
-            meta = client.call(
-                "import_seq", tokens=tokens, first_token=first_token,
-                num_pages=len(pages), max_new_tokens=max_new_tokens,
-                src_rank=plan.rank_of(
-                    os.environ.get(C.ENV_ROLE_INSTANCE_NAME, "")),
-                temperature=temperature,
-                arrival_time=seq.arrival_time)
+            res = client.call(
+                "import_seq",
+                src_rank=plan.rank_of(self.my_instance), **meta)
             self.transfer.send_pages(self.engine.runner.cache, pages,
                                      plan.rank_of(decode_instance))
         finally:
-            self.engine.runner.cache.free(pages)
+            self._release_parked(parked)
             client.close()
-        return {"decode_seq_id": meta["seq_id"], "first_token": first_token,
+        return {"decode_seq_id": res["seq_id"], "first_token": first_token,
                 "ttft_s": seq.ttft()}
 
     # -- decode role ---------------------------------------------------------
 
-    def _rpc_import_seq(self, tokens: List[int], first_token: int,
-                        num_pages: int, max_new_tokens: int, src_rank: int,
-                        temperature: float = 0.0,
-                        arrival_time: float = 0.0) -> Dict[str, Any]:
+    def _import_alloc(self, tokens, num_pages, max_new_tokens,
+                      temperature) -> Sequence:
         from .kv_cache import BlockTable
         with self.lock:
             seq = Sequence(list(tokens),
@@ -355,52 +473,90 @@ class ServeWorker:
             bt = BlockTable(self.engine.runner.cache)
             bt.pages = self.engine.runner.cache.alloc(num_pages)
             seq.block_table = bt
-            # visible to poll() immediately; stays unfinished until the
-            # recv thread enqueues it as running
             self.results[seq.seq_id] = seq
-        # recv synchronously: prefill sends right after this RPC returns…
-        # post the recv in a worker thread, THEN return so the send can start
-        done = threading.Event()
+        return seq
+
+    def _import_finish(self, seq: Sequence, first_token: int,
+                       max_new_tokens: int, arrival_time: float,
+                       recv_from: int) -> None:
+        """Receive the KV shard into the allocated pages and enqueue as
+        running.  Runs inline (lockstep apply) or in a recv thread."""
+        if self.gcomm is not None:
+            self._recv_pages(seq.block_table.pages, recv_from)
+        else:
+            self.transfer.recv_pages(self.engine.runner.cache,
+                                     seq.block_table.pages, recv_from)
+        with self.lock:
+            seq.append_token(int(first_token))
+            if arrival_time:
+                seq.arrival_time = arrival_time
+            seq.block_table.ensure(seq.num_tokens + max_new_tokens)
+            seq.status = "running"
+            self.engine.scheduler.running.append(seq)
+
+    def _rpc_import_seq(self, tokens: List[int], first_token: int,
+                        num_pages: int, max_new_tokens: int,
+                        src_rank: int = -1, src_instance: str = "",
+                        temperature: float = 0.0,
+                        arrival_time: float = 0.0) -> Dict[str, Any]:
+        if self.tp is not None:
+            # lockstep import: every decode rank allocates identical pages
+            # and receives from its prefill counterpart inside the apply
+            import uuid
+            ticket = uuid.uuid4().hex[:10]
+            with self.lock:
+                self._pending_ops.append({
+                    "kind": "import", "ticket": ticket, "tokens": tokens,
+                    "first_token": first_token, "num_pages": num_pages,
+                    "max_new_tokens": max_new_tokens,
+                    "temperature": temperature,
+                    "arrival_time": arrival_time,
+                    "src_instance": src_instance})
+            return {"ticket": ticket}
+        recv_from = (self._peer_rank(src_instance)
+                     if self.gcomm is not None else src_rank)
+        seq = self._import_alloc(tokens, num_pages, max_new_tokens,
+                                 temperature)
 
         def do_recv():
             try:
-                self.transfer.recv_pages(self.engine.runner.cache, bt.pages,
-                                         src_rank)
-                with self.lock:
-                    seq.append_token(int(first_token))
-                    if arrival_time:
-                        seq.arrival_time = arrival_time
-                    bt.ensure(seq.num_tokens + max_new_tokens)
-                    seq.status = "running"
-                    self.engine.scheduler.running.append(seq)
-                    self.results[seq.seq_id] = seq
-            finally:
-                done.set()
+                self._import_finish(seq, first_token, max_new_tokens,
+                                    arrival_time, recv_from)
+            except Exception:  # noqa: BLE001
+                log.exception("KV import failed")
+        # post the recv asynchronously, THEN return so the send can start
         threading.Thread(target=do_recv, daemon=True).start()
         return {"seq_id": seq.seq_id}
+
+    def _rpc_resolve_ticket(self, ticket: str) -> Dict[str, Any]:
+        sid = self._tickets.get(ticket)
+        return {"seq_id": sid}
 
     # -- engine loop ---------------------------------------------------------
 
     def run(self) -> None:
         if self.rpc:
             self.rpc.start()
-        if self.tp is not None:
-            self._run_tp_lockstep()
-            return
         if self.mode == "prefill":
             # prefill keeps finished sequences' pages for migration: patch
-            # the scheduler's release with a park list
+            # the scheduler's release with a park list (pages + how many of
+            # them belong to the prefix cache)
             sched = self.engine.scheduler
             orig = sched._retire_finished
 
             def retire_keep_pages():
                 for s in list(sched.running):
                     if s.should_stop() and s.block_table is not None:
-                        self._finished_pages[s.seq_id] = list(
-                            s.block_table.pages)
+                        self._finished_pages[s.seq_id] = (
+                            list(s.block_table.pages),
+                            s.block_table.num_shared)
                         s.block_table.pages = []   # keep pages alive
+                        s.block_table.num_shared = 0
                 orig()
             sched._retire_finished = retire_keep_pages
+        if self.tp is not None:
+            self._run_tp_lockstep()
+            return
         self.ctx.set_ready(rpc_port=self.rpc.port, mode=self.mode,
                            device=self.cfg.device,
                            model=self.cfg.model.name)

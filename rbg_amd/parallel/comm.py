@@ -80,3 +80,67 @@ def warmup_collectives(ctx: CommContext, sizes=(1 << 10, 1 << 20)) -> None:
 def destroy() -> None:
     if dist.is_initialized():
         dist.destroy_process_group()
+
+
+# ---------------------------------------------------------------------------
+# RBG-wide communicator world (the gang-scheduling realization):
+# the controller assigns every llm-engine worker of the group a GLOBAL rank
+# and publishes the instance-level groupings; engines join ONE default
+# process group (blocking until the whole gang is up — all-or-nothing
+# startup), carve their TP communicator as a subgroup, and run KV migration
+# as point-to-point sends on the default group.  One world also means one
+# rendezvous port per RBG instead of ad-hoc transfer ports.
+# ---------------------------------------------------------------------------
+
+ENV_GLOBAL_RANK = "RBG_GLOBAL_RANK"
+ENV_GLOBAL_WORLD = "RBG_GLOBAL_WORLD"
+ENV_COMM_GROUPS = "RBG_COMM_GROUPS"      # JSON [[ranks of instance], ...]
+ENV_COMM_MEMBERS = "RBG_COMM_MEMBERS"    # JSON {instance_name: [ranks]}
+
+
+@dataclass
+class GlobalComm:
+    rank: int
+    world_size: int
+    backend: str
+    tp_group: Optional[object] = None
+    tp_rank: int = 0
+    tp_size: int = 1
+    members: Optional[dict] = None       # instance -> [global ranks]
+    device: Optional["torch.device"] = None
+
+    def ranks_of(self, instance: str):
+        return list((self.members or {}).get(instance, []))
+
+
+def init_global_from_env(backend: Optional[str] = None,
+                         timeout_s: float = 300.0) -> Optional[GlobalComm]:
+    """Join the RBG-wide world described by RBG_GLOBAL_* env; returns None
+    when the group has a single member (no collectives needed)."""
+    import json
+    world = int(os.environ.get(ENV_GLOBAL_WORLD, "1") or 1)
+    if world <= 1:
+        return None
+    rank = int(os.environ.get(ENV_GLOBAL_RANK, "0") or 0)
+    host = os.environ.get(C.ENV_MASTER_ADDR, "127.0.0.1")
+    port = os.environ.get(C.ENV_MASTER_PORT, "29500")
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    dist.init_process_group(
+        backend=backend, rank=rank, world_size=world,
+        init_method=f"tcp://{host}:{port}",
+        timeout=datetime.timedelta(seconds=timeout_s))
+    groups = json.loads(os.environ.get(ENV_COMM_GROUPS, "[]") or "[]")
+    members = json.loads(os.environ.get(ENV_COMM_MEMBERS, "{}") or "{}")
+    comm = GlobalComm(rank=rank, world_size=world, backend=backend,
+                      members=members)
+    if backend == "nccl":
+        comm.device = torch.device("cuda", torch.cuda.current_device())
+    # every rank must create every subgroup, in the same order
+    for ranks in groups:
+        g = dist.new_group(ranks=ranks)
+        if rank in ranks:
+            comm.tp_group = g
+            comm.tp_rank = ranks.index(rank)
+            comm.tp_size = len(ranks)
+    return comm

@@ -168,8 +168,10 @@ def main(argv: Optional[List[str]] = None) -> int:
     try:
         runner(ctx)
     except Exception as e:  # noqa: BLE001 — any engine error fails the worker
+        import traceback
         write_status(run_dir, name, "Failed", error=repr(e))
-        print(f"worker {name}: runner failed: {e!r}", file=sys.stderr)
+        print(f"worker {name}: runner failed: {e!r}\n"
+              f"{traceback.format_exc()}", file=sys.stderr)
         return 1
     finally:
         ctx.stop_event.set()

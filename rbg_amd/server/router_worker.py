@@ -105,8 +105,19 @@ class Router:
             "prefill", tokens=tokens, max_new_tokens=max_new_tokens,
             decode_instance=decode["name"], temperature=temperature)
         ttft = time.monotonic() - t0
-        seq_id = pres["decode_seq_id"]
         dclient = self._client(decode)
+        seq_id = pres.get("decode_seq_id")
+        if seq_id is None:
+            # TP decode groups return a ticket that resolves once the
+            # lockstep import lands
+            deadline = time.monotonic() + 120
+            while seq_id is None and time.monotonic() < deadline:
+                seq_id = dclient.call("resolve_ticket",
+                                      ticket=pres["ticket"])["seq_id"]
+                if seq_id is None:
+                    time.sleep(0.01)
+            if seq_id is None:
+                raise RuntimeError("KV import ticket never resolved")
         while True:
             res = dclient.call("poll", seq_id=seq_id)
             if res["finished"]:

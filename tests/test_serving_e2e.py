@@ -121,9 +121,9 @@ def test_colocated_serving(mgr):
 
 @pytest.mark.timeout(420)
 def test_pd_disaggregated_serving(mgr):
-    xfer_port = _free_port()
-    shared = {"transfer_port": xfer_port, "transfer_world": 2,
-              "transfer_backend": "gloo",
+    # the controller's comm plan assigns global ranks + rendezvous port;
+    # engines only pick the backend (gloo on CPU)
+    shared = {"comm_backend": "gloo",
               "prefill_roles": ["prefill"], "decode_roles": ["decode"]}
     rbg = RoleBasedGroup(
         metadata=ObjectMeta(name="pd"),
@@ -189,4 +189,47 @@ def test_tp2_leader_worker_serving(mgr):
                      {"prompt_tokens": prompt, "max_new_tokens": 5},
                      timeout=240)
     assert len(res["tokens"]) == 5
+    assert res["tokens"] == _local_reference_tokens(prompt, 5, model="tiny-tp")
+
+
+@pytest.mark.timeout(600)
+def test_tp_pd_disaggregated_serving(mgr):
+    """Config-4 shape on CPU: prefill TP=2 + decode TP=2 leaderWorker roles
+    in ONE controller-assigned communicator world (gloo); per-rank KV shards
+    migrate point-to-point; output must exactly match a TP=1 colocated
+    engine (degree-invariant weights)."""
+    from rbg_amd.api.types import LeaderWorkerPattern
+    args = dict(ENGINE_ARGS, model="tiny-tp", cpu_model="tiny-tp",
+                comm_backend="gloo")
+
+    def tp_role(name, mode):
+        return RoleSpec(
+            name=name, replicas=1, dependencies=["router"],
+            pattern=C.PATTERN_LEADER_WORKER,
+            leader_worker_pattern=LeaderWorkerPattern(size=2),
+            template=EngineTemplate(engines=[EngineSpec(
+                name="engine", runner="llm-engine",
+                args=dict(args, mode=mode),
+                resources=EngineResources(cpu_only=True))]))
+
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="tppd"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("pd", {"prefill_roles": ["prefill"],
+                               "decode_roles": ["decode"],
+                               "vocab_size": 500}),
+            tp_role("prefill", "prefill"),
+            tp_role("decode", "decode"),
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "tppd"), timeout=240)
+    assert mgr.wait_for(
+        lambda: (_router_http_port(mgr, "tppd") is not None), timeout=30)
+    port = _router_http_port(mgr, "tppd")
+    torch.manual_seed(31)
+    prompt = torch.randint(0, 500, (21,)).tolist()
+    res = _http_post(port, "/generate",
+                     {"prompt_tokens": prompt, "max_new_tokens": 5},
+                     timeout=240)
+    assert len(res["tokens"]) == 5, res
     assert res["tokens"] == _local_reference_tokens(prompt, 5, model="tiny-tp")
