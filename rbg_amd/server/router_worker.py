@@ -90,12 +90,19 @@ class Router:
         if self.mode == "pd":
             return self._generate_pd(tokens, max_new_tokens, temperature, t0)
         inst = self._pick_role(self.worker_roles)
-        res = self._client(inst).call("generate", tokens=tokens,
-                                      max_new_tokens=max_new_tokens,
-                                      temperature=temperature)
-        res["wall_s"] = time.monotonic() - t0
-        res["instance"] = inst["name"]
-        return res
+        client = self._client(inst)
+        # submit + poll (never a long-blocking call: the per-instance RPC
+        # socket is shared by every concurrent request)
+        sid = client.call("submit", tokens=tokens,
+                          max_new_tokens=max_new_tokens,
+                          temperature=temperature)
+        while True:
+            res = client.call("poll", seq_id=sid)
+            if res["finished"]:
+                res["wall_s"] = time.monotonic() - t0
+                res["instance"] = inst["name"]
+                return res
+            time.sleep(0.005)
 
     def _generate_pd(self, tokens: List[int], max_new_tokens: int,
                      temperature: float, t0: float) -> Dict[str, Any]:

@@ -1,0 +1,198 @@
+#!/usr/bin/env python3
+"""Request-rate serving benchmark — the reference's published methodology
+(reference keps/74-mooncake-integration: sglang.bench_serving, 300 prompts,
+2048 in / 512 out, request-rate 10).
+
+Drives the FULL stack: Manager -> RBG (colocated or P/D roles, real engine
+processes) -> router HTTP; requests arrive at a Poisson-ish fixed rate from
+client threads; reports total token throughput, mean/p50/p99 TTFT and mean
+ITL, next to the reference's numbers.
+
+  python tools/bench_serving.py --mode colocated --prompts 300 --rate 10
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import tempfile
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+
+from rbg_amd.api import constants as KC
+from rbg_amd.api.types import (EngineResources, EngineSpec, EngineTemplate,
+                               ObjectMeta, RoleBasedGroup, RoleBasedGroupSpec,
+                               RoleSpec, get_condition)
+from rbg_amd.controller.manager import Manager, ManagerOptions
+from rbg_amd.server.rpc import RpcClient
+
+
+def build_rbg(args, engine_args):
+    def role(name, runner, rargs, gpus, deps=()):
+        res = (EngineResources(cpu_only=True) if gpus == 0 else
+               EngineResources(gpus=1, hbm_bytes=args.hbm_gb << 30))
+        return RoleSpec(name=name, replicas=1, dependencies=list(deps),
+                        template=EngineTemplate(engines=[EngineSpec(
+                            name="engine", runner=runner, args=rargs,
+                            resources=res)]))
+
+    if args.mode == "pd":
+        shared = dict(engine_args,
+                      prefill_roles=["prefill"], decode_roles=["decode"])
+        roles = [
+            role("router", "router",
+                 {"dispatch": "pd", "prefill_roles": ["prefill"],
+                  "decode_roles": ["decode"],
+                  "vocab_size": args.vocab}, 0),
+            role("prefill", "llm-engine", dict(shared, mode="prefill"),
+                 args.gpus_per_engine, deps=("router",)),
+            role("decode", "llm-engine", dict(shared, mode="decode"),
+                 args.gpus_per_engine, deps=("router",)),
+        ]
+    else:
+        roles = [
+            role("router", "router",
+                 {"dispatch": "colocated", "worker_roles": ["worker"],
+                  "vocab_size": args.vocab}, 0),
+            role("worker", "llm-engine", dict(engine_args, mode="colocated"),
+                 args.gpus_per_engine, deps=("router",)),
+        ]
+    return RoleBasedGroup(metadata=ObjectMeta(name="bench"),
+                          spec=RoleBasedGroupSpec(roles=roles))
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--mode", choices=["colocated", "pd"], default="colocated")
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--device", default="cuda")
+    ap.add_argument("--prompts", type=int, default=300)
+    ap.add_argument("--rate", type=float, default=10.0)
+    ap.add_argument("--in-len", type=int, default=2048)
+    ap.add_argument("--out-len", type=int, default=128)
+    ap.add_argument("--gpus-per-engine", type=int, default=1)
+    ap.add_argument("--hbm-gb", type=int, default=100)
+    ap.add_argument("--timeout", type=float, default=900.0)
+    args = ap.parse_args()
+    from rbg_amd.engine.config import ModelConfig
+    model_cfg = ModelConfig.preset(args.model)
+    args.vocab = min(model_cfg.vocab_size, 128000)
+
+    engine_args = {"model": args.model, "device": args.device,
+                   "max_batch_size": 256,
+                   "kv_pool_tokens": args.prompts *
+                   (args.in_len + args.out_len + 32) + 8192}
+    if args.device != "cuda":
+        engine_args["cpu_model"] = args.model
+    if args.mode == "pd" and args.device != "cuda":
+        engine_args["comm_backend"] = "gloo"
+
+    run_root = tempfile.mkdtemp(prefix="rbg-bench-serve-")
+    mgr = Manager(ManagerOptions(run_root=run_root,
+                                 num_gpus=0 if args.device == "cuda" else 8))
+    mgr.start()
+    results = []
+    try:
+        mgr.store.create(build_rbg(args, engine_args))
+
+        def ready():
+            rbg = mgr.store.try_get(KC.KIND_RBG, "bench")
+            c = rbg and get_condition(rbg.status.conditions, KC.COND_READY)
+            return bool(c and c.status == "True")
+        assert mgr.wait_for(ready, timeout=600), "group never became Ready"
+
+        # find the router's RPC-free HTTP port
+        port = None
+        deadline = time.time() + 60
+        while port is None and time.time() < deadline:
+            for inst in mgr.store.list(KC.KIND_ROLE_INSTANCE, selector={
+                    KC.LABEL_GROUP_NAME: "bench",
+                    KC.LABEL_ROLE_NAME: "router"}):
+                for w in inst.status.workers:
+                    if w.ports:
+                        port = w.ports[0]
+            time.sleep(0.2)
+        assert port, "router port not published"
+
+        torch.manual_seed(99)
+        prompts = [torch.randint(0, args.vocab,
+                                 (args.in_len,)).tolist()
+                   for _ in range(args.prompts)]
+        lock = threading.Lock()
+
+        def fire(idx, tokens):
+            import urllib.request
+            t0 = time.monotonic()
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{port}/generate",
+                data=json.dumps({"prompt_tokens": tokens,
+                                 "max_new_tokens": args.out_len}).encode(),
+                headers={"Content-Type": "application/json"})
+            try:
+                with urllib.request.urlopen(req, timeout=args.timeout) as r:
+                    res = json.loads(r.read())
+                wall = time.monotonic() - t0
+                ttft = res.get("ttft_s") or wall
+                n_out = len(res["tokens"])
+                itl = (wall - ttft) / max(1, n_out - 1)
+                with lock:
+                    results.append({"ttft": ttft, "itl": itl, "wall": wall,
+                                    "out": n_out})
+            except Exception as e:  # noqa: BLE001
+                with lock:
+                    results.append({"error": repr(e)})
+
+        threads = []
+        bench_start = time.monotonic()
+        for i, tokens in enumerate(prompts):
+            target = bench_start + i / args.rate
+            now = time.monotonic()
+            if target > now:
+                time.sleep(target - now)
+            t = threading.Thread(target=fire, args=(i, tokens))
+            t.start()
+            threads.append(t)
+        for t in threads:
+            t.join(timeout=args.timeout)
+        wall = time.monotonic() - bench_start
+    finally:
+        mgr.stop()
+
+    ok = [r for r in results if "error" not in r]
+    errs = [r for r in results if "error" in r]
+    ttfts = sorted(r["ttft"] for r in ok)
+    total_tokens = sum(r["out"] for r in ok) + args.in_len * len(ok)
+    out_tokens = sum(r["out"] for r in ok)
+    report = {
+        "config": {"mode": args.mode, "model": model_cfg.name,
+                   "prompts": args.prompts, "rate": args.rate,
+                   "in_len": args.in_len, "out_len": args.out_len},
+        "completed": len(ok), "errors": len(errs),
+        "total_token_throughput_tok_s": round(total_tokens / wall, 1),
+        "output_tok_s": round(out_tokens / wall, 1),
+        "mean_ttft_ms": round(1000 * statistics.mean(ttfts), 1) if ok else 0,
+        "p50_ttft_ms": round(1000 * ttfts[len(ttfts) // 2], 1) if ok else 0,
+        "p99_ttft_ms": round(
+            1000 * ttfts[min(len(ttfts) - 1,
+                             int(0.99 * len(ttfts)))], 1) if ok else 0,
+        "mean_itl_ms": round(1000 * statistics.mean(
+            r["itl"] for r in ok), 2) if ok else 0,
+        "wall_s": round(wall, 1),
+        "reference_baseline": {"total_token_throughput_tok_s": 1300.41,
+                               "mean_ttft_ms": 4808.37,
+                               "mean_itl_ms": 162.69,
+                               "note": "Qwen3-32B on NVIDIA (BASELINE.md)"},
+    }
+    if errs:
+        report["first_error"] = errs[0]["error"]
+    print(json.dumps(report, indent=1), flush=True)
+    return 0 if not errs else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
