@@ -100,19 +100,22 @@ class ModelRunner:
         admission — engine/scheduler.py), so the only host->device traffic
         per step is nothing at all; sampled tokens stay on-device as the
         next step's input."""
+        import numpy as np
         dev = self.device
         positions = self._i32([seq.num_tokens - 1 for seq in seqs])
         ctx = self._i32([seq.num_tokens for seq in seqs])
         tokens = torch.tensor([seq.last_token for seq in seqs],
                               dtype=torch.int64, device=dev)
         max_pages = max(len(seq.block_table.pages) for seq in seqs)
-        bt = torch.zeros(len(seqs), max_pages, dtype=torch.int32)
+        # numpy staging: batch composition churns every step under
+        # continuous serving, so the rebuild is on the step critical path
+        bt_np = np.zeros((len(seqs), max_pages), dtype=np.int32)
         for i, seq in enumerate(seqs):
-            bt[i, :len(seq.block_table.pages)] = torch.tensor(
-                seq.block_table.pages, dtype=torch.int32)
+            pages = seq.block_table.pages
+            bt_np[i, :len(pages)] = pages
         return {"ids": tuple(s.seq_id for s in seqs),
                 "positions": positions, "ctx": ctx, "tokens": tokens,
-                "bt": bt.to(dev), "fresh": True}
+                "bt": torch.from_numpy(bt_np).to(dev), "fresh": True}
 
     def _decode_slots(self, state: dict) -> torch.Tensor:
         ps = self.cfg.page_size
