@@ -315,6 +315,7 @@ class ServeWorker:
         self.rpc.register("generate", self._rpc_generate)
         self.rpc.register("submit", self._rpc_submit)
         self.rpc.register("poll", self._rpc_poll)
+        self.rpc.register("poll_many", self._rpc_poll_many)
         self.rpc.register("reload_weights", self._rpc_reload)
         self.rpc.register("apply_update", self._rpc_apply_update)
         if self.mode == "prefill":
@@ -374,6 +375,12 @@ class ServeWorker:
         return {"tokens": list(seq.output_tokens),
                 "finished": seq.status == FINISHED,
                 "ttft_s": seq.ttft()}
+
+    def _rpc_poll_many(self, seq_ids: List[int]) -> List[Dict[str, Any]]:
+        """One round trip for a whole batch of outstanding sequences — the
+        router polls through a single watcher instead of per-request loops
+        (hundreds of per-request pollers starve the engine loop's GIL)."""
+        return [self._rpc_poll(sid) for sid in seq_ids]
 
     def _rpc_generate(self, tokens: List[int], max_new_tokens: int = 64,
                       temperature: float = 0.0) -> Dict[str, Any]:

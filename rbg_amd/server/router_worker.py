@@ -34,6 +34,56 @@ def simple_tokenize(text: str, vocab_size: int = 128000) -> List[int]:
             for i, b in enumerate(text.encode())]
 
 
+class _InstanceWatcher:
+    """One poller per engine instance: batches every outstanding request
+    into a single poll_many RPC per tick, so N concurrent requests cost one
+    round trip instead of N."""
+
+    def __init__(self, client: RpcClient, tick: float = 0.01):
+        self.client = client
+        self.tick = tick
+        self._lock = threading.Lock()
+        self._waiting: Dict[int, threading.Event] = {}
+        self._results: Dict[int, Dict[str, Any]] = {}
+        self._thread: threading.Thread = None
+
+    def wait_for(self, seq_id: int, timeout: float) -> Dict[str, Any]:
+        ev = threading.Event()
+        with self._lock:
+            self._waiting[seq_id] = ev
+            if self._thread is None or not self._thread.is_alive():
+                self._thread = threading.Thread(target=self._loop, daemon=True)
+                self._thread.start()
+        if not ev.wait(timeout):
+            with self._lock:
+                self._waiting.pop(seq_id, None)
+            raise TimeoutError(f"seq {seq_id} did not finish in {timeout}s")
+        with self._lock:
+            return self._results.pop(seq_id)
+
+    def _loop(self):
+        while True:
+            with self._lock:
+                ids = list(self._waiting.keys())
+            if not ids:
+                time.sleep(self.tick)
+                with self._lock:
+                    if not self._waiting:
+                        continue
+                continue
+            try:
+                results = self.client.call("poll_many", seq_ids=ids)
+            except Exception:  # noqa: BLE001 — instance restarting; retry
+                time.sleep(0.2)
+                continue
+            with self._lock:
+                for sid, res in zip(ids, results):
+                    if res.get("finished") and sid in self._waiting:
+                        self._results[sid] = res
+                        self._waiting.pop(sid).set()
+            time.sleep(self.tick)
+
+
 class Router:
     def __init__(self, ctx: WorkerContext):
         self.ctx = ctx
@@ -44,6 +94,7 @@ class Router:
         self.prefill_roles = args.get("prefill_roles", ["prefill"])
         self.decode_roles = args.get("decode_roles", ["decode"])
         self._clients: Dict[str, RpcClient] = {}
+        self._watchers: Dict[str, _InstanceWatcher] = {}
         self._rr = itertools.count()
         self._lock = threading.Lock()
 
@@ -69,6 +120,18 @@ class Router:
                 self._clients[key] = c
             return c
 
+    def _watcher(self, inst: Dict[str, Any]) -> _InstanceWatcher:
+        key = inst["name"]
+        with self._lock:
+            w = self._watchers.get(key)
+            if w is None:
+                # dedicated polling connection (the control socket stays
+                # free for submits)
+                w = _InstanceWatcher(RpcClient(
+                    inst.get("address", "127.0.0.1"), int(inst["ports"][0])))
+                self._watchers[key] = w
+            return w
+
     def _pick_role(self, roles: List[str], timeout: float = 15.0
                    ) -> Dict[str, Any]:
         """Round-robin over ready instances; brief retry covers the window
@@ -90,19 +153,13 @@ class Router:
         if self.mode == "pd":
             return self._generate_pd(tokens, max_new_tokens, temperature, t0)
         inst = self._pick_role(self.worker_roles)
-        client = self._client(inst)
-        # submit + poll (never a long-blocking call: the per-instance RPC
-        # socket is shared by every concurrent request)
-        sid = client.call("submit", tokens=tokens,
-                          max_new_tokens=max_new_tokens,
-                          temperature=temperature)
-        while True:
-            res = client.call("poll", seq_id=sid)
-            if res["finished"]:
-                res["wall_s"] = time.monotonic() - t0
-                res["instance"] = inst["name"]
-                return res
-            time.sleep(0.005)
+        sid = self._client(inst).call("submit", tokens=tokens,
+                                      max_new_tokens=max_new_tokens,
+                                      temperature=temperature)
+        res = self._watcher(inst).wait_for(sid, timeout=900.0)
+        res["wall_s"] = time.monotonic() - t0
+        res["instance"] = inst["name"]
+        return res
 
     def _generate_pd(self, tokens: List[int], max_new_tokens: int,
                      temperature: float, t0: float) -> Dict[str, Any]:
@@ -125,15 +182,12 @@ class Router:
                     time.sleep(0.01)
             if seq_id is None:
                 raise RuntimeError("KV import ticket never resolved")
-        while True:
-            res = dclient.call("poll", seq_id=seq_id)
-            if res["finished"]:
-                res["ttft_s"] = ttft
-                res["wall_s"] = time.monotonic() - t0
-                res["prefill_instance"] = prefill["name"]
-                res["decode_instance"] = decode["name"]
-                return res
-            time.sleep(0.005)
+        res = self._watcher(decode).wait_for(seq_id, timeout=900.0)
+        res["ttft_s"] = ttft
+        res["wall_s"] = time.monotonic() - t0
+        res["prefill_instance"] = prefill["name"]
+        res["decode_instance"] = decode["name"]
+        return res
 
     def stats(self) -> Dict[str, Any]:
         out: Dict[str, Any] = {"mode": self.mode, "instances": {}}
