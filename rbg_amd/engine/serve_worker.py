@@ -53,6 +53,8 @@ def _engine_cfg(ctx: WorkerContext) -> EngineConfig:
         max_seq_len=int(args.get("max_seq_len", 8192)),
         kv_pool_tokens=int(args.get("kv_pool_tokens", 0)),
         gpu_memory_utilization=float(args.get("gpu_memory_utilization", 0.85)),
+        enable_prefix_cache=bool(args.get("enable_prefix_cache", True)),
+        max_prefill_tokens=int(args.get("max_prefill_tokens", 8192)),
         enforce_eager=bool(args.get("enforce_eager", device != "cuda")),
     )
 
