@@ -29,6 +29,10 @@ class EngineStats:
     decode_tokens: int = 0
     prefill_steps: int = 0
     decode_steps: int = 0
+    prefill_wall_s: float = 0.0
+    decode_wall_s: float = 0.0
+    idle_wall_s: float = 0.0
+    sched_wall_s: float = 0.0
     finished: int = 0
     ttfts: List[float] = field(default_factory=list)
     started: float = field(default_factory=time.monotonic)
@@ -40,6 +44,12 @@ class EngineStats:
         return {
             "prefill_tokens": self.prefill_tokens,
             "decode_tokens": self.decode_tokens,
+            "prefill_steps": self.prefill_steps,
+            "decode_steps": self.decode_steps,
+            "prefill_wall_s": round(self.prefill_wall_s, 2),
+            "decode_wall_s": round(self.decode_wall_s, 2),
+            "idle_wall_s": round(self.idle_wall_s, 2),
+            "sched_wall_s": round(self.sched_wall_s, 2),
             "output_tok_per_s": self.decode_tokens / wall,
             "finished": self.finished,
             "p50_ttft_s": p50,
@@ -64,7 +74,10 @@ class LLMEngine:
 
     def step(self) -> str:
         """Run one engine iteration; returns the mode executed."""
+        t0 = time.monotonic()
         mode, seqs = self.scheduler.schedule()
+        self.stats.sched_wall_s += time.monotonic() - t0
+        t0 = time.monotonic()
         if mode == "prefill":
             self.runner.prefill(seqs)
             self.stats.prefill_tokens += sum(s.chunk_len for s in seqs)
@@ -82,11 +95,15 @@ class LLMEngine:
                         s.prompt_tokens, s.block_table.pages,
                         s.block_table.num_shared)
             self.scheduler.finish_prefill(completed)
+            self.stats.prefill_wall_s += time.monotonic() - t0
         elif mode == "decode":
             self.runner.decode(seqs)
             self.stats.decode_tokens += len(seqs)
             self.stats.decode_steps += 1
             self.scheduler.finish_decode()
+            self.stats.decode_wall_s += time.monotonic() - t0
+        else:
+            self.stats.idle_wall_s += time.monotonic() - t0
         self._collect_finished()
         return mode
 

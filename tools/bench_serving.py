@@ -160,6 +160,19 @@ def main() -> int:
         for t in threads:
             t.join(timeout=args.timeout)
         wall = time.monotonic() - bench_start
+        # engine-side step accounting (per worker instance)
+        engine_stats = {}
+        role = "worker" if args.mode == "colocated" else "decode"
+        for inst in mgr.store.list(KC.KIND_ROLE_INSTANCE, selector={
+                KC.LABEL_GROUP_NAME: "bench", KC.LABEL_ROLE_NAME: role}):
+            for w in inst.status.workers:
+                if w.ports:
+                    try:
+                        c = RpcClient("127.0.0.1", w.ports[0])
+                        engine_stats[w.name] = c.call("stats")
+                        c.close()
+                    except Exception as e:  # noqa: BLE001
+                        engine_stats[w.name] = {"error": repr(e)}
     finally:
         mgr.stop()
 
@@ -188,6 +201,7 @@ def main() -> int:
                                "mean_itl_ms": 162.69,
                                "note": "Qwen3-32B on NVIDIA (BASELINE.md)"},
     }
+    report["engine_stats"] = engine_stats
     if errs:
         report["first_error"] = errs[0]["error"]
     print(json.dumps(report, indent=1), flush=True)
