@@ -40,6 +40,8 @@ class PagedKVCache:
         # page 0 is reserved scratch: hipGraph decode padding rows write
         # their (dead) KV slot there (model_runner._decode_graph)
         self._free: List[int] = list(range(num_pages - 1, 0, -1))
+        import threading
+        self._lock = threading.Lock()   # alloc/free may race import threads
         self.prefix: Optional["PrefixCache"] = (
             PrefixCache(self) if cfg.enable_prefix_cache else None)
 
@@ -70,16 +72,17 @@ class PagedKVCache:
         return len(self._free) + extra
 
     def alloc(self, n: int) -> List[int]:
-        if n > len(self._free) and self.prefix is not None:
-            # reclaim idle cached-prefix pages (LRU) under pressure
-            self._free.extend(self.prefix.evict(n - len(self._free)))
-        if n > len(self._free):
-            raise OutOfPages(f"need {n} pages, {len(self._free)} free")
-        out = [self._free.pop() for _ in range(n)]
-        return out
+        with self._lock:
+            if n > len(self._free) and self.prefix is not None:
+                # reclaim idle cached-prefix pages (LRU) under pressure
+                self._free.extend(self.prefix.evict(n - len(self._free)))
+            if n > len(self._free):
+                raise OutOfPages(f"need {n} pages, {len(self._free)} free")
+            return [self._free.pop() for _ in range(n)]
 
     def free(self, pages: List[int]) -> None:
-        self._free.extend(pages)
+        with self._lock:
+            self._free.extend(pages)
 
     # -- kernel views --------------------------------------------------------
 

@@ -128,6 +128,10 @@ class ServeWorker:
         self._pending_ops: List[Dict[str, Any]] = []
         self.rpc = RpcServer(port=_rpc_port(ctx)) if self.is_leader else None
         self.lock = threading.Lock()
+        # rare exclusive operations (weight reload) raise this flag so the
+        # engine loop yields the lock — a busy loop re-acquiring its own
+        # lock otherwise starves waiters indefinitely (lock convoy)
+        self._pause = threading.Event()
         self.results: Dict[int, Sequence] = {}
         self._finished_pages: Dict[int, List[int]] = {}
         self._plan: Optional[_TransferPlan] = None
@@ -329,9 +333,13 @@ class ServeWorker:
     def _rpc_reload(self, seed) -> None:
         if self.tp is not None:
             self._tp_submit_op({"kind": "reload", "seed": int(seed)})
-        else:
+            return
+        self._pause.set()       # make the engine loop yield the lock
+        try:
             with self.lock:
                 self.engine.reload_weights(int(seed))
+        finally:
+            self._pause.clear()
 
     def _rpc_apply_update(self, args: Dict[str, Any]) -> Dict[str, Any]:
         """Live in-place update: apply the arg diffs an engine can absorb
@@ -363,11 +371,13 @@ class ServeWorker:
             if seq is None:
                 raise RuntimeError("TP lockstep loop did not apply the op")
             return seq.seq_id
-        with self.lock:
-            seq = self.engine.add_request(
-                tokens, SamplingParams(max_new_tokens=max_new_tokens,
-                                       temperature=temperature))
-            self.results[seq.seq_id] = seq
+        # lock-free: scheduler.waiting is a single-consumer deque (appends
+        # are GIL-atomic); taking the engine lock here would convoy behind
+        # the never-idle engine loop and serialize admissions
+        seq = self.engine.add_request(
+            tokens, SamplingParams(max_new_tokens=max_new_tokens,
+                                   temperature=temperature))
+        self.results[seq.seq_id] = seq
         return seq.seq_id
 
     def _rpc_poll(self, seq_id: int) -> Dict[str, Any]:
@@ -415,10 +425,9 @@ class ServeWorker:
                                       "max_new_tokens": 1,
                                       "temperature": 0.0})
         else:
-            with self.lock:
-                seq = self.engine.add_request(
-                    tokens, SamplingParams(max_new_tokens=1))
-                self.results[seq.seq_id] = seq
+            seq = self.engine.add_request(
+                tokens, SamplingParams(max_new_tokens=1))
+            self.results[seq.seq_id] = seq
         while self._rpc_poll(seq.seq_id)["finished"] is False:
             time.sleep(0.002)
         if self.tp is not None:
@@ -474,15 +483,16 @@ class ServeWorker:
     def _import_alloc(self, tokens, num_pages, max_new_tokens,
                       temperature) -> Sequence:
         from .kv_cache import BlockTable
-        with self.lock:
-            seq = Sequence(list(tokens),
-                           SamplingParams(max_new_tokens=max_new_tokens,
-                                          temperature=temperature))
-            seq.imported_kv = True
-            bt = BlockTable(self.engine.runner.cache)
-            bt.pages = self.engine.runner.cache.alloc(num_pages)
-            seq.block_table = bt
-            self.results[seq.seq_id] = seq
+        # page allocation is internally locked; everything else here is
+        # GIL-atomic against the engine loop
+        seq = Sequence(list(tokens),
+                       SamplingParams(max_new_tokens=max_new_tokens,
+                                      temperature=temperature))
+        seq.imported_kv = True
+        bt = BlockTable(self.engine.runner.cache)
+        bt.pages = self.engine.runner.cache.alloc(num_pages)
+        seq.block_table = bt
+        self.results[seq.seq_id] = seq
         return seq
 
     def _import_finish(self, seq: Sequence, first_token: int,
@@ -495,13 +505,19 @@ class ServeWorker:
         else:
             self.transfer.recv_pages(self.engine.runner.cache,
                                      seq.block_table.pages, recv_from)
-        with self.lock:
-            seq.append_token(int(first_token))
-            if arrival_time:
-                seq.arrival_time = arrival_time
-            seq.block_table.ensure(seq.num_tokens + max_new_tokens)
-            seq.status = "running"
-            self.engine.scheduler.running.append(seq)
+        # running-list append must not race finish_decode; pause first so
+        # the busy engine loop actually yields the lock (convoy avoidance)
+        self._pause.set()
+        try:
+            with self.lock:
+                seq.append_token(int(first_token))
+                if arrival_time:
+                    seq.arrival_time = arrival_time
+                seq.block_table.ensure(seq.num_tokens + max_new_tokens)
+                seq.status = "running"
+                self.engine.scheduler.running.append(seq)
+        finally:
+            self._pause.clear()
 
     def _rpc_import_seq(self, tokens: List[int], first_token: int,
                         num_pages: int, max_new_tokens: int,
@@ -571,6 +587,9 @@ class ServeWorker:
                            model=self.cfg.model.name)
         try:
             while not self.ctx.should_stop():
+                if self._pause.is_set():
+                    time.sleep(0.002)
+                    continue
                 with self.lock:
                     mode = self.engine.step()
                 if mode == "idle":
