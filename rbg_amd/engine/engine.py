@@ -69,6 +69,12 @@ class LLMEngine:
     def add_request(self, prompt_tokens: List[int],
                     sampling: Optional[SamplingParams] = None) -> Sequence:
         seq = Sequence(prompt_tokens, sampling)
+        limit = min(self.cfg.max_seq_len, self.cfg.model.max_position)
+        total = seq.num_prompt_tokens + seq.sampling.max_new_tokens
+        if total > limit:
+            raise ValueError(
+                f"prompt+max_new_tokens = {total} exceeds the engine limit "
+                f"{limit} (max_seq_len/model.max_position)")
         self.scheduler.add(seq)
         return seq
 
