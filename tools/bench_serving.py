@@ -89,12 +89,15 @@ def main() -> int:
                    (args.in_len + args.out_len + 32) + 8192}
     if args.device != "cuda":
         engine_args["cpu_model"] = args.model
-    if args.mode == "pd" and args.device != "cuda":
-        engine_args["comm_backend"] = "gloo"
-
     run_root = tempfile.mkdtemp(prefix="rbg-bench-serve-")
     mgr = Manager(ManagerOptions(run_root=run_root,
                                  num_gpus=0 if args.device == "cuda" else 8))
+    if args.mode == "pd":
+        # RCCL cannot place two ranks on one device: engines sharing a GPU
+        # (or running on CPU) must carry the comm world over gloo
+        engines_needed = 2 * args.gpus_per_engine
+        if args.device != "cuda" or mgr.topo.num_gpus < engines_needed:
+            engine_args["comm_backend"] = "gloo"
     mgr.start()
     results = []
     try:
