@@ -25,7 +25,7 @@ constexpr float NEG_INF = -1e30f;
 // VARIANT 1: packed-bf16 v_dot2 scores (96 VGPRs, 5 waves/SIMD for QPG<=4).
 // Runtime-selected (RBG_DECODE_VARIANT) for within-probe A/B.
 template <int QPG, int VARIANT>
-__global__ __launch_bounds__(256, (VARIANT == 1 && QPG <= 4) ? 5 : 2)
+__global__ __launch_bounds__(256, (VARIANT >= 1 && QPG <= 4) ? 5 : 2)
 void decode_attn_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
@@ -60,7 +60,7 @@ void decode_attn_kernel(
   // q fragment: VARIANT 1 keeps it packed bf16 (v_dot2 pairs, half the
   // registers); VARIANT 0 pre-converts to f32 with the scale folded in
   typedef __attribute__((ext_vector_type(2))) __bf16 bfpair;
-  bfpair qp[VARIANT == 1 ? QPG : 1][4];
+  bfpair qp[VARIANT >= 1 ? QPG : 1][4];
   float qf[VARIANT == 0 ? QPG : 1][8];
   {
     const __hip_bfloat16* qrow =
@@ -69,7 +69,7 @@ void decode_attn_kernel(
     for (int h = 0; h < QPG; ++h) {
       Bf16x8U qv;
       qv.u = *reinterpret_cast<const uint4*>(qrow + h * HEAD_DIM + dbase);
-      if (VARIANT == 1) {
+      if (VARIANT >= 1) {
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           qp[h][j] = reinterpret_cast<const bfpair*>(&qv)[j];
@@ -116,7 +116,7 @@ void decode_attn_kernel(
 #pragma unroll
     for (int h = 0; h < QPG; ++h) {
       float s = 0.f;
-      if (VARIANT == 1) {
+      if (VARIANT >= 1) {
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           s = __builtin_amdgcn_fdot2_f32_bf16(qp[h][j], kp[j], s, false);
@@ -155,10 +155,30 @@ void decode_attn_kernel(
     vd.u = *reinterpret_cast<const uint4*>(val_cache + off + dbase);
   };
 
-  // 4 keys per wave-iteration with 1-ahead prefetch.  (Wider 8-key and
-  // 2-ahead-ring variants both measured null-to-negative —
-  // profiles/decode_breakdown.md.)
-  {
+  // 4 keys per wave-iteration with 1-ahead data prefetch.  (Wider 8-key
+  // and 2-ahead-ring variants measured null-to-negative —
+  // profiles/decode_breakdown.md.)  VARIANT 2 additionally computes the
+  // page-table address one MORE iteration ahead so the btab read never
+  // serializes in front of the K/V loads.
+  if (VARIANT == 2) {
+    int base = key_begin + wave * 4;
+    Bf16x8U k_pref, v_pref;
+    size_t off_next = 0;
+    if (base < key_end) {
+      const size_t off0 = row_offset(base);
+      k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off0 + dbase);
+      v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off0 + dbase);
+      off_next = row_offset(base + 16);
+    }
+    for (; base < key_end; base += 16) {
+      const bool valid = base + group < key_end;
+      Bf16x8U kv = k_pref, vv = v_pref;
+      k_pref.u = *reinterpret_cast<const uint4*>(key_cache + off_next + dbase);
+      v_pref.u = *reinterpret_cast<const uint4*>(val_cache + off_next + dbase);
+      off_next = row_offset(base + 32);
+      process4(kv, vv, valid);
+    }
+  } else {
     int base = key_begin + wave * 4;
     Bf16x8U k_pref, v_pref;
     if (base < key_end)
@@ -274,15 +294,20 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                      (const __hip_bfloat16*)val_cache,                        \
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits)
-  switch (qpg * 2 + (variant ? 1 : 0)) {
-    case 2: LAUNCH_QPG(1, 0); break;
-    case 3: LAUNCH_QPG(1, 1); break;
-    case 4: LAUNCH_QPG(2, 0); break;
-    case 5: LAUNCH_QPG(2, 1); break;
-    case 8: LAUNCH_QPG(4, 0); break;
-    case 9: LAUNCH_QPG(4, 1); break;
-    case 16: LAUNCH_QPG(8, 0); break;
-    case 17: LAUNCH_QPG(8, 1); break;
+  const int v = variant < 0 ? 1 : (variant > 2 ? 2 : variant);
+  switch (qpg * 4 + v) {
+    case 4: LAUNCH_QPG(1, 0); break;
+    case 5: LAUNCH_QPG(1, 1); break;
+    case 6: LAUNCH_QPG(1, 2); break;
+    case 8: LAUNCH_QPG(2, 0); break;
+    case 9: LAUNCH_QPG(2, 1); break;
+    case 10: LAUNCH_QPG(2, 2); break;
+    case 16: LAUNCH_QPG(4, 0); break;
+    case 17: LAUNCH_QPG(4, 1); break;
+    case 18: LAUNCH_QPG(4, 2); break;
+    case 32: LAUNCH_QPG(8, 0); break;
+    case 33: LAUNCH_QPG(8, 1); break;
+    case 34: LAUNCH_QPG(8, 2); break;
     default: return;   // validated host-side
   }
 #undef LAUNCH_QPG

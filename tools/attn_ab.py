@@ -36,13 +36,14 @@ def run(variant):
     torch.cuda.synchronize()
     return (time.monotonic() - t0) / a.iters
 
-for v in (0, 1):
-    run(v)   # warmup both
-results = {0: [], 1: []}
+VARIANTS = (0, 1, 2)
+for v in VARIANTS:
+    run(v)   # warmup
+results = {v: [] for v in VARIANTS}
 for r in range(a.rounds):
-    for v in (0, 1):
+    for v in VARIANTS:
         results[v].append(run(v))
-for v in (0, 1):
+for v in VARIANTS:
     med = statistics.median(results[v])
     best = min(results[v])
     print(f"variant {v}: median {med*1e3:.3f} ms ({gb/med:.2f} GB/s), "
