@@ -103,8 +103,9 @@ def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
 def prefill_attention(q, k, v, cu_seqlens, scale: float,
                       cu_seqlens_k=None) -> torch.Tensor:
     if _on_gpu(q):
+        qtile = 128 if (PREFILL_SWZ & 4) else 64   # bit 2 = 8-wave blocks
         block_info, seq_lens = reference.prefill_block_info(
-            cu_seqlens.cpu(),
+            cu_seqlens.cpu(), qtile=qtile,
             cu_seqlens_k=None if cu_seqlens_k is None else cu_seqlens_k.cpu())
         return _require_hip().prefill_attention(
             q, k, v, block_info.to(q.device), seq_lens.to(q.device), scale,

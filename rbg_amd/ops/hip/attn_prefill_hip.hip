@@ -21,7 +21,6 @@
 namespace {
 
 constexpr int HEAD_DIM = 128;
-constexpr int QTILE = 64;      // q rows per block (16 per wave)
 constexpr int KTILE = 64;      // keys per LDS tile
 constexpr int PAD = 8;         // bf16 row padding (16 B) — bank-conflict fix
 constexpr float NEG_INF = -1e30f;
@@ -62,8 +61,10 @@ DEV_INLINE int v_img_off(int key, int dim) {
 // DB=1: T14 register staging (issue tile t+1's global loads BEFORE tile
 // t's compute, write them to LDS after the barrier) — HBM latency hides
 // under the MFMAs at the cost of ~32 staging VGPRs (guide G15/T14).
-template <int SWZ, int DB>
-__global__ __launch_bounds__(256) void prefill_attn_kernel(
+// NW: waves per block (4 -> 64 q rows, 8 -> 128 q rows; the bigger tile
+// halves K/V staging traffic per output row).
+template <int SWZ, int DB, int NW>
+__global__ __launch_bounds__(NW * 64) void prefill_attn_kernel(
     __hip_bfloat16* __restrict__ out,        // [T, QH, D]
     const __hip_bfloat16* __restrict__ q,    // [T, QH, D]
     const __hip_bfloat16* __restrict__ k,    // [T, KVH, D]
@@ -106,6 +107,8 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   const int q_offset = block_info[blk * 4 + 3];
   const int seq_len = seq_lens[blk];     // total KV tokens
 
+  constexpr int QTILE = NW * 16;     // q rows per block (16 per wave)
+  constexpr int NTHREADS = NW * 64;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -114,7 +117,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
 
   __shared__ __hip_bfloat16 k_lds[KTILE][HEAD_DIM + PAD];
   __shared__ __hip_bfloat16 v_img[KTILE * HEAD_DIM];   // tr-read image
-  __shared__ __hip_bfloat16 p_lds[4][16][KTILE + PAD];
+  __shared__ __hip_bfloat16 p_lds[NW][16][KTILE + PAD];
 
   // ---- Q fragments: wave's 16 rows, 4 k-steps of 32 dims ---------------
   const int num_q_rows = seq_len - q_offset;    // rows in the q tensor
@@ -145,11 +148,12 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   // UNCONDITIONAL clamped loads: a per-piece `if (krow < seq_len)` guard
   // makes hipcc branch around each load and drain vmcnt(0) per element
   // (guide §5 trap 4c); out-of-range rows are masked in softmax anyway.
-  uint4 kreg[4], vreg[4];
+  constexpr int PIECES = KTILE * (HEAD_DIM / 8) / NTHREADS;  // per thread
+  uint4 kreg[PIECES], vreg[PIECES];
   auto issue_loads = [&](int kt2) {
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const int i = tid + s * 256;
+    for (int s = 0; s < PIECES; ++s) {
+      const int i = tid + s * NTHREADS;
       const int key = i >> 4;
       const int chunk = (i & 15) * 8;
       const int krow = min(kt2 * KTILE + key, seq_len - 1);
@@ -161,8 +165,8 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   };
   auto write_tile = [&]() {
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const int i = tid + s * 256;
+    for (int s = 0; s < PIECES; ++s) {
+      const int i = tid + s * NTHREADS;
       const int key = i >> 4;
       const int chunk = (i & 15) * 8;
       *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = kreg[s];
@@ -320,19 +324,25 @@ void launch_prefill_attention(void* out, const void* q, const void* k,
                               const void* seq_lens, float scale, int nblocks,
                               int num_q_heads, int num_kv_heads, int swz,
                               hipStream_t stream) {
-  dim3 grid(nblocks * num_q_heads), block(256);
-  // swz bit 0: XCD-affine GQA swizzle; bit 1: T14 double-buffer staging
-#define PF_LAUNCH(S, D)                                                     \
-  hipLaunchKernelGGL((prefill_attn_kernel<S, D>), grid, block, 0, stream,   \
+  dim3 grid(nblocks * num_q_heads);
+  // swz bit 0: XCD-affine GQA swizzle; bit 1: T14 double-buffer staging;
+  // bit 2: 8-wave blocks (host must build block_info with qtile=128)
+#define PF_LAUNCH(S, D, NW)                                                 \
+  hipLaunchKernelGGL((prefill_attn_kernel<S, D, NW>), grid, dim3(NW * 64),  \
+                     0, stream,                                             \
                      (__hip_bfloat16*)out, (const __hip_bfloat16*)q,        \
                      (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,    \
                      (const int*)block_info, (const int*)seq_lens, scale,   \
                      num_q_heads, num_kv_heads, nblocks)
-  switch (swz & 3) {
-    case 0: PF_LAUNCH(0, 0); break;
-    case 1: PF_LAUNCH(1, 0); break;
-    case 2: PF_LAUNCH(0, 1); break;
-    case 3: PF_LAUNCH(1, 1); break;
+  switch (swz & 7) {
+    case 0: PF_LAUNCH(0, 0, 4); break;
+    case 1: PF_LAUNCH(1, 0, 4); break;
+    case 2: PF_LAUNCH(0, 1, 4); break;
+    case 3: PF_LAUNCH(1, 1, 4); break;
+    case 4: PF_LAUNCH(0, 0, 8); break;
+    case 5: PF_LAUNCH(1, 0, 8); break;
+    case 6: PF_LAUNCH(0, 1, 8); break;
+    case 7: PF_LAUNCH(1, 1, 8); break;
   }
 #undef PF_LAUNCH
 }

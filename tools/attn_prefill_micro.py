@@ -27,22 +27,25 @@ k = torch.randn(T, a.kvh, D, dtype=torch.bfloat16, device=dev)
 v = torch.randn_like(k)
 cu = torch.arange(0, T + 1, a.slen, dtype=torch.int32, device=dev)
 from rbg_amd.ops import reference as refmod
-block_info, seq_lens = refmod.prefill_block_info(cu.cpu())
-block_info, seq_lens = block_info.to(dev), seq_lens.to(dev)
+_binfo = {}
+for qt in (64, 128):
+    bi, sl = refmod.prefill_block_info(cu.cpu(), qtile=qt)
+    _binfo[qt] = (bi.to(dev), sl.to(dev))
 
 def run(swz):
+    bi, sl = _binfo[128 if swz & 4 else 64]
     torch.cuda.synchronize()
     t0 = time.monotonic()
     for _ in range(a.iters):
-        ops._hip.prefill_attention(q, k, v, block_info, seq_lens, 0.088, swz)
+        ops._hip.prefill_attention(q, k, v, bi, sl, 0.088, swz)
     torch.cuda.synchronize()
     return (time.monotonic() - t0) / a.iters
 
 flops = a.seqs * 2 * 2 * (a.slen * a.slen / 2) * D * a.qh
-for swz in (0, 2):
+for swz in variants:
     run(swz)
 import statistics
-variants = (0, 2)
+variants = (0, 4)
 res = {v: [] for v in variants}
 for _ in range(a.rounds):
     for swz in variants:
