@@ -42,11 +42,11 @@ def run(swz):
     return (time.monotonic() - t0) / a.iters
 
 flops = a.seqs * 2 * 2 * (a.slen * a.slen / 2) * D * a.qh
-for swz in variants:
-    run(swz)
 import statistics
 variants = (0, 4)
 res = {v: [] for v in variants}
+for swz in variants:
+    run(swz)
 for _ in range(a.rounds):
     for swz in variants:
         res[swz].append(run(swz))
