@@ -10,7 +10,7 @@ import copy
 import dataclasses
 import functools
 import typing
-from typing import Any, Dict, Optional, Type, TypeVar, get_args, get_origin
+from typing import Any, Dict, Type, TypeVar, get_args, get_origin
 
 T = TypeVar("T")
 

@@ -16,7 +16,6 @@ from __future__ import annotations
 
 import argparse
 import difflib
-import json
 import sys
 from typing import List, Optional
 

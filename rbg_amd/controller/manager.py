@@ -11,7 +11,6 @@ drive `reconcile_all()` synchronously.
 from __future__ import annotations
 
 import logging
-import queue
 import threading
 import time
 from dataclasses import dataclass

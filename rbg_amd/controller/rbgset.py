@@ -9,7 +9,7 @@ from __future__ import annotations
 from typing import List
 
 from ..api import constants as C
-from ..api.serde import asdict, clone, fromdict
+from ..api.serde import asdict, fromdict
 from ..api.types import (Condition, ObjectMeta, RoleBasedGroup,
                          RoleBasedGroupSet, RoleBasedGroupSpec,
                          get_condition, set_condition)

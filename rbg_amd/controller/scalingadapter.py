@@ -10,8 +10,6 @@ group spec.
 from __future__ import annotations
 
 import time
-from typing import Optional
-
 from ..api import constants as C
 from ..api.types import RoleBasedGroupScalingAdapter
 from ..store.store import Store

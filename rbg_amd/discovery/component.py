@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import json
 import logging
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..api import constants as C
 

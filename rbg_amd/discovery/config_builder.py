@@ -10,11 +10,10 @@ groups directly from the topology file.
 from __future__ import annotations
 
 import os
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 import yaml
 
-from ..api import constants as C
 
 
 def instance_name(rbg_name: str, role: str, index: int) -> str:

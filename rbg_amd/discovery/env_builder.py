@@ -8,7 +8,7 @@ its RCCL communicator without any service discovery round-trip.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..api import constants as C
 from ..api.types import EngineSpec, EnvVar

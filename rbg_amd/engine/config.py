@@ -7,7 +7,6 @@ benchmark config, llama-3-70b the TP=4+ config, tiny the CPU test config.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Optional
 
 
 @dataclass

@@ -12,8 +12,6 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
-import torch
-
 from ..models.llama import TPContext
 from .config import EngineConfig
 from .model_runner import ModelRunner

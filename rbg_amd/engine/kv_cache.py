@@ -17,7 +17,7 @@ from typing import Dict, List, Optional, Tuple
 
 import torch
 
-from .config import EngineConfig, ModelConfig
+from .config import EngineConfig
 
 
 class OutOfPages(RuntimeError):

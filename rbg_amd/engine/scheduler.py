@@ -11,11 +11,11 @@ pools).
 from __future__ import annotations
 
 from collections import deque
-from typing import Deque, List, Optional, Tuple
+from typing import Deque, List, Tuple
 
 from .config import EngineConfig
-from .kv_cache import BlockTable, OutOfPages, PagedKVCache
-from .sequence import FINISHED, RUNNING, WAITING, Sequence
+from .kv_cache import BlockTable, PagedKVCache
+from .sequence import FINISHED, RUNNING, Sequence
 
 
 class Scheduler:

@@ -18,7 +18,6 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Tuple
 
-from ..api import constants as C
 from .topology import NodeTopology, fully_connected
 
 

@@ -11,7 +11,7 @@ warm HBM pages and avoids re-pathing xGMI peers.  Evicted O(1) on RBG delete.
 from __future__ import annotations
 
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 MODE_PREFERRED = "preferred"
 MODE_REQUIRED = "required"

@@ -14,8 +14,8 @@ from __future__ import annotations
 import json
 import socket
 import threading
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import Dict, List
 
 from ..api import constants as C
 

@@ -10,10 +10,9 @@ per-pair bandwidth feeds KV-transfer chunk sizing.
 """
 from __future__ import annotations
 
-import os
 import subprocess
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from ..api import constants as C
 

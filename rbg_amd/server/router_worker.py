@@ -19,7 +19,7 @@ import os
 import threading
 import time
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 from ..runtime.worker import WorkerContext, register_runner
 from .rpc import RpcClient
