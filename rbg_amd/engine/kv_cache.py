@@ -155,6 +155,10 @@ class PrefixCache:
         self.lru: "OrderedDict[int, None]" = OrderedDict()
         self.hits = 0
         self.misses = 0
+        import threading
+        # refcounts are touched from the engine thread AND prefill RPC
+        # threads (_release_parked); compound read-modify-writes need a lock
+        self._lock = threading.Lock()
 
     def _chain(self, tokens: List[int]):
         ps = self.cache.page_size
@@ -166,53 +170,57 @@ class PrefixCache:
     def match(self, prompt: List[int]) -> List[int]:
         """Longest cached page chain for prompt[:-1] (at least one token is
         always left to prefill so logits exist); acquires the pages."""
-        got: List[int] = []
-        for h in self._chain(prompt[:-1]):
-            pg = self.by_hash.get(h)
-            if pg is None:
-                break
-            got.append(pg)
-        for pg in got:
-            h, refs = self.page_info[pg]
-            self.page_info[pg] = (h, refs + 1)
-            self.lru.pop(pg, None)
-        self.hits += len(got)
-        self.misses += max(0, len(prompt[:-1]) // self.cache.page_size
-                           - len(got))
-        return got
+        with self._lock:
+            got: List[int] = []
+            for h in self._chain(prompt[:-1]):
+                pg = self.by_hash.get(h)
+                if pg is None:
+                    break
+                got.append(pg)
+            for pg in got:
+                h, refs = self.page_info[pg]
+                self.page_info[pg] = (h, refs + 1)
+                self.lru.pop(pg, None)
+            self.hits += len(got)
+            self.misses += max(0, len(prompt[:-1]) // self.cache.page_size
+                               - len(got))
+            return got
 
     def register(self, prompt: List[int], pages: List[int],
                  already_shared: int) -> int:
         """After prefill: publish the prompt's full pages.  Returns the new
         shared-page count (callers update BlockTable.num_shared)."""
-        shared = already_shared
-        for idx, h in enumerate(self._chain(prompt)):
-            if idx < already_shared:
-                continue
-            if h in self.by_hash:
-                break     # someone registered concurrently; keep ours private
-            pg = pages[idx]
-            self.by_hash[h] = pg
-            self.page_info[pg] = (h, 1)
-            shared = idx + 1
-        return shared
+        with self._lock:
+            shared = already_shared
+            for idx, h in enumerate(self._chain(prompt)):
+                if idx < already_shared:
+                    continue
+                if h in self.by_hash:
+                    break   # someone registered concurrently; keep ours
+                pg = pages[idx]
+                self.by_hash[h] = pg
+                self.page_info[pg] = (h, 1)
+                shared = idx + 1
+            return shared
 
     def release_page(self, pg: int) -> None:
-        h, refs = self.page_info[pg]
-        if refs <= 1:
-            self.page_info[pg] = (h, 0)
-            self.lru[pg] = None
-        else:
-            self.page_info[pg] = (h, refs - 1)
+        with self._lock:
+            h, refs = self.page_info[pg]
+            if refs <= 1:
+                self.page_info[pg] = (h, 0)
+                self.lru[pg] = None
+            else:
+                self.page_info[pg] = (h, refs - 1)
 
     def evict(self, n: int) -> List[int]:
-        out = []
-        while self.lru and len(out) < n:
-            pg, _ = self.lru.popitem(last=False)
-            h, _refs = self.page_info.pop(pg)
-            self.by_hash.pop(h, None)
-            out.append(pg)
-        return out
+        with self._lock:
+            out = []
+            while self.lru and len(out) < n:
+                pg, _ = self.lru.popitem(last=False)
+                h, _refs = self.page_info.pop(pg)
+                self.by_hash.pop(h, None)
+                out.append(pg)
+            return out
 
     def stats(self) -> Dict[str, int]:
         return {"hits": self.hits, "misses": self.misses,
