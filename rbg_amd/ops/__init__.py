@@ -80,7 +80,7 @@ def pick_decode_splits(num_seqs: int, num_kv_heads: int,
 
 
 DECODE_VARIANT = int(__import__("os").environ.get("RBG_DECODE_VARIANT", "1"))
-PREFILL_SWZ = int(__import__("os").environ.get("RBG_PREFILL_SWZ", "4"))  # 8-wave blocks (+37% A/B)
+PREFILL_SWZ = int(__import__("os").environ.get("RBG_PREFILL_SWZ", "6"))  # 8-wave + K/V double-buffer (305 TF/s vs 252 @ swz=4, within-probe)
 
 
 def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
