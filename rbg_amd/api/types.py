@@ -272,6 +272,7 @@ class RoleInstanceSetSpec:
     pod_management_policy: str = C.POD_MANAGEMENT_PARALLEL
     update_strategy: InstanceUpdateStrategy = field(default_factory=InstanceUpdateStrategy)
     revision_history_limit: int = 10
+    min_ready_seconds: int = 0   # reference rolebasedgroup_types.go MinReadySeconds
 
 
 @dataclass
@@ -540,7 +541,12 @@ KIND_TO_TYPE = {
 
 
 def load_object(data: Dict[str, Any]):
-    """Build a typed object from a parsed YAML/JSON dict (kind-dispatched)."""
+    """Build a typed object from a parsed YAML/JSON dict (kind-dispatched).
+    v1alpha1 RoleBasedGroup docs are converted to v1alpha2 on the way in
+    (the conversion-webhook analog, reference rolebasedgroup_conversion.go)."""
+    from . import v1alpha1 as _legacy
+    if _legacy.is_v1alpha1(data) and data.get("kind") == C.KIND_RBG:
+        data = _legacy.to_v2(data)
     kind = data.get("kind", "")
     cls = KIND_TO_TYPE.get(kind)
     if cls is None:

@@ -153,6 +153,7 @@ class RoleBasedGroupController:
             return 0.0
         revision = self.revisions.ensure_current(rbg, asdict(rbg.spec))
         statuses = self._construct_role_statuses(rbg)
+        self._migrate_legacy_coordination(rbg)
         policy = self.store.try_get(C.KIND_COORDINATED_POLICY,
                                     rbg.metadata.name, rbg.metadata.namespace)
         scale_caps = self._coordination_scale_caps(rbg, policy, statuses)
@@ -167,6 +168,52 @@ class RoleBasedGroupController:
         return requeue
 
     # ------------------------------------------------------------------
+
+    def _migrate_legacy_coordination(self, rbg: RoleBasedGroup) -> None:
+        """v1alpha1 `spec.coordination[]` → CoordinatedPolicy CR (reference
+        coordinatedpolicy_migration_controller.go:59-156).  The conversion
+        layer preserved the legacy rules in an annotation; synthesize the
+        v1alpha2 object once, owned by the RBG, if none exists yet."""
+        import json as _json
+        from ..api.v1alpha1 import ANNO_COORDINATION
+        raw = rbg.metadata.annotations.get(ANNO_COORDINATION)
+        if not raw or self.store.try_get(C.KIND_COORDINATED_POLICY,
+                                         rbg.metadata.name,
+                                         rbg.metadata.namespace) is not None:
+            return
+        from ..api.types import (CoordinatedPolicy, CoordinatedPolicySpec,
+                                 CoordinatedRollingUpdate, CoordinatedScaling,
+                                 CoordinationRule, CoordinationStrategy)
+
+        def _pct(v, default):
+            if v is None:
+                return default
+            return int(str(v).rstrip("%"))
+
+        rules = []
+        for entry in _json.loads(raw):
+            strat = entry.get("strategy") or {}
+            cs = CoordinationStrategy()
+            ru = strat.get("rollingUpdate")
+            if ru:
+                cs.rolling_update = CoordinatedRollingUpdate(
+                    max_skew=_pct(ru.get("maxSkew"), 10),
+                    partition=_pct(ru.get("partition"), 0),
+                    max_unavailable=_pct(ru.get("maxUnavailable"), 1))
+            sc = strat.get("scaling")
+            if sc:
+                cs.scaling = CoordinatedScaling(
+                    max_skew=_pct(sc.get("maxSkew"), 10),
+                    progression=sc.get("progression",
+                                       C.PROGRESSION_ORDER_SCHEDULED))
+            rules.append(CoordinationRule(roles=entry.get("roles", []),
+                                          strategy=cs))
+        policy = CoordinatedPolicy(
+            metadata=ObjectMeta(name=rbg.metadata.name,
+                                namespace=rbg.metadata.namespace),
+            spec=CoordinatedPolicySpec(rules=rules))
+        set_owner(policy, rbg)
+        self.store.create(policy)
 
     def _apply_scaling_adapter_override(self, rbg: RoleBasedGroup) -> RoleBasedGroup:
         """Adapter-driven replicas win over spec (reference
@@ -310,6 +357,7 @@ class RoleBasedGroupController:
             ris.spec.selector = {C.LABEL_GROUP_NAME: rbg.metadata.name,
                                  C.LABEL_ROLE_NAME: role.name}
             ris.spec.pod_management_policy = role.pod_management_policy
+            ris.spec.min_ready_seconds = role.min_ready_seconds
             ris.spec.update_strategy = InstanceUpdateStrategy(
                 type=role.update_strategy_type,
                 partition=part,

@@ -37,6 +37,19 @@ def _is_ready(inst: RoleInstance) -> bool:
     return c is not None and c.status == "True"
 
 
+def _is_available(inst: RoleInstance, min_ready_seconds: int) -> bool:
+    """Ready AND has stayed Ready for minReadySeconds (reference
+    minReadySeconds availability semantics: a newly-Ready instance only
+    counts after the stability window)."""
+    c = get_condition(inst.status.conditions, C.COND_READY)
+    if c is None or c.status != "True":
+        return False
+    if min_ready_seconds <= 0:
+        return True
+    import time
+    return time.time() - c.last_transition_time >= min_ready_seconds
+
+
 class RoleInstanceSetController:
     def __init__(self, store: Store):
         self.store = store
@@ -62,7 +75,8 @@ class RoleInstanceSetController:
             requeue = self._scale(ris, instances, update_hash)
             requeue = max(requeue,
                           self._rolling_update(ris, instances, update_hash))
-        self._update_status(ris, instances, update_hash)
+        requeue = max(requeue,
+                      self._update_status(ris, instances, update_hash))
         return requeue
 
     # ------------------------------------------------------------------
@@ -301,10 +315,14 @@ class RoleInstanceSetController:
 
     def _update_status(self, ris: RoleInstanceSet,
                        instances: Dict[str, RoleInstance],
-                       update_hash: str) -> None:
+                       update_hash: str) -> float:
         live = [i for i in self._owned(ris)
                 if i.metadata.deletion_timestamp is None]
-        ready = sum(1 for i in live if _is_ready(i))
+        ready = sum(1 for i in live
+                    if _is_available(i, ris.spec.min_ready_seconds))
+        # ready-but-inside-the-stability-window instances converge on their
+        # own clock; requeue to observe availability without an event
+        pending_window = sum(1 for i in live if _is_ready(i)) - ready
         updated = sum(1 for i in live
                       if i.metadata.labels.get(C.LABEL_REVISION_HASH) == update_hash)
 
@@ -327,3 +345,5 @@ class RoleInstanceSetController:
                              mutate, ris.metadata.namespace, subresource="status")
         except KeyError:
             pass
+        return min(1.0, ris.spec.min_ready_seconds / 2.0) \
+            if pending_window > 0 else 0.0

@@ -1,0 +1,137 @@
+"""v1alpha1 legacy API: conversion, round-trip, apply-path, migration.
+
+Covers reference api/workloads/v1alpha1/rolebasedgroup_conversion.go and
+coordinatedpolicy_migration_controller.go behaviors.
+"""
+import copy
+
+from rbg_amd.api import constants as C
+from rbg_amd.api import v1alpha1 as legacy
+from rbg_amd.api.types import load_object
+
+LEGACY_DOC = {
+    "apiVersion": "workloads.x-k8s.io/v1alpha1",
+    "kind": "RoleBasedGroup",
+    "metadata": {"name": "pd", "namespace": "default"},
+    "spec": {
+        "podGroupPolicy": {"kubeScheduling": {"scheduleTimeoutSeconds": 90}},
+        "coordination": [
+            {"name": "pd-scale", "roles": ["prefill", "decode"],
+             "strategy": {"scaling": {"maxSkew": "5%",
+                                      "progression": "OrderReady"},
+                          "rollingUpdate": {"maxSkew": "20%",
+                                            "maxUnavailable": "10%"}}},
+        ],
+        "roles": [
+            {"name": "router", "replicas": 1,
+             "workload": {"apiVersion": "apps/v1", "kind": "Deployment"},
+             "template": {"engines": [
+                 {"name": "router", "runner": "echo", "args": {}}]}},
+            {"name": "prefill", "replicas": 2,
+             "workload": {"apiVersion": "leaderworkerset.x-k8s.io/v1",
+                          "kind": "LeaderWorkerSet"},
+             "restartPolicy": "RecreateRoleInstanceOnPodRestart",
+             "dependencies": ["router"],
+             "leaderWorkerSet": {
+                 "size": 2,
+                 "patchLeaderTemplate": {"engines": [
+                     {"name": "eng", "args": {"rank": "leader"}}]}},
+             "template": {"engines": [
+                 {"name": "eng", "runner": "echo",
+                  "args": {"mode": "prefill"}}]},
+             "engineRuntimes": [{"profileName": "metrics-sidecar"}],
+             "minReadySeconds": 3},
+            {"name": "decode", "replicas": 2,
+             "components": [
+                 {"name": "engine", "size": 1,
+                  "template": {"engines": [
+                      {"name": "eng", "runner": "echo", "args": {}}]}}]},
+        ],
+    },
+}
+
+
+def test_to_v2_patterns_and_fields():
+    doc = legacy.to_v2(copy.deepcopy(LEGACY_DOC))
+    assert doc["apiVersion"] == C.API_VERSION
+    roles = {r["name"]: r for r in doc["spec"]["roles"]}
+    assert roles["router"]["pattern"] == C.PATTERN_STANDALONE
+    assert roles["prefill"]["pattern"] == C.PATTERN_LEADER_WORKER
+    lwp = roles["prefill"]["leaderWorkerPattern"]
+    assert lwp["size"] == 2
+    # patchLeaderTemplate merged over the role template
+    assert lwp["leaderTemplate"]["engines"][0]["args"] == {
+        "mode": "prefill", "rank": "leader"}
+    assert roles["prefill"]["engineRuntimes"] == ["metrics-sidecar"]
+    assert roles["prefill"]["minReadySeconds"] == 3
+    assert roles["decode"]["pattern"] == C.PATTERN_CUSTOM_COMPONENTS
+    annos = doc["metadata"]["annotations"]
+    assert annos[C.ANNO_GANG_SCHEDULING] == "true"
+    assert annos[C.ANNO_GANG_TIMEOUT] == "90"
+    assert legacy.ANNO_COORDINATION in annos
+
+
+def test_load_object_accepts_v1alpha1():
+    rbg = load_object(copy.deepcopy(LEGACY_DOC))
+    assert rbg.kind == C.KIND_RBG
+    assert rbg.spec.role("prefill").leader_worker_pattern.size == 2
+    assert rbg.spec.role("prefill").dependencies == ["router"]
+
+
+def test_round_trip_from_v2():
+    v2 = legacy.to_v2(copy.deepcopy(LEGACY_DOC))
+    back = legacy.from_v2(v2)
+    assert back["apiVersion"] == legacy.API_VERSION_V1ALPHA1
+    spec = back["spec"]
+    assert spec["podGroupPolicy"] == LEGACY_DOC["spec"]["podGroupPolicy"]
+    assert spec["coordination"] == LEGACY_DOC["spec"]["coordination"]
+    roles = {r["name"]: r for r in spec["roles"]}
+    assert roles["router"]["workload"]["kind"] == "Deployment"
+    assert roles["prefill"]["workload"]["kind"] == "LeaderWorkerSet"
+    assert roles["prefill"]["leaderWorkerSet"]["size"] == 2
+    assert roles["decode"]["components"][0]["name"] == "engine"
+    # conversion-only annotations removed from the legacy view
+    annos = back["metadata"].get("annotations", {})
+    assert legacy.ANNO_COORDINATION not in annos
+    assert legacy.ANNO_WORKLOAD_TYPE not in annos
+
+
+def test_migration_synthesizes_coordinated_policy():
+    from rbg_amd.controller.rbg_controller import RoleBasedGroupController
+    from rbg_amd.store.store import Store
+    store = Store()
+    rbg = load_object(copy.deepcopy(LEGACY_DOC))
+    store.create(rbg)
+    ctrl = RoleBasedGroupController(store)
+    ctrl.reconcile("pd")
+    pol = store.try_get(C.KIND_COORDINATED_POLICY, "pd")
+    assert pol is not None
+    rule = pol.spec.rules[0]
+    assert rule.roles == ["prefill", "decode"]
+    assert rule.strategy.scaling.max_skew == 5
+    assert rule.strategy.scaling.progression == "OrderReady"
+    assert rule.strategy.rolling_update.max_skew == 20
+    assert rule.strategy.rolling_update.max_unavailable == 10
+    # owned by the RBG → torn down with it
+    assert pol.metadata.owner_references[0].uid == rbg.metadata.uid
+    # idempotent: second reconcile does not duplicate / conflict
+    ctrl.reconcile("pd")
+    assert store.try_get(C.KIND_COORDINATED_POLICY, "pd") is not None
+
+
+def test_min_ready_seconds_gates_availability(monkeypatch):
+    """minReadySeconds: a just-Ready instance is not counted available until
+    the stability window elapses (reference availability semantics)."""
+    import time as _time
+    from rbg_amd.api.types import (Condition, ObjectMeta, RoleInstance,
+                                   RoleInstanceSet, set_condition)
+    from rbg_amd.controller.roleinstanceset import (_is_available, _is_ready)
+    inst = RoleInstance(metadata=ObjectMeta(name="i0"))
+    set_condition(inst.status.conditions,
+                  Condition.new(C.COND_READY, True, "Up", ""))
+    assert _is_ready(inst)
+    assert _is_available(inst, 0)
+    assert not _is_available(inst, 5)
+    # age the transition past the window
+    inst.status.conditions[0].last_transition_time = _time.time() - 6
+    assert _is_available(inst, 5)
