@@ -356,10 +356,31 @@ class RoleInstanceController:
                 self.bindings.record(rbg_uid, GpuBindingStore.key(
                     inst.metadata.name, suffix), gpus)
 
-    def _stop_all(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
-        for h in rt.handles.values():
-            self.runner.stop(h)
+    def _stop_ordered(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
+        """Stop workers in reverse dependency order (reference
+        component_lifecycle.go reverse deletion gates): dependents first,
+        their dependencies only once nothing running needs them."""
+        from ..discovery import component as comp_disc
+        deps = comp_disc.parse_depends_on(inst.metadata.annotations)
+        for comp in inst.spec.components:
+            deps.update(comp_disc.parse_depends_on(comp.annotations))
+        comp_names = [c.name for c in inst.spec.components]
+        prefix = len(inst.metadata.name) + 1
+        by_comp = {}
+        for wname, h in rt.handles.items():
+            cname = wname[prefix:].rsplit("-", 1)[0]
+            by_comp.setdefault(cname, []).append(h)
+        for wave in comp_disc.delete_waves(comp_names, deps):
+            for cname in wave:
+                for h in by_comp.pop(cname, []):
+                    self.runner.stop(h)
+        for hs in by_comp.values():      # workers of unlisted components
+            for h in hs:
+                self.runner.stop(h)
         rt.handles.clear()
+
+    def _stop_all(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
+        self._stop_ordered(inst, rt)
         self.gang.release(rt.gang_id)
 
     def teardown(self, inst: RoleInstance) -> None:
@@ -367,8 +388,7 @@ class RoleInstanceController:
         if rt is None:
             return
         self._record_bindings(inst, rt)
-        for h in rt.handles.values():
-            self.runner.stop(h)
+        self._stop_ordered(inst, rt)
         self.gang.release(rt.gang_id)
         self.ports.release(f"lwp-{inst.metadata.uid}")
         self.restarts.evict(inst.metadata.uid)

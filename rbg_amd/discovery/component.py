@@ -74,6 +74,32 @@ def start_gate(component: str, deps: Dict[str, List[str]],
     return all(component_ready.get(d, False) for d in deps.get(component, []))
 
 
+def delete_waves(components: List[str],
+                 deps: Dict[str, List[str]]) -> List[List[str]]:
+    """Reverse deletion order (reference component_lifecycle.go:66-262):
+    a component stops only after every component that depends on it has
+    stopped — i.e. reverse-topological waves of the depends-on graph.
+    Cycles (or no deps) degrade to one parallel wave, like start_gate."""
+    if not deps or has_cycle(deps):
+        return [list(components)]
+    remaining = set(components)
+    dependents: Dict[str, List[str]] = {}
+    for comp, ds in deps.items():
+        for d in ds:
+            dependents.setdefault(d, []).append(comp)
+    waves: List[List[str]] = []
+    while remaining:
+        # stop everything nothing-still-running depends on
+        wave = sorted(c for c in remaining
+                      if not any(d in remaining
+                                 for d in dependents.get(c, [])))
+        if not wave:            # defensive: shouldn't happen (acyclic)
+            wave = sorted(remaining)
+        waves.append(wave)
+        remaining -= set(wave)
+    return waves
+
+
 def sibling_env(component: str, specs: List[Dict[str, str]],
                 sibling_ports: Dict[str, List[int]]) -> Dict[str, str]:
     """Env for `component` from its discovery spec: each entry names a

@@ -137,3 +137,69 @@ def test_engine_runtime_profile_injection(mgr):
     eng = ris.spec.template.components[0].template.engines[0]
     assert {"RUNTIME_FLAG": "on"} == {e.name: e.value for e in eng.env}
     assert eng.args["ready_delay"] == 0.1
+
+
+def test_delete_waves_reverse_order():
+    """Reverse deletion gates: dependents stop before their dependencies
+    (reference component_lifecycle.go:66-262)."""
+    from rbg_amd.discovery.component import delete_waves
+    comps = ["cache", "engine", "frontend"]
+    deps = {"engine": ["cache"], "frontend": ["engine"]}
+    waves = delete_waves(comps, deps)
+    assert waves == [["frontend"], ["engine"], ["cache"]]
+    # cycle → single parallel wave
+    assert delete_waves(comps, {"a": ["b"], "b": ["a"]}) == [comps]
+    # no deps → single parallel wave
+    assert delete_waves(comps, {}) == [comps]
+
+
+def test_teardown_stops_in_reverse_dependency_order():
+    """An instance with component deps tears down dependents-first."""
+    import json
+    from rbg_amd.api import constants as C
+    from rbg_amd.api.types import (ComponentSpec, EngineSpec, EngineTemplate,
+                                   ObjectMeta, RoleInstance, RoleInstanceSpec)
+    from rbg_amd.controller.roleinstance import RoleInstanceController
+    from rbg_amd.store.store import Store
+
+    stopped = []
+
+    class FakeHandle:
+        def __init__(self, name):
+            self.name = name
+            self.gpu_ids = []
+
+        def phase(self):
+            return "Ready"
+
+    class FakeRunner:
+        def stop(self, h, grace=5.0):
+            stopped.append(h.name)
+
+    from rbg_amd.runtime.process import ProcessRunner
+    from rbg_amd.scheduler.gang import GangAllocator
+    from rbg_amd.scheduler.placement import GpuBindingStore
+    from rbg_amd.scheduler.ports import PortAllocator
+    from rbg_amd.scheduler.topology import fully_connected
+    store = Store()
+    tmpl = EngineTemplate(engines=[EngineSpec(name="e", runner="echo")])
+    inst = RoleInstance(
+        metadata=ObjectMeta(
+            name="g-r-0",
+            annotations={C.ANNO_COMPONENT_DEPENDS_ON: json.dumps(
+                {"engine": ["cache"], "frontend": ["engine"]})}),
+        spec=RoleInstanceSpec(components=[
+            ComponentSpec(name="cache", size=1, template=tmpl),
+            ComponentSpec(name="engine", size=1, template=tmpl),
+            ComponentSpec(name="frontend", size=1, template=tmpl)]))
+    store.create(inst)
+    ctrl = RoleInstanceController(store, GangAllocator(fully_connected()),
+                                  FakeRunner(), PortAllocator(),
+                                  GpuBindingStore())
+    from rbg_amd.controller.roleinstance import InstanceRuntime
+    rt = ctrl._runtimes.setdefault(inst.metadata.uid, InstanceRuntime())
+    for cname in ("cache", "engine", "frontend"):
+        rt.handles[f"g-r-0-{cname}-0"] = FakeHandle(f"g-r-0-{cname}-0")
+    ctrl.teardown(inst)
+    assert stopped == ["g-r-0-frontend-0", "g-r-0-engine-0",
+                       "g-r-0-cache-0"]
