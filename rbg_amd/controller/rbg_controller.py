@@ -97,6 +97,7 @@ class RoleBasedGroupController:
         import json as _json
         rank = 0
         groups: List[List[int]] = []
+        subgroups: List[List[int]] = []   # TP-stage groups for TP x PP roles
         members: Dict[str, List[int]] = {}
         rank_map: Dict[str, int] = {}
         for role in rbg.spec.roles:
@@ -121,18 +122,32 @@ class RoleBasedGroupController:
                         rank += 1
                 groups.append(inst_ranks)
                 members[inst] = inst_ranks
+                # TP x PP: the instance's n ranks split into pp stages of
+                # n/pp tensor-parallel ranks each (stage = idx // tp);
+                # emit each stage's TP group so workers can build the
+                # all-reduce communicator separate from the lockstep group
+                pp = max(int(engines[0][1].args.get("pp", 1) or 1), 1)
+                if pp > 1 and len(inst_ranks) % pp == 0 and \
+                        len(inst_ranks) > pp:
+                    tp_deg = len(inst_ranks) // pp
+                    for s in range(pp):
+                        subgroups.append(
+                            inst_ranks[s * tp_deg:(s + 1) * tp_deg])
         if rank <= 1 and len(groups) <= 1:
             return {}
         port = 29500
         if self.ports is not None:
             port = self.ports.allocate(f"comm-{rbg.metadata.uid}", 1)[0]
-        return {
+        annos = {
             "rbg.comm-world": str(rank),
             "rbg.comm-port": str(port),
             "rbg.comm-groups": _json.dumps(groups),
             "rbg.comm-members": _json.dumps(members),
             "rbg.comm-rank-map": _json.dumps(rank_map),
         }
+        if subgroups:
+            annos["rbg.comm-subgroups"] = _json.dumps(subgroups)
+        return annos
 
     # ------------------------------------------------------------------
 

@@ -322,6 +322,8 @@ class RoleInstanceController:
                 env[C.ENV_MASTER_PORT] = annos.get("rbg.comm-port", "29500")
                 env["RBG_COMM_GROUPS"] = annos.get("rbg.comm-groups", "[]")
                 env["RBG_COMM_MEMBERS"] = annos.get("rbg.comm-members", "{}")
+                env["RBG_COMM_SUBGROUPS"] = annos.get(
+                    "rbg.comm-subgroups", "[]")
         # sibling-component discovery env (discovery/component.py)
         from ..discovery import component as comp_disc
         disc = comp_disc.parse_discovery(comp.annotations) or \

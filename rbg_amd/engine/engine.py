@@ -55,9 +55,10 @@ class EngineStats:
 
 
 class LLMEngine:
-    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None):
+    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
+                 pp=None):
         self.cfg = cfg
-        self.runner = ModelRunner(cfg, tp)
+        self.runner = ModelRunner(cfg, tp, pp)
         self.scheduler = Scheduler(cfg, self.runner.cache)
         self.stats = EngineStats()
         self._finished: List[Sequence] = []

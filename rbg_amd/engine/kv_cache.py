@@ -26,16 +26,19 @@ class OutOfPages(RuntimeError):
 
 class PagedKVCache:
     def __init__(self, cfg: EngineConfig, device: torch.device,
-                 num_pages: Optional[int] = None):
+                 num_pages: Optional[int] = None,
+                 num_layers: Optional[int] = None):
         self.cfg = cfg
         m = cfg.model
         self.page_size = cfg.page_size
+        layers = num_layers or m.num_layers      # PP: only local layers
+        self.num_layers = layers
         if num_pages is None:
-            num_pages = self._size_pool(cfg, device)
+            num_pages = self._size_pool(cfg, device, layers)
         self.num_pages = num_pages
         kvh = m.num_kv_heads // max(1, cfg.tp_size)   # TP shards the heads
         self.kv = torch.zeros(
-            (m.num_layers, 2, num_pages, kvh, cfg.page_size, m.head_dim),
+            (layers, 2, num_pages, kvh, cfg.page_size, m.head_dim),
             dtype=torch.bfloat16, device=device)
         # page 0 is reserved scratch: hipGraph decode padding rows write
         # their (dead) KV slot there (model_runner._decode_graph)
@@ -46,11 +49,12 @@ class PagedKVCache:
             PrefixCache(self) if cfg.enable_prefix_cache else None)
 
     @staticmethod
-    def _size_pool(cfg: EngineConfig, device: torch.device) -> int:
+    def _size_pool(cfg: EngineConfig, device: torch.device,
+                   num_layers: Optional[int] = None) -> int:
         m = cfg.model
         kvh = m.num_kv_heads // max(1, cfg.tp_size)
-        page_bytes = (m.num_layers * 2 * kvh * cfg.page_size *
-                      m.head_dim * 2)
+        page_bytes = ((num_layers or m.num_layers) * 2 * kvh *
+                      cfg.page_size * m.head_dim * 2)
         if device.type == "cuda" and torch.cuda.is_available():
             free, _total = torch.cuda.mem_get_info(device)
             budget = int(free * cfg.gpu_memory_utilization)

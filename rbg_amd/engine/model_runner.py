@@ -13,7 +13,8 @@ from typing import Dict, List, Optional, Tuple
 
 import torch
 
-from ..models.llama import ForwardBatch, LlamaForCausalLM, TPContext
+from ..models.llama import (ForwardBatch, LlamaForCausalLM, PPContext,
+                            TPContext)
 from .config import EngineConfig
 from .kv_cache import PagedKVCache
 from .sequence import Sequence
@@ -24,12 +25,15 @@ GRAPH_BATCH_SIZES = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256)
 
 
 class ModelRunner:
-    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None):
+    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
+                 pp: Optional[PPContext] = None):
         self.cfg = cfg
         self.device = torch.device(cfg.device)
         torch.manual_seed(cfg.seed)
-        self.model = LlamaForCausalLM(cfg.model, self.device, tp)
-        self.cache = PagedKVCache(cfg, self.device)
+        self.pp = pp or PPContext()
+        self.model = LlamaForCausalLM(cfg.model, self.device, tp, pp=self.pp)
+        self.cache = PagedKVCache(cfg, self.device,
+                                  num_layers=self.model.num_local_layers)
         self.max_pages_per_seq = (cfg.max_seq_len + cfg.page_size - 1) // \
             cfg.page_size
         self._graphs: Dict[int, Tuple] = {}
@@ -80,13 +84,27 @@ class ModelRunner:
             cu_seqlens=self._i32(cu),
             cu_seqlens_k=self._i32(cu_k) if any_past else None,
             kv_gather_slots=self._i32(gather) if any_past else None)
+        hidden_in = None
+        if not self.pp.first:
+            hidden_in = self.pp.recv_prev(
+                (len(tokens), self.cfg.model.hidden_size),
+                self.model.dtype, self.device)
         hidden = self.model.forward(
             torch.tensor(tokens, dtype=torch.int64, device=self.device),
-            batch, self.cache)
+            batch, self.cache, hidden=hidden_in)
+        if not self.pp.last:
+            self.pp.send_next(hidden)
         if not completing:
             return []
-        logits = self.model.logits(hidden, self._i32(last_idx))
-        next_tokens = self.sample(logits, completing)
+        if self.pp.last:
+            logits = self.model.logits(hidden, self._i32(last_idx))
+            next_dev = self.sample_device(logits, completing)[0]
+        else:
+            next_dev = torch.empty(len(completing), dtype=torch.int64,
+                                   device=self.device)
+        if self.pp.size > 1:
+            next_dev = self.pp.broadcast_tokens(next_dev)
+        next_tokens = next_dev.tolist()
         for seq, tok in zip(completing, next_tokens):
             seq.append_token(tok)
         return next_tokens
@@ -139,6 +157,7 @@ class ModelRunner:
         slots = self._decode_slots(state)
         use_graph = (not self.cfg.enforce_eager and
                      self.device.type == "cuda" and
+                     self.pp.size == 1 and           # p2p hops: eager path
                      n <= GRAPH_BATCH_SIZES[-1])
         if use_graph:
             logits = self._decode_graph(state, slots, n)
@@ -150,9 +169,24 @@ class ModelRunner:
                 block_tables=state["bt"],
                 context_lens=state["ctx"],
                 decode_num_splits=self._splits_for(n))
-            hidden = self.model.forward(state["tokens"], batch, self.cache)
-            logits = self.model.logits(hidden)
-        next_dev, next_tokens = self.sample_device(logits[:n], seqs)
+            hidden_in = None
+            if not self.pp.first:
+                hidden_in = self.pp.recv_prev(
+                    (n, self.cfg.model.hidden_size),
+                    self.model.dtype, self.device)
+            hidden = self.model.forward(state["tokens"], batch, self.cache,
+                                        hidden=hidden_in)
+            if self.pp.last:
+                logits = self.model.logits(hidden)
+            else:
+                self.pp.send_next(hidden)
+        if self.pp.last:
+            next_dev, next_tokens = self.sample_device(logits[:n], seqs)
+        else:
+            next_dev = torch.empty(n, dtype=torch.int64, device=self.device)
+        if self.pp.size > 1:
+            next_dev = self.pp.broadcast_tokens(next_dev)
+            next_tokens = next_dev.tolist()
         state["tokens"] = next_dev
         for seq, tok in zip(seqs, next_tokens):
             seq.append_token(tok)

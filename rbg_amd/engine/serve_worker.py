@@ -113,15 +113,39 @@ class ServeWorker:
                        ctx.args.get("tp_backend") or
                        ctx.args.get("transfer_backend"))
             self.gcomm = commmod.init_global_from_env(backend=backend)
+        self.model_tp = None
+        self.model_pp = None
         if self.gcomm is not None and self.gcomm.tp_size > 1:
-            from ..models.llama import TPContext
-            self.cfg.tp_size = self.gcomm.tp_size
-            self.cfg.tp_rank = self.gcomm.tp_rank
+            from ..models.llama import PPContext, TPContext
+            # self.tp is the INSTANCE-wide context (lockstep identity);
+            # with pp>1 the model computes over the per-stage TP subgroup
+            # and a PPContext carries the stage-to-stage hops
             self.tp = TPContext(size=self.gcomm.tp_size,
                                 rank=self.gcomm.tp_rank,
                                 group=self.gcomm.tp_group)
+            pp_deg = max(int(ctx.args.get("pp", 1) or 1), 1)
+            if pp_deg > 1 and self.gcomm.stage_group is not None:
+                self.cfg.tp_size = self.gcomm.stage_size
+                self.cfg.tp_rank = self.gcomm.stage_rank
+                self.model_tp = TPContext(size=self.gcomm.stage_size,
+                                          rank=self.gcomm.stage_rank,
+                                          group=self.gcomm.stage_group)
+                import json as _json
+                inst_ranks = _json.loads(
+                    os.environ.get("RBG_COMM_MEMBERS", "{}")).get(
+                    os.environ.get(C.ENV_ROLE_INSTANCE_NAME, ""), [])
+                my_idx = inst_ranks.index(self.gcomm.rank)
+                self.model_pp = PPContext(
+                    size=pp_deg, stage=my_idx // self.gcomm.stage_size,
+                    instance_ranks=inst_ranks,
+                    tp_size=self.gcomm.stage_size,
+                    group=self.gcomm.tp_group)
+            else:
+                self.cfg.tp_size = self.gcomm.tp_size
+                self.cfg.tp_rank = self.gcomm.tp_rank
+                self.model_tp = self.tp
         self.my_instance = os.environ.get(C.ENV_ROLE_INSTANCE_NAME, "")
-        self.engine = LLMEngine(self.cfg, self.tp)
+        self.engine = LLMEngine(self.cfg, self.model_tp, self.model_pp)
         self._xfer_lock = threading.Lock()
         self._tickets: Dict[str, int] = {}
         self.is_leader = self.tp is None or self.tp.rank == 0

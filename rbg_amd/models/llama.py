@@ -73,6 +73,59 @@ class TPContext:
         return t
 
 
+class PPContext:
+    """Pipeline-parallel context over the instance's global ranks.
+
+    The reference expresses PP through the same leaderWorker rank env as TP
+    ("PP vs TP is the engine's choice"); here the engine makes
+    that choice concrete: the instance's n ranks split into pp stages of
+    tp ranks each (stage = idx // tp).  Hidden states hop stage-to-stage as
+    point-to-point sends between same-tp-lane ranks over xGMI (each lane's
+    post-all-reduce activations are replicated, so lane-to-lane is exact);
+    sampled tokens broadcast from the last stage over the instance group.
+    """
+
+    def __init__(self, size: int = 1, stage: int = 0,
+                 instance_ranks=None, tp_size: int = 1, group=None):
+        self.size = size
+        self.stage = stage
+        self.ranks = list(instance_ranks or [])
+        self.tp_size = tp_size
+        self.group = group              # instance-wide group (broadcasts)
+
+    @property
+    def first(self) -> bool:
+        return self.stage == 0
+
+    @property
+    def last(self) -> bool:
+        return self.stage == self.size - 1
+
+    def _lane_rank(self, stage: int) -> int:
+        my_idx = self.ranks.index(torch.distributed.get_rank())
+        lane = my_idx % self.tp_size
+        return self.ranks[stage * self.tp_size + lane]
+
+    def send_next(self, t: torch.Tensor) -> None:
+        t = t.contiguous()
+        if t.device.type == "cpu" and t.dtype == torch.bfloat16:
+            t = t.view(torch.int16)   # gloo lacks bf16; bit-exact reinterp
+        torch.distributed.send(t, dst=self._lane_rank(self.stage + 1))
+
+    def recv_prev(self, shape, dtype, device) -> torch.Tensor:
+        wire = torch.int16 if (device.type == "cpu" and
+                               dtype == torch.bfloat16) else dtype
+        t = torch.empty(shape, dtype=wire, device=device)
+        torch.distributed.recv(t, src=self._lane_rank(self.stage - 1))
+        return t.view(dtype)
+
+    def broadcast_tokens(self, t: torch.Tensor) -> torch.Tensor:
+        """Deliver last-stage sampled ids to every rank of the instance."""
+        src = self.ranks[(self.size - 1) * self.tp_size]
+        torch.distributed.broadcast(t, src=src, group=self.group)
+        return t
+
+
 class LlamaAttention(nn.Module):
     def __init__(self, cfg: ModelConfig, layer_idx: int, tp: TPContext,
                  device, dtype, gen):
@@ -172,6 +225,8 @@ class LlamaLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, layer_idx: int, tp: TPContext,
                  device, dtype, gen):
         super().__init__()
+        # layer_idx here is the KV-cache index LOCAL to this PP stage; the
+        # weight generator is already seeded by the GLOBAL layer index
         self.attn = LlamaAttention(cfg, layer_idx, tp, device, dtype, gen)
         self.mlp = LlamaMLP(cfg, tp, device, dtype, gen)
         self.input_norm = nn.Parameter(
@@ -200,39 +255,78 @@ class LlamaLayer(nn.Module):
 
 class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: ModelConfig, device: torch.device,
-                 tp: Optional[TPContext] = None, base_seed: int = 1234):
+                 tp: Optional[TPContext] = None, base_seed: int = 1234,
+                 pp: Optional[PPContext] = None):
         super().__init__()
         self.cfg = cfg
         self.tp = tp or TPContext()
+        self.pp = pp or PPContext()
         self.base_seed = base_seed
         self.device = device
         dtype = torch.bfloat16
-        gen = torch.Generator(device=device)
-        gen.manual_seed(base_seed)
-        self.embed = nn.Parameter(
-            torch.empty(cfg.vocab_size, cfg.hidden_size, device=device,
-                        dtype=dtype).normal_(0, 0.02, generator=gen),
-            requires_grad=False)
+        self.dtype = dtype
+
+        # Per-tensor generators keyed by (base_seed, tag): weights are
+        # identical for every TP degree (full-then-shard) AND every PP
+        # degree (a stage seeds its layers by GLOBAL index, independent of
+        # which ranks materialize them).
+        def _gen(tag: int) -> torch.Generator:
+            g = torch.Generator(device=device)
+            g.manual_seed(base_seed * 1000003 + tag)
+            return g
+
+        L = cfg.num_layers
+        assert L % self.pp.size == 0, "PP degree must divide layer count"
+        per = L // self.pp.size
+        self.layer_start = self.pp.stage * per
+        self.layer_end = self.layer_start + per
+        self.embed = None
+        if self.pp.first:
+            self.embed = nn.Parameter(
+                torch.empty(cfg.vocab_size, cfg.hidden_size, device=device,
+                            dtype=dtype).normal_(0, 0.02, generator=_gen(0)),
+                requires_grad=False)
         self.layers = nn.ModuleList([
-            LlamaLayer(cfg, i, self.tp, device, dtype, gen)
-            for i in range(cfg.num_layers)])
-        self.final_norm = nn.Parameter(
-            torch.ones(cfg.hidden_size, device=device, dtype=dtype),
-            requires_grad=False)
-        self.lm_head = _linear_weight(cfg.vocab_size, cfg.hidden_size,
-                                      device, dtype, gen)
+            LlamaLayer(cfg, i - self.layer_start, self.tp, device, dtype,
+                       _gen(1 + i))
+            for i in range(self.layer_start, self.layer_end)])
+        self.final_norm = None
+        self.lm_head = None
+        if self.pp.last:
+            self.final_norm = nn.Parameter(
+                torch.ones(cfg.hidden_size, device=device, dtype=dtype),
+                requires_grad=False)
+            self.lm_head = _linear_weight(cfg.vocab_size, cfg.hidden_size,
+                                          device, dtype, _gen(L + 1))
         self.cos_sin = ops.build_cos_sin_table(
             cfg.head_dim, cfg.max_position, cfg.rope_theta,
             device=device)
 
+    @property
+    def num_local_layers(self) -> int:
+        return self.layer_end - self.layer_start
+
     @torch.inference_mode()
     def forward(self, tokens: torch.Tensor, batch: ForwardBatch,
-                kv: PagedKVCache) -> torch.Tensor:
-        x = self.embed[tokens.long()]
+                kv: PagedKVCache,
+                hidden: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """First PP stage embeds `tokens`; later stages take `hidden` from
+        the previous stage (the combined h+residual stream — re-splitting it
+        as (residual=x, h=rmsnorm(x)) in the first local layer is exactly
+        the fused_add_rmsnorm the single-stage model would have run).
+        Non-last stages return the combined stream to send onward; the last
+        stage returns final-normed hidden states ready for logits()."""
+        if self.pp.first:
+            x = self.embed[tokens.long()]
+        else:
+            assert hidden is not None, "non-first PP stage needs hidden"
+            x = hidden
         residual = None
         for layer in self.layers:
             x, residual = layer(x, residual, batch, kv, self.cos_sin)
         x = x + residual
+        if not self.pp.last:
+            return x
         x = ops.rmsnorm(x, self.final_norm, self.cfg.rms_eps)
         return x
 
@@ -250,7 +344,7 @@ class LlamaForCausalLM(nn.Module):
         across ranks, then copies into the live parameters (the KV pool and
         any captured hipGraphs keep their addresses)."""
         fresh = LlamaForCausalLM(self.cfg, self.device, self.tp,
-                                 base_seed=seed)
+                                 base_seed=seed, pp=self.pp)
         with torch.no_grad():
             for p, q in zip(self.parameters(), fresh.parameters()):
                 p.copy_(q)

@@ -96,6 +96,9 @@ ENV_GLOBAL_RANK = "RBG_GLOBAL_RANK"
 ENV_GLOBAL_WORLD = "RBG_GLOBAL_WORLD"
 ENV_COMM_GROUPS = "RBG_COMM_GROUPS"      # JSON [[ranks of instance], ...]
 ENV_COMM_MEMBERS = "RBG_COMM_MEMBERS"    # JSON {instance_name: [ranks]}
+ENV_COMM_SUBGROUPS = "RBG_COMM_SUBGROUPS"  # JSON [[ranks of TP stage], ...]
+#   present when a role runs TP x PP inside one instance: each entry is the
+#   tensor-parallel group of ONE pipeline stage (controller _comm_plan)
 
 
 @dataclass
@@ -103,11 +106,14 @@ class GlobalComm:
     rank: int
     world_size: int
     backend: str
-    tp_group: Optional[object] = None
+    tp_group: Optional[object] = None    # instance-wide group (lockstep)
     tp_rank: int = 0
     tp_size: int = 1
     members: Optional[dict] = None       # instance -> [global ranks]
     device: Optional["torch.device"] = None
+    stage_group: Optional[object] = None  # my TP-stage subgroup (TP x PP)
+    stage_rank: int = 0
+    stage_size: int = 0
 
     def ranks_of(self, instance: str):
         return list((self.members or {}).get(instance, []))
@@ -143,4 +149,11 @@ def init_global_from_env(backend: Optional[str] = None,
             comm.tp_group = g
             comm.tp_rank = ranks.index(rank)
             comm.tp_size = len(ranks)
+    subgroups = json.loads(os.environ.get(ENV_COMM_SUBGROUPS, "[]") or "[]")
+    for ranks in subgroups:
+        g = dist.new_group(ranks=ranks)
+        if rank in ranks:
+            comm.stage_group = g
+            comm.stage_rank = ranks.index(rank)
+            comm.stage_size = len(ranks)
     return comm
