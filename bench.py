@@ -32,7 +32,7 @@ def parse_args():
                    help="synthetic prompt length")
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--device", default="cuda")
-    p.add_argument("--parallel", choices=["dp", "tp"], default="dp",
+    p.add_argument("--parallel", choices=["dp", "tp", "pp"], default="dp",
                    help="dp: one engine replica per rank (weak scaling); "
                         "tp: all ranks form one tensor-parallel engine")
     p.add_argument("--eager", action="store_true",
@@ -62,28 +62,32 @@ def main() -> int:
     from rbg_amd.engine.config import EngineConfig, ModelConfig
     from rbg_amd.engine.engine import LLMEngine
     from rbg_amd.engine.sequence import SamplingParams
-    from rbg_amd.models.llama import TPContext
+    from rbg_amd.models.llama import PPContext, TPContext
 
     tp_mode = args.parallel == "tp" and world > 1
+    pp_mode = args.parallel == "pp" and world > 1
     model_cfg = ModelConfig.preset(args.model)
     cfg = EngineConfig(
         model=model_cfg, device=device,
         max_batch_size=max(args.batch, 8),
         max_seq_len=args.seq_len + args.steps + args.warmup + 64,
         max_prefill_tokens=8192,
-        # TP: collectives inside hipGraph capture are deferred to round 2
-        enforce_eager=args.eager or device != "cuda" or tp_mode,
+        # TP/PP: collectives inside hipGraph capture are deferred to round 2
+        enforce_eager=args.eager or device != "cuda" or tp_mode or pp_mode,
         kv_pool_tokens=(args.batch *
                         (args.seq_len + args.steps + args.warmup + 64) + 4096),
         tp_size=world if tp_mode else 1,
         tp_rank=rank if tp_mode else 0,
     )
     tp = TPContext(size=world, rank=rank) if tp_mode else None
-    eng = LLMEngine(cfg, tp)
+    pp = PPContext(size=world, stage=rank,
+                   instance_ranks=list(range(world)),
+                   tp_size=1) if pp_mode else None
+    eng = LLMEngine(cfg, tp, pp)
 
-    # TP lockstep: every rank must build the IDENTICAL schedule, so the
+    # TP/PP lockstep: every rank must build the IDENTICAL schedule, so the
     # prompt seed must not depend on rank
-    torch.manual_seed(123 if tp_mode else 123 + rank)
+    torch.manual_seed(123 if (tp_mode or pp_mode) else 123 + rank)
     prompts = [torch.randint(0, model_cfg.vocab_size, (args.seq_len,)).tolist()
                for _ in range(args.batch)]
     sampling = SamplingParams(
@@ -132,7 +136,7 @@ def main() -> int:
         elapsed = float(t.item())
 
     ms_per_step = elapsed / args.steps * 1000.0
-    replicas = 1 if tp_mode else n_gpus
+    replicas = 1 if (tp_mode or pp_mode) else n_gpus
     total_tok_s = args.batch * replicas * args.steps / elapsed
     ttfts = sorted(eng.stats.ttfts)
     p50_ttft_ms = (ttfts[len(ttfts) // 2] * 1000.0) if ttfts else 0.0
@@ -158,6 +162,7 @@ def main() -> int:
                 "global_batch": args.batch * replicas,
                 "seq_len": args.seq_len,
                 "parallelism": (f"tp{n_gpus}" if tp_mode else
+                                f"pp{n_gpus}" if pp_mode else
                                 f"dp{n_gpus}"),
                 "p50_ttft_ms": round(p50_ttft_ms, 1),
                 "prefill_wall_s": round(prefill_wall, 3),
