@@ -103,7 +103,8 @@ def main():
         prompts = [torch.randint(0, m.vocab_size, (CTX,)).tolist()
                    for _ in range(B)]
         for p in prompts:
-            eng.add_request(p, SamplingParams(max_new_tokens=512))
+            eng.add_request(p, SamplingParams(max_new_tokens=200,
+                                              ignore_eos=True))
         while eng.scheduler.waiting:
             eng.step()
         torch.cuda.synchronize()
