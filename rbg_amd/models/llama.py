@@ -162,7 +162,7 @@ class LlamaAttention(nn.Module):
     def forward(self, x: torch.Tensor, batch: ForwardBatch,
                 kv: PagedKVCache, cos_sin: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
-        qkv = torch.nn.functional.linear(x, self.wqkv)
+        qkv = ops.linear(x, self.wqkv)
         q = qkv[:, :self.q_out].contiguous()
         k = qkv[:, self.q_out:self.q_out + self.kv_out].contiguous()
         v = qkv[:, self.q_out + self.kv_out:].contiguous()
@@ -192,7 +192,7 @@ class LlamaAttention(nn.Module):
                                      batch.block_tables, batch.context_lens,
                                      self.scale,
                                      num_splits=batch.decode_num_splits)
-        out = torch.nn.functional.linear(o.view(T, self.q_out), self.wo)
+        out = ops.linear(o.view(T, self.q_out).contiguous(), self.wo)
         return self.tp.all_reduce(out)
 
 
@@ -216,9 +216,9 @@ class LlamaMLP(nn.Module):
         self.inter = inter
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        gu = torch.nn.functional.linear(x, self.w_gate_up)
+        gu = ops.linear(x, self.w_gate_up)
         h = ops.silu_mul(gu)
-        return self.tp.all_reduce(torch.nn.functional.linear(h, self.w_down))
+        return self.tp.all_reduce(ops.linear(h, self.w_down))
 
 
 class LlamaLayer(nn.Module):

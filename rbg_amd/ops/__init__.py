@@ -35,6 +35,22 @@ def _on_gpu(t: torch.Tensor) -> bool:
     return t.is_cuda
 
 
+SKINNY_GEMM = int(__import__("os").environ.get("RBG_SKINNY_GEMM", "1"))
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Projection GEMM with decode-shape dispatch: for M <= 128 rows the
+    hand-written split-K streaming kernel (hip/skinny_gemm.hip) replaces
+    hipBLASLt, which leaves 2-5x on narrow-N skinny shapes
+    (profiles/gemm_ab_b128.json).  Prefill-sized M goes to the library."""
+    if (SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and
+            x.dtype == torch.bfloat16 and x.shape[0] <= 128 and
+            w.shape[0] % 32 == 0 and x.shape[1] % 256 == 0):
+        return _require_hip().skinny_gemm(
+            x if x.is_contiguous() else x.contiguous(), w)
+    return torch.nn.functional.linear(x, w)
+
+
 def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
     if _on_gpu(x):
         return _require_hip().rmsnorm(x, weight, eps)

@@ -20,6 +20,9 @@ void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const void*, const void*, float, int, int, int,
                               int, hipStream_t);
 void launch_mfma_probe(void*, const void*, const void*, int, int, hipStream_t);
+void launch_skinny_gemm(void*, void*, const void*, const void*, int, int,
+                        int, int, hipStream_t);
+void launch_reduce_splits(void*, const void*, int, long, hipStream_t);
 }
 
 namespace {
@@ -161,6 +164,46 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b, int64_t a_split,
 
 }  // namespace
 
+static int pick_gemm_splits(int N, int K) {
+  // fill the chip with 16-wave workgroups (>= ~256 WGs); each split's K
+  // range divides into 8 x 32-k wave slices (K % (splits*256) == 0)
+  const int n_groups = N >> 5;
+  for (int s : {8, 4, 2})
+    if (n_groups * s <= 512 && n_groups * s >= 192 && K % (s * 256) == 0)
+      return s;
+  for (int s : {4, 2})
+    if (n_groups * s >= 128 && K % (s * 256) == 0) return s;
+  return 1;
+}
+
+bool skinny_gemm_supported(int64_t M, int64_t N, int64_t K) {
+  // M <= 128: the cross-slice LDS reduce holds 16 waves x M_TILES x 1 KB
+  // (128 KB at M_TILES=8); larger decode batches fall back to hipBLASLt
+  return M >= 1 && M <= 128 && (N % 32) == 0 && (K % 256) == 0;
+}
+
+torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w) {
+  check_bf16_contig(a, "a");
+  check_bf16_contig(w, "w");
+  const int64_t M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(skinny_gemm_supported(M, N, K), "unsupported skinny shape");
+  auto c = torch::empty({M, N}, a.options());
+  const int splits = pick_gemm_splits((int)N, (int)K);
+  torch::Tensor ws;
+  void* wsp = nullptr;
+  if (splits > 1) {
+    ws = torch::empty({splits, M, N}, a.options().dtype(torch::kFloat32));
+    wsp = ws.data_ptr();
+  }
+  launch_skinny_gemm(c.data_ptr(), wsp, a.data_ptr(), w.data_ptr(), (int)M,
+                     (int)N, (int)K, splits, current_stream());
+  if (splits > 1)
+    launch_reduce_splits(c.data_ptr(), wsp, splits, (long)(M * N),
+                         current_stream());
+  return c;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "x,residual += ; rmsnorm");
@@ -169,4 +212,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attention", &decode_attention, "paged flash-decode");
   m.def("prefill_attention", &prefill_attention, "varlen causal flash prefill");
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe");
+  m.def("skinny_gemm", &skinny_gemm, "decode-shape GEMM (M<=128)");
 }

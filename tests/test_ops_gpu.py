@@ -180,3 +180,24 @@ def test_prefill_attention_with_prefix(dev, lens):
     want = ref.prefill_attention(q, k, v, cu_q, scale, cu_k)
     err = (got.float() - want.float()).abs().max().item()
     assert err < 3e-2, f"lens={lens} err={err}"
+
+
+@pytest.mark.gpu
+def test_skinny_gemm_matches_fp32():
+    """skinny_gemm vs fp32 matmul on every decode projection shape of
+    llama-3-8b plus edge batches (M=1, 7, 33, 128) and all split picks."""
+    from rbg_amd import ops
+    dev = torch.device("cuda:0")
+    H, I, QH, KVH, D = 4096, 14336, 32, 8, 128
+    shapes = [(H, (QH + 2 * KVH) * D), (QH * D, H), (H, 2 * I), (I, H)]
+    torch.manual_seed(7)
+    for M in (1, 7, 33, 64, 128):
+        for (k, n) in shapes:
+            x = torch.randn(M, k, dtype=torch.bfloat16, device=dev) * 0.5
+            w = torch.randn(n, k, dtype=torch.bfloat16, device=dev) * 0.05
+            got = ops._hip.skinny_gemm(x, w).float()
+            ref = x.float() @ w.float().t()
+            # bf16 inputs, f32 accumulate: error comes from input rounding
+            tol = ref.abs().max().item() * 3e-2 + 0.02
+            assert (got - ref).abs().max().item() < tol, \
+                (M, k, n, (got - ref).abs().max().item(), tol)
