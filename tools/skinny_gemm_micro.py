@@ -2,6 +2,10 @@
 """Within-probe A/B: skinny_gemm vs hipBLASLt on decode projection shapes."""
 import argparse
 import json
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
