@@ -45,7 +45,8 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     (profiles/gemm_ab_b128.json).  Prefill-sized M goes to the library."""
     if (SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and
             x.dtype == torch.bfloat16 and x.shape[0] <= 128 and
-            w.shape[0] % 32 == 0 and x.shape[1] % 256 == 0):
+            w.shape[0] % 32 == 0 and w.shape[0] <= 8192 and
+            x.shape[1] % 256 == 0):
         return _require_hip().skinny_gemm(
             x if x.is_contiguous() else x.contiguous(), w)
     return torch.nn.functional.linear(x, w)

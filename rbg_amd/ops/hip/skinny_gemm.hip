@@ -6,37 +6,43 @@
 // 2-5x on the narrow-N shapes (profiles/gemm_ab_b128.json: hipBLASLt and
 // rocBLAS both ~1.7-2.4 TB/s on o/qkv/down vs ~6 TB/s on the wide gate_up;
 // composable-kernel far worse).  Library tilings are built for big M; a
-// skinny GEMM instead wants every CU streaming DISTINCT weight rows with
-// the whole A operand riding in L2.
+// skinny GEMM wants every CU streaming DISTINCT weight rows with the whole
+// A operand riding in L2.
 //
-// Shape of the kernel (CDNA4, 64-lane waves, v_mfma_f32_16x16x32_bf16):
-//   grid  = (N/32) * SPLITS workgroups, 16 waves each (1024 threads)
-//   wave  = (k_slice 0..7) x (n_tile 0..1): 16 waves cover a 32-column
-//           C panel, splitting this split's K range 8 ways
-//   loop  = per 32-k step: one W fragment (16 rows x 32 k, the MFMA B
-//           operand straight from global: lane n = l&15, k = (l>>4)*8 --
-//           the contiguous-8 per-lane layout matches row-major [N,K]) and
-//           M_TILES A fragments (same layout over A rows, clamped), one
-//           MFMA each, 1-ahead prefetched so the weight stream stays
-//           ahead of the math
-//   end   = the 8 k-slices of each n-tile reduce through LDS (128 KB:
-//           16 waves x 8 KB of f32 partials), then one bf16 store --
-//           no global atomics, deterministic, single kernel when SPLITS=1
-//   SPLITS>1 writes f32 partials to a workspace; reduce_splits() folds
-//           them to bf16 (two tiny extra launches only when K is deep)
+// Kernel shape (CDNA4, 64-lane waves, v_mfma_f32_16x16x32_bf16):
+//   grid  = (N/32) * SPLITS workgroups, 16 waves each
+//   wave  = (k_slice 0..7) x (n_tile 0..1): the WG owns a 32-column C
+//           panel and walks its split's K range 256 k per step, the 8
+//           k-slices covering 32 k each
+//   stage = W[32 x 256] and A[M x 256] land in LDS through COALESCED
+//           16-lane x 16 B row pieces (direct per-lane fragment gathers
+//           from global were 3-5x slower: 16 discontiguous 16 B requests
+//           per instruction — measured, first cut of this kernel), with
+//           the next step's pieces prefetched into registers under the
+//           current step's MFMAs (attn_prefill.hip staging discipline)
+//   math  = per step each wave reads its fragment window from LDS
+//           (1 W frag + M_TILES A frags, contiguous-8 per-lane layout)
+//           and issues M_TILES MFMAs
+//   end   = the 8 k-slices of each n-tile fold through the SAME LDS
+//           buffer (re-used as f32 scratch), one bf16 store; SPLITS>1
+//           writes f32 partials + reduce_splits() folds them
 //
-// A-traffic note: computing a 32-column panel reads all of A (M x K);
-// with M=128 that is ~1 MB for K=4096 -- resident in each XCD's L2 after
-// the first panel, so HBM sees ~W + 8 x A bytes total.
+// A-traffic note: a 32-column panel reads all of A (M x K, ~1 MB at
+// M=128/K=4096) — L2-resident after the first panel per XCD, so HBM sees
+// ~W + 8 x A bytes; LDS sees ~2x A (two waves share a k-window), well
+// under the 128 B/clk LDS budget needed to keep W streaming at HBM rate.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 gg_bf8;
 
-DEV_INLINE gg_bf8 g_load_bf8(const __hip_bfloat16* p) {
+DEV_INLINE gg_bf8 lds_frag(const __hip_bfloat16* p) {
   union { uint4 u; gg_bf8 v; } cvt;
   cvt.u = *reinterpret_cast<const uint4*>(p);
   return cvt.v;
 }
+
+#define SG_KSTEP 256              // k consumed per WG step
+#define SG_LDS_PITCH (SG_KSTEP + 8)   // +8 bf16: bank-conflict pad
 
 template <int M_TILES, int SPLITS>
 __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
@@ -55,74 +61,101 @@ __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
   const int gslice = lane >> 4;    // fragment k sub-chunk
 
   const int n_tile = wave & 1;               // 0..1 within the 32-col panel
-  const int k_slice = wave >> 1;             // 0..7
-  const int k_per_split = K / SPLITS;
-  const int k_per_wave = k_per_split >> 3;   // 8 slices per split
-  const int k0 = split * k_per_split + k_slice * k_per_wave;
-  const int iters = k_per_wave >> 5;         // 32 k per step
+  const int k_slice = wave >> 1;             // 0..7, 32 k each per step
+  const int k_wg = K / SPLITS;               // this WG's k span
+  const int k0 = split * k_wg;
+  const int nsteps = k_wg / SG_KSTEP;
 
-  const int n_base = (n_group << 5) + (n_tile << 4);
-  const __hip_bfloat16* wrow = w + (size_t)(n_base + gl) * K + k0 +
-                               gslice * 8;
-  // A rows clamp to M-1: rows past M compute garbage that the store masks
-  const __hip_bfloat16* arow[M_TILES];
+  // LDS: staged W/A panels for the current step, re-used afterwards as the
+  // f32 cross-slice reduction scratch (128 KB at M_TILES=8)
+  constexpr int A_ROWS = M_TILES * 16;
+  __shared__ union {
+    __hip_bfloat16 stage[(32 + A_ROWS) * SG_LDS_PITCH];
+    float red[16][M_TILES * 256];
+  } lds;
+  __hip_bfloat16* w_lds = lds.stage;                        // [32][pitch]
+  __hip_bfloat16* a_lds = lds.stage + 32 * SG_LDS_PITCH;    // [A_ROWS][pitch]
+
+  // staging pieces: 16 B per (row, chunk16); W = 32x16 = 512 pieces
+  // (threads 0..511), A = A_ROWS x 16 pieces, 1..2 per thread
+  const int wrow_st = tid >> 4, wchk = (tid & 15) * 8;      // W piece
+  constexpr int A_PIECES = (A_ROWS * 16 + 1023) / 1024;
+  uint4 wreg, areg[A_PIECES];
+  auto issue_loads = [&](int step) {
+    const int kb = k0 + step * SG_KSTEP;
+    if (wrow_st < 32) {
+      const int n = (n_group << 5) + wrow_st;
+      wreg = *reinterpret_cast<const uint4*>(w + (size_t)n * K + kb + wchk);
+    }
 #pragma unroll
-  for (int mt = 0; mt < M_TILES; ++mt) {
-    const int m = min(mt * 16 + gl, M - 1);
-    arow[mt] = a + (size_t)m * K + k0 + gslice * 8;
-  }
+    for (int s = 0; s < A_PIECES; ++s) {
+      const int i = tid + s * 1024;
+      const int row = min(i >> 4, M - 1);       // clamp: garbage rows masked
+      const int chk = (i & 15) * 8;
+      areg[s] = *reinterpret_cast<const uint4*>(a + (size_t)row * K + kb + chk);
+    }
+  };
+  auto write_tile = [&]() {
+    if (wrow_st < 32)
+      *reinterpret_cast<uint4*>(&w_lds[wrow_st * SG_LDS_PITCH + wchk]) = wreg;
+#pragma unroll
+    for (int s = 0; s < A_PIECES; ++s) {
+      const int i = tid + s * 1024;
+      if (i < A_ROWS * 16)
+        *reinterpret_cast<uint4*>(
+            &a_lds[(i >> 4) * SG_LDS_PITCH + (i & 15) * 8]) = areg[s];
+    }
+  };
 
   f32x4 acc[M_TILES];
 #pragma unroll
   for (int mt = 0; mt < M_TILES; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
 
-  // 1-ahead software pipeline: weights + A for step i+1 issue before the
-  // MFMAs of step i retire (unconditional loads; the last iteration
-  // re-reads its own step, harmless and branch-free)
-  gg_bf8 wf = g_load_bf8(wrow);
-  gg_bf8 af[M_TILES];
-#pragma unroll
-  for (int mt = 0; mt < M_TILES; ++mt) af[mt] = g_load_bf8(arow[mt]);
+  issue_loads(0);
+  write_tile();
 
-  for (int it = 0; it < iters; ++it) {
-    const int nxt = (it + 1 < iters) ? (it + 1) << 5 : it << 5;
-    gg_bf8 wn = g_load_bf8(wrow + nxt);
-    gg_bf8 an[M_TILES];
+  for (int step = 0; step < nsteps; ++step) {
+    __syncthreads();                     // tile `step` visible to all waves
+    if (step + 1 < nsteps)
+      issue_loads(step + 1);             // in flight under the MFMAs
+
+    const int kf = k_slice * 32 + gslice * 8;     // fragment k offset
+    const gg_bf8 wf = lds_frag(
+        &w_lds[((n_tile << 4) + gl) * SG_LDS_PITCH + kf]);
 #pragma unroll
-    for (int mt = 0; mt < M_TILES; ++mt)
-      an[mt] = g_load_bf8(arow[mt] + nxt);
-#pragma unroll
-    for (int mt = 0; mt < M_TILES; ++mt)
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          af[mt], wf, acc[mt], 0, 0, 0);
-    wf = wn;
-#pragma unroll
-    for (int mt = 0; mt < M_TILES; ++mt) af[mt] = an[mt];
+    for (int mt = 0; mt < M_TILES; ++mt) {
+      const gg_bf8 af = lds_frag(
+          &a_lds[(mt * 16 + gl) * SG_LDS_PITCH + kf]);
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, wf, acc[mt],
+                                                        0, 0, 0);
+    }
+    __syncthreads();                     // all reads done before overwrite
+    if (step + 1 < nsteps)
+      write_tile();
   }
 
-  // ---- cross-slice reduction through LDS --------------------------------
-  // wave's partial panel: M_TILES x (16 rows x 16 cols) f32, 4 regs/lane
-  __shared__ float red[16][M_TILES * 256];
+  // ---- cross-slice reduction through LDS (buffer re-use) ----------------
+  __syncthreads();
 #pragma unroll
   for (int mt = 0; mt < M_TILES; ++mt) {
     F32x4U u;
     u.v = acc[mt];
 #pragma unroll
     for (int r = 0; r < 4; ++r)
-      red[wave][mt * 256 + r * 64 + lane] = u.e[r];
+      lds.red[wave][mt * 256 + r * 64 + lane] = u.e[r];
   }
   __syncthreads();
 
-  // each thread folds 8 k-slices for its share of the two n-tiles.
-  // elem e of tile nt: lane64 = e & 63, reg = (e >> 6) & 3, mt = e >> 8
-  //   -> m = mt*16 + 4*(lane64>>4) + reg, n = n_base' + (lane64 & 15)
-  const int elems = M_TILES * 256 * 2;              // both n-tiles
+  // fold 8 k-slices; elem e of n-tile nt: lane64 = e & 63,
+  // reg = (e >> 6) & 3, mt = e >> 8 -> m = mt*16 + 4*(lane64>>4) + reg,
+  // n = panel + nt*16 + (lane64 & 15)
+  const int elems = M_TILES * 256 * 2;
   for (int e = tid; e < elems; e += 1024) {
     const int nt = e >= M_TILES * 256;
     const int ee = e - nt * M_TILES * 256;
     float v = 0.f;
 #pragma unroll
-    for (int ks = 0; ks < 8; ++ks) v += red[ks * 2 + nt][ee];
+    for (int ks = 0; ks < 8; ++ks) v += lds.red[ks * 2 + nt][ee];
     const int l64 = ee & 63;
     const int reg = (ee >> 6) & 3;
     const int mt = ee >> 8;
