@@ -76,34 +76,34 @@ __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
   __hip_bfloat16* w_lds = lds.stage;                        // [32][pitch]
   __hip_bfloat16* a_lds = lds.stage + 32 * SG_LDS_PITCH;    // [A_ROWS][pitch]
 
-  // staging pieces: 16 B per (row, chunk16); W = 32x16 = 512 pieces
-  // (threads 0..511), A = A_ROWS x 16 pieces, 1..2 per thread
-  const int wrow_st = tid >> 4, wchk = (tid & 15) * 8;      // W piece
-  constexpr int A_PIECES = (A_ROWS * 16 + 1023) / 1024;
+  // staging pieces: 16 B per (row, chunk); a 256-k row is 32 chunks, so
+  // W = 32 x 32 = 1024 pieces (one per thread, 32 lanes x 16 B = 512 B
+  // contiguous per row) and A = A_ROWS x 32 pieces (up to 4 per thread)
+  const int wrow_st = tid >> 5, wchk = (tid & 31) * 8;      // W piece
+  constexpr int A_PIECES = (A_ROWS * 32 + 1023) / 1024;
   uint4 wreg, areg[A_PIECES];
   auto issue_loads = [&](int step) {
     const int kb = k0 + step * SG_KSTEP;
-    if (wrow_st < 32) {
+    {
       const int n = (n_group << 5) + wrow_st;
       wreg = *reinterpret_cast<const uint4*>(w + (size_t)n * K + kb + wchk);
     }
 #pragma unroll
     for (int s = 0; s < A_PIECES; ++s) {
       const int i = tid + s * 1024;
-      const int row = min(i >> 4, M - 1);       // clamp: garbage rows masked
-      const int chk = (i & 15) * 8;
+      const int row = min(i >> 5, M - 1);       // clamp: garbage rows masked
+      const int chk = (i & 31) * 8;
       areg[s] = *reinterpret_cast<const uint4*>(a + (size_t)row * K + kb + chk);
     }
   };
   auto write_tile = [&]() {
-    if (wrow_st < 32)
-      *reinterpret_cast<uint4*>(&w_lds[wrow_st * SG_LDS_PITCH + wchk]) = wreg;
+    *reinterpret_cast<uint4*>(&w_lds[wrow_st * SG_LDS_PITCH + wchk]) = wreg;
 #pragma unroll
     for (int s = 0; s < A_PIECES; ++s) {
       const int i = tid + s * 1024;
-      if (i < A_ROWS * 16)
+      if (i < A_ROWS * 32)
         *reinterpret_cast<uint4*>(
-            &a_lds[(i >> 4) * SG_LDS_PITCH + (i & 15) * 8]) = areg[s];
+            &a_lds[(i >> 5) * SG_LDS_PITCH + (i & 31) * 8]) = areg[s];
     }
   };
 
