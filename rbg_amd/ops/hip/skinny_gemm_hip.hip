@@ -11,10 +11,15 @@
 // A operand riding in L2.
 //
 // Kernel shape (CDNA4, 64-lane waves, v_mfma_f32_16x16x32_bf16):
-//   grid  = (N/32) * SPLITS workgroups, 16 waves each
-//   wave  = (k_slice 0..7) x (n_tile 0..1): the WG owns a 32-column C
-//           panel and walks its split's K range 256 k per step, the 8
-//           k-slices covering 32 k each
+//   grid  = (N/32) * SPLITS workgroups, 8 waves each
+//   wave  = (k_slice 0..3) x (n_tile 0..1): the WG owns a 32-column C
+//           panel and walks its split's K range 128 k per step, the 4
+//           k-slices covering 32 k each.  8 waves + 64 KB LDS keep TWO
+//           workgroups resident per CU, so one WG's MFMAs and global
+//           loads run while the other sits in its staging barrier — with
+//           a single 16-wave WG (v2 of this kernel) every wave stalled
+//           on the same barrier and the step time collapsed to the raw
+//           HBM latency (measured 5x slower than the library)
 //   stage = W[32 x 256] and A[M x 256] land in LDS through COALESCED
 //           16-lane x 16 B row pieces (direct per-lane fragment gathers
 //           from global were 3-5x slower: 16 discontiguous 16 B requests
@@ -42,11 +47,11 @@ DEV_INLINE gg_bf8 lds_frag(const __hip_bfloat16* p) {
   return cvt.v;
 }
 
-#define SG_KSTEP 256              // k consumed per WG step
+#define SG_KSTEP 128              // k consumed per WG step
 #define SG_LDS_PITCH (SG_KSTEP + 8)   // +8 bf16: bank-conflict pad
 
 template <int M_TILES, int SPLITS>
-__global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
+__global__ __launch_bounds__(512, 2) void skinny_gemm_kernel(
     const __hip_bfloat16* __restrict__ a,   // [M, K]
     const __hip_bfloat16* __restrict__ w,   // [N, K]
     __hip_bfloat16* __restrict__ c,         // [M, N]   (SPLITS == 1)
@@ -62,26 +67,26 @@ __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
   const int gslice = lane >> 4;    // fragment k sub-chunk
 
   const int n_tile = wave & 1;               // 0..1 within the 32-col panel
-  const int k_slice = wave >> 1;             // 0..7, 32 k each per step
+  const int k_slice = wave >> 1;             // 0..3, 32 k each per step
   const int k_wg = K / SPLITS;               // this WG's k span
   const int k0 = split * k_wg;
   const int nsteps = k_wg / SG_KSTEP;
 
   // LDS: staged W/A panels for the current step, re-used afterwards as the
-  // f32 cross-slice reduction scratch (128 KB at M_TILES=8)
+  // f32 cross-slice reduction scratch (64 KB at M_TILES=8 -> 2 WGs/CU)
   constexpr int A_ROWS = M_TILES * 16;
   __shared__ union {
     __hip_bfloat16 stage[(32 + A_ROWS) * SG_LDS_PITCH];
-    float red[16][M_TILES * 256];
+    float red[8][M_TILES * 256];
   } lds;
   __hip_bfloat16* w_lds = lds.stage;                        // [32][pitch]
   __hip_bfloat16* a_lds = lds.stage + 32 * SG_LDS_PITCH;    // [A_ROWS][pitch]
 
-  // staging pieces: 16 B per (row, chunk); a 256-k row is 32 chunks, so
-  // W = 32 x 32 = 1024 pieces (one per thread, 32 lanes x 16 B = 512 B
-  // contiguous per row) and A = A_ROWS x 32 pieces (up to 4 per thread)
-  const int wrow_st = tid >> 5, wchk = (tid & 31) * 8;      // W piece
-  constexpr int A_PIECES = (A_ROWS * 32 + 1023) / 1024;
+  // staging pieces: 16 B per (row, chunk); a 128-k row is 16 chunks, so
+  // W = 32 x 16 = 512 pieces (one per thread, 16 lanes x 16 B = 256 B
+  // contiguous per row) and A = A_ROWS x 16 pieces (up to 4 per thread)
+  const int wrow_st = tid >> 4, wchk = (tid & 15) * 8;      // W piece
+  constexpr int A_PIECES = (A_ROWS * 16 + 511) / 512;
   uint4 wreg, areg[A_PIECES];
   auto issue_loads = [&](int step) {
     const int kb = k0 + step * SG_KSTEP;
@@ -91,9 +96,9 @@ __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
     }
 #pragma unroll
     for (int s = 0; s < A_PIECES; ++s) {
-      const int i = tid + s * 1024;
-      const int row = min(i >> 5, M - 1);       // clamp: garbage rows masked
-      const int chk = (i & 31) * 8;
+      const int i = tid + s * 512;
+      const int row = min(i >> 4, M - 1);       // clamp: garbage rows masked
+      const int chk = (i & 15) * 8;
       areg[s] = *reinterpret_cast<const uint4*>(a + (size_t)row * K + kb + chk);
     }
   };
@@ -101,10 +106,10 @@ __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
     *reinterpret_cast<uint4*>(&w_lds[wrow_st * SG_LDS_PITCH + wchk]) = wreg;
 #pragma unroll
     for (int s = 0; s < A_PIECES; ++s) {
-      const int i = tid + s * 1024;
-      if (i < A_ROWS * 32)
+      const int i = tid + s * 512;
+      if (i < A_ROWS * 16)
         *reinterpret_cast<uint4*>(
-            &a_lds[(i >> 5) * SG_LDS_PITCH + (i & 31) * 8]) = areg[s];
+            &a_lds[(i >> 4) * SG_LDS_PITCH + (i & 15) * 8]) = areg[s];
     }
   };
 
@@ -147,16 +152,16 @@ __global__ __launch_bounds__(1024, 1) void skinny_gemm_kernel(
   }
   __syncthreads();
 
-  // fold 8 k-slices; elem e of n-tile nt: lane64 = e & 63,
+  // fold 4 k-slices; elem e of n-tile nt: lane64 = e & 63,
   // reg = (e >> 6) & 3, mt = e >> 8 -> m = mt*16 + 4*(lane64>>4) + reg,
   // n = panel + nt*16 + (lane64 & 15)
   const int elems = M_TILES * 256 * 2;
-  for (int e = tid; e < elems; e += 1024) {
+  for (int e = tid; e < elems; e += 512) {
     const int nt = e >= M_TILES * 256;
     const int ee = e - nt * M_TILES * 256;
     float v = 0.f;
 #pragma unroll
-    for (int ks = 0; ks < 8; ++ks) v += lds.red[ks * 2 + nt][ee];
+    for (int ks = 0; ks < 4; ++ks) v += lds.red[ks * 2 + nt][ee];
     const int l64 = ee & 63;
     const int reg = (ee >> 6) & 3;
     const int mt = ee >> 8;
@@ -187,7 +192,7 @@ __global__ void reduce_splits_kernel(const float* __restrict__ ws,
 extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
                                    const void* w, int M, int N, int K,
                                    int splits, hipStream_t stream) {
-  const dim3 grid((N >> 5) * splits), block(1024);
+  const dim3 grid((N >> 5) * splits), block(512);
   const __hip_bfloat16* ap = (const __hip_bfloat16*)a;
   const __hip_bfloat16* wp = (const __hip_bfloat16*)w;
   __hip_bfloat16* cp = (__hip_bfloat16*)c;
