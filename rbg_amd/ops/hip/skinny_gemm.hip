@@ -10,32 +10,26 @@
 // A operand riding in L2.
 //
 // Kernel shape (CDNA4, 64-lane waves, v_mfma_f32_16x16x32_bf16):
-//   grid  = (N/32) * SPLITS workgroups, 8 waves each
-//   wave  = (k_slice 0..3) x (n_tile 0..1): the WG owns a 32-column C
-//           panel and walks its split's K range 128 k per step, the 4
-//           k-slices covering 32 k each.  8 waves + 64 KB LDS keep TWO
-//           workgroups resident per CU, so one WG's MFMAs and global
-//           loads run while the other sits in its staging barrier — with
-//           a single 16-wave WG (v2 of this kernel) every wave stalled
-//           on the same barrier and the step time collapsed to the raw
-//           HBM latency (measured 5x slower than the library)
-//   stage = W[32 x 256] and A[M x 256] land in LDS through COALESCED
-//           16-lane x 16 B row pieces (direct per-lane fragment gathers
-//           from global were 3-5x slower: 16 discontiguous 16 B requests
-//           per instruction — measured, first cut of this kernel), with
-//           the next step's pieces prefetched into registers under the
-//           current step's MFMAs (attn_prefill.hip staging discipline)
-//   math  = per step each wave reads its fragment window from LDS
-//           (1 W frag + M_TILES A frags, contiguous-8 per-lane layout)
-//           and issues M_TILES MFMAs
-//   end   = the 8 k-slices of each n-tile fold through the SAME LDS
-//           buffer (re-used as f32 scratch), one bf16 store; SPLITS>1
-//           writes f32 partials + reduce_splits() folds them
+//   grid  = (N/32) * SPLITS workgroups of 4 waves: (k_slice 0..1) x
+//           (n_tile 0..1) over a 32-column C panel, 64 k per step
+//   stage = W[32 x 64] + A[M x 64] through LDS via coalesced 16-lane row
+//           pieces (direct per-lane fragment gathers measured 3-5x slow)
+//   pipe  = DOUBLE-buffered LDS + TWO register load-sets issuing 2-3
+//           steps ahead, ONE barrier per step.  Earlier cuts of this
+//           kernel staged 1 step ahead behind 2 barriers: every step then
+//           serialized on full HBM latency (rocprof: SQ_WAIT_ANY = 90%
+//           of cycles, 42 us for a 33 MB weight stream).  Small WGs
+//           (23 KB stage + 32 KB reduce scratch) keep ~3 WGs resident
+//           per CU so stalled WGs are covered by runnable ones.
+//   math  = per step each wave reads 1 W + M_TILES A fragments from LDS
+//           (contiguous-8 per-lane MFMA layout) and issues M_TILES MFMAs
+//   end   = the 2 k-slices of each n-tile fold through the LDS reduce
+//           scratch, one bf16 store; SPLITS>1 writes f32 partials +
+//           reduce_splits() folds them
 //
 // A-traffic note: a 32-column panel reads all of A (M x K, ~1 MB at
 // M=128/K=4096) — L2-resident after the first panel per XCD, so HBM sees
-// ~W + 8 x A bytes; LDS sees ~2x A (two waves share a k-window), well
-// under the 128 B/clk LDS budget needed to keep W streaming at HBM rate.
+// ~W + 8 x A bytes.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 gg_bf8;
@@ -46,11 +40,12 @@ DEV_INLINE gg_bf8 lds_frag(const __hip_bfloat16* p) {
   return cvt.v;
 }
 
-#define SG_KSTEP 128              // k consumed per WG step
+#define SG_KSTEP 64                   // k consumed per WG step
 #define SG_LDS_PITCH (SG_KSTEP + 8)   // +8 bf16: bank-conflict pad
+#define SG_NT 256                     // threads (4 waves)
 
 template <int M_TILES, int SPLITS>
-__global__ __launch_bounds__(512, 2) void skinny_gemm_kernel(
+__global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
     const __hip_bfloat16* __restrict__ a,   // [M, K]
     const __hip_bfloat16* __restrict__ w,   // [N, K]
     __hip_bfloat16* __restrict__ c,         // [M, N]   (SPLITS == 1)
@@ -66,49 +61,48 @@ __global__ __launch_bounds__(512, 2) void skinny_gemm_kernel(
   const int gslice = lane >> 4;    // fragment k sub-chunk
 
   const int n_tile = wave & 1;               // 0..1 within the 32-col panel
-  const int k_slice = wave >> 1;             // 0..3, 32 k each per step
+  const int k_slice = wave >> 1;             // 0..1, 32 k each per step
   const int k_wg = K / SPLITS;               // this WG's k span
   const int k0 = split * k_wg;
   const int nsteps = k_wg / SG_KSTEP;
 
-  // LDS: staged W/A panels for the current step, re-used afterwards as the
-  // f32 cross-slice reduction scratch (64 KB at M_TILES=8 -> 2 WGs/CU)
   constexpr int A_ROWS = M_TILES * 16;
+  constexpr int BUF = (32 + A_ROWS) * SG_LDS_PITCH;
   __shared__ union {
-    __hip_bfloat16 stage[(32 + A_ROWS) * SG_LDS_PITCH];
-    float red[8][M_TILES * 256];
+    __hip_bfloat16 stage[2][BUF];
+    float red[4][M_TILES * 256];
   } lds;
-  __hip_bfloat16* w_lds = lds.stage;                        // [32][pitch]
-  __hip_bfloat16* a_lds = lds.stage + 32 * SG_LDS_PITCH;    // [A_ROWS][pitch]
 
-  // staging pieces: 16 B per (row, chunk); a 128-k row is 16 chunks, so
-  // W = 32 x 16 = 512 pieces (one per thread, 16 lanes x 16 B = 256 B
-  // contiguous per row) and A = A_ROWS x 16 pieces (up to 4 per thread)
-  const int wrow_st = tid >> 4, wchk = (tid & 15) * 8;      // W piece
-  constexpr int A_PIECES = (A_ROWS * 16 + 511) / 512;
-  uint4 wreg, areg[A_PIECES];
-  auto issue_loads = [&](int step) {
-    const int kb = k0 + step * SG_KSTEP;
-    {
-      const int n = (n_group << 5) + wrow_st;
-      wreg = *reinterpret_cast<const uint4*>(w + (size_t)n * K + kb + wchk);
-    }
+  // staging pieces, 16 B per (row, chunk): a 64-k row is 8 chunks, so
+  // W = 32 x 8 = 256 pieces (one per thread) and A = A_ROWS x 8 pieces
+  // (up to 4 per thread); 8 lanes x 16 B = 128 B contiguous per row
+  const int wrow_st = tid >> 3, wchk = (tid & 7) * 8;
+  constexpr int A_PIECES = (A_ROWS * 8 + SG_NT - 1) / SG_NT;
+  struct LoadSet { uint4 wreg; uint4 areg[A_PIECES]; };
+  LoadSet set[2];
+  auto issue_loads = [&](int step, int si) {
+    const int kb = k0 + min(step, nsteps - 1) * SG_KSTEP;
+    set[si].wreg = *reinterpret_cast<const uint4*>(
+        w + (size_t)((n_group << 5) + wrow_st) * K + kb + wchk);
 #pragma unroll
     for (int s = 0; s < A_PIECES; ++s) {
-      const int i = tid + s * 512;
-      const int row = min(i >> 4, M - 1);       // clamp: garbage rows masked
-      const int chk = (i & 15) * 8;
-      areg[s] = *reinterpret_cast<const uint4*>(a + (size_t)row * K + kb + chk);
+      const int i = tid + s * SG_NT;
+      const int row = min(i >> 3, M - 1);     // clamp: garbage rows masked
+      set[si].areg[s] = *reinterpret_cast<const uint4*>(
+          a + (size_t)row * K + kb + (i & 7) * 8);
     }
   };
-  auto write_tile = [&]() {
-    *reinterpret_cast<uint4*>(&w_lds[wrow_st * SG_LDS_PITCH + wchk]) = wreg;
+  auto write_tile = [&](int buf, int si) {
+    __hip_bfloat16* st = lds.stage[buf];
+    *reinterpret_cast<uint4*>(&st[wrow_st * SG_LDS_PITCH + wchk]) =
+        set[si].wreg;
 #pragma unroll
     for (int s = 0; s < A_PIECES; ++s) {
-      const int i = tid + s * 512;
-      if (i < A_ROWS * 16)
+      const int i = tid + s * SG_NT;
+      if (i < A_ROWS * 8)
         *reinterpret_cast<uint4*>(
-            &a_lds[(i >> 4) * SG_LDS_PITCH + (i & 15) * 8]) = areg[s];
+            &st[(32 + (i >> 3)) * SG_LDS_PITCH + (i & 7) * 8]) =
+            set[si].areg[s];
     }
   };
 
@@ -116,27 +110,30 @@ __global__ __launch_bounds__(512, 2) void skinny_gemm_kernel(
 #pragma unroll
   for (int mt = 0; mt < M_TILES; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
 
-  issue_loads(0);
-  write_tile();
+  // pipeline prologue: tiles 0,1 in flight; tile 0 lands in buffer 0
+  issue_loads(0, 0);
+  issue_loads(1, 1);
+  write_tile(0, 0);
+  issue_loads(2, 0);
 
   for (int step = 0; step < nsteps; ++step) {
-    __syncthreads();                     // tile `step` visible to all waves
-    if (step + 1 < nsteps)
-      issue_loads(step + 1);             // in flight under the MFMAs
-
-    const int kf = k_slice * 32 + gslice * 8;     // fragment k offset
+    __syncthreads();     // buf[step&1] holds tile `step` for every wave
+    const int kf = k_slice * 32 + gslice * 8;
+    const __hip_bfloat16* st = lds.stage[step & 1];
     const gg_bf8 wf = lds_frag(
-        &w_lds[((n_tile << 4) + gl) * SG_LDS_PITCH + kf]);
+        &st[((n_tile << 4) + gl) * SG_LDS_PITCH + kf]);
 #pragma unroll
     for (int mt = 0; mt < M_TILES; ++mt) {
       const gg_bf8 af = lds_frag(
-          &a_lds[(mt * 16 + gl) * SG_LDS_PITCH + kf]);
+          &st[(32 + mt * 16 + gl) * SG_LDS_PITCH + kf]);
       acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, wf, acc[mt],
                                                         0, 0, 0);
     }
-    __syncthreads();                     // all reads done before overwrite
-    if (step + 1 < nsteps)
-      write_tile();
+    // tile step+1 was issued 2 steps ago -> its vmcnt wait is covered;
+    // its buffer's previous tile (step-1) was fully read before the
+    // barrier above, so no second barrier is needed
+    write_tile((step + 1) & 1, (step + 1) & 1);
+    issue_loads(step + 3, (step + 1) & 1);
   }
 
   // ---- cross-slice reduction through LDS (buffer re-use) ----------------
@@ -151,16 +148,14 @@ __global__ __launch_bounds__(512, 2) void skinny_gemm_kernel(
   }
   __syncthreads();
 
-  // fold 4 k-slices; elem e of n-tile nt: lane64 = e & 63,
+  // fold 2 k-slices; elem e of n-tile nt: lane64 = e & 63,
   // reg = (e >> 6) & 3, mt = e >> 8 -> m = mt*16 + 4*(lane64>>4) + reg,
   // n = panel + nt*16 + (lane64 & 15)
   const int elems = M_TILES * 256 * 2;
-  for (int e = tid; e < elems; e += 512) {
+  for (int e = tid; e < elems; e += SG_NT) {
     const int nt = e >= M_TILES * 256;
     const int ee = e - nt * M_TILES * 256;
-    float v = 0.f;
-#pragma unroll
-    for (int ks = 0; ks < 4; ++ks) v += lds.red[ks * 2 + nt][ee];
+    const float v = lds.red[nt][ee] + lds.red[2 + nt][ee];
     const int l64 = ee & 63;
     const int reg = (ee >> 6) & 3;
     const int mt = ee >> 8;
@@ -191,7 +186,7 @@ __global__ void reduce_splits_kernel(const float* __restrict__ ws,
 extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
                                    const void* w, int M, int N, int K,
                                    int splits, hipStream_t stream) {
-  const dim3 grid((N >> 5) * splits), block(512);
+  const dim3 grid((N >> 5) * splits), block(SG_NT);
   const __hip_bfloat16* ap = (const __hip_bfloat16*)a;
   const __hip_bfloat16* wp = (const __hip_bfloat16*)w;
   __hip_bfloat16* cp = (__hip_bfloat16*)c;
