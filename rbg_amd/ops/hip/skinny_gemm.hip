@@ -75,51 +75,70 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
 
   // staging pieces, 16 B per (row, chunk): a 64-k row is 8 chunks, so
   // W = 32 x 8 = 256 pieces (one per thread) and A = A_ROWS x 8 pieces
-  // (up to 4 per thread); 8 lanes x 16 B = 128 B contiguous per row
+  // (up to 4 per thread); 8 lanes x 16 B = 128 B contiguous per row.
+  // The two in-flight load sets are NAMED variables and the loop is
+  // unrolled by 2: a runtime-indexed set[2] array was demoted to scratch
+  // memory and every prefetch round-tripped through HBM (seen in the ISA
+  // as scratch_load/store_dwordx4 in the hot loop — 5x slowdown)
   const int wrow_st = tid >> 3, wchk = (tid & 7) * 8;
   constexpr int A_PIECES = (A_ROWS * 8 + SG_NT - 1) / SG_NT;
-  struct LoadSet { uint4 wreg; uint4 areg[A_PIECES]; };
-  LoadSet set[2];
-  auto issue_loads = [&](int step, int si) {
-    const int kb = k0 + min(step, nsteps - 1) * SG_KSTEP;
-    set[si].wreg = *reinterpret_cast<const uint4*>(
-        w + (size_t)((n_group << 5) + wrow_st) * K + kb + wchk);
-#pragma unroll
-    for (int s = 0; s < A_PIECES; ++s) {
-      const int i = tid + s * SG_NT;
-      const int row = min(i >> 3, M - 1);     // clamp: garbage rows masked
-      set[si].areg[s] = *reinterpret_cast<const uint4*>(
-          a + (size_t)row * K + kb + (i & 7) * 8);
-    }
-  };
-  auto write_tile = [&](int buf, int si) {
-    __hip_bfloat16* st = lds.stage[buf];
-    *reinterpret_cast<uint4*>(&st[wrow_st * SG_LDS_PITCH + wchk]) =
-        set[si].wreg;
-#pragma unroll
-    for (int s = 0; s < A_PIECES; ++s) {
-      const int i = tid + s * SG_NT;
-      if (i < A_ROWS * 8)
-        *reinterpret_cast<uint4*>(
-            &st[(32 + (i >> 3)) * SG_LDS_PITCH + (i & 7) * 8]) =
-            set[si].areg[s];
-    }
-  };
+  // the two in-flight load sets are SEPARATE named arrays: a struct array
+  // passed through lambda references was demoted to scratch memory and
+  // every prefetch round-tripped through HBM (ISA: scratch_*_dwordx4 in
+  // the hot loop)
+  // fully scalar load sets (a0..a3 per set): any array or struct here,
+  // even with unrolled constant indices, was demoted to scratch memory by
+  // the compiler and every prefetch round-tripped through HBM
+  uint4 wreg0, wreg1;
+  uint4 a00 = {}, a01 = {}, a02 = {}, a03 = {};
+  uint4 a10 = {}, a11 = {}, a12 = {}, a13 = {};
+  const __hip_bfloat16* wbase =
+      w + (size_t)((n_group << 5) + wrow_st) * K + wchk;
+  const __hip_bfloat16* ab0 = nullptr;
+  const __hip_bfloat16* ab1 = nullptr;
+  const __hip_bfloat16* ab2 = nullptr;
+  const __hip_bfloat16* ab3 = nullptr;
+#define SG_ABASE(s) (a + (size_t)min((tid + (s) * SG_NT) >> 3, M - 1) * K + \
+                     ((tid + (s) * SG_NT) & 7) * 8)
+  ab0 = SG_ABASE(0);
+  if (A_PIECES > 1) ab1 = SG_ABASE(1);
+  if (A_PIECES > 2) ab2 = SG_ABASE(2);
+  if (A_PIECES > 3) ab3 = SG_ABASE(3);
+  static_assert(A_PIECES <= 4, "A_PIECES grew past the scalar sets");
+
+#define SG_LD(p, kb) (*reinterpret_cast<const uint4*>((p) + (kb)))
+#define SG_ISSUE(step, wreg, a0, a1, a2, a3)                                \
+  do {                                                                      \
+    const int kb = k0 + min(step, nsteps - 1) * SG_KSTEP;                   \
+    wreg = SG_LD(wbase, kb);                                                \
+    a0 = SG_LD(ab0, kb);                                                    \
+    if (A_PIECES > 1) a1 = SG_LD(ab1, kb);                                  \
+    if (A_PIECES > 2) a2 = SG_LD(ab2, kb);                                  \
+    if (A_PIECES > 3) a3 = SG_LD(ab3, kb);                                  \
+  } while (0)
+
+  // LDS store address for A piece s (rows past A_ROWS never occur: the
+  // last partial piece only exists when A_PIECES*SG_NT > A_ROWS*8, and
+  // then piece s covers i = tid + s*SG_NT < A_ROWS*8 by construction of
+  // A_PIECES for the supported M_TILES set)
+#define SG_AST(st, s) reinterpret_cast<uint4*>(                             \
+    &st[(32 + ((tid + (s) * SG_NT) >> 3)) * SG_LDS_PITCH +                  \
+        ((tid + (s) * SG_NT) & 7) * 8])
+#define SG_WRITE(st, wreg, a0, a1, a2, a3)                                  \
+  do {                                                                      \
+    *reinterpret_cast<uint4*>(&st[wrow_st * SG_LDS_PITCH + wchk]) = wreg;   \
+    if (tid < A_ROWS * 8) *SG_AST(st, 0) = a0;                              \
+    if (A_PIECES > 1) *SG_AST(st, 1) = a1;                                  \
+    if (A_PIECES > 2) *SG_AST(st, 2) = a2;                                  \
+    if (A_PIECES > 3) *SG_AST(st, 3) = a3;                                  \
+  } while (0)
 
   f32x4 acc[M_TILES];
 #pragma unroll
   for (int mt = 0; mt < M_TILES; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
 
-  // pipeline prologue: tiles 0,1 in flight; tile 0 lands in buffer 0
-  issue_loads(0, 0);
-  issue_loads(1, 1);
-  write_tile(0, 0);
-  issue_loads(2, 0);
-
-  for (int step = 0; step < nsteps; ++step) {
-    __syncthreads();     // buf[step&1] holds tile `step` for every wave
-    const int kf = k_slice * 32 + gslice * 8;
-    const __hip_bfloat16* st = lds.stage[step & 1];
+  const int kf = k_slice * 32 + gslice * 8;
+  auto compute = [&](const __hip_bfloat16* st) {
     const gg_bf8 wf = lds_frag(
         &st[((n_tile << 4) + gl) * SG_LDS_PITCH + kf]);
 #pragma unroll
@@ -129,12 +148,36 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
       acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, wf, acc[mt],
                                                         0, 0, 0);
     }
-    // tile step+1 was issued 2 steps ago -> its vmcnt wait is covered;
-    // its buffer's previous tile (step-1) was fully read before the
-    // barrier above, so no second barrier is needed
-    write_tile((step + 1) & 1, (step + 1) & 1);
-    issue_loads(step + 3, (step + 1) & 1);
+  };
+
+  __hip_bfloat16* buf0 = lds.stage[0];
+  __hip_bfloat16* buf1 = lds.stage[1];
+
+  // pipeline prologue: tiles 0,1 in flight; tile 0 lands in buffer 0
+  SG_ISSUE(0, wreg0, a00, a01, a02, a03);
+  SG_ISSUE(1, wreg1, a10, a11, a12, a13);
+  SG_WRITE(buf0, wreg0, a00, a01, a02, a03);
+  SG_ISSUE(2, wreg0, a00, a01, a02, a03);
+
+  // nsteps is even (K % (SPLITS * 2*SG_KSTEP) == 0, enforced host-side);
+  // tile t+1 was issued 2 steps ago so its vmcnt wait is covered, and a
+  // buffer's previous tile was fully read before the barrier, so each
+  // step needs exactly one barrier
+  for (int step = 0; step < nsteps; step += 2) {
+    __syncthreads();
+    compute(buf0);                                 // tile step
+    SG_WRITE(buf1, wreg1, a10, a11, a12, a13);     // tile step+1
+    SG_ISSUE(step + 3, wreg1, a10, a11, a12, a13);
+    __syncthreads();
+    compute(buf1);                                 // tile step+1
+    SG_WRITE(buf0, wreg0, a00, a01, a02, a03);     // tile step+2
+    SG_ISSUE(step + 4, wreg0, a00, a01, a02, a03);
   }
+#undef SG_ISSUE
+#undef SG_WRITE
+#undef SG_ABASE
+#undef SG_LD
+#undef SG_AST
 
   // ---- cross-slice reduction through LDS (buffer re-use) ----------------
   __syncthreads();
