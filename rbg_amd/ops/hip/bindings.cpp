@@ -182,14 +182,17 @@ bool skinny_gemm_supported(int64_t M, int64_t N, int64_t K) {
   return M >= 1 && M <= 128 && (N % 32) == 0 && (K % 256) == 0;
 }
 
-torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w) {
+torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w,
+                          int64_t force_splits = 0) {
   check_bf16_contig(a, "a");
   check_bf16_contig(w, "w");
   const int64_t M = a.size(0), K = a.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
   TORCH_CHECK(skinny_gemm_supported(M, N, K), "unsupported skinny shape");
   auto c = torch::empty({M, N}, a.options());
-  const int splits = pick_gemm_splits((int)N, (int)K);
+  int splits = force_splits > 0 ? (int)force_splits
+                                : pick_gemm_splits((int)N, (int)K);
+  if (K % (splits * 256) != 0) splits = pick_gemm_splits((int)N, (int)K);
   torch::Tensor ws;
   void* wsp = nullptr;
   if (splits > 1) {
@@ -212,5 +215,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attention", &decode_attention, "paged flash-decode");
   m.def("prefill_attention", &prefill_attention, "varlen causal flash prefill");
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe");
-  m.def("skinny_gemm", &skinny_gemm, "decode-shape GEMM (M<=128)");
+  m.def("skinny_gemm", &skinny_gemm, "decode-shape GEMM (M<=128)",
+        py::arg("a"), py::arg("w"), py::arg("force_splits") = 0);
 }

@@ -41,8 +41,12 @@ def main():
         for name, (k, n) in shapes.items():
             x = torch.randn(M, k, dtype=torch.bfloat16, device=dev)
             w = torch.randn(n, k, dtype=torch.bfloat16, device=dev)
-            for impl, fn in (("lib", lambda: torch.nn.functional.linear(x, w)),
-                             ("skinny", lambda: ops._hip.skinny_gemm(x, w))):
+            cands = [("lib", lambda: torch.nn.functional.linear(x, w))]
+            for s in (1, 2, 4, 8):
+                if k % (s * 256) == 0:
+                    cands.append((f"s{s}",
+                                  lambda s=s: ops._hip.skinny_gemm(x, w, s)))
+            for impl, fn in cands:
                 ms = timeit(fn)
                 key = f"{name}/{impl}"
                 best[key] = min(best.get(key, 1e9), ms)
