@@ -43,10 +43,15 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     hand-written split-K streaming kernel (hip/skinny_gemm.hip) replaces
     hipBLASLt, which leaves 2-5x on narrow-N skinny shapes
     (profiles/gemm_ab_b128.json).  Prefill-sized M goes to the library."""
+    # measured policy (profiles/sg_micro7_b*.json): the custom kernel wins
+    # on narrow-N shapes (o/down, N<=4096) at any decode batch, on mid-N
+    # (qkv, 6144) only up to M=64, and never on wide-N (gate_up 28672,
+    # where hipBLASLt already streams at ~6 TB/s)
     if (SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and
             x.dtype == torch.bfloat16 and x.shape[0] <= 128 and
-            w.shape[0] % 32 == 0 and w.shape[0] <= 8192 and
-            x.shape[1] % 256 == 0):
+            w.shape[0] % 32 == 0 and x.shape[1] % 512 == 0 and
+            (w.shape[0] <= 4608 or
+             (x.shape[0] <= 64 and w.shape[0] <= 8192))):
         return _require_hip().skinny_gemm(
             x if x.is_contiguous() else x.contiguous(), w)
     return torch.nn.functional.linear(x, w)
