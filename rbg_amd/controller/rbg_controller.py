@@ -128,7 +128,7 @@ class RoleBasedGroupController:
                 # all-reduce communicator separate from the lockstep group
                 pp = max(int(engines[0][1].args.get("pp", 1) or 1), 1)
                 if pp > 1 and len(inst_ranks) % pp == 0 and \
-                        len(inst_ranks) > pp:
+                        len(inst_ranks) >= pp:
                     tp_deg = len(inst_ranks) // pp
                     for s in range(pp):
                         subgroups.append(

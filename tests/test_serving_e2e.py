@@ -233,3 +233,42 @@ def test_tp_pd_disaggregated_serving(mgr):
                      timeout=240)
     assert len(res["tokens"]) == 5, res
     assert res["tokens"] == _local_reference_tokens(prompt, 5, model="tiny-tp")
+
+
+@pytest.mark.timeout(420)
+def test_pp2_leader_worker_serving(mgr):
+    """leaderWorker pattern with engine arg pp=2 (pure pipeline parallel,
+    gloo on CPU): the controller's comm plan emits per-stage subgroups;
+    stage 0 embeds + first half of layers, stage 1 finishes and samples.
+    Per-layer weight seeding makes output exactly equal the single-process
+    reference."""
+    from rbg_amd.api.types import LeaderWorkerPattern
+    args = dict(ENGINE_ARGS, mode="colocated", model="tiny",
+                cpu_model="tiny", tp_from_env=True, tp_backend="gloo",
+                pp=2)
+    role = RoleSpec(
+        name="worker", replicas=1, dependencies=["router"],
+        pattern=C.PATTERN_LEADER_WORKER,
+        leader_worker_pattern=LeaderWorkerPattern(size=2),
+        template=EngineTemplate(engines=[EngineSpec(
+            name="engine", runner="llm-engine", args=args,
+            resources=EngineResources(cpu_only=True))]))
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="pp"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("colocated", {"worker_roles": ["worker"],
+                                      "vocab_size": 500}),
+            role,
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "pp"), timeout=180)
+    assert mgr.wait_for(
+        lambda: (_router_http_port(mgr, "pp") is not None), timeout=30)
+    port = _router_http_port(mgr, "pp")
+    torch.manual_seed(33)
+    prompt = torch.randint(0, 500, (13,)).tolist()
+    res = _http_post(port, "/generate",
+                     {"prompt_tokens": prompt, "max_new_tokens": 5},
+                     timeout=240)
+    assert len(res["tokens"]) == 5
+    assert res["tokens"] == _local_reference_tokens(prompt, 5)
