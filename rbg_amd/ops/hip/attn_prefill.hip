@@ -148,29 +148,41 @@ __global__ __launch_bounds__(NW * 64) void prefill_attn_kernel(
   // makes hipcc branch around each load and drain vmcnt(0) per element
   // (guide §5 trap 4c); out-of-range rows are masked in softmax anyway.
   constexpr int PIECES = KTILE * (HEAD_DIM / 8) / NTHREADS;  // per thread
-  uint4 kreg[PIECES], vreg[PIECES];
+  static_assert(PIECES <= 4, "scalar staging sets cover 4 pieces");
+  // SCALAR staging registers: `uint4 kreg[PIECES]` arrays were demoted to
+  // scratch memory by the compiler (ISA: scratch_store/load_dwordx4 in
+  // the k-tile loop), silently tripling the staged tile's HBM traffic —
+  // the same demotion found while building skinny_gemm.hip
+  uint4 kr0 = {}, kr1 = {}, kr2 = {}, kr3 = {};
+  uint4 vr0 = {}, vr1 = {}, vr2 = {}, vr3 = {};
+#define PF_ROW(s, kt2)                                                      \
+  (((size_t)(kv_start + min((kt2) * KTILE + ((tid + (s) * NTHREADS) >> 4),  \
+                            seq_len - 1)) * num_kv_heads + kvh) *           \
+       HEAD_DIM + ((tid + (s) * NTHREADS) & 15) * 8)
+#define PF_LD(s, kt2, kr, vr)                                               \
+  do {                                                                      \
+    const size_t row = PF_ROW(s, kt2);                                      \
+    kr = *reinterpret_cast<const uint4*>(k + row);                          \
+    vr = *reinterpret_cast<const uint4*>(v + row);                          \
+  } while (0)
   auto issue_loads = [&](int kt2) {
-#pragma unroll
-    for (int s = 0; s < PIECES; ++s) {
-      const int i = tid + s * NTHREADS;
-      const int key = i >> 4;
-      const int chunk = (i & 15) * 8;
-      const int krow = min(kt2 * KTILE + key, seq_len - 1);
-      const size_t row = ((size_t)(kv_start + krow) * num_kv_heads + kvh) *
-                         HEAD_DIM + chunk;
-      kreg[s] = *reinterpret_cast<const uint4*>(k + row);
-      vreg[s] = *reinterpret_cast<const uint4*>(v + row);
-    }
+    PF_LD(0, kt2, kr0, vr0);
+    if (PIECES > 1) PF_LD(1, kt2, kr1, vr1);
+    if (PIECES > 2) PF_LD(2, kt2, kr2, vr2);
+    if (PIECES > 3) PF_LD(3, kt2, kr3, vr3);
   };
+#define PF_ST(s, kr, vr)                                                    \
+  do {                                                                      \
+    const int key = (tid + (s) * NTHREADS) >> 4;                            \
+    const int chunk = ((tid + (s) * NTHREADS) & 15) * 8;                    \
+    *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = kr;                     \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) = vr;          \
+  } while (0)
   auto write_tile = [&]() {
-#pragma unroll
-    for (int s = 0; s < PIECES; ++s) {
-      const int i = tid + s * NTHREADS;
-      const int key = i >> 4;
-      const int chunk = (i & 15) * 8;
-      *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = kreg[s];
-      *reinterpret_cast<uint4*>(&v_img[v_img_off(key, chunk)]) = vreg[s];
-    }
+    PF_ST(0, kr0, vr0);
+    if (PIECES > 1) PF_ST(1, kr1, vr1);
+    if (PIECES > 2) PF_ST(2, kr2, vr2);
+    if (PIECES > 3) PF_ST(3, kr3, vr3);
   };
 
   if (DB) {   // prologue: tile 0 resident before the loop
