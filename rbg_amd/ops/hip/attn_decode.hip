@@ -25,7 +25,10 @@ constexpr float NEG_INF = -1e30f;
 // VARIANT 1: packed-bf16 v_dot2 scores (96 VGPRs, 5 waves/SIMD for QPG<=4).
 // Runtime-selected (RBG_DECODE_VARIANT) for within-probe A/B.
 template <int QPG, int VARIANT>
-__global__ __launch_bounds__(256, (VARIANT >= 1 && QPG <= 4) ? 5 : 2)
+__global__ __launch_bounds__(256,
+    (VARIANT == 3) ? 4 : ((VARIANT >= 1 && QPG <= 4) ? 5 : 2))
+// V3 doubles the in-flight K/V state: 4 waves/SIMD (128-VGPR budget) so
+// the extra pairs stay in registers instead of spilling
 void decode_attn_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
@@ -178,6 +181,26 @@ void decode_attn_kernel(
       off_next = row_offset(base + 32);
       process4(kv, vv, valid);
     }
+  } else if (VARIANT == 3) {
+    // 2-ahead prefetch with SCALAR named register pairs, unrolled by 2.
+    // An earlier 2-ahead used a ring array indexed by (i & 1) — which the
+    // compiler demoted to scratch memory (docs/cdna_lessons.md §1), so it
+    // measured -22%; with named pairs both iterations' loads stay in
+    // flight and the per-iteration vmcnt slack doubles.
+    int base = key_begin + wave * 4;
+    Bf16x8U k0, v0, k1, v1;
+    load_pair(base, k0, v0);          // row_offset clamps out-of-range
+    load_pair(base + 16, k1, v1);
+    for (; base < key_end; base += 32) {
+      const bool valid0 = base + group < key_end;
+      Bf16x8U a = k0, b = v0;
+      load_pair(base + 32, k0, v0);
+      process4(a, b, valid0);
+      const bool valid1 = base + 16 + group < key_end;
+      Bf16x8U c = k1, d = v1;
+      load_pair(base + 48, k1, v1);
+      process4(c, d, valid1);
+    }
   } else {
     int base = key_begin + wave * 4;
     Bf16x8U k_pref, v_pref;
@@ -294,20 +317,24 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                      (const __hip_bfloat16*)val_cache,                        \
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits)
-  const int v = variant < 0 ? 1 : (variant > 2 ? 2 : variant);
-  switch (qpg * 4 + v) {
-    case 4: LAUNCH_QPG(1, 0); break;
-    case 5: LAUNCH_QPG(1, 1); break;
-    case 6: LAUNCH_QPG(1, 2); break;
-    case 8: LAUNCH_QPG(2, 0); break;
-    case 9: LAUNCH_QPG(2, 1); break;
-    case 10: LAUNCH_QPG(2, 2); break;
-    case 16: LAUNCH_QPG(4, 0); break;
-    case 17: LAUNCH_QPG(4, 1); break;
-    case 18: LAUNCH_QPG(4, 2); break;
-    case 32: LAUNCH_QPG(8, 0); break;
-    case 33: LAUNCH_QPG(8, 1); break;
-    case 34: LAUNCH_QPG(8, 2); break;
+  const int v = variant < 0 ? 1 : (variant > 3 ? 3 : variant);
+  switch (qpg * 8 + v) {
+    case 8: LAUNCH_QPG(1, 0); break;
+    case 9: LAUNCH_QPG(1, 1); break;
+    case 10: LAUNCH_QPG(1, 2); break;
+    case 11: LAUNCH_QPG(1, 3); break;
+    case 16: LAUNCH_QPG(2, 0); break;
+    case 17: LAUNCH_QPG(2, 1); break;
+    case 18: LAUNCH_QPG(2, 2); break;
+    case 19: LAUNCH_QPG(2, 3); break;
+    case 32: LAUNCH_QPG(4, 0); break;
+    case 33: LAUNCH_QPG(4, 1); break;
+    case 34: LAUNCH_QPG(4, 2); break;
+    case 35: LAUNCH_QPG(4, 3); break;
+    case 64: LAUNCH_QPG(8, 0); break;
+    case 65: LAUNCH_QPG(8, 1); break;
+    case 66: LAUNCH_QPG(8, 2); break;
+    case 67: LAUNCH_QPG(8, 3); break;
     default: return;   // validated host-side
   }
 #undef LAUNCH_QPG
