@@ -21,11 +21,12 @@ KIND_SCALING_ADAPTER = "RoleBasedGroupScalingAdapter"
 KIND_WARMUP = "RoleBasedGroupWarmup"
 KIND_ENGINE_RUNTIME_PROFILE = "ClusterEngineRuntimeProfile"
 KIND_CONTROLLER_REVISION = "ControllerRevision"
+KIND_EVENT = "Event"
 
 ALL_KINDS = (
     KIND_RBG, KIND_RBG_SET, KIND_ROLE_INSTANCE_SET, KIND_ROLE_INSTANCE,
     KIND_COORDINATED_POLICY, KIND_SCALING_ADAPTER, KIND_WARMUP,
-    KIND_ENGINE_RUNTIME_PROFILE, KIND_CONTROLLER_REVISION,
+    KIND_ENGINE_RUNTIME_PROFILE, KIND_CONTROLLER_REVISION, KIND_EVENT,
 )
 
 # ---- Labels (reference constants/label.go) ---------------------------------

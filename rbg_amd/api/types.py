@@ -519,6 +519,30 @@ class ClusterEngineRuntimeProfile:
 
 
 @dataclass
+class ObjectRef:
+    kind: str = ""
+    name: str = ""
+    namespace: str = "default"
+
+
+@dataclass
+class Event:
+    """K8s-Event analog: controllers record one at every decision point
+    (reference emits recorder.Event at e.g. rolebasedgroup_controller.go:
+    242,296,549); identical repeats dedupe by bumping `count`."""
+    api_version: str = f"{C.API_GROUP}/{C.API_VERSION}"
+    kind: str = C.KIND_EVENT
+    metadata: ObjectMeta = field(default_factory=ObjectMeta)
+    involved_object: ObjectRef = field(default_factory=ObjectRef)
+    type: str = "Normal"            # Normal | Warning
+    reason: str = ""
+    message: str = ""
+    count: int = 1
+    first_timestamp: float = 0.0
+    last_timestamp: float = 0.0
+
+
+@dataclass
 class ControllerRevision:
     api_version: str = f"{C.API_GROUP}/{C.API_VERSION}"
     kind: str = C.KIND_CONTROLLER_REVISION
@@ -537,6 +561,7 @@ KIND_TO_TYPE = {
     C.KIND_WARMUP: RoleBasedGroupWarmup,
     C.KIND_ENGINE_RUNTIME_PROFILE: ClusterEngineRuntimeProfile,
     C.KIND_CONTROLLER_REVISION: ControllerRevision,
+    C.KIND_EVENT: Event,
 }
 
 

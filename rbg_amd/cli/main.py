@@ -64,6 +64,16 @@ def cmd_get(client: BaseClient, args) -> int:
             return 1
         print(yaml.safe_dump(asdict(obj), sort_keys=False))
         return 0
+    if kind == C.KIND_EVENT:
+        import time as _t
+        evs = sorted(client.list(kind, args.namespace),
+                     key=lambda e: e.last_timestamp)
+        rows = [[f"{_t.time() - e.last_timestamp:.0f}s", e.type, e.reason,
+                 f"{e.involved_object.kind}/{e.involved_object.name}",
+                 e.count, e.message[:60]] for e in evs]
+        _print_table(rows, ["AGE", "TYPE", "REASON", "OBJECT", "COUNT",
+                            "MESSAGE"])
+        return 0
     rows = []
     for obj in client.list(kind, args.namespace):
         ready = ""
@@ -171,6 +181,7 @@ _KIND_ALIASES = {
     "coordinatedpolicy": C.KIND_COORDINATED_POLICY,
     "warmup": C.KIND_WARMUP,
     "revision": C.KIND_CONTROLLER_REVISION,
+    "event": C.KIND_EVENT, "events": C.KIND_EVENT, "ev": C.KIND_EVENT,
 }
 
 

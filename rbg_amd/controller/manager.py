@@ -134,13 +134,17 @@ class Manager:
         self.registry = TopologyRegistry(self.opts.run_root + "/discovery")
         self.restarts = RestartRegistry()
 
+        from ..store.events import EventRecorder
+        self.recorder = EventRecorder(self.store)
         self.rbg = RoleBasedGroupController(self.store, self.registry,
                                             self.opts.history_limit,
-                                            ports=self.ports)
+                                            ports=self.ports,
+                                            recorder=self.recorder)
         self.ris = RoleInstanceSetController(self.store)
         self.instance = RoleInstanceController(
             self.store, self.gang, self.runner, self.ports, self.bindings,
-            self.restarts, gang_timeout=self.opts.gang_timeout)
+            self.restarts, gang_timeout=self.opts.gang_timeout,
+            recorder=self.recorder)
         self.adapter = ScalingAdapterController(self.store)
         self.rbgset = RoleBasedGroupSetController(self.store)
         self.warmup = WarmupController(self.store, self.topo.num_gpus)

@@ -76,11 +76,13 @@ def _merge_template(base: EngineTemplate, patch: Dict) -> EngineTemplate:
 
 class RoleBasedGroupController:
     def __init__(self, store: Store, registry: Optional[TopologyRegistry] = None,
-                 history_limit: int = 10, ports=None):
+                 history_limit: int = 10, ports=None, recorder=None):
+        from ..store.events import recorder_or_null
         self.store = store
         self.registry = registry
         self.revisions = RevisionManager(store, history_limit)
         self.ports = ports        # PortAllocator (comm rendezvous ports)
+        self.recorder = recorder_or_null(recorder)
 
     # ------------------------------------------------------------------
 
@@ -165,6 +167,7 @@ class RoleBasedGroupController:
         except ValidationError as e:
             self._set_group_condition(rbg, C.COND_READY, False,
                                       "ValidationFailed", str(e))
+            self.recorder.warning(rbg, "ValidationFailed", str(e))
             return 0.0
         revision = self.revisions.ensure_current(rbg, asdict(rbg.spec))
         statuses = self._construct_role_statuses(rbg)
@@ -471,6 +474,14 @@ class RoleBasedGroupController:
         statuses = self._construct_role_statuses(rbg)
         all_ready = all(self._role_ready(rbg, r, statuses)
                         for r in rbg.spec.roles)
+        was = get_condition(rbg.status.conditions, C.COND_READY)
+        was_ready = was is not None and was.status == "True"
+        if all_ready and not was_ready:
+            self.recorder.normal(rbg, "GroupReady",
+                                 "all roles ready")
+        elif was_ready and not all_ready:
+            self.recorder.warning(rbg, "GroupNotReady",
+                                  "one or more roles lost readiness")
 
         def mutate(cur: RoleBasedGroup):
             cur.status.observed_generation = cur.metadata.generation

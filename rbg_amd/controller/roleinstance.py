@@ -55,7 +55,9 @@ class RoleInstanceController:
                  runner: ProcessRunner, ports: PortAllocator,
                  bindings: GpuBindingStore,
                  restart_registry: Optional[RestartRegistry] = None,
-                 gang_timeout: float = 30.0):
+                 gang_timeout: float = 30.0, recorder=None):
+        from ..store.events import recorder_or_null
+        self.recorder = recorder_or_null(recorder)
         self.store = store
         self.gang = gang
         self.runner = runner
@@ -149,6 +151,9 @@ class RoleInstanceController:
             inst.status.restart_count = tracker.restart_count
             inst.status.last_restart_time = now
             self._set_restarting(inst, True, f"restart #{tracker.restart_count}")
+            self.recorder.warning(
+                inst, "InstanceRestart",
+                f"engine failure; gang recreate #{tracker.restart_count}")
         elif failed_fatal and policy == C.RESTART_POLICY_NONE:
             return 1.0   # leave failed workers visible in status
 
@@ -161,6 +166,7 @@ class RoleInstanceController:
                 self._ensure_gang(inst, rt, desired)
             except GangUnschedulable as e:
                 self._set_condition(inst, C.COND_READY, False, "Unschedulable", str(e))
+                self.recorder.warning(inst, "Unschedulable", str(e))
                 return 1.0
             from ..discovery import component as comp_disc
             deps = comp_disc.parse_depends_on(inst.metadata.annotations)
@@ -188,6 +194,8 @@ class RoleInstanceController:
     # ------------------------------------------------------------------
 
     def _record_recovery(self, inst: RoleInstance, duration: float) -> None:
+        self.recorder.normal(inst, "Recovered",
+                             f"group recovered in {duration:.2f}s")
         def mutate(cur: RoleInstance):
             cur.status.last_recovery_duration = duration
             return cur

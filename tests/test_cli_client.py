@@ -138,3 +138,31 @@ def test_daemon_metrics(daemon):
     assert "rbg_groups 1" in text
     assert 'rbg_group_ready{group="metrics-demo"}' in text
     assert "rbg_free_gpus" in text
+
+
+def test_events_recorded_and_listed(tmp_run_dir):
+    """Controllers record Events at decision points; `rbgctl get events`
+    renders them (reference recorder.Event + kubectl get events analog)."""
+    from rbg_amd.api import constants as C
+    from rbg_amd.cli.main import main as ctl_main
+    from rbg_amd.client.client import InProcessClient
+    from rbg_amd.controller.manager import Manager, ManagerOptions
+    from tests.test_controller_e2e import router_worker_rbg, rbg_ready
+    m = Manager(ManagerOptions(run_root=tmp_run_dir, num_gpus=8,
+                               resync_period=0.1))
+    m.start()
+    try:
+        m.store.create(router_worker_rbg(name="evt"))
+        assert m.wait_for(lambda: rbg_ready(m, "evt"), timeout=60)
+        evs = m.store.list(C.KIND_EVENT)
+        assert any(e.reason == "GroupReady" for e in evs), \
+            [e.reason for e in evs]
+        # dedupe: repeated identical event bumps count, not cardinality
+        m.recorder.normal(("RoleBasedGroup", "evt", "default"), "Ping", "x")
+        m.recorder.normal(("RoleBasedGroup", "evt", "default"), "Ping", "x")
+        pings = [e for e in m.store.list(C.KIND_EVENT) if e.reason == "Ping"]
+        assert len(pings) == 1 and pings[0].count == 2
+        rc = ctl_main(["get", "events"], client=InProcessClient(m.store))
+        assert rc == 0
+    finally:
+        m.stop()
