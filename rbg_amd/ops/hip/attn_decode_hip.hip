@@ -272,6 +272,270 @@ void decode_attn_kernel(
   }
 }
 
+
+// V image offset for the decode PV tr-read path (32-key pair image):
+// identical sub-tiling to attn_prefill.hip's v_img_off at ks == 0
+DEV_INLINE int v_img_off(int key, int dim) {
+  const int q = key >> 3, jj = key & 7;
+  return (dim >> 4) * 512 + (jj >> 2) * 256 + q * 64 + (jj & 3) * 16 +
+         (dim & 15);
+}
+
+
+// ---------------------------------------------------------------------------
+// MFMA-tiled decode (VARIANT 4 dispatch): one wave per sequence, 32-key
+// page-pair tiles on v_mfma_f32_16x16x32_bf16 — the prefill kernel's
+// engine applied to decode.  The dot2 kernel (V1) is latency-bound on its
+// serial cross-lane shuffle chain (a group16 reduction round every 4
+// keys, SQ_WAIT_ANY 68%); here one MFMA chain scores 32 keys and the
+// reduction round runs once per PAIR of pages (8x fewer), with PV
+// through the same 32-k fragment + ds_read_b64_tr_b16 V image as
+// attn_prefill.hip.
+//
+// Layout notes:
+//  * page_size must be 16: a KV page slice [16 tokens][128 d] is 4 KB
+//    contiguous; two pages form one 32-key MFMA step.
+//  * q A-fragment rows = the kv-head's QPG query heads (rows >= QPG
+//    replicate the last head; their outputs are never written).
+//  * S is C-layout: key col = lane&15, q row = 4*(lane>>4)+reg; the two
+//    pages' columns combine lane-locally before ONE group16 max/sum.
+//  * 2-wave workgroups, ~18 KB LDS per wave -> 4 WGs (8 waves) per CU;
+//    MFMA depth inside the wave covers the low waves/SIMD (the same
+//    budget shape as the 8-wave prefill kernel).
+template <int QPG>
+__global__ __launch_bounds__(128, 4) void decode_attn_mfma_kernel(
+    float* __restrict__ partial_o,        // [splits, seqs, QH, D]
+    float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
+    __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
+    const __hip_bfloat16* __restrict__ q, // [seqs, QH, D]
+    const __hip_bfloat16* __restrict__ key_cache,  // [pages, KVH, 16, D]
+    const __hip_bfloat16* __restrict__ val_cache,
+    const int* __restrict__ block_tables, // [seqs, max_pages]
+    const int* __restrict__ context_lens, // [seqs]
+    const float scale, const int num_kv_heads, const int max_pages,
+    const int num_splits, const int num_seqs) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 dm_bf8;
+  const int kvh = blockIdx.x;
+  const int wave = threadIdx.x >> 6;
+  const int seq = blockIdx.y * 2 + wave;
+  const int split = blockIdx.z;
+  const int lane = threadIdx.x & 63;
+  const int gl = lane & 15;
+  const int gslice = lane >> 4;
+  const int num_q_heads = num_kv_heads * QPG;
+
+  // per-wave LDS slices (no cross-wave sharing, no barriers)
+  constexpr int KP = 132;                        // k_lds pitch (bf16)
+  __shared__ __hip_bfloat16 k_lds_all[2][32 * KP];
+  __shared__ __hip_bfloat16 v_img_all[2][8192];  // 32-key tr image
+  __shared__ __hip_bfloat16 p_lds_all[2][16 * 40];
+  __hip_bfloat16* k_lds = k_lds_all[wave];
+  __hip_bfloat16* v_img = v_img_all[wave];
+  __hip_bfloat16* p_lds = p_lds_all[wave];
+  // zero V image once: elements past the last valid key are never
+  // written (odd page counts / masked tails) and 0 x P(=0) must be 0,
+  // never garbage-NaN x 0
+  {
+    const uint4 z = {0, 0, 0, 0};
+    for (int i = lane; i < 8192 / 8; i += 64)
+      *reinterpret_cast<uint4*>(&v_img[i * 8]) = z;
+  }
+  if (seq >= num_seqs) return;
+  const int ctx = context_lens[seq];
+
+  const int chunk = 16;
+  const int nchunks = (ctx + chunk - 1) / chunk;
+  const int per_split = (nchunks + num_splits - 1) / num_splits;
+  const int key_begin = split * per_split * chunk;
+  const int key_end = min(ctx, (split + 1) * per_split * chunk);
+  const int pg_begin = key_begin >> 4;
+  const int pg_end = (key_end + 15) >> 4;
+
+  // q fragments: row = head (clamped), 4 k-steps of 32 dims
+  dm_bf8 qa[4];
+  {
+    const int h = min(gl, QPG - 1);
+    const __hip_bfloat16* qrow =
+        q + ((size_t)seq * num_q_heads + kvh * QPG + h) * HEAD_DIM;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      union { uint4 u; dm_bf8 v; } cvt;
+      cvt.u = *reinterpret_cast<const uint4*>(qrow + ks * 32 + gslice * 8);
+      qa[ks] = cvt.v;
+    }
+  }
+
+  const int* btab = block_tables + (size_t)seq * max_pages;
+  // staging: lane covers K/V key row = lane>>1 (32 rows), 64-elem chunk
+  // = (lane&1)*64; 2 lanes x 128 B = one contiguous page row
+  const int srow = lane >> 1, schunk = (lane & 1) * 64;
+  const int pg_last = pg_end - 1;
+  auto page_base = [&](int pg, int local) -> size_t {
+    // pair page `local` (0/1): clamp to the last real page
+    const int p = btab[min(pg + local, pg_last)];
+    return (((size_t)p * num_kv_heads + kvh) * 16 + (srow & 15)) * HEAD_DIM +
+           schunk;
+  };
+  const int myloc = srow >> 4;     // which page of the pair this lane stages
+
+  // SCALAR staging registers (docs/cdna_lessons.md §1: local arrays are
+  // demoted to scratch memory, round-tripping every staged page via HBM)
+  uint4 ka0, ka1, ka2, ka3, va0, va1, va2, va3;
+#define DM_ISSUE(pg)                                                        \
+  do {                                                                      \
+    const size_t off = page_base(pg, myloc);                                \
+    ka0 = *reinterpret_cast<const uint4*>(key_cache + off);                 \
+    ka1 = *reinterpret_cast<const uint4*>(key_cache + off + 8);             \
+    ka2 = *reinterpret_cast<const uint4*>(key_cache + off + 16);            \
+    ka3 = *reinterpret_cast<const uint4*>(key_cache + off + 24);            \
+    va0 = *reinterpret_cast<const uint4*>(val_cache + off);                 \
+    va1 = *reinterpret_cast<const uint4*>(val_cache + off + 8);             \
+    va2 = *reinterpret_cast<const uint4*>(val_cache + off + 16);            \
+    va3 = *reinterpret_cast<const uint4*>(val_cache + off + 24);            \
+  } while (0)
+#define DM_WRITE()                                                          \
+  do {                                                                      \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk]) = ka0;            \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 8]) = ka1;        \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 16]) = ka2;       \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 24]) = ka3;       \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk)]) = va0;       \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 8)]) = va1;   \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 16)]) = va2;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 24)]) = va3;  \
+  } while (0)
+
+  float m[4], l[4];
+  f32x4 acc_o[8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = NEG_INF; l[r] = 0.f; }
+#pragma unroll
+  for (int dt = 0; dt < 8; ++dt) acc_o[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  DM_ISSUE(pg_begin);
+
+  for (int pg = pg_begin; pg < pg_end; pg += 2) {
+    DM_WRITE();
+    DM_ISSUE(pg + 2);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // S = q K^T over the pair's 32 keys (B-frag: key row = gl / gl+16)
+    f32x4 sA = {0.f, 0.f, 0.f, 0.f}, sB = sA;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      union { uint4 u; dm_bf8 v; } kfA, kfB;
+      kfA.u = *reinterpret_cast<const uint4*>(
+          &k_lds[gl * KP + ks * 32 + gslice * 8]);
+      kfB.u = *reinterpret_cast<const uint4*>(
+          &k_lds[(16 + gl) * KP + ks * 32 + gslice * 8]);
+      sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[ks], kfA.v, sA, 0, 0, 0);
+      sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[ks], kfB.v, sB, 0, 0, 0);
+    }
+
+    // online softmax: lane holds rows 4*gslice+r for key cols
+    // (pg*16+gl, pg*16+16+gl); ONE group16 round covers both pages
+    const int keyA = pg * 16 + gl;
+    const int keyB = keyA + 16;
+    const bool vA = keyA >= key_begin && keyA < key_end;
+    const bool vB = keyB >= key_begin && keyB < key_end;
+    float pA[4], pB[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float a = vA ? sA[r] * scale : NEG_INF;
+      const float b = vB ? sB[r] * scale : NEG_INF;
+      const float tm = group16_max(fmaxf(a, b));
+      const float m_new = fmaxf(m[r], tm);
+      const float alpha = __expf(m[r] - m_new);
+      pA[r] = vA ? __expf(a - m_new) : 0.f;
+      pB[r] = vB ? __expf(b - m_new) : 0.f;
+      l[r] = l[r] * alpha + group16_sum(pA[r] + pB[r]);
+      m[r] = m_new;
+      acc_o[0][r] *= alpha; acc_o[1][r] *= alpha;
+      acc_o[2][r] *= alpha; acc_o[3][r] *= alpha;
+      acc_o[4][r] *= alpha; acc_o[5][r] *= alpha;
+      acc_o[6][r] *= alpha; acc_o[7][r] *= alpha;
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      p_lds[(4 * gslice + r) * 40 + gl] = f2bf(pA[r]);
+      p_lds[(4 * gslice + r) * 40 + 16 + gl] = f2bf(pB[r]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // O += P V via the 32-k fragment + tr-read V image (attn_prefill PV)
+    {
+      union { uint4 u; dm_bf8 v; } paf;
+      const int pk = gslice * 8;
+      paf.u.x = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk]);
+      paf.u.y = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 2]);
+      paf.u.z = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 4]);
+      paf.u.w = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 6]);
+      const unsigned vaddr = (unsigned)(unsigned long long)(
+          &v_img[gslice * 64 + gl * 4]);
+      unsigned long long vlo[8], vhi[8];
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %16\n\t"
+          "ds_read_b64_tr_b16 %1, %16 offset:512\n\t"
+          "ds_read_b64_tr_b16 %2, %16 offset:1024\n\t"
+          "ds_read_b64_tr_b16 %3, %16 offset:1536\n\t"
+          "ds_read_b64_tr_b16 %4, %16 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %5, %16 offset:2560\n\t"
+          "ds_read_b64_tr_b16 %6, %16 offset:3072\n\t"
+          "ds_read_b64_tr_b16 %7, %16 offset:3584\n\t"
+          "ds_read_b64_tr_b16 %8, %16 offset:4096\n\t"
+          "ds_read_b64_tr_b16 %9, %16 offset:4608\n\t"
+          "ds_read_b64_tr_b16 %10, %16 offset:5120\n\t"
+          "ds_read_b64_tr_b16 %11, %16 offset:5632\n\t"
+          "ds_read_b64_tr_b16 %12, %16 offset:6144\n\t"
+          "ds_read_b64_tr_b16 %13, %16 offset:6656\n\t"
+          "ds_read_b64_tr_b16 %14, %16 offset:7168\n\t"
+          "ds_read_b64_tr_b16 %15, %16 offset:7680\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(vlo[0]), "=&v"(vhi[0]), "=&v"(vlo[1]), "=&v"(vhi[1]),
+            "=&v"(vlo[2]), "=&v"(vhi[2]), "=&v"(vlo[3]), "=&v"(vhi[3]),
+            "=&v"(vlo[4]), "=&v"(vhi[4]), "=&v"(vlo[5]), "=&v"(vhi[5]),
+            "=&v"(vlo[6]), "=&v"(vhi[6]), "=&v"(vlo[7]), "=&v"(vhi[7])
+          : "v"(vaddr)
+          : "memory");
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) {
+        union { struct { unsigned long long lo, hi; } u; dm_bf8 vf2; } vf;
+        vf.u.lo = vlo[dt];
+        vf.u.hi = vhi[dt];
+        acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            paf.v, vf.vf2, acc_o[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: lane holds O rows 4*gslice+r at dim cols dt*16+gl
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = 4 * gslice + r;
+    if (row >= QPG) continue;
+    const int qh = kvh * QPG + row;
+    if (num_splits == 1) {
+      const float inv_l = l[r] > 0.f ? 1.f / l[r] : 0.f;
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt)
+        out[((size_t)seq * num_q_heads + qh) * HEAD_DIM + dt * 16 + gl] =
+            f2bf(acc_o[dt][r] * inv_l);
+    } else {
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt)
+        partial_o[(((size_t)split * num_seqs + seq) * num_q_heads + qh) *
+                      HEAD_DIM + dt * 16 + gl] = acc_o[dt][r];
+      if (gl == 0) {
+        float* ml = partial_ml +
+            (((size_t)split * num_seqs + seq) * num_q_heads + qh) * 2;
+        ml[0] = m[r];
+        ml[1] = l[r];
+      }
+    }
+  }
+}
+
 // Merge split partials: one block per (seq, q_head).
 __global__ void decode_combine_kernel(
     __hip_bfloat16* __restrict__ out,       // [seqs, QH, D]
@@ -318,6 +582,30 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                      (const __hip_bfloat16*)val_cache,                        \
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits)
+  if (variant == 4 && page_size == 16) {
+    dim3 mgrid(num_kv_heads, (num_seqs + 1) / 2, num_splits), mblock(128);
+#define LAUNCH_MFMA(QPG)                                                     \
+    hipLaunchKernelGGL((decode_attn_mfma_kernel<QPG>), mgrid, mblock, 0,     \
+                       stream, (float*)partial_o, (float*)partial_ml,        \
+                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,       \
+                       (const __hip_bfloat16*)key_cache,                     \
+                       (const __hip_bfloat16*)val_cache,                     \
+                       (const int*)block_tables, (const int*)context_lens,   \
+                       scale, num_kv_heads, max_pages, num_splits, num_seqs)
+    if (qpg == 1) LAUNCH_MFMA(1);
+    else if (qpg == 2) LAUNCH_MFMA(2);
+    else if (qpg == 4) LAUNCH_MFMA(4);
+    else if (qpg == 8) LAUNCH_MFMA(8);
+    else return;
+#undef LAUNCH_MFMA
+    if (num_splits > 1) {
+      dim3 cgrid(num_seqs, num_q_heads), cblock(HEAD_DIM);
+      hipLaunchKernelGGL(decode_combine_kernel, cgrid, cblock, 0, stream,
+                         (__hip_bfloat16*)out, (const float*)partial_o,
+                         (const float*)partial_ml, num_splits, num_q_heads);
+    }
+    return;
+  }
   const int v = variant < 0 ? 1 : (variant > 3 ? 3 : variant);
   switch (qpg * 8 + v) {
     case 8: LAUNCH_QPG(1, 0); break;

@@ -201,3 +201,35 @@ def test_skinny_gemm_matches_fp32():
             tol = ref.abs().max().item() * 3e-2 + 0.02
             assert (got - ref).abs().max().item() < tol, \
                 (M, k, n, (got - ref).abs().max().item(), tol)
+
+
+@pytest.mark.gpu
+def test_decode_attention_mfma_variant4():
+    """MFMA-tiled decode (variant 4) vs fp32 reference, including ragged
+    contexts, multiple splits, and odd page tails."""
+    import torch
+    from rbg_amd import ops
+    from rbg_amd.ops import reference
+    dev = torch.device("cuda:0")
+    torch.manual_seed(3)
+    B, KVH, QH, D, page = 5, 8, 32, 128, 16
+    ctxs = [1, 17, 33, 256, 2048]
+    max_pages = (max(ctxs) + page - 1) // page
+    npages = sum((c + page - 1) // page for c in ctxs) + 1
+    kc = torch.randn(npages, KVH, page, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(B, max_pages, dtype=torch.int32, device=dev)
+    nxt = 1
+    for i, c in enumerate(ctxs):
+        n = (c + page - 1) // page
+        bt[i, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    ctx = torch.tensor(ctxs, dtype=torch.int32, device=dev)
+    q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev)
+    ref = reference.decode_attention(
+        q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(), ctx.cpu(), 0.088).to(dev)
+    for splits in (1, 2, 4):
+        got = ops._hip.decode_attention(q, kc, vc, bt, ctx, 0.088, splits, 4)
+        assert torch.allclose(got.float(), ref.float(), atol=2e-2,
+                              rtol=2e-2), \
+            (splits, (got.float() - ref.float()).abs().max().item())
