@@ -303,7 +303,7 @@ DEV_INLINE int v_img_off(int key, int dim) {
 //    MFMA depth inside the wave covers the low waves/SIMD (the same
 //    budget shape as the 8-wave prefill kernel).
 template <int QPG>
-__global__ __launch_bounds__(128, 4) void decode_attn_mfma_kernel(
+__global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
     __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
@@ -324,22 +324,18 @@ __global__ __launch_bounds__(128, 4) void decode_attn_mfma_kernel(
   const int gslice = lane >> 4;
   const int num_q_heads = num_kv_heads * QPG;
 
-  // per-wave LDS slices (no cross-wave sharing, no barriers)
-  constexpr int KP = 132;                        // k_lds pitch (bf16)
-  __shared__ __hip_bfloat16 k_lds_all[2][32 * KP];
-  __shared__ __hip_bfloat16 v_img_all[2][8192];  // 32-key tr image
+  // per-wave LDS (no cross-wave sharing, no barriers).  K and V SHARE one
+  // buffer: the pair's K panel serves the S MFMAs, then the V image
+  // overwrites it for PV (same-wave DS ops retire in order, so the
+  // write-after-read needs no fence) — halving LDS doubles resident
+  // waves.  V is fully rewritten every pair (tail pages clamp to real
+  // pages and are masked through P=0), so no zero-init is needed.
+  constexpr int KP = 132;                        // k panel pitch (bf16)
+  __shared__ __hip_bfloat16 kv_all[2][32 * KP];  // K panel / V tr image
   __shared__ __hip_bfloat16 p_lds_all[2][16 * 40];
-  __hip_bfloat16* k_lds = k_lds_all[wave];
-  __hip_bfloat16* v_img = v_img_all[wave];
+  __hip_bfloat16* k_lds = kv_all[wave];
+  __hip_bfloat16* v_img = kv_all[wave];
   __hip_bfloat16* p_lds = p_lds_all[wave];
-  // zero V image once: elements past the last valid key are never
-  // written (odd page counts / masked tails) and 0 x P(=0) must be 0,
-  // never garbage-NaN x 0
-  {
-    const uint4 z = {0, 0, 0, 0};
-    for (int i = lane; i < 8192 / 8; i += 64)
-      *reinterpret_cast<uint4*>(&v_img[i * 8]) = z;
-  }
   if (seq >= num_seqs) return;
   const int ctx = context_lens[seq];
 
@@ -379,30 +375,46 @@ __global__ __launch_bounds__(128, 4) void decode_attn_mfma_kernel(
   const int myloc = srow >> 4;     // which page of the pair this lane stages
 
   // SCALAR staging registers (docs/cdna_lessons.md §1: local arrays are
-  // demoted to scratch memory, round-tripping every staged page via HBM)
-  uint4 ka0, ka1, ka2, ka3, va0, va1, va2, va3;
-#define DM_ISSUE(pg)                                                        \
+  // demoted to scratch memory, round-tripping every staged page via HBM).
+  // Each lane owns HALF a page row = 64 elements = 8 x 16 B pieces.
+  uint4 ka0, ka1, ka2, ka3, ka4, ka5, ka6, ka7;
+  uint4 va0, va1, va2, va3, va4, va5, va6, va7;
+#define DM_LOAD8(base, r0, r1, r2, r3, r4, r5, r6, r7, off)                 \
   do {                                                                      \
-    const size_t off = page_base(pg, myloc);                                \
-    ka0 = *reinterpret_cast<const uint4*>(key_cache + off);                 \
-    ka1 = *reinterpret_cast<const uint4*>(key_cache + off + 8);             \
-    ka2 = *reinterpret_cast<const uint4*>(key_cache + off + 16);            \
-    ka3 = *reinterpret_cast<const uint4*>(key_cache + off + 24);            \
-    va0 = *reinterpret_cast<const uint4*>(val_cache + off);                 \
-    va1 = *reinterpret_cast<const uint4*>(val_cache + off + 8);             \
-    va2 = *reinterpret_cast<const uint4*>(val_cache + off + 16);            \
-    va3 = *reinterpret_cast<const uint4*>(val_cache + off + 24);            \
+    r0 = *reinterpret_cast<const uint4*>(base + off);                       \
+    r1 = *reinterpret_cast<const uint4*>(base + off + 8);                   \
+    r2 = *reinterpret_cast<const uint4*>(base + off + 16);                  \
+    r3 = *reinterpret_cast<const uint4*>(base + off + 24);                  \
+    r4 = *reinterpret_cast<const uint4*>(base + off + 32);                  \
+    r5 = *reinterpret_cast<const uint4*>(base + off + 40);                  \
+    r6 = *reinterpret_cast<const uint4*>(base + off + 48);                  \
+    r7 = *reinterpret_cast<const uint4*>(base + off + 56);                  \
   } while (0)
-#define DM_WRITE()                                                          \
+#define DM_ISSUE_K(pg) DM_LOAD8(key_cache, ka0, ka1, ka2, ka3, ka4, ka5,    \
+                                ka6, ka7, page_base(pg, myloc))
+#define DM_ISSUE_V(pg) DM_LOAD8(val_cache, va0, va1, va2, va3, va4, va5,    \
+                                va6, va7, page_base(pg, myloc))
+#define DM_WRITE_K()                                                        \
   do {                                                                      \
     *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk]) = ka0;            \
     *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 8]) = ka1;        \
     *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 16]) = ka2;       \
     *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 24]) = ka3;       \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 32]) = ka4;       \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 40]) = ka5;       \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 48]) = ka6;       \
+    *reinterpret_cast<uint4*>(&k_lds[srow * KP + schunk + 56]) = ka7;       \
+  } while (0)
+#define DM_WRITE_V()                                                        \
+  do {                                                                      \
     *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk)]) = va0;       \
     *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 8)]) = va1;   \
     *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 16)]) = va2;  \
     *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 24)]) = va3;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 32)]) = va4;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 40)]) = va5;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 48)]) = va6;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 56)]) = va7;  \
   } while (0)
 
   float m[4], l[4];
@@ -412,11 +424,11 @@ __global__ __launch_bounds__(128, 4) void decode_attn_mfma_kernel(
 #pragma unroll
   for (int dt = 0; dt < 8; ++dt) acc_o[dt] = {0.f, 0.f, 0.f, 0.f};
 
-  DM_ISSUE(pg_begin);
+  DM_ISSUE_K(pg_begin);
 
   for (int pg = pg_begin; pg < pg_end; pg += 2) {
-    DM_WRITE();
-    DM_ISSUE(pg + 2);
+    DM_WRITE_K();
+    DM_ISSUE_V(pg);          // V latency hides under S + softmax
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // S = q K^T over the pair's 32 keys (B-frag: key row = gl / gl+16)
@@ -460,6 +472,10 @@ __global__ __launch_bounds__(128, 4) void decode_attn_mfma_kernel(
       p_lds[(4 * gslice + r) * 40 + gl] = f2bf(pA[r]);
       p_lds[(4 * gslice + r) * 40 + 16 + gl] = f2bf(pB[r]);
     }
+    // V overwrites the K panel: same-wave DS ordering protects the S
+    // fragment reads issued above
+    DM_WRITE_V();
+    DM_ISSUE_K(pg + 2);      // next pair's K hides under PV
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // O += P V via the 32-k fragment + tr-read V image (attn_prefill PV)
