@@ -101,7 +101,7 @@ def pick_decode_splits(num_seqs: int, num_kv_heads: int,
     return int(min(want, by_ctx, 16))
 
 
-DECODE_VARIANT = int(__import__("os").environ.get("RBG_DECODE_VARIANT", "1"))
+DECODE_VARIANT = int(__import__("os").environ.get("RBG_DECODE_VARIANT", "4"))  # MFMA page-pair decode: 5.6 TB/s vs 4.0 dot2 (B128, within-probe)
 PREFILL_SWZ = int(__import__("os").environ.get("RBG_PREFILL_SWZ", "6"))  # 8-wave + K/V double-buffer (305 TF/s vs 252 @ swz=4, within-probe)
 
 

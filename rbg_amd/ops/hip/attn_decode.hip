@@ -621,7 +621,8 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
     }
     return;
   }
-  const int v = variant < 0 ? 1 : (variant > 3 ? 3 : variant);
+  // variant 4 with a non-16 page size falls back to the dot2 kernel
+  const int v = variant < 0 ? 1 : (variant > 3 ? 1 : variant);
   switch (qpg * 8 + v) {
     case 8: LAUNCH_QPG(1, 0); break;
     case 9: LAUNCH_QPG(1, 1); break;
