@@ -88,16 +88,23 @@ def rope_store_kv(q, k, v, key_cache, value_cache, cos_sin, positions,
 
 
 def pick_decode_splits(num_seqs: int, num_kv_heads: int,
-                       max_context: int) -> int:
-    """Fill the chip: MI355X has 256 CUs / 8 XCDs; measured occupancy keeps
-    improving up to ~1024 workgroups (profiles/decode_breakdown: splits 1->2
-    at batch 64 x 8 kv-heads = 1.86 -> 2.97 TB/s), so target >=1024 unless
-    the context is too short to split into 256-key chunks."""
-    base = num_seqs * num_kv_heads
+                       max_context: int, variant: int = -1) -> int:
+    """Fill the chip: MI355X has 256 CUs / 8 XCDs; measured optimum is
+    ~1024 workgroups for BOTH kernels (profiles/decode_breakdown r59
+    sweep: V4 B128 peaks at splits=2 = 1024 WGs, B64 at splits=4 = 1024).
+    The MFMA kernel packs 2 sequences per WG, so its base WG count is
+    half the dot2 kernel's at equal splits."""
+    if variant < 0:
+        variant = DECODE_VARIANT
+    if variant == 4:
+        base = num_kv_heads * ((num_seqs + 1) // 2)
+        by_ctx = max(1, max_context // 32)      # >= one page pair per split
+    else:
+        base = num_seqs * num_kv_heads
+        by_ctx = max(1, max_context // 256)
     if base >= 1024:
         return 1
     want = max(1, 1024 // max(base, 1))
-    by_ctx = max(1, max_context // 256)
     return int(min(want, by_ctx, 16))
 
 
