@@ -23,6 +23,7 @@ class ModelConfig:
     rms_eps: float = 1e-5
     max_position: int = 16384
     dtype: str = "bfloat16"
+    qk_norm: bool = False      # Qwen3-style per-head RMSNorm on q/k
 
     @property
     def q_size(self) -> int:
@@ -43,6 +44,28 @@ class ModelConfig:
             return ModelConfig(name="llama-3-70b", hidden_size=8192,
                                intermediate_size=28672, num_layers=80,
                                num_heads=64, num_kv_heads=8, head_dim=128)
+        if name in ("qwen3-32b", "qwen3", "32b"):
+            # the reference KEP's benchmark model:
+            # Qwen3-32B shapes with per-head q/k RMSNorm
+            return ModelConfig(name="qwen3-32b", hidden_size=5120,
+                               intermediate_size=25600, num_layers=64,
+                               num_heads=64, num_kv_heads=8, head_dim=128,
+                               vocab_size=151936, rope_theta=1e6,
+                               rms_eps=1e-6, max_position=40960,
+                               qk_norm=True)
+        if name in ("qwen3-8b",):
+            return ModelConfig(name="qwen3-8b", hidden_size=4096,
+                               intermediate_size=12288, num_layers=36,
+                               num_heads=32, num_kv_heads=8, head_dim=128,
+                               vocab_size=151936, rope_theta=1e6,
+                               rms_eps=1e-6, max_position=40960,
+                               qk_norm=True)
+        if name == "tiny-qwen":   # qk-norm CPU test shape
+            return ModelConfig(name="tiny-qwen", hidden_size=256,
+                               intermediate_size=512, num_layers=2,
+                               num_heads=2, num_kv_heads=1, head_dim=128,
+                               vocab_size=512, max_position=2048,
+                               qk_norm=True)
         if name == "tiny":
             return ModelConfig(name="tiny", hidden_size=256,
                                intermediate_size=512, num_layers=2,

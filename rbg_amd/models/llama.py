@@ -158,6 +158,15 @@ class LlamaAttention(nn.Module):
                                  shard_dim=1, tp_size=tp.size,
                                  tp_rank=tp.rank)
         self.q_out, self.kv_out = q_out, kv_out
+        self.q_norm = self.k_norm = None
+        if cfg.qk_norm:
+            # Qwen3: per-head RMSNorm on q and k before RoPE
+            self.q_norm = nn.Parameter(torch.ones(cfg.head_dim, device=device,
+                                                  dtype=dtype),
+                                       requires_grad=False)
+            self.k_norm = nn.Parameter(torch.ones(cfg.head_dim, device=device,
+                                                  dtype=dtype),
+                                       requires_grad=False)
 
     def forward(self, x: torch.Tensor, batch: ForwardBatch,
                 kv: PagedKVCache, cos_sin: torch.Tensor) -> torch.Tensor:
@@ -166,6 +175,11 @@ class LlamaAttention(nn.Module):
         q = qkv[:, :self.q_out].contiguous()
         k = qkv[:, self.q_out:self.q_out + self.kv_out].contiguous()
         v = qkv[:, self.q_out + self.kv_out:].contiguous()
+        if self.q_norm is not None:
+            q = ops.rmsnorm(q.view(-1, self.head_dim), self.q_norm,
+                            self.cfg.rms_eps).view(T, self.q_out)
+            k = ops.rmsnorm(k.view(-1, self.head_dim), self.k_norm,
+                            self.cfg.rms_eps).view(T, self.kv_out)
         key_cache = kv.key_cache(self.layer_idx)
         value_cache = kv.value_cache(self.layer_idx)
         # fused RoPE + paged KV write (q,k rotated in place)

@@ -95,3 +95,40 @@ def test_stats_snapshot():
     snap = eng.stats.snapshot()
     assert snap["decode_tokens"] >= 2
     assert snap["p50_ttft_s"] >= 0.0
+
+
+def test_qwen3_family_qk_norm():
+    """Qwen3-style models (per-head q/k RMSNorm — the reference KEP's
+    benchmark family): engine generates, and batched == sequential."""
+    import torch
+    from rbg_amd.engine.config import EngineConfig, ModelConfig
+    from rbg_amd.engine.engine import LLMEngine
+    from rbg_amd.engine.sequence import SamplingParams
+    cfg = EngineConfig(model=ModelConfig.preset("tiny-qwen"), device="cpu",
+                       kv_pool_tokens=2048)
+    eng = LLMEngine(cfg)
+    torch.manual_seed(5)
+    prompts = [torch.randint(0, 500, (n,)).tolist() for n in (9, 14)]
+    seqs = [eng.add_request(p, SamplingParams(max_new_tokens=5,
+                                              temperature=0.0))
+            for p in prompts]
+    for _ in range(30):
+        eng.step()
+        if all(s.status == "finished" for s in seqs):
+            break
+    outs = [s.output_tokens for s in seqs]
+    assert all(len(o) == 5 for o in outs)
+    # sequential reference
+    eng2 = LLMEngine(EngineConfig(model=ModelConfig.preset("tiny-qwen"),
+                                  device="cpu", kv_pool_tokens=2048))
+    for p, want in zip(prompts, outs):
+        s = eng2.add_request(p, SamplingParams(max_new_tokens=5,
+                                               temperature=0.0))
+        for _ in range(30):
+            eng2.step()
+            if s.status == "finished":
+                break
+        assert s.output_tokens == want
+    # qwen3-32b preset has the reference benchmark shapes
+    q32 = ModelConfig.preset("qwen3-32b")
+    assert q32.qk_norm and q32.num_layers == 64 and q32.hidden_size == 5120
