@@ -177,9 +177,9 @@ static int pick_gemm_splits(int N, int K) {
 }
 
 bool skinny_gemm_supported(int64_t M, int64_t N, int64_t K) {
-  // M <= 128: the cross-slice LDS reduce holds 16 waves x M_TILES x 1 KB
-  // (128 KB at M_TILES=8); larger decode batches fall back to hipBLASLt
-  return M >= 1 && M <= 128 && (N % 32) == 0 && (K % 256) == 0;
+  // M <= 256: the cross-slice LDS reduce holds 4 waves x M_TILES x 1 KB
+  // (64 KB at M_TILES=16); larger batches fall back to hipBLASLt
+  return M >= 1 && M <= 256 && (N % 32) == 0 && (K % 256) == 0;
 }
 
 torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w,

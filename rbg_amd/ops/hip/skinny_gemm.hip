@@ -1,5 +1,5 @@
 // Skinny GEMM for decode-shape projections: C[M,N] = A[M,K] x W[N,K]^T,
-// bf16 in / f32 accumulate / bf16 out, M <= 128 (a decode batch).
+// bf16 in / f32 accumulate / bf16 out, M <= 256 (a decode batch).
 //
 // Why hand-written: at M=128 the layer projections are pure weight streams
 // (ideal time = W bytes / HBM bandwidth), yet every library backend leaves
@@ -91,23 +91,33 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
   // the compiler and every prefetch round-tripped through HBM
   uint4 wreg0, wreg1;
   uint4 a00 = {}, a01 = {}, a02 = {}, a03 = {};
+  uint4 a04 = {}, a05 = {}, a06 = {}, a07 = {};
   uint4 a10 = {}, a11 = {}, a12 = {}, a13 = {};
+  uint4 a14 = {}, a15 = {}, a16 = {}, a17 = {};
   const __hip_bfloat16* wbase =
       w + (size_t)((n_group << 5) + wrow_st) * K + wchk;
   const __hip_bfloat16* ab0 = nullptr;
   const __hip_bfloat16* ab1 = nullptr;
   const __hip_bfloat16* ab2 = nullptr;
   const __hip_bfloat16* ab3 = nullptr;
+  const __hip_bfloat16* ab4 = nullptr;
+  const __hip_bfloat16* ab5 = nullptr;
+  const __hip_bfloat16* ab6 = nullptr;
+  const __hip_bfloat16* ab7 = nullptr;
 #define SG_ABASE(s) (a + (size_t)min((tid + (s) * SG_NT) >> 3, M - 1) * K + \
                      ((tid + (s) * SG_NT) & 7) * 8)
   ab0 = SG_ABASE(0);
   if (A_PIECES > 1) ab1 = SG_ABASE(1);
   if (A_PIECES > 2) ab2 = SG_ABASE(2);
   if (A_PIECES > 3) ab3 = SG_ABASE(3);
-  static_assert(A_PIECES <= 4, "A_PIECES grew past the scalar sets");
+  if (A_PIECES > 4) ab4 = SG_ABASE(4);
+  if (A_PIECES > 5) ab5 = SG_ABASE(5);
+  if (A_PIECES > 6) ab6 = SG_ABASE(6);
+  if (A_PIECES > 7) ab7 = SG_ABASE(7);
+  static_assert(A_PIECES <= 8, "A_PIECES grew past the scalar sets");
 
 #define SG_LD(p, kb) (*reinterpret_cast<const uint4*>((p) + (kb)))
-#define SG_ISSUE(step, wreg, a0, a1, a2, a3)                                \
+#define SG_ISSUE(step, wreg, a0, a1, a2, a3, a4, a5, a6, a7)                \
   do {                                                                      \
     const int kb = k0 + min(step, nsteps - 1) * SG_KSTEP;                   \
     wreg = SG_LD(wbase, kb);                                                \
@@ -115,6 +125,10 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
     if (A_PIECES > 1) a1 = SG_LD(ab1, kb);                                  \
     if (A_PIECES > 2) a2 = SG_LD(ab2, kb);                                  \
     if (A_PIECES > 3) a3 = SG_LD(ab3, kb);                                  \
+    if (A_PIECES > 4) a4 = SG_LD(ab4, kb);                                  \
+    if (A_PIECES > 5) a5 = SG_LD(ab5, kb);                                  \
+    if (A_PIECES > 6) a6 = SG_LD(ab6, kb);                                  \
+    if (A_PIECES > 7) a7 = SG_LD(ab7, kb);                                  \
   } while (0)
 
   // LDS store address for A piece s (rows past A_ROWS never occur: the
@@ -124,13 +138,17 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
 #define SG_AST(st, s) reinterpret_cast<uint4*>(                             \
     &st[(32 + ((tid + (s) * SG_NT) >> 3)) * SG_LDS_PITCH +                  \
         ((tid + (s) * SG_NT) & 7) * 8])
-#define SG_WRITE(st, wreg, a0, a1, a2, a3)                                  \
+#define SG_WRITE(st, wreg, a0, a1, a2, a3, a4, a5, a6, a7)                  \
   do {                                                                      \
     *reinterpret_cast<uint4*>(&st[wrow_st * SG_LDS_PITCH + wchk]) = wreg;   \
     if (tid < A_ROWS * 8) *SG_AST(st, 0) = a0;                              \
     if (A_PIECES > 1) *SG_AST(st, 1) = a1;                                  \
     if (A_PIECES > 2) *SG_AST(st, 2) = a2;                                  \
     if (A_PIECES > 3) *SG_AST(st, 3) = a3;                                  \
+    if (A_PIECES > 4) *SG_AST(st, 4) = a4;                                  \
+    if (A_PIECES > 5) *SG_AST(st, 5) = a5;                                  \
+    if (A_PIECES > 6) *SG_AST(st, 6) = a6;                                  \
+    if (A_PIECES > 7) *SG_AST(st, 7) = a7;                                  \
   } while (0)
 
   f32x4 acc[M_TILES];
@@ -154,10 +172,10 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
   __hip_bfloat16* buf1 = lds.stage[1];
 
   // pipeline prologue: tiles 0,1 in flight; tile 0 lands in buffer 0
-  SG_ISSUE(0, wreg0, a00, a01, a02, a03);
-  SG_ISSUE(1, wreg1, a10, a11, a12, a13);
-  SG_WRITE(buf0, wreg0, a00, a01, a02, a03);
-  SG_ISSUE(2, wreg0, a00, a01, a02, a03);
+  SG_ISSUE(0, wreg0, a00, a01, a02, a03, a04, a05, a06, a07);
+  SG_ISSUE(1, wreg1, a10, a11, a12, a13, a14, a15, a16, a17);
+  SG_WRITE(buf0, wreg0, a00, a01, a02, a03, a04, a05, a06, a07);
+  SG_ISSUE(2, wreg0, a00, a01, a02, a03, a04, a05, a06, a07);
 
   // nsteps is even (K % (SPLITS * 2*SG_KSTEP) == 0, enforced host-side);
   // tile t+1 was issued 2 steps ago so its vmcnt wait is covered, and a
@@ -166,12 +184,12 @@ __global__ __launch_bounds__(SG_NT, 3) void skinny_gemm_kernel(
   for (int step = 0; step < nsteps; step += 2) {
     __syncthreads();
     compute(buf0);                                 // tile step
-    SG_WRITE(buf1, wreg1, a10, a11, a12, a13);     // tile step+1
-    SG_ISSUE(step + 3, wreg1, a10, a11, a12, a13);
+    SG_WRITE(buf1, wreg1, a10, a11, a12, a13, a14, a15, a16, a17);
+    SG_ISSUE(step + 3, wreg1, a10, a11, a12, a13, a14, a15, a16, a17);
     __syncthreads();
     compute(buf1);                                 // tile step+1
-    SG_WRITE(buf0, wreg0, a00, a01, a02, a03);     // tile step+2
-    SG_ISSUE(step + 4, wreg0, a00, a01, a02, a03);
+    SG_WRITE(buf0, wreg0, a00, a01, a02, a03, a04, a05, a06, a07);
+    SG_ISSUE(step + 4, wreg0, a00, a01, a02, a03, a04, a05, a06, a07);
   }
 #undef SG_ISSUE
 #undef SG_WRITE
@@ -255,7 +273,8 @@ extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
   if (m_tiles <= 1) LAUNCH_MT(1);
   else if (m_tiles <= 2) LAUNCH_MT(2);
   else if (m_tiles <= 4) LAUNCH_MT(4);
-  else LAUNCH_MT(8);
+  else if (m_tiles <= 8) LAUNCH_MT(8);
+  else LAUNCH_MT(16);
 #undef LAUNCH_MT
 }
 

@@ -191,7 +191,7 @@ def test_skinny_gemm_matches_fp32():
     H, I, QH, KVH, D = 4096, 14336, 32, 8, 128
     shapes = [(H, (QH + 2 * KVH) * D), (QH * D, H), (H, 2 * I), (I, H)]
     torch.manual_seed(7)
-    for M in (1, 7, 33, 64, 128):
+    for M in (1, 7, 33, 64, 128, 192, 256):
         for (k, n) in shapes:
             x = torch.randn(M, k, dtype=torch.bfloat16, device=dev) * 0.5
             w = torch.randn(n, k, dtype=torch.bfloat16, device=dev) * 0.05
