@@ -88,6 +88,18 @@ class Daemon:
                     lines.append(
                         f'rbg_instance_recovery_seconds{{instance="{n}"}} '
                         f"{inst.status.last_recovery_duration:.3f}")
+            lines.append("# TYPE rbg_reconcile_seconds_total counter")
+            lines.append("# TYPE rbg_reconcile_count counter")
+            lines.append("# TYPE rbg_reconcile_max_seconds gauge")
+            for kind, st in sorted(self.manager.reconcile_stats.items()):
+                lines.append(
+                    f'rbg_reconcile_seconds_total{{kind="{kind}"}} '
+                    f"{st['total_s']:.4f}")
+                lines.append(
+                    f'rbg_reconcile_count{{kind="{kind}"}} {st["count"]}')
+                lines.append(
+                    f'rbg_reconcile_max_seconds{{kind="{kind}"}} '
+                    f"{st['max_s']:.4f}")
             return "\n".join(lines) + "\n"
         reg("metrics", metrics)
 

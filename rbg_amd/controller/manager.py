@@ -136,6 +136,7 @@ class Manager:
 
         from ..store.events import EventRecorder
         self.recorder = EventRecorder(self.store)
+        self.reconcile_stats: Dict[str, Dict[str, float]] = {}
         self.rbg = RoleBasedGroupController(self.store, self.registry,
                                             self.opts.history_limit,
                                             ports=self.ports,
@@ -226,11 +227,21 @@ class Manager:
                 key = q.get(timeout=0.2)
                 if key is None:
                     continue
+                t0 = time.monotonic()
                 try:
                     requeue = fn(key[0], key[1])
                 except Exception:
                     log.exception("reconcile %s %s failed", kind, key)
                     requeue = 1.0
+                # per-reconcile wall accounting (reference logs reconcile
+                # wall per loop; exposed as rbg_reconcile_* gauges)
+                dt = time.monotonic() - t0
+                st = self.reconcile_stats.setdefault(
+                    kind, {"count": 0, "total_s": 0.0, "max_s": 0.0})
+                st["count"] += 1
+                st["total_s"] += dt
+                if dt > st["max_s"]:
+                    st["max_s"] = dt
                 q.done(key, requeue_after=requeue if requeue else 0.0)
 
         self._threads = [threading.Thread(target=watch_loop, daemon=True),
