@@ -137,6 +137,7 @@ def test_daemon_metrics(daemon):
     text = c.call("metrics")
     assert "rbg_groups 1" in text
     assert 'rbg_group_ready{group="metrics-demo"}' in text
+    assert "rbg_reconcile_seconds_total" in text
     assert "rbg_free_gpus" in text
 
 
