@@ -570,8 +570,11 @@ def load_object(data: Dict[str, Any]):
     v1alpha1 RoleBasedGroup docs are converted to v1alpha2 on the way in
     (the conversion-webhook analog, reference rolebasedgroup_conversion.go)."""
     from . import v1alpha1 as _legacy
-    if _legacy.is_v1alpha1(data) and data.get("kind") == C.KIND_RBG:
-        data = _legacy.to_v2(data)
+    if _legacy.is_v1alpha1(data):
+        if data.get("kind") == C.KIND_RBG:
+            data = _legacy.to_v2(data)
+        elif data.get("kind") == C.KIND_RBG_SET:
+            data = _legacy.set_to_v2(data)
     kind = data.get("kind", "")
     cls = KIND_TO_TYPE.get(kind)
     if cls is None:

@@ -56,6 +56,21 @@ def is_v1alpha1(doc: Dict[str, Any]) -> bool:
     return doc.get("apiVersion", "") == API_VERSION_V1ALPHA1
 
 
+def set_to_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
+    """v1alpha1 RoleBasedGroupSet doc → v1alpha2 (the template is an RBG
+    spec; reference rolebasedgroupset_conversion.go)."""
+    spec = doc.get("spec") or {}
+    inner = {"apiVersion": API_VERSION_V1ALPHA1, "kind": C.KIND_RBG,
+             "metadata": dict(doc.get("metadata") or {}),
+             "spec": spec.get("template") or {}}
+    conv = to_v2(inner)
+    out_meta = conv["metadata"]
+    return {"apiVersion": C.API_VERSION, "kind": C.KIND_RBG_SET,
+            "metadata": out_meta,
+            "spec": {"replicas": spec.get("replicas", 1),
+                     "template": conv["spec"]}}
+
+
 def _merge_patch(base: Dict[str, Any], patch: Dict[str, Any]) -> Dict[str, Any]:
     """JSON-merge-patch-lite used for patchLeaderTemplate/patchWorkerTemplate
     over the role template (good enough for EngineTemplate docs: dicts merge

@@ -135,3 +135,23 @@ def test_min_ready_seconds_gates_availability(monkeypatch):
     # age the transition past the window
     inst.status.conditions[0].last_transition_time = _time.time() - 6
     assert _is_available(inst, 5)
+
+
+def test_v1alpha1_rbgset_conversion():
+    """Legacy RoleBasedGroupSet wraps an RBG spec template (reference
+    rolebasedgroupset_conversion.go): conversion reuses the role mapping."""
+    doc = {"apiVersion": "workloads.x-k8s.io/v1alpha1",
+           "kind": "RoleBasedGroupSet",
+           "metadata": {"name": "fleet"},
+           "spec": {"replicas": 3,
+                    "template": {"roles": [
+                        {"name": "w", "replicas": 2,
+                         "workload": {"apiVersion": "apps/v1",
+                                      "kind": "StatefulSet"},
+                         "template": {"engines": [
+                             {"name": "e", "runner": "echo"}]}}]}}}
+    obj = load_object(copy.deepcopy(doc))
+    assert obj.kind == C.KIND_RBG_SET
+    assert obj.spec.replicas == 3
+    role = obj.spec.template.roles[0]
+    assert role.name == "w" and role.pattern == C.PATTERN_STANDALONE
