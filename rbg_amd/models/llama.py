@@ -172,14 +172,21 @@ class LlamaAttention(nn.Module):
                 kv: PagedKVCache, cos_sin: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
         qkv = ops.linear(x, self.wqkv)
-        q = qkv[:, :self.q_out].contiguous()
-        k = qkv[:, self.q_out:self.q_out + self.kv_out].contiguous()
-        v = qkv[:, self.q_out + self.kv_out:].contiguous()
+        # q/k/v stay SLICES of the fused projection: the rope and attention
+        # kernels take a token-row stride, so no .contiguous() copies
+        # (~73 ms/bench of pure copy kernels before this)
+        q = qkv[:, :self.q_out]
+        k = qkv[:, self.q_out:self.q_out + self.kv_out]
+        v = qkv[:, self.q_out + self.kv_out:]
         if self.q_norm is not None:
-            q = ops.rmsnorm(q.view(-1, self.head_dim), self.q_norm,
+            # qwen per-head norm reshapes across rows: materialize
+            q = ops.rmsnorm(q.contiguous().view(-1, self.head_dim),
+                            self.q_norm,
                             self.cfg.rms_eps).view(T, self.q_out)
-            k = ops.rmsnorm(k.view(-1, self.head_dim), self.k_norm,
+            k = ops.rmsnorm(k.contiguous().view(-1, self.head_dim),
+                            self.k_norm,
                             self.cfg.rms_eps).view(T, self.kv_out)
+            v = v.contiguous()
         key_cache = kv.key_cache(self.layer_idx)
         value_cache = kv.value_cache(self.layer_idx)
         # fused RoPE + paged KV write (q,k rotated in place)

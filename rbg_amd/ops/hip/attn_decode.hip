@@ -39,7 +39,7 @@ void decode_attn_kernel(
     const int* __restrict__ block_tables, // [seqs, max_pages]
     const int* __restrict__ context_lens, // [seqs]
     const float scale, const int num_kv_heads, const int page_size,
-    const int max_pages, const int num_splits) {
+    const int max_pages, const int num_splits, const int q_stride) {
   const int kvh = blockIdx.x;
   const int seq = blockIdx.y;
   const int split = blockIdx.z;
@@ -67,7 +67,7 @@ void decode_attn_kernel(
   float qf[VARIANT == 0 ? QPG : 1][8];
   {
     const __hip_bfloat16* qrow =
-        q + ((size_t)seq * num_q_heads + kvh * QPG) * HEAD_DIM;
+        q + (size_t)seq * q_stride + (size_t)kvh * QPG * HEAD_DIM;
 #pragma unroll
     for (int h = 0; h < QPG; ++h) {
       Bf16x8U qv;
@@ -312,7 +312,7 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
     const int* __restrict__ block_tables, // [seqs, max_pages]
     const int* __restrict__ context_lens, // [seqs]
     const float scale, const int num_kv_heads, const int max_pages,
-    const int num_splits, const int num_seqs) {
+    const int num_splits, const int num_seqs, const int q_stride) {
   typedef __attribute__((ext_vector_type(8))) __bf16 dm_bf8;
   const int kvh = blockIdx.x;
   const int wave = threadIdx.x >> 6;
@@ -351,7 +351,7 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
   {
     const int h = min(gl, QPG - 1);
     const __hip_bfloat16* qrow =
-        q + ((size_t)seq * num_q_heads + kvh * QPG + h) * HEAD_DIM;
+        q + (size_t)seq * q_stride + (size_t)(kvh * QPG + h) * HEAD_DIM;
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
       union { uint4 u; dm_bf8 v; } cvt;
@@ -586,7 +586,7 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                              const void* context_lens, float scale,
                              int num_seqs, int num_q_heads, int num_kv_heads,
                              int page_size, int max_pages, int num_splits,
-                             int variant, hipStream_t stream) {
+                             int variant, int q_stride, hipStream_t stream) {
   const int qpg = num_q_heads / num_kv_heads;
   dim3 grid(num_kv_heads, num_seqs, num_splits), block(256);
 #define LAUNCH_QPG(QPG, VAR)                                                  \
@@ -596,7 +596,8 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                      (const __hip_bfloat16*)key_cache,                        \
                      (const __hip_bfloat16*)val_cache,                        \
                      (const int*)block_tables, (const int*)context_lens,      \
-                     scale, num_kv_heads, page_size, max_pages, num_splits)
+                     scale, num_kv_heads, page_size, max_pages, num_splits,  \
+                     q_stride)
   if (variant == 4 && page_size == 16) {
     dim3 mgrid(num_kv_heads, (num_seqs + 1) / 2, num_splits), mblock(128);
 #define LAUNCH_MFMA(QPG)                                                     \
@@ -606,7 +607,8 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                        (const __hip_bfloat16*)key_cache,                     \
                        (const __hip_bfloat16*)val_cache,                     \
                        (const int*)block_tables, (const int*)context_lens,   \
-                       scale, num_kv_heads, max_pages, num_splits, num_seqs)
+                       scale, num_kv_heads, max_pages, num_splits,         \
+                       num_seqs, q_stride)
     if (qpg == 1) LAUNCH_MFMA(1);
     else if (qpg == 2) LAUNCH_MFMA(2);
     else if (qpg == 4) LAUNCH_MFMA(4);

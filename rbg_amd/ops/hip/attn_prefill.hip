@@ -74,6 +74,7 @@ __global__ __launch_bounds__(NW * 64) void prefill_attn_kernel(
     //   prefill attend over [past; new] K/V)
     const int* __restrict__ seq_lens,        // [nblocks] KV length of the seq
     const float scale, const int num_q_heads, const int num_kv_heads,
+    const int q_stride, const int kv_stride,
     const int nblocks) {
   const int qpg_n = num_q_heads / num_kv_heads;
   int blk, qh;
@@ -125,8 +126,8 @@ __global__ __launch_bounds__(NW * 64) void prefill_attn_kernel(
   mfma_bf8 q_frag[4];
 #pragma unroll
   for (int ks = 0; ks < 4; ++ks)
-    q_frag[ks] = load_bf8(q + ((size_t)q_row * num_q_heads + qh) * HEAD_DIM +
-                          ks * 32 + gslice * 8);
+    q_frag[ks] = load_bf8(q + (size_t)q_row * q_stride +
+                          (size_t)qh * HEAD_DIM + ks * 32 + gslice * 8);
 
   // per-lane softmax state: 4 q-rows (rows 4*gslice + r of the wave tile)
   float m_run[4], l_run[4];
@@ -156,9 +157,9 @@ __global__ __launch_bounds__(NW * 64) void prefill_attn_kernel(
   uint4 kr0 = {}, kr1 = {}, kr2 = {}, kr3 = {};
   uint4 vr0 = {}, vr1 = {}, vr2 = {}, vr3 = {};
 #define PF_ROW(s, kt2)                                                      \
-  (((size_t)(kv_start + min((kt2) * KTILE + ((tid + (s) * NTHREADS) >> 4),  \
-                            seq_len - 1)) * num_kv_heads + kvh) *           \
-       HEAD_DIM + ((tid + (s) * NTHREADS) & 15) * 8)
+  ((size_t)(kv_start + min((kt2) * KTILE + ((tid + (s) * NTHREADS) >> 4),  \
+                           seq_len - 1)) * kv_stride +                      \
+   (size_t)kvh * HEAD_DIM + ((tid + (s) * NTHREADS) & 15) * 8)
 #define PF_LD(s, kt2, kr, vr)                                               \
   do {                                                                      \
     const size_t row = PF_ROW(s, kt2);                                      \
@@ -334,6 +335,7 @@ void launch_prefill_attention(void* out, const void* q, const void* k,
                               const void* v, const void* block_info,
                               const void* seq_lens, float scale, int nblocks,
                               int num_q_heads, int num_kv_heads, int swz,
+                              int q_stride, int kv_stride,
                               hipStream_t stream) {
   dim3 grid(nblocks * num_q_heads);
   // swz bit 0: XCD-affine GQA swizzle; bit 1: T14 double-buffer staging;
@@ -344,7 +346,7 @@ void launch_prefill_attention(void* out, const void* q, const void* k,
                      (__hip_bfloat16*)out, (const __hip_bfloat16*)q,        \
                      (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,    \
                      (const int*)block_info, (const int*)seq_lens, scale,   \
-                     num_q_heads, num_kv_heads, nblocks)
+                     num_q_heads, num_kv_heads, q_stride, kv_stride, nblocks)
   switch (swz & 7) {
     case 0: PF_LAUNCH(0, 0, 4); break;
     case 1: PF_LAUNCH(1, 0, 4); break;

@@ -12,13 +12,14 @@ void launch_fused_add_rmsnorm(void*, void*, const void*, float, int, int,
 void launch_silu_mul(void*, const void*, int, int, hipStream_t);
 void launch_rope_store_kv(void*, void*, const void*, void*, void*,
                           const void*, const void*, const void*, int, int,
-                          int, int, int, hipStream_t);
+                          int, int, int, int, int, hipStream_t);
 void launch_decode_attention(void*, void*, void*, const void*, const void*,
                              const void*, const void*, const void*, float,
-                             int, int, int, int, int, int, int, hipStream_t);
+                             int, int, int, int, int, int, int, int,
+                             hipStream_t);
 void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const void*, const void*, float, int, int, int,
-                              int, hipStream_t);
+                              int, int, int, hipStream_t);
 void launch_mfma_probe(void*, const void*, const void*, int, int, hipStream_t);
 void launch_skinny_gemm(void*, void*, const void*, const void*, int, int,
                         int, int, hipStream_t);
@@ -73,13 +74,23 @@ torch::Tensor silu_mul(torch::Tensor x) {
   return out;
 }
 
+static void check_bf16_rowslice(const torch::Tensor& t, const char* name) {
+  // 2-D bf16 on GPU whose LAST dim is contiguous; the row stride may be
+  // wider (a column slice of the fused QKV projection)
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 2 && t.stride(1) == 1, name,
+              " must be 2-D with contiguous rows");
+}
+
 void rope_store_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                    torch::Tensor key_cache, torch::Tensor value_cache,
                    torch::Tensor cos_sin, torch::Tensor positions,
                    torch::Tensor slot_mapping) {
-  check_bf16_contig(q, "q");
-  check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
+  check_bf16_rowslice(q, "q");
+  check_bf16_rowslice(k, "k");
+  check_bf16_rowslice(v, "v");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v stride mismatch");
   TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32, "cos_sin fp32");
   TORCH_CHECK(positions.scalar_type() == torch::kInt32, "positions int32");
   TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt32, "slots int32");
@@ -87,12 +98,14 @@ void rope_store_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   const int head_dim = key_cache.size(3);
   const int num_kv_heads = key_cache.size(1);
   const int page_size = key_cache.size(2);
-  const int num_q_heads = q.numel() / tokens / head_dim;
+  const int num_q_heads = q.size(1) / head_dim;
   launch_rope_store_kv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                        key_cache.data_ptr(), value_cache.data_ptr(),
                        cos_sin.data_ptr(), positions.data_ptr(),
                        slot_mapping.data_ptr(), tokens, num_q_heads,
-                       num_kv_heads, head_dim, page_size, current_stream());
+                       num_kv_heads, head_dim, page_size,
+                       (int)q.stride(0), (int)k.stride(0),
+                       current_stream());
 }
 
 torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
@@ -100,7 +113,11 @@ torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
                                torch::Tensor block_tables,
                                torch::Tensor context_lens, double scale,
                                int64_t num_splits, int64_t wide) {
-  check_bf16_contig(q, "q");
+  // q may be a slice of the fused QKV projection: 3-D bf16 with
+  // contiguous (head, dim) rows and an arbitrary token stride
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16 &&
+              q.dim() == 3 && q.stride(2) == 1 &&
+              q.stride(1) == q.size(2), "q must be [T, QH, D] row-sliced");
   const int num_seqs = q.size(0);
   const int num_q_heads = q.size(1);
   const int head_dim = q.size(2);
@@ -113,7 +130,7 @@ torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
               "GQA group must be 1/2/4/8");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32, "tables int32");
   TORCH_CHECK(context_lens.scalar_type() == torch::kInt32, "lens int32");
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({num_seqs, num_q_heads, head_dim}, q.options());
   torch::Tensor partial_o, partial_ml;
   void *po = nullptr, *pml = nullptr;
   if (num_splits > 1) {
@@ -128,7 +145,7 @@ torch::Tensor decode_attention(torch::Tensor q, torch::Tensor key_cache,
                           block_tables.data_ptr(), context_lens.data_ptr(),
                           (float)scale, num_seqs, num_q_heads, num_kv_heads,
                           page_size, max_pages, (int)num_splits, (int)wide,
-                          current_stream());
+                          (int)q.stride(0), current_stream());
   return out;
 }
 
@@ -136,20 +153,29 @@ torch::Tensor prefill_attention(torch::Tensor q, torch::Tensor k,
                                 torch::Tensor v, torch::Tensor block_info,
                                 torch::Tensor seq_lens, double scale,
                                 int64_t swz) {
-  check_bf16_contig(q, "q");
-  check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
+  auto rowsliced3 = [](const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
+                t.dim() == 3 && t.stride(2) == 1 &&
+                t.stride(1) == t.size(2), name,
+                " must be [T, H, D] row-sliced");
+  };
+  rowsliced3(q, "q");
+  rowsliced3(k, "k");
+  rowsliced3(v, "v");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v stride mismatch");
   TORCH_CHECK(q.size(2) == 128, "head_dim must be 128");
   TORCH_CHECK(block_info.scalar_type() == torch::kInt32, "block_info int32");
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32, "seq_lens int32");
   const int nblocks = block_info.size(0);
   const int num_q_heads = q.size(1);
   const int num_kv_heads = k.size(1);
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({q.size(0), (int64_t)num_q_heads, (int64_t)128},
+                          q.options());
   launch_prefill_attention(out.data_ptr(), q.data_ptr(), k.data_ptr(),
                            v.data_ptr(), block_info.data_ptr(),
                            seq_lens.data_ptr(), (float)scale, nblocks,
                            num_q_heads, num_kv_heads, (int)swz,
+                           (int)q.stride(0), (int)k.stride(0),
                            current_stream());
   return out;
 }

@@ -26,7 +26,11 @@ __global__ void rope_store_kv_kernel(
     const int* __restrict__ positions,       // [T]
     const int* __restrict__ slot_mapping,    // [T] page*page_size+off; -1 skip
     const int num_q_heads, const int num_kv_heads, const int head_dim,
-    const int page_size) {
+    const int page_size,
+    const int q_stride, const int kv_stride) { // token-row strides (elems):
+  // q/k/v may be SLICES of the fused QKV projection (row pitch = the full
+  // qkv width) — strided access here removes three .contiguous() copies
+  // per layer (profiles: ~73 ms/bench of pure copy kernels)
   const int token = blockIdx.x;
   const int tid = threadIdx.x;
   const int pos = positions[token];
@@ -35,7 +39,7 @@ __global__ void rope_store_kv_kernel(
   const float* cs = cos_sin + (size_t)pos * head_dim;
 
   // ---- rotate Q: num_q_heads * half pairs -------------------------------
-  __hip_bfloat16* q_tok = q + (size_t)token * num_q_heads * head_dim;
+  __hip_bfloat16* q_tok = q + (size_t)token * q_stride;
   const int q_pairs = num_q_heads * half;
   for (int i = tid; i < q_pairs; i += blockDim.x) {
     const int h = i / half, d = i % half;
@@ -47,8 +51,8 @@ __global__ void rope_store_kv_kernel(
   }
 
   // ---- rotate K and scatter K,V into the paged pool ---------------------
-  __hip_bfloat16* k_tok = k + (size_t)token * num_kv_heads * head_dim;
-  const __hip_bfloat16* v_tok = v + (size_t)token * num_kv_heads * head_dim;
+  __hip_bfloat16* k_tok = k + (size_t)token * kv_stride;
+  const __hip_bfloat16* v_tok = v + (size_t)token * kv_stride;
   const int page = slot >= 0 ? slot / page_size : 0;
   const int poff = slot >= 0 ? slot % page_size : 0;
   const int kv_pairs = num_kv_heads * half;
@@ -90,14 +94,16 @@ void launch_rope_store_kv(void* q, void* k, const void* v, void* key_cache,
                           void* val_cache, const void* cos_sin,
                           const void* positions, const void* slot_mapping,
                           int tokens, int num_q_heads, int num_kv_heads,
-                          int head_dim, int page_size, hipStream_t stream) {
+                          int head_dim, int page_size, int q_stride,
+                          int kv_stride, hipStream_t stream) {
   dim3 grid(tokens), block(256);
   hipLaunchKernelGGL(rope_store_kv_kernel, grid, block, 0, stream,
                      (__hip_bfloat16*)q, (__hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (__hip_bfloat16*)key_cache,
                      (__hip_bfloat16*)val_cache, (const float*)cos_sin,
                      (const int*)positions, (const int*)slot_mapping,
-                     num_q_heads, num_kv_heads, head_dim, page_size);
+                     num_q_heads, num_kv_heads, head_dim, page_size,
+                     q_stride, kv_stride);
 }
 
 }  // extern "C"

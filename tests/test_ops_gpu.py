@@ -233,3 +233,60 @@ def test_decode_attention_mfma_variant4():
         assert torch.allclose(got.float(), ref.float(), atol=2e-2,
                               rtol=2e-2), \
             (splits, (got.float() - ref.float()).abs().max().item())
+
+
+@pytest.mark.gpu
+def test_strided_qkv_slices_match_contiguous():
+    """rope + both attention kernels accept fused-QKV slices (wide row
+    stride): results must equal the contiguous-copy path exactly."""
+    import torch
+    from rbg_amd import ops
+    dev = torch.device("cuda:0")
+    torch.manual_seed(11)
+    T, QH, KVH, D, page = 32, 32, 8, 128, 16
+    width = (QH + 2 * KVH) * D
+    kv_pages = 16
+    for mode in ("prefill", "decode"):
+        qkv = torch.randn(T, width, dtype=torch.bfloat16, device=dev)
+        qkv2 = qkv.clone()
+        kc1 = torch.zeros(kv_pages, KVH, page, D, dtype=torch.bfloat16,
+                          device=dev)
+        vc1 = torch.zeros_like(kc1)
+        kc2, vc2 = kc1.clone(), vc1.clone()
+        cs = ops.build_cos_sin_table(D, 4096, 5e5, device=dev)
+        pos = torch.arange(T, dtype=torch.int32, device=dev)
+        slots = torch.arange(16, 16 + T, dtype=torch.int32, device=dev)
+        # strided path
+        q = qkv[:, :QH * D]
+        k = qkv[:, QH * D:(QH + KVH) * D]
+        v = qkv[:, (QH + KVH) * D:]
+        ops._hip.rope_store_kv(q, k, v, kc1, vc1, cs, pos, slots)
+        # contiguous path
+        q2 = qkv2[:, :QH * D].contiguous()
+        k2 = qkv2[:, QH * D:(QH + KVH) * D].contiguous()
+        v2 = qkv2[:, (QH + KVH) * D:].contiguous()
+        ops._hip.rope_store_kv(q2, k2, v2, kc2, vc2, cs, pos, slots)
+        assert torch.equal(kc1, kc2) and torch.equal(vc1, vc2)
+        assert torch.equal(q.contiguous(), q2)
+        if mode == "prefill":
+            from rbg_amd.ops import reference
+            cu = torch.tensor([0, T], dtype=torch.int32)
+            bi, sl = reference.prefill_block_info(cu, qtile=128)
+            o1 = ops._hip.prefill_attention(
+                q.view(T, QH, D), k.view(T, KVH, D), v.view(T, KVH, D),
+                bi.to(dev), sl.to(dev), 0.088, 6)
+            o2 = ops._hip.prefill_attention(
+                q2.view(T, QH, D), k2.view(T, KVH, D), v2.view(T, KVH, D),
+                bi.to(dev), sl.to(dev), 0.088, 6)
+            assert torch.equal(o1, o2)
+        else:
+            bt = torch.arange(1, 3, dtype=torch.int32,
+                              device=dev).repeat(T, 1)
+            bt = torch.arange(1, 1 + 2, dtype=torch.int32,
+                              device=dev).unsqueeze(0).repeat(T, 1)
+            ctx = torch.full((T,), 2, dtype=torch.int32, device=dev)
+            o1 = ops._hip.decode_attention(q.view(T, QH, D), kc1, vc1, bt,
+                                           ctx, 0.088, 1, 4)
+            o2 = ops._hip.decode_attention(q2.view(T, QH, D), kc2, vc2, bt,
+                                           ctx, 0.088, 1, 4)
+            assert torch.equal(o1, o2)
