@@ -75,12 +75,17 @@ torch::Tensor silu_mul(torch::Tensor x) {
 }
 
 static void check_bf16_rowslice(const torch::Tensor& t, const char* name) {
-  // 2-D bf16 on GPU whose LAST dim is contiguous; the row stride may be
-  // wider (a column slice of the fused QKV projection)
+  // bf16 on GPU, [T, W] or [T, H, D], innermost dims contiguous; the
+  // TOKEN stride may be wider (a column slice of the fused QKV buffer)
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
-  TORCH_CHECK(t.dim() == 2 && t.stride(1) == 1, name,
-              " must be 2-D with contiguous rows");
+  if (t.dim() == 2) {
+    TORCH_CHECK(t.stride(1) == 1, name, " rows must be contiguous");
+  } else {
+    TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 &&
+                t.stride(1) == t.size(2), name,
+                " must be [T, H, D] with contiguous (H, D) rows");
+  }
 }
 
 void rope_store_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -98,7 +103,7 @@ void rope_store_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   const int head_dim = key_cache.size(3);
   const int num_kv_heads = key_cache.size(1);
   const int page_size = key_cache.size(2);
-  const int num_q_heads = q.size(1) / head_dim;
+  const int num_q_heads = (int)(q.numel() / tokens / head_dim);
   launch_rope_store_kv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                        key_cache.data_ptr(), value_cache.data_ptr(),
                        cos_sin.data_ptr(), positions.data_ptr(),
