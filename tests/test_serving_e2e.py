@@ -272,3 +272,42 @@ def test_pp2_leader_worker_serving(mgr):
                      timeout=240)
     assert len(res["tokens"]) == 5
     assert res["tokens"] == _local_reference_tokens(prompt, 5)
+
+
+@pytest.mark.timeout(420)
+def test_tp2_pp2_leader_worker_serving(mgr):
+    """TP x PP composition: leaderWorker size 4 with engine arg pp=2 ->
+    2 pipeline stages x TP-2 stage subgroups (controller emits
+    rbg.comm-subgroups).  Output must exactly equal the single-process
+    reference (weights degree-invariant in both dimensions)."""
+    from rbg_amd.api.types import LeaderWorkerPattern
+    args = dict(ENGINE_ARGS, mode="colocated", model="tiny-tp",
+                cpu_model="tiny-tp", tp_from_env=True, tp_backend="gloo",
+                pp=2)
+    role = RoleSpec(
+        name="worker", replicas=1, dependencies=["router"],
+        pattern=C.PATTERN_LEADER_WORKER,
+        leader_worker_pattern=LeaderWorkerPattern(size=4),
+        template=EngineTemplate(engines=[EngineSpec(
+            name="engine", runner="llm-engine", args=args,
+            resources=EngineResources(cpu_only=True))]))
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="tp-pp"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("colocated", {"worker_roles": ["worker"],
+                                      "vocab_size": 500}),
+            role,
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "tp-pp"), timeout=240)
+    assert mgr.wait_for(
+        lambda: (_router_http_port(mgr, "tp-pp") is not None), timeout=30)
+    port = _router_http_port(mgr, "tp-pp")
+    torch.manual_seed(27)
+    prompt = torch.randint(0, 500, (11,)).tolist()
+    res = _http_post(port, "/generate",
+                     {"prompt_tokens": prompt, "max_new_tokens": 5},
+                     timeout=240)
+    assert len(res["tokens"]) == 5
+    assert res["tokens"] == _local_reference_tokens(prompt, 5,
+                                                    model="tiny-tp")
