@@ -31,12 +31,42 @@ class Daemon:
         self.manager = Manager(opts)
         self.rpc = RpcServer(port=port)
         self.port = self.rpc.port
+        # watch ring buffer: remote clients long-poll `watch_events` with
+        # the last sequence they saw (client-go informer analog)
+        from collections import deque
+        self._watch_seq = 0
+        self._watch_buf = deque(maxlen=2000)
+        self._watch_lock = threading.Lock()
+        self._watch_thread = threading.Thread(target=self._watch_pump,
+                                              daemon=True)
         self._register()
+
+    def _watch_pump(self) -> None:
+        for ev in self._watch_stream:
+            with self._watch_lock:
+                self._watch_seq += 1
+                self._watch_buf.append((self._watch_seq, {
+                    "type": ev.type, "kind": ev.obj.kind,
+                    "name": ev.obj.metadata.name,
+                    "namespace": ev.obj.metadata.namespace,
+                    "resourceVersion": ev.obj.metadata.resource_version,
+                }))
 
     def _register(self) -> None:
         store = self.manager.store
+        self._watch_stream = store.watch(replay=False)
+        self._watch_thread.start()
         reg = self.rpc.register
         reg("ping", lambda: "pong")
+
+        def watch_events(since: int = 0, kinds: Optional[List[str]] = None):
+            """Events after sequence `since` (ring-buffered; a client that
+            falls > buffer-size behind should relist)."""
+            with self._watch_lock:
+                evs = [{"seq": s, **e} for s, e in self._watch_buf
+                       if s > since and (not kinds or e["kind"] in kinds)]
+                return {"next": self._watch_seq, "events": evs}
+        reg("watch_events", watch_events)
         reg("healthz", lambda: {"status": "ok",
                                 "gpus": self.manager.topo.num_gpus,
                                 "free_gpus": self.manager.gang.free_gpus()})

@@ -129,3 +129,19 @@ class RemoteClient(BaseClient):
     def delete_raw(self, kind, name, namespace):
         return self.rpc.call("store_delete", kind=kind, name=name,
                              namespace=namespace)
+
+
+def remote_watch(client: "RemoteClient", kinds=None, poll_s: float = 0.3):
+    """Generator of watch events from a remote daemon (informer analog):
+    long-polls `watch_events`, yielding dicts with type/kind/name/
+    namespace/resourceVersion.  Falls back to relisting is the caller's
+    job if it lags the daemon's ring buffer."""
+    import time as _time
+    since = client.rpc.call("watch_events", since=0)["next"]
+    while True:
+        res = client.rpc.call("watch_events", since=since, kinds=kinds)
+        for ev in res["events"]:
+            since = ev["seq"]
+            yield ev
+        since = max(since, res["next"]) if not res["events"] else since
+        _time.sleep(poll_s)

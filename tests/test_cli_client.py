@@ -167,3 +167,23 @@ def test_events_recorded_and_listed(tmp_run_dir):
         assert rc == 0
     finally:
         m.stop()
+
+
+def test_remote_watch_events(daemon):
+    """Remote watch (informer analog): long-poll ring buffer surfaces
+    create/update events with kind/name/resourceVersion."""
+    from rbg_amd.client.client import RemoteClient
+    c = RemoteClient("127.0.0.1", daemon.port)
+    base = c.rpc.call("watch_events", since=0)["next"]
+    daemon.manager.store.create(make_rbg("watch-demo"))
+    import time
+    deadline = time.time() + 10
+    seen = []
+    while time.time() < deadline:
+        res = c.rpc.call("watch_events", since=base)
+        seen = [e for e in res["events"]
+                if e["name"] == "watch-demo" and e["kind"] == "RoleBasedGroup"]
+        if seen:
+            break
+        time.sleep(0.1)
+    assert seen and seen[0]["type"] == "ADDED"
