@@ -313,6 +313,10 @@ class WorkerStatus:
 class RoleInstanceSpec:
     components: List[ComponentSpec] = field(default_factory=list)
     restart_policy: str = C.RESTART_POLICY_RECREATE_INSTANCE
+    # reference roleinstance_types.go readinessGates: extra condition types
+    # that must be True before the instance counts Ready (the in-place
+    # update engine gates on InPlaceUpdateReady this way)
+    readiness_gates: List[str] = field(default_factory=list)
 
 
 @dataclass

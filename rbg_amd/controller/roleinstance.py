@@ -467,10 +467,17 @@ class RoleInstanceController:
                 f"{ready}/{len(desired)} workers ready"))
             restarting = get_condition(cur.status.conditions, C.COND_RESTARTING)
             is_restarting = restarting is not None and restarting.status == "True"
+            gates_ok = True
+            for gate in cur.spec.readiness_gates:
+                gc = get_condition(cur.status.conditions, gate)
+                if gc is None or gc.status != "True":
+                    gates_ok = False
             set_condition(cur.status.conditions, Condition.new(
-                C.COND_READY, all_ready and not is_restarting,
-                "Ready" if all_ready else "NotReady",
-                f"{ready}/{len(desired)} workers ready"))
+                C.COND_READY, all_ready and not is_restarting and gates_ok,
+                "Ready" if (all_ready and gates_ok) else
+                ("GatesNotReady" if all_ready else "NotReady"),
+                f"{ready}/{len(desired)} workers ready" +
+                ("" if gates_ok else "; readiness gates pending")))
             return cur
         try:
             self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mutate,

@@ -116,3 +116,36 @@ def test_external_kill_recovers(mgr):
         return victim_pid not in pids and all(
             w.phase == "Ready" for i in insts2 for w in i.status.workers)
     assert mgr.wait_for(recovered, timeout=90)
+
+
+def test_readiness_gates_hold_ready(mgr):
+    """spec.readinessGates (reference roleinstance_types.go): the instance
+    stays NotReady until every gate condition is True."""
+    import time
+    from rbg_amd.api import constants as C
+    from rbg_amd.api.types import Condition, get_condition, set_condition
+    from tests.test_controller_e2e import router_worker_rbg, rbg_ready
+    mgr.store.create(router_worker_rbg(name="gated"))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "gated"), timeout=60)
+    insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                           selector={C.LABEL_GROUP_NAME: "gated",
+                                     C.LABEL_ROLE_NAME: "worker"})
+    name = insts[0].metadata.name
+
+    def add_gate(cur):
+        cur.spec.readiness_gates = ["CacheWarm"]
+        return cur
+    mgr.store.apply(C.KIND_ROLE_INSTANCE, name, add_gate)
+
+    def inst_ready():
+        i = mgr.store.get(C.KIND_ROLE_INSTANCE, name)
+        c = get_condition(i.status.conditions, C.COND_READY)
+        return c is not None and c.status == "True"
+    assert mgr.wait_for(lambda: not inst_ready(), timeout=30)
+
+    def satisfy(cur):
+        set_condition(cur.status.conditions,
+                      Condition.new("CacheWarm", True, "Warm", ""))
+        return cur
+    mgr.store.apply(C.KIND_ROLE_INSTANCE, name, satisfy, subresource="status")
+    assert mgr.wait_for(inst_ready, timeout=30)
