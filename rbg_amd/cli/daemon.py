@@ -156,7 +156,14 @@ class Daemon:
         signal.signal(signal.SIGTERM, on_sig)
         signal.signal(signal.SIGINT, on_sig)
         stop.wait()
+        self.close()
+
+    def close(self) -> None:
         self.rpc.stop()
+        try:
+            self._watch_stream.stop()
+        except Exception:
+            pass
         self.manager.stop()
 
 

@@ -24,8 +24,7 @@ def daemon(tmp_run_dir):
     d.manager.start()
     d.rpc.start()
     yield d
-    d.rpc.stop()
-    d.manager.stop()
+    d.close()
 
 
 def run_cli(daemon, *argv):
