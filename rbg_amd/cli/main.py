@@ -115,6 +115,35 @@ def cmd_status(client: BaseClient, args) -> int:
     return 0
 
 
+def cmd_describe(client: BaseClient, args) -> int:
+    """kubectl-describe analog: spec summary + conditions + recent events
+    about the object."""
+    kind = _resolve_kind(args.kind)
+    obj = client.get(kind, args.name, args.namespace)
+    if obj is None:
+        print(f"{args.kind}/{args.name} not found", file=sys.stderr)
+        return 1
+    print(f"Name:      {obj.metadata.name}")
+    print(f"Namespace: {obj.metadata.namespace}")
+    print(f"Kind:      {obj.kind}")
+    if obj.metadata.labels:
+        print(f"Labels:    {obj.metadata.labels}")
+    for cond in getattr(obj.status, "conditions", []):
+        print(f"Condition: {cond.type}={cond.status} ({cond.reason}) "
+              f"{cond.message}")
+    import time as _t
+    evs = [e for e in client.list(C.KIND_EVENT, args.namespace)
+           if e.involved_object.kind == kind and
+           e.involved_object.name == args.name]
+    evs.sort(key=lambda e: e.last_timestamp)
+    if evs:
+        print("Events:")
+        rows = [[f"{_t.time() - e.last_timestamp:.0f}s", e.type, e.reason,
+                 e.count, e.message[:60]] for e in evs[-15:]]
+        _print_table(rows, ["AGE", "TYPE", "REASON", "COUNT", "MESSAGE"])
+    return 0
+
+
 def cmd_scale(client: BaseClient, args) -> int:
     client.scale(args.name, args.replicas, args.namespace)
     print(f"scalingadapter/{args.name} scaled to {args.replicas}")
@@ -207,6 +236,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("name", nargs="?")
     p = sub.add_parser("status")
     p.add_argument("name")
+    p = sub.add_parser("describe")
+    p.add_argument("kind")
+    p.add_argument("name")
     p = sub.add_parser("scale")
     p.add_argument("name")
     p.add_argument("--replicas", type=int, required=True)
@@ -227,7 +259,8 @@ def main(argv: Optional[List[str]] = None,
         client = RemoteClient(args.host, args.port)
     return {
         "apply": cmd_apply, "get": cmd_get, "status": cmd_status,
-        "scale": cmd_scale, "delete": cmd_delete, "rollout": cmd_rollout,
+        "describe": cmd_describe, "scale": cmd_scale, "delete": cmd_delete,
+        "rollout": cmd_rollout,
     }[args.cmd](client, args)
 
 

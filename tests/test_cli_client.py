@@ -164,6 +164,9 @@ def test_events_recorded_and_listed(tmp_run_dir):
         assert len(pings) == 1 and pings[0].count == 2
         rc = ctl_main(["get", "events"], client=InProcessClient(m.store))
         assert rc == 0
+        rc = ctl_main(["describe", "rbg", "evt"],
+                      client=InProcessClient(m.store))
+        assert rc == 0
     finally:
         m.stop()
 
