@@ -171,6 +171,8 @@ def main(argv: Optional[List[str]] = None) -> int:
     ap = argparse.ArgumentParser(prog="rbgd")
     ap.add_argument("--port", type=int, default=DEFAULT_PORT)
     ap.add_argument("--run-root", default="/tmp/rbg-run")
+    ap.add_argument("--persist-dir", default="",
+                    help="mirror objects to JSON for restart recovery")
     ap.add_argument("--gpus", type=int, default=0, help="0 = discover")
     ap.add_argument("--gang-timeout", type=float, default=30.0)
     ap.add_argument("--history-limit", type=int, default=10)
@@ -181,7 +183,8 @@ def main(argv: Optional[List[str]] = None) -> int:
         format="%(asctime)s %(name)s %(levelname)s %(message)s")
     daemon = Daemon(ManagerOptions(
         run_root=args.run_root, num_gpus=args.gpus,
-        gang_timeout=args.gang_timeout, history_limit=args.history_limit),
+        gang_timeout=args.gang_timeout, history_limit=args.history_limit,
+        persist_dir=args.persist_dir),
         port=args.port)
     daemon.run_forever()
     return 0

@@ -113,6 +113,7 @@ class ManagerOptions:
     port_range: Tuple[int, int] = (30000, 40000)
     history_limit: int = 10
     concurrency: int = 4          # reconcile workers per controller
+    persist_dir: str = ""         # "" = in-memory only (etcd analog off)
 
 
 class Manager:
@@ -120,6 +121,11 @@ class Manager:
                  topo: Optional[NodeTopology] = None):
         self.opts = opts or ManagerOptions()
         self.store = Store()
+        self.persister = None
+        if self.opts.persist_dir:
+            from ..store.persist import StorePersister
+            self.persister = StorePersister(self.store, self.opts.persist_dir)
+            self.persister.restore()   # before controllers see the store
         if topo is not None:
             self.topo = topo
         elif self.opts.num_gpus:
@@ -202,6 +208,8 @@ class Manager:
 
     def start(self) -> None:
         self._stop.clear()
+        if self.persister is not None:
+            self.persister.start()
         watch = self.store.watch(replay=True)
 
         def watch_loop():
@@ -255,6 +263,8 @@ class Manager:
 
     def stop(self, teardown: bool = True) -> None:
         self._stop.set()
+        if self.persister is not None:
+            self.persister.stop()
         for t in self._threads:
             t.join(timeout=2.0)
         self._threads.clear()
