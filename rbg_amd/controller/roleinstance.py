@@ -48,6 +48,7 @@ class InstanceRuntime:
     master_port: int = 0
     recovery_started: float = 0.0        # failure detected, not Ready yet
     applied_args: Dict[str, str] = field(default_factory=dict)  # in-place
+    unschedulable: str = ""              # last gang-reserve failure message
 
 
 class RoleInstanceController:
@@ -165,9 +166,11 @@ class RoleInstanceController:
             try:
                 self._ensure_gang(inst, rt, desired)
             except GangUnschedulable as e:
+                rt.unschedulable = str(e)
                 self._set_condition(inst, C.COND_READY, False, "Unschedulable", str(e))
                 self.recorder.warning(inst, "Unschedulable", str(e))
                 return 1.0
+            rt.unschedulable = ""
             from ..discovery import component as comp_disc
             deps = comp_disc.parse_depends_on(inst.metadata.annotations)
             for comp in inst.spec.components:
@@ -472,6 +475,10 @@ class RoleInstanceController:
                 gc = get_condition(cur.status.conditions, gate)
                 if gc is None or gc.status != "True":
                     gates_ok = False
+            if rt.unschedulable and not all_ready:
+                set_condition(cur.status.conditions, Condition.new(
+                    C.COND_READY, False, "Unschedulable", rt.unschedulable))
+                return cur
             set_condition(cur.status.conditions, Condition.new(
                 C.COND_READY, all_ready and not is_restarting and gates_ok,
                 "Ready" if (all_ready and gates_ok) else

@@ -264,3 +264,40 @@ def _debug_dump(m):
             for w in getattr(o.status, "workers", []):
                 lines.append(f"   worker {w.name} phase={w.phase} pid={w.pid}")
     return "\n".join(lines)
+
+
+def test_unschedulable_gang_sets_condition_and_event(mgr):
+    """A role demanding more GPUs than the node has: the gang allocator
+    refuses atomically, the instance reports Unschedulable, and a Warning
+    event is recorded (reference: PodGroup stays Pending)."""
+    from rbg_amd.api.types import (EngineResources, EngineSpec,
+                                   EngineTemplate, RoleBasedGroup,
+                                   RoleBasedGroupSpec, RoleSpec, ObjectMeta,
+                                   get_condition)
+    # each role alone is feasible (5 <= 8) but together they oversubscribe
+    # the node, so the second gang reservation must time out
+    def gpu_role(name):
+        return RoleSpec(
+            name=name, replicas=1,
+            template=EngineTemplate(engines=[EngineSpec(
+                name="e", runner="echo",
+                resources=EngineResources(gpus=5))]))
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="toobig"),
+        spec=RoleBasedGroupSpec(roles=[gpu_role("a"), gpu_role("b")]))
+    mgr.store.create(rbg)
+
+    def unschedulable():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                               selector={C.LABEL_GROUP_NAME: "toobig"})
+        for i in insts:
+            c = get_condition(i.status.conditions, C.COND_READY)
+            if c is not None and c.reason == "Unschedulable":
+                return True
+        return False
+    assert mgr.wait_for(unschedulable, timeout=90)
+
+    def warned():
+        return any(e.reason == "Unschedulable" and e.type == "Warning"
+                   for e in mgr.store.list(C.KIND_EVENT))
+    assert mgr.wait_for(warned, timeout=30)
