@@ -66,6 +66,8 @@ def test_scaling_adapter_drives_role_replicas(mgr):
                                          C.LABEL_ROLE_NAME: "worker"})
         return len(insts) == 3
     assert mgr.wait_for(scaled, timeout=60)
-    cur = mgr.store.get(C.KIND_SCALING_ADAPTER, "auto-worker")
-    assert cur.status.replicas == 3
-    assert cur.status.last_scale_time > 0
+
+    def status_converged():
+        cur = mgr.store.get(C.KIND_SCALING_ADAPTER, "auto-worker")
+        return cur.status.replicas == 3 and cur.status.last_scale_time > 0
+    assert mgr.wait_for(status_converged, timeout=30)
