@@ -155,3 +155,34 @@ def test_v1alpha1_rbgset_conversion():
     assert obj.spec.replicas == 3
     role = obj.spec.template.roles[0]
     assert role.name == "w" and role.pattern == C.PATTERN_STANDALONE
+
+
+def test_cli_apply_legacy_yaml(tmp_path, tmp_run_dir):
+    """rbgctl apply -f legacy-v1alpha1.yaml converts at the API boundary
+    and the controllers run the converted group to Ready."""
+    import yaml as _yaml
+    from rbg_amd.cli.main import main as ctl_main
+    from rbg_amd.client.client import InProcessClient
+    from rbg_amd.controller.manager import Manager, ManagerOptions
+    from tests.test_controller_e2e import rbg_ready
+    doc = {"apiVersion": "workloads.x-k8s.io/v1alpha1",
+           "kind": "RoleBasedGroup",
+           "metadata": {"name": "legacy"},
+           "spec": {"roles": [
+               {"name": "w", "replicas": 1,
+                "workload": {"apiVersion": "apps/v1", "kind": "StatefulSet"},
+                "template": {"engines": [
+                    {"name": "e", "runner": "echo",
+                     "resources": {"cpuOnly": True}}]}}]}}
+    f = tmp_path / "legacy.yaml"
+    f.write_text(_yaml.safe_dump(doc))
+    m = Manager(ManagerOptions(run_root=tmp_run_dir, num_gpus=8,
+                               resync_period=0.1))
+    m.start()
+    try:
+        rc = ctl_main(["apply", "-f", str(f)],
+                      client=InProcessClient(m.store))
+        assert rc == 0
+        assert m.wait_for(lambda: rbg_ready(m, "legacy"), timeout=60)
+    finally:
+        m.stop()
