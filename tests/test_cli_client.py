@@ -189,3 +189,26 @@ def test_remote_watch_events(daemon):
             break
         time.sleep(0.1)
     assert seen and seen[0]["type"] == "ADDED"
+
+
+def test_remote_watch_generator(daemon):
+    """remote_watch yields events as they happen (informer loop)."""
+    import threading
+    from rbg_amd.client.client import RemoteClient, remote_watch
+    c = RemoteClient("127.0.0.1", daemon.port)
+    got = []
+    done = threading.Event()
+
+    def consume():
+        for ev in remote_watch(c, kinds=["RoleBasedGroup"], poll_s=0.05):
+            got.append(ev)
+            if ev["name"] == "watch-gen":
+                done.set()
+                return
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    import time
+    time.sleep(0.3)
+    daemon.manager.store.create(make_rbg("watch-gen"))
+    assert done.wait(timeout=10), got
+    assert got[-1]["kind"] == "RoleBasedGroup"
