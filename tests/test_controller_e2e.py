@@ -301,3 +301,48 @@ def test_unschedulable_gang_sets_condition_and_event(mgr):
         return any(e.reason == "Unschedulable" and e.type == "Warning"
                    for e in mgr.store.list(C.KIND_EVENT))
     assert mgr.wait_for(warned, timeout=30)
+
+
+def test_rolling_recreate_respects_max_unavailable(mgr):
+    """A non-in-place-able template change (component size change) rolls
+    through recreates with maxUnavailable=1: ready replicas never dip
+    below replicas-1, and every instance converges to the new revision."""
+    from rbg_amd.api.types import get_condition
+    rbg = router_worker_rbg(name="roll", worker_replicas=3)
+    role = rbg.spec.role("worker")
+    role.update_strategy_type = C.UPDATE_RECREATE
+    role.rollout_strategy.rolling_update.max_unavailable = 1
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "roll"), timeout=90)
+
+    def worker_instances():
+        return mgr.store.list(C.KIND_ROLE_INSTANCE,
+                              selector={C.LABEL_GROUP_NAME: "roll",
+                                        C.LABEL_ROLE_NAME: "worker"})
+    old_uids = {i.metadata.uid for i in worker_instances()}
+
+    def bump(cur):
+        eng = cur.spec.role("worker").template.engines[0]
+        eng.args = dict(eng.args, generation="2")
+        return cur
+    mgr.store.apply(C.KIND_RBG, "roll", bump)
+
+    import time as _t
+    min_ready = 3
+    deadline = _t.time() + 120
+    while _t.time() < deadline:
+        insts = worker_instances()
+        ready = sum(1 for i in insts
+                    if (get_condition(i.status.conditions, C.COND_READY)
+                        or None) is not None and
+                    get_condition(i.status.conditions,
+                                  C.COND_READY).status == "True")
+        min_ready = min(min_ready, ready)
+        if all(i.metadata.uid not in old_uids for i in insts) and \
+                len(insts) == 3 and ready == 3:
+            break
+        _t.sleep(0.05)
+    insts = worker_instances()
+    assert all(i.metadata.uid not in old_uids for i in insts), \
+        "rollout did not complete"
+    assert min_ready >= 2, f"availability dipped to {min_ready}"
