@@ -210,3 +210,50 @@ def test_pp2_matches_pp1(tmp_path):
     out0 = torch.load(str(tmp_path / "pp_rank0.pt"))
     out1 = torch.load(str(tmp_path / "pp_rank1.pt"))
     assert out0 == out1 == ref, (ref, out0, out1)
+
+
+def _pp_chunked_worker(rank, world, port, result_dir):
+    """PP=2 with chunked prefill (tiny max_prefill_tokens): hidden states
+    hop per CHUNK and the final tokens still match single-process."""
+    dist = _init(rank, world, port)
+    from rbg_amd.engine.config import EngineConfig, ModelConfig
+    from rbg_amd.engine.engine import LLMEngine
+    from rbg_amd.engine.sequence import SamplingParams
+    from rbg_amd.models.llama import PPContext
+    cfg = ModelConfig.preset("tiny")
+    pp = PPContext(size=world, stage=rank, instance_ranks=list(range(world)),
+                   tp_size=1, group=None)
+    ecfg = EngineConfig(model=cfg, device="cpu", kv_pool_tokens=2048,
+                        enforce_eager=True, max_prefill_tokens=16)
+    eng = LLMEngine(ecfg, None, pp)
+    prompt = list(range(1, 41))          # 40 tokens -> 3 chunks of <=16
+    s = eng.add_request(prompt, SamplingParams(max_new_tokens=4,
+                                               temperature=0.0))
+    for _ in range(40):
+        eng.step()
+        if s.status == "finished":
+            break
+    torch.save(s.output_tokens, os.path.join(result_dir,
+                                             f"ppc_rank{rank}.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_pp2_chunked_prefill_matches(tmp_path):
+    _run_dist(_pp_chunked_worker, 2, _free_port(), str(tmp_path))
+    from rbg_amd.engine.config import EngineConfig, ModelConfig
+    from rbg_amd.engine.engine import LLMEngine
+    from rbg_amd.engine.sequence import SamplingParams
+    eng = LLMEngine(EngineConfig(model=ModelConfig.preset("tiny"),
+                                 device="cpu", kv_pool_tokens=2048,
+                                 enforce_eager=True, max_prefill_tokens=16))
+    prompt = list(range(1, 41))
+    s = eng.add_request(prompt, SamplingParams(max_new_tokens=4,
+                                               temperature=0.0))
+    for _ in range(40):
+        eng.step()
+        if s.status == "finished":
+            break
+    ref = s.output_tokens
+    assert torch.load(str(tmp_path / "ppc_rank0.pt")) == ref
+    assert torch.load(str(tmp_path / "ppc_rank1.pt")) == ref
