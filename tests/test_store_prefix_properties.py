@@ -131,3 +131,37 @@ def test_gang_allocator_concurrent_invariants():
         t.join()
     assert not violations, violations[:3]
     assert sorted(alloc.free_gpus()) == list(range(8))
+
+
+@settings(max_examples=40, deadline=None)
+@given(seed=st.integers(min_value=0, max_value=10**6))
+def test_serde_roundtrip_random_rbg(seed):
+    """asdict/fromdict round-trip over randomized RoleBasedGroups: the
+    serialized form (camelCase YAML shape) must rebuild identically."""
+    import random
+    from rbg_amd.api.serde import asdict, fromdict
+    from rbg_amd.api.types import (EngineResources, EngineSpec,
+                                   EngineTemplate, EnvVar,
+                                   LeaderWorkerPattern, RoleSpec)
+    rng = random.Random(seed)
+    roles = []
+    for i in range(rng.randint(1, 4)):
+        eng = EngineSpec(
+            name=f"e{i}", runner=rng.choice(["echo", "llm-engine"]),
+            args={"k": rng.randint(0, 9), "mode": "colocated"},
+            env=[EnvVar(name="A", value=str(rng.random()))],
+            resources=EngineResources(gpus=rng.randint(0, 4),
+                                      hbm_bytes=rng.randint(0, 1 << 30)))
+        role = RoleSpec(name=f"r{i}", replicas=rng.randint(1, 5),
+                        template=EngineTemplate(engines=[eng]),
+                        min_ready_seconds=rng.randint(0, 9))
+        if rng.random() < 0.4:
+            role.pattern = C.PATTERN_LEADER_WORKER
+            role.leader_worker_pattern = LeaderWorkerPattern(
+                size=rng.randint(1, 4))
+        roles.append(role)
+    rbg = RoleBasedGroup(metadata=ObjectMeta(name=f"x{seed % 97}"),
+                         spec=RoleBasedGroupSpec(roles=roles))
+    doc = asdict(rbg)
+    back = fromdict(RoleBasedGroup, doc)
+    assert asdict(back) == doc
