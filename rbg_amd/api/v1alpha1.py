@@ -71,6 +71,19 @@ def set_to_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
                      "template": conv["spec"]}}
 
 
+def set_from_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
+    """v1alpha2 RoleBasedGroupSet doc → v1alpha1 for legacy clients."""
+    spec = doc.get("spec") or {}
+    inner = {"apiVersion": C.API_VERSION, "kind": C.KIND_RBG,
+             "metadata": dict(doc.get("metadata") or {}),
+             "spec": spec.get("template") or {}}
+    conv = from_v2(inner)
+    return {"apiVersion": API_VERSION_V1ALPHA1, "kind": C.KIND_RBG_SET,
+            "metadata": conv["metadata"],
+            "spec": {"replicas": spec.get("replicas", 1),
+                     "template": conv["spec"]}}
+
+
 def _merge_patch(base: Dict[str, Any], patch: Dict[str, Any]) -> Dict[str, Any]:
     """JSON-merge-patch-lite used for patchLeaderTemplate/patchWorkerTemplate
     over the role template (good enough for EngineTemplate docs: dicts merge

@@ -186,3 +186,22 @@ def test_cli_apply_legacy_yaml(tmp_path, tmp_run_dir):
         assert m.wait_for(lambda: rbg_ready(m, "legacy"), timeout=60)
     finally:
         m.stop()
+
+
+def test_rbgset_round_trip():
+    from rbg_amd.api.v1alpha1 import set_from_v2, set_to_v2
+    doc = {"apiVersion": "workloads.x-k8s.io/v1alpha1",
+           "kind": "RoleBasedGroupSet",
+           "metadata": {"name": "fleet"},
+           "spec": {"replicas": 2,
+                    "template": {"roles": [
+                        {"name": "w", "replicas": 1,
+                         "workload": {"apiVersion": "apps/v1",
+                                      "kind": "Deployment"},
+                         "template": {"engines": [
+                             {"name": "e", "runner": "echo"}]}}]}}}
+    v2 = set_to_v2(copy.deepcopy(doc))
+    back = set_from_v2(v2)
+    assert back["spec"]["replicas"] == 2
+    assert back["spec"]["template"]["roles"][0]["workload"]["kind"] == \
+        "Deployment"
