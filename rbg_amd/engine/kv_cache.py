@@ -37,9 +37,22 @@ class PagedKVCache:
             num_pages = self._size_pool(cfg, device, layers)
         self.num_pages = num_pages
         kvh = m.num_kv_heads // max(1, cfg.tp_size)   # TP shards the heads
-        self.kv = torch.zeros(
-            (layers, 2, num_pages, kvh, cfg.page_size, m.head_dim),
-            dtype=torch.bfloat16, device=device)
+        shape = (layers, 2, num_pages, kvh, cfg.page_size, m.head_dim)
+        self.ipc_exportable = False
+        if device.type == "cuda":
+            # hipMalloc base allocation (not the caching allocator) so the
+            # pool is hipIpc-exportable: prefill engines map it and push
+            # migrated pages straight over xGMI (parallel/kv_peer.py)
+            from .. import ops
+            if ops.HAVE_HIP:
+                self.kv = ops._require_hip().ipc_alloc_bf16(list(shape))
+                self.kv.zero_()
+                self.ipc_exportable = True
+            else:
+                self.kv = torch.zeros(shape, dtype=torch.bfloat16,
+                                      device=device)
+        else:
+            self.kv = torch.zeros(shape, dtype=torch.bfloat16, device=device)
         # page 0 is reserved scratch: hipGraph decode padding rows write
         # their (dead) KV slot there (model_runner._decode_graph)
         self._free: List[int] = list(range(num_pages - 1, 0, -1))

@@ -160,7 +160,31 @@ class ServeWorker:
         self._finished_pages: Dict[int, List[int]] = {}
         self._plan: Optional[_TransferPlan] = None
         self._transfer = None
+        # xGMI fast path (parallel/kv_peer.py): direct hipIpc page push for
+        # the non-TP P/D pair; negotiated per sequence in import_seq.
+        self._peer_pusher = None
+        self._pending_imports: Dict[int, tuple] = {}
         self._register_rpc()
+
+    # -- xGMI peer push ------------------------------------------------------
+
+    def _peer_possible(self) -> bool:
+        from ..parallel.kv_peer import peer_capable
+        return (self.tp is None and
+                peer_capable(self.engine.runner.cache))
+
+    def _peer_push(self, pages: List[int], peer_meta: Dict[str, Any],
+                   dst_pages: List[int]) -> None:
+        """Push my pages into the decode pool over xGMI and wait for the
+        copy event.  Runs in the RPC handler thread; the engine loop keeps
+        launching decode work on the default stream meanwhile."""
+        from ..parallel.kv_peer import PeerKVPusher
+        if self._peer_pusher is None:
+            self._peer_pusher = PeerKVPusher(
+                self.engine.runner.cache.kv.device)
+        pending = self._peer_pusher.push(
+            self.engine.runner.cache, pages, peer_meta, dst_pages)
+        pending.wait()
 
     # lazy: the discovery config lists peer instances only once the
     # controller has created them (same dependency wave) — re-read until
@@ -204,23 +228,29 @@ class ServeWorker:
 
     def _send_pages(self, pages: List[int], dst_rank: int) -> None:
         import torch.distributed as dist
+
+        from ..parallel.comm import to_wire
         cache = self.engine.runner.cache
         with self._xfer_lock:
             idx = torch.tensor(pages, dtype=torch.int64, device=cache.kv.device)
             buf = cache.kv.index_select(2, idx).contiguous()
             if buf.device != self._xfer_buf_device():
                 buf = buf.to(self._xfer_buf_device())
-            dist.send(buf, dst_rank)
+            dist.send(to_wire(buf), dst_rank)
 
     def _recv_pages(self, pages: List[int], src_rank: int) -> None:
         import torch.distributed as dist
+
+        from ..parallel.comm import from_wire, wire_dtype
         cache = self.engine.runner.cache
         with self._xfer_lock:
             m = cache.kv.shape
+            dev = self._xfer_buf_device()
             buf = torch.empty((m[0], m[1], len(pages), m[3], m[4], m[5]),
-                              dtype=cache.kv.dtype,
-                              device=self._xfer_buf_device())
+                              dtype=wire_dtype(cache.kv.dtype, dev),
+                              device=dev)
             dist.recv(buf, src_rank)
+            buf = from_wire(buf, cache.kv.dtype)
             idx = torch.tensor(pages, dtype=torch.int64, device=cache.kv.device)
             cache.kv.index_copy_(2, idx, buf.to(cache.kv.device))
 
@@ -352,6 +382,7 @@ class ServeWorker:
             self.rpc.register("prefill", self._rpc_prefill)
         if self.mode == "decode":
             self.rpc.register("import_seq", self._rpc_import_seq)
+            self.rpc.register("import_commit", self._rpc_import_commit)
             self.rpc.register("resolve_ticket", self._rpc_resolve_ticket)
 
     def _rpc_reload(self, seed) -> None:
@@ -464,13 +495,21 @@ class ServeWorker:
         meta = dict(tokens=tokens, first_token=first_token,
                     num_pages=len(pages), max_new_tokens=max_new_tokens,
                     temperature=temperature, arrival_time=seq.arrival_time)
+        peer_ok = self._peer_possible()
         if self.gcomm is not None:
             client = RpcClient("127.0.0.1",
                                self._decode_port(decode_instance))
             try:
-                res = client.call("import_seq",
+                res = client.call("import_seq", peer_ok=peer_ok,
                                   src_instance=self.my_instance, **meta)
-                if self.tp is not None:
+                if res.get("peer") is not None:
+                    # xGMI fast path (non-TP only — _peer_possible): push
+                    # straight into the decode pool, then commit; no
+                    # collective in the path
+                    self._peer_push(pages, res["peer"], res["dst_pages"])
+                    client.call("import_commit", seq_id=res["seq_id"])
+                    self._release_parked(parked)
+                elif self.tp is not None:
                     # every rank sends its KV shard to its counterpart;
                     # followers park identical page ids (lockstep)
                     self._tp_submit_op({"kind": "migrate",
@@ -492,10 +531,14 @@ class ServeWorker:
         client = RpcClient("127.0.0.1", int(dinst["ports"][0]))
         try:
             res = client.call(
-                "import_seq",
+                "import_seq", peer_ok=peer_ok,
                 src_rank=plan.rank_of(self.my_instance), **meta)
-            self.transfer.send_pages(self.engine.runner.cache, pages,
-                                     plan.rank_of(decode_instance))
+            if res.get("peer") is not None:
+                self._peer_push(pages, res["peer"], res["dst_pages"])
+                client.call("import_commit", seq_id=res["seq_id"])
+            else:
+                self.transfer.send_pages(self.engine.runner.cache, pages,
+                                         plan.rank_of(decode_instance))
         finally:
             self._release_parked(parked)
             client.close()
@@ -529,6 +572,10 @@ class ServeWorker:
         else:
             self.transfer.recv_pages(self.engine.runner.cache,
                                      seq.block_table.pages, recv_from)
+        self._import_enqueue(seq, first_token, max_new_tokens, arrival_time)
+
+    def _import_enqueue(self, seq: Sequence, first_token: int,
+                        max_new_tokens: int, arrival_time: float) -> None:
         # running-list append must not race finish_decode; pause first so
         # the busy engine loop actually yields the lock (convoy avoidance)
         self._pause.set()
@@ -547,7 +594,20 @@ class ServeWorker:
                         num_pages: int, max_new_tokens: int,
                         src_rank: int = -1, src_instance: str = "",
                         temperature: float = 0.0,
-                        arrival_time: float = 0.0) -> Dict[str, Any]:
+                        arrival_time: float = 0.0,
+                        peer_ok: bool = False) -> Dict[str, Any]:
+        if peer_ok and self._peer_possible():
+            # xGMI fast path: hand the sender my pool handle + target pages;
+            # it pushes directly and then calls import_commit
+            from ..parallel import kv_peer
+            seq = self._import_alloc(tokens, num_pages, max_new_tokens,
+                                     temperature)
+            self._pending_imports[seq.seq_id] = (
+                seq, int(first_token), int(max_new_tokens),
+                float(arrival_time))
+            return {"seq_id": seq.seq_id,
+                    "peer": kv_peer.export_meta(self.engine.runner.cache),
+                    "dst_pages": list(seq.block_table.pages)}
         if self.tp is not None:
             # lockstep import: every decode rank allocates identical pages
             # and receives from its prefill counterpart inside the apply
@@ -575,6 +635,15 @@ class ServeWorker:
                 log.exception("KV import failed")
         # post the recv asynchronously, THEN return so the send can start
         threading.Thread(target=do_recv, daemon=True).start()
+        return {"seq_id": seq.seq_id}
+
+    def _rpc_import_commit(self, seq_id: int) -> Dict[str, Any]:
+        """Sender finished its xGMI push (copy event resolved on its side):
+        enqueue the sequence.  The pages were written remotely; no local
+        copy or CU work happened here."""
+        seq, first_token, max_new_tokens, arrival_time = \
+            self._pending_imports.pop(int(seq_id))
+        self._import_enqueue(seq, first_token, max_new_tokens, arrival_time)
         return {"seq_id": seq.seq_id}
 
     def _rpc_resolve_ticket(self, ticket: str) -> Dict[str, Any]:

@@ -107,17 +107,16 @@ class PPContext:
         return self.ranks[stage * self.tp_size + lane]
 
     def send_next(self, t: torch.Tensor) -> None:
-        t = t.contiguous()
-        if t.device.type == "cpu" and t.dtype == torch.bfloat16:
-            t = t.view(torch.int16)   # gloo lacks bf16; bit-exact reinterp
-        torch.distributed.send(t, dst=self._lane_rank(self.stage + 1))
+        from ..parallel.comm import to_wire
+        torch.distributed.send(to_wire(t.contiguous()),
+                               dst=self._lane_rank(self.stage + 1))
 
     def recv_prev(self, shape, dtype, device) -> torch.Tensor:
-        wire = torch.int16 if (device.type == "cpu" and
-                               dtype == torch.bfloat16) else dtype
-        t = torch.empty(shape, dtype=wire, device=device)
+        from ..parallel.comm import from_wire, wire_dtype
+        t = torch.empty(shape, dtype=wire_dtype(dtype, device),
+                        device=device)
         torch.distributed.recv(t, src=self._lane_rank(self.stage - 1))
-        return t.view(dtype)
+        return from_wire(t, dtype)
 
     def broadcast_tokens(self, t: torch.Tensor) -> torch.Tensor:
         """Deliver last-stage sampled ids to every rank of the instance."""

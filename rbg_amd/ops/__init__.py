@@ -22,6 +22,37 @@ except Exception as e:  # noqa: BLE001
 HAVE_HIP = _hip is not None
 
 
+def _check_stale() -> None:
+    """Fail loudly on a GPU box if the shipped .so predates the kernel
+    sources (stale-binary guard; the stamp is written by
+    __graft_entry__.build())."""
+    import hashlib
+    import os
+    here = os.path.dirname(os.path.abspath(__file__))
+    stamp = os.path.join(here, "_hip_ops.srchash")
+    if not os.path.exists(stamp):
+        return
+    h = hashlib.sha256()
+    hip_dir = os.path.join(here, "hip")
+    for name in sorted(os.listdir(hip_dir)):
+        if name.endswith("_hip.hip"):
+            continue
+        if name.endswith((".hip", ".cpp", ".h")):
+            with open(os.path.join(hip_dir, name), "rb") as f:
+                h.update(name.encode() + b"\0" + f.read() + b"\0")
+    with open(stamp) as f:
+        recorded = f.read().strip()
+    if recorded and recorded != h.hexdigest():
+        raise RuntimeError(
+            "rbg_amd HIP extension is STALE: kernel sources changed after "
+            "the last build.  Rebuild in-tree: PYTORCH_ROCM_ARCH=gfx950 "
+            "python setup.py build_ext --inplace")
+
+
+if HAVE_HIP and torch.cuda.is_available():
+    _check_stale()
+
+
 def _require_hip():
     if _hip is None:
         raise RuntimeError(

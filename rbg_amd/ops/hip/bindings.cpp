@@ -5,6 +5,9 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstring>
+#include <vector>
+
 extern "C" {
 void launch_rmsnorm(void*, void*, const void*, float, int, int, hipStream_t);
 void launch_fused_add_rmsnorm(void*, void*, const void*, float, int, int,
@@ -24,6 +27,8 @@ void launch_mfma_probe(void*, const void*, const void*, int, int, hipStream_t);
 void launch_skinny_gemm(void*, void*, const void*, const void*, int, int,
                         int, int, hipStream_t);
 void launch_reduce_splits(void*, const void*, int, long, hipStream_t);
+void launch_kv_peer_copy(void*, const void*, const void*, const void*, int,
+                         int, int, long, long, long, long, hipStream_t);
 }
 
 namespace {
@@ -238,6 +243,83 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w,
   return c;
 }
 
+// ---- KV-pool IPC export + xGMI peer push (parallel/kv_peer.py) -----------
+// The decode engine's KV pool is allocated with hipMalloc directly (not the
+// caching allocator) so data_ptr() IS the allocation base hipIpcGetMemHandle
+// requires; the prefill process maps it with lazy peer access (the xGMI
+// route) and pushes pages with the kv_peer_copy kernel.
+
+torch::Tensor ipc_alloc_bf16(std::vector<int64_t> sizes) {
+  int64_t numel = 1;
+  for (auto s : sizes) numel *= s;
+  void* ptr = nullptr;
+  hipError_t err = hipMalloc(&ptr, numel * 2);
+  TORCH_CHECK(err == hipSuccess, "hipMalloc(", numel * 2,
+              " B) failed: ", hipGetErrorString(err));
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  auto opts = torch::TensorOptions()
+                  .dtype(torch::kBFloat16)
+                  .device(torch::kCUDA, dev);
+  return torch::from_blob(
+      ptr, sizes, [](void* p) { (void)hipFree(p); }, opts);
+}
+
+py::bytes kv_ipc_export(torch::Tensor pool) {
+  TORCH_CHECK(pool.is_cuda(), "pool must be on GPU");
+  hipIpcMemHandle_t handle;
+  hipError_t err = hipIpcGetMemHandle(&handle, pool.data_ptr());
+  TORCH_CHECK(err == hipSuccess,
+              "hipIpcGetMemHandle failed (pool must be an ipc_alloc_bf16 "
+              "base allocation): ", hipGetErrorString(err));
+  return py::bytes(reinterpret_cast<const char*>(&handle), sizeof(handle));
+}
+
+int64_t kv_ipc_open(py::bytes handle_bytes) {
+  std::string raw = handle_bytes;
+  TORCH_CHECK(raw.size() == sizeof(hipIpcMemHandle_t), "bad handle size");
+  hipIpcMemHandle_t handle;
+  memcpy(&handle, raw.data(), sizeof(handle));
+  void* ptr = nullptr;
+  hipError_t err =
+      hipIpcOpenMemHandle(&ptr, handle, hipIpcMemLazyEnablePeerAccess);
+  TORCH_CHECK(err == hipSuccess,
+              "hipIpcOpenMemHandle failed: ", hipGetErrorString(err));
+  return reinterpret_cast<int64_t>(ptr);
+}
+
+void kv_ipc_close(int64_t ptr) {
+  hipError_t err = hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr));
+  TORCH_CHECK(err == hipSuccess,
+              "hipIpcCloseMemHandle failed: ", hipGetErrorString(err));
+}
+
+void kv_peer_copy(int64_t dst_base, torch::Tensor src_kv,
+                  torch::Tensor src_pages, torch::Tensor dst_pages,
+                  int64_t dst_num_pages) {
+  // src_kv: [L, 2, Ps, kvh, ps, hd] bf16 contiguous; dst pool has the SAME
+  // per-page layout but its own page count (dst_num_pages)
+  check_bf16_contig(src_kv, "src_kv");
+  TORCH_CHECK(src_kv.dim() == 6 && src_kv.size(1) == 2, "kv must be 6-D");
+  TORCH_CHECK(src_pages.is_cuda() && dst_pages.is_cuda() &&
+                  src_pages.scalar_type() == torch::kInt32 &&
+                  dst_pages.scalar_type() == torch::kInt32,
+              "page lists must be int32 on GPU");
+  TORCH_CHECK(src_pages.numel() == dst_pages.numel(), "page count mismatch");
+  const int n_pages = src_pages.numel();
+  if (n_pages == 0) return;
+  const int layers = src_kv.size(0);
+  const long Ps = src_kv.size(2);
+  const long chunk_elems = src_kv.size(3) * src_kv.size(4) * src_kv.size(5);
+  TORCH_CHECK(chunk_elems % 8 == 0, "chunk must be 16B-aligned");
+  const long chunk_vec = chunk_elems / 8;  // uint4 = 8 bf16
+  launch_kv_peer_copy(reinterpret_cast<void*>(dst_base), src_kv.data_ptr(),
+                      src_pages.data_ptr(), dst_pages.data_ptr(), n_pages,
+                      2 * layers, (int)chunk_vec, chunk_vec,
+                      chunk_vec, Ps * chunk_vec, dst_num_pages * chunk_vec,
+                      current_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "x,residual += ; rmsnorm");
@@ -248,4 +330,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "decode-shape GEMM (M<=128)",
         py::arg("a"), py::arg("w"), py::arg("force_splits") = 0);
+  m.def("ipc_alloc_bf16", &ipc_alloc_bf16,
+        "hipMalloc-backed bf16 tensor (IPC-exportable base allocation)");
+  m.def("kv_ipc_export", &kv_ipc_export, "hipIpcGetMemHandle of a KV pool");
+  m.def("kv_ipc_open", &kv_ipc_open,
+        "map a peer KV pool (lazy xGMI peer access)");
+  m.def("kv_ipc_close", &kv_ipc_close, "unmap a peer KV pool");
+  m.def("kv_peer_copy", &kv_peer_copy,
+        "gather local pages -> scatter into peer pool over xGMI");
 }

@@ -63,6 +63,28 @@ def init_from_env(backend: Optional[str] = None,
                        device=device)
 
 
+# -- bf16-over-gloo wire format ---------------------------------------------
+# gloo has no bf16 point-to-point support; every CPU-path send of bf16
+# tensors bit-reinterprets to int16 (exact) and the receiver views back.
+# Shared by PPContext hops and the KV-migration fallback so the two paths
+# can never disagree about the wire dtype.
+
+def to_wire(t: torch.Tensor) -> torch.Tensor:
+    if t.device.type == "cpu" and t.dtype == torch.bfloat16:
+        return t.view(torch.int16)
+    return t
+
+
+def wire_dtype(dtype: torch.dtype, device: torch.device) -> torch.dtype:
+    if device.type == "cpu" and dtype == torch.bfloat16:
+        return torch.int16
+    return dtype
+
+
+def from_wire(t: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    return t.view(dtype) if t.dtype != dtype else t
+
+
 def warmup_collectives(ctx: CommContext, sizes=(1 << 10, 1 << 20)) -> None:
     """Prime RCCL channels so the first real collective pays no setup cost
     (the warmup controller's rccl-ring action at group scope)."""

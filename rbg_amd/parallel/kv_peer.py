@@ -1,0 +1,119 @@
+"""Direct xGMI KV-page push between engine processes (prefill -> decode).
+
+The MI355X realization of the reference's Mooncake transfer-engine role
+(reference keps/74-mooncake-integration/README.md:45-140) on its fast path:
+
+  1. the decode engine allocates its KV pool as one hipMalloc base
+     allocation and exports it ONCE with hipIpcGetMemHandle
+     (`export_meta`); the 64-byte handle rides the import_seq RPC reply,
+  2. the prefill engine maps the pool with hipIpcOpenMemHandle
+     (hipIpcMemLazyEnablePeerAccess — this is what routes the mapping over
+     the point-to-point xGMI link) and caches the mapping per peer,
+  3. `push` launches the fused gather->scatter kernel
+     (ops/hip/kv_peer_copy.hip) on a dedicated HIP stream: local pages are
+     read once and stored straight into the decode pool's target pages.
+     No intermediate gather copy, no collective library, and the decode
+     GPU spends zero CU cycles receiving — its engine loop keeps stepping
+     while the bytes stream in (the "overlapped with decode" property of
+     BASELINE.json's north star),
+  4. the returned PendingPush resolves via hipEvent (query/synchronize);
+     the caller then RPCs import_commit so decode enqueues the sequence.
+
+The RCCL send/recv path (kv_transfer.py / serve_worker._send_pages) remains
+the fallback for CPU test boxes and for TP-sharded roles (per-rank shard
+exchange), negotiated per sequence via the import_seq handshake.
+"""
+from __future__ import annotations
+
+import base64
+import threading
+from typing import Any, Dict, List
+
+import torch
+
+from .. import ops
+from ..engine.kv_cache import PagedKVCache
+
+
+def peer_capable(cache: PagedKVCache) -> bool:
+    """True when this engine can export / push via hipIpc: GPU-resident
+    pool that IS a hipMalloc base allocation (kv_cache ipc_alloc path)."""
+    return bool(cache.kv.is_cuda and ops.HAVE_HIP
+                and getattr(cache, "ipc_exportable", False))
+
+
+def export_meta(cache: PagedKVCache) -> Dict[str, Any]:
+    """Decode side: one-time description of my pool for peers.  Safe to
+    call repeatedly (the handle is stable for a given allocation)."""
+    handle = ops._require_hip().kv_ipc_export(cache.kv)
+    return {
+        "handle": base64.b64encode(handle).decode("ascii"),
+        "num_pages": int(cache.kv.shape[2]),
+        "shape": list(cache.kv.shape),
+    }
+
+
+class PendingPush:
+    def __init__(self, event: torch.cuda.Event, nbytes: int):
+        self._event = event
+        self.nbytes = nbytes
+
+    def done(self) -> bool:
+        return self._event.query()
+
+    def wait(self) -> None:
+        self._event.synchronize()
+
+
+class PeerKVPusher:
+    """Prefill side: maps peer pools on first use and pushes page sets.
+    One instance per engine process; thread-safe (RPC handler threads)."""
+
+    def __init__(self, device: torch.device):
+        self.device = device
+        self._stream = torch.cuda.Stream(device)
+        self._open: Dict[str, int] = {}     # handle b64 -> mapped base ptr
+        self._lock = threading.Lock()
+
+    def _map(self, meta: Dict[str, Any]) -> int:
+        key = meta["handle"]
+        with self._lock:
+            ptr = self._open.get(key)
+            if ptr is None:
+                raw = base64.b64decode(key)
+                ptr = ops._require_hip().kv_ipc_open(raw)
+                self._open[key] = ptr
+            return ptr
+
+    def push(self, cache: PagedKVCache, src_pages: List[int],
+             dst_meta: Dict[str, Any], dst_pages: List[int]) -> PendingPush:
+        """Launch the peer copy; returns immediately with a PendingPush.
+        The copy runs on this pusher's dedicated stream so it overlaps any
+        compute the caller's engine keeps issuing on the default stream."""
+        assert len(src_pages) == len(dst_pages)
+        shape = cache.kv.shape
+        assert list(shape)[3:] == list(dst_meta["shape"])[3:], \
+            "page layout mismatch between src and dst pools"
+        dst_base = self._map(dst_meta)
+        src_t = torch.tensor(src_pages, dtype=torch.int32,
+                             device=cache.kv.device)
+        dst_t = torch.tensor(dst_pages, dtype=torch.int32,
+                             device=cache.kv.device)
+        chunk_bytes = shape[3] * shape[4] * shape[5] * 2
+        nbytes = len(src_pages) * 2 * shape[0] * chunk_bytes
+        ev = torch.cuda.Event()
+        with torch.cuda.stream(self._stream):
+            ops._require_hip().kv_peer_copy(
+                dst_base, cache.kv, src_t, dst_t,
+                int(dst_meta["num_pages"]))
+            ev.record(self._stream)
+        return PendingPush(ev, nbytes)
+
+    def close(self) -> None:
+        with self._lock:
+            for ptr in self._open.values():
+                try:
+                    ops._require_hip().kv_ipc_close(ptr)
+                except Exception:  # noqa: BLE001
+                    pass
+            self._open.clear()

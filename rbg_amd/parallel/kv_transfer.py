@@ -85,6 +85,7 @@ class TransferEngine:
 
     def send_pages(self, cache: PagedKVCache, pages: List[int],
                    dst_rank: int) -> None:
+        from .comm import to_wire
         self.connect()
         with self._lock:
             idx = torch.tensor(pages, dtype=torch.int64,
@@ -92,6 +93,7 @@ class TransferEngine:
             buf = cache.kv.index_select(2, idx).contiguous()
             if self.backend == "gloo" and buf.is_cuda:
                 buf = buf.cpu()
+            buf = to_wire(buf)
             if self._stream is not None:
                 # dedicated stream: the copy/collective overlaps decode work
                 with torch.cuda.stream(self._stream):
@@ -102,18 +104,22 @@ class TransferEngine:
 
     def recv_pages(self, cache: PagedKVCache, pages: List[int],
                    src_rank: int) -> None:
+        from .comm import from_wire, wire_dtype
         self.connect()
         with self._lock:
             m = cache.kv.shape
             shape = (m[0], m[1], len(pages), m[3], m[4], m[5])
-            dev = cache.kv.device if self.backend == "nccl" else "cpu"
-            buf = torch.empty(shape, dtype=cache.kv.dtype, device=dev)
+            dev = (cache.kv.device if self.backend == "nccl"
+                   else torch.device("cpu"))
+            buf = torch.empty(shape, dtype=wire_dtype(cache.kv.dtype, dev),
+                              device=dev)
             if self._stream is not None:
                 with torch.cuda.stream(self._stream):
                     dist.recv(buf, src_rank)
                 self._stream.synchronize()
             else:
                 dist.recv(buf, src_rank)
+            buf = from_wire(buf, cache.kv.dtype)
             idx = torch.tensor(pages, dtype=torch.int64,
                                device=cache.kv.device)
             cache.kv.index_copy_(2, idx, buf.to(cache.kv.device))

@@ -105,7 +105,11 @@ class Store:
             stored = clone(obj)
             self._objects[key] = stored
             out = clone(stored)
-        self._notify(Event("ADDED", obj.kind, clone(stored)))
+            # enqueue while still holding the lock: watchers must observe
+            # events in resourceVersion order (advisor round-1 finding —
+            # an out-of-order MODIFIED could otherwise be the version the
+            # write-behind persister durably records)
+            self._notify(Event("ADDED", obj.kind, clone(stored)))
         return out
 
     def get(self, kind: str, name: str, namespace: str = "default") -> Any:
@@ -157,7 +161,7 @@ class Store:
                 m.generation = cur.metadata.generation
             self._objects[key] = stored
             out = clone(stored)
-        self._notify(Event("MODIFIED", obj.kind, clone(stored)))
+            self._notify(Event("MODIFIED", obj.kind, clone(stored)))
         return out
 
     def apply(self, kind: str, name: str, mutate: Callable[[Any], Any],
@@ -179,7 +183,7 @@ class Store:
             obj = self._objects.pop(key, None)
             if obj is None:
                 raise NotFound(f"{kind} {namespace}/{name}")
-        self._notify(Event("DELETED", kind, clone(obj)))
+            self._notify(Event("DELETED", kind, clone(obj)))
         return obj
 
     def try_delete(self, kind: str, name: str, namespace: str = "default") -> bool:
@@ -234,9 +238,10 @@ class Store:
                 self._watches.remove(w)
 
     def _notify(self, ev: Event) -> None:
-        with self._lock:
-            watches = list(self._watches)
-        for w in watches:
+        # called with self._lock held (RLock): Watch._offer is a plain
+        # queue put, so holding the lock across the fan-out is non-blocking
+        # and makes delivery order == resourceVersion order per object
+        for w in list(self._watches):
             w._offer(ev)
 
 

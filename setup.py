@@ -16,8 +16,21 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(ROOT, "rbg_amd", "ops", "hip")
 
-sources = [os.path.join(HIP_DIR, f) for f in sorted(os.listdir(HIP_DIR))
-           if f.endswith((".hip", ".cpp"))]
+# Canonical sources, listed explicitly: the glob previously also picked up
+# hipify build artifacts (*_hip.hip) that torch's cpp_extension emits next to
+# the originals, risking duplicate extern-C symbols from drifted copies.
+HIP_SOURCES = [
+    "attn_decode.hip",
+    "attn_prefill.hip",
+    "mfma_probe.hip",
+    "rmsnorm.hip",
+    "rope_kv.hip",
+    "silu_mul.hip",
+    "skinny_gemm.hip",
+    "kv_peer_copy.hip",
+    "bindings.cpp",
+]
+sources = [os.path.join(HIP_DIR, f) for f in HIP_SOURCES]
 
 ext = CUDAExtension(
     name="rbg_amd.ops._hip_ops",
