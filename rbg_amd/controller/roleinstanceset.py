@@ -51,8 +51,12 @@ def _is_available(inst: RoleInstance, min_ready_seconds: int) -> bool:
 
 
 class RoleInstanceSetController:
-    def __init__(self, store: Store):
+    def __init__(self, store: Store, recorder=None):
         self.store = store
+        if recorder is None:
+            from ..store.events import NullRecorder
+            recorder = NullRecorder()
+        self.recorder = recorder
 
     def reconcile(self, name: str, namespace: str = "default") -> float:
         ris = self.store.try_get(C.KIND_ROLE_INSTANCE_SET, name, namespace)
@@ -145,10 +149,19 @@ class RoleInstanceSetController:
                     return 0.2   # OrderedReady: one at a time
             elif ordered and not _is_ready(inst):
                 return 0.2       # gate the next ordinal on readiness
-        # condemned: ordinals >= want, deleted descending (monotonic guard)
+        # condemned: ordinals >= want, deleted descending (monotonic guard).
+        # During a surge rolling update (maxSurge>0 with updatable stale
+        # ordinals) the surge ordinals [want, want+maxSurge) are legitimate
+        # temporary capacity — exempt them until the update completes
+        # (reference sts_reconciler.go maxUnavailable emulation +
+        # statefulmode progressUpdate "maxUnavailable(+surge)").
+        allowed = want
+        if ris.spec.update_strategy.max_surge > 0 and \
+                self._updatable_stale(ris, instances, update_hash):
+            allowed = want + ris.spec.update_strategy.max_surge
         condemned = sorted(
             (i for i in instances.values()
-             if self._ordinal(i) is not None and self._ordinal(i) >= want
+             if self._ordinal(i) is not None and self._ordinal(i) >= allowed
              and i.metadata.deletion_timestamp is None),
             key=lambda i: -self._ordinal(i))
         for name in to_delete:
@@ -181,6 +194,12 @@ class RoleInstanceSetController:
             inst = self._make_instance_named(ris, name, update_hash)
             self.store.create(inst)
             requeue = 0.2
+        # surge surplus created by _rolling_update_stateless is legitimate
+        # while stale instances remain; only trim past want+maxSurge then
+        if ris.spec.update_strategy.max_surge > 0 and any(
+                i.metadata.labels.get(C.LABEL_REVISION_HASH) != update_hash
+                for i in live):
+            want = ris.spec.replicas + ris.spec.update_strategy.max_surge
         if len(live) > want:
             to_delete = {n.strip() for n in ris.metadata.annotations.get(
                 C.ANNO_ROLE_INSTANCE_TO_DELETE, "").split(",") if n.strip()}
@@ -206,11 +225,27 @@ class RoleInstanceSetController:
                  if i.metadata.labels.get(C.LABEL_REVISION_HASH) != update_hash]
         if not stale:
             return 0.0
-        not_ready = sum(1 for i in live if not _is_ready(i))
-        budget = max(0, strat.max_unavailable - not_ready)
+        # maxSurge: spawn up to `surge` extra new-revision instances first
+        # (CloneSet-style, reference statelessmode/sync/update.go:38-300);
+        # _scale_stateless leaves the surplus alone because live counts
+        # here already include them and shrink as stale ones are deleted
+        if strat.max_surge > 0 and len(live) < ris.spec.replicas + \
+                strat.max_surge:
+            import uuid
+            for _ in range(ris.spec.replicas + strat.max_surge - len(live)):
+                name = f"{ris.metadata.name}-{uuid.uuid4().hex[:5]}"
+                inst = self._make_instance_named(ris, name, update_hash)
+                self.store.create(inst)
+                instances[name] = inst
+        # availability invariant: ready - deletions >= want - maxUnavailable;
+        # a ready surge instance raises `ready` and so buys budget even at
+        # maxUnavailable=0
+        ready = sum(1 for i in live if _is_ready(i))
+        budget = max(0, ready - ris.spec.replicas + strat.max_unavailable)
         # unready stale first (free progress), then oldest
         stale.sort(key=lambda i: (_is_ready(i), i.metadata.creation_timestamp))
-        for inst in stale[:budget]:
+        for inst in stale[:max(budget,
+                               sum(1 for i in stale if not _is_ready(i)))]:
             if self._can_update_in_place(ris, inst):
                 self._in_place_update(ris, inst, update_hash)
             else:
@@ -230,13 +265,32 @@ class RoleInstanceSetController:
 
     # ------------------------------------------------------------------
 
+    def _updatable_stale(self, ris: RoleInstanceSet,
+                         instances: Dict[str, RoleInstance],
+                         update_hash: str) -> bool:
+        """Any ordinal >= partition still at an old revision?  (Stale
+        ordinals below the partition stay old by design and must not keep
+        surge capacity alive forever.)"""
+        strat = ris.spec.update_strategy
+        for i in range(strat.partition, ris.spec.replicas):
+            inst = instances.get(instance_ordinal_name(ris.metadata.name, i))
+            if inst is not None and inst.metadata.deletion_timestamp is None \
+                    and inst.metadata.labels.get(
+                        C.LABEL_REVISION_HASH) != update_hash:
+                return True
+        return False
+
     def _rolling_update(self, ris: RoleInstanceSet,
                         instances: Dict[str, RoleInstance],
                         update_hash: str) -> float:
-        """Monotonic rolling update honoring partition + maxUnavailable
-        (reference statefulmode progressUpdate:553-633): walk ordinals
-        descending, update instances above the partition, never exceeding
-        maxUnavailable simultaneously-not-ready instances."""
+        """Monotonic rolling update honoring partition + maxUnavailable +
+        maxSurge (reference statefulmode progressUpdate:553-633): walk
+        ordinals descending, update instances above the partition, never
+        exceeding maxUnavailable simultaneously-not-ready instances.  With
+        maxSurge>0, temporary surge ordinals [want, want+surge) are created
+        at the new revision first; each READY surge instance buys one extra
+        recreate of the budget (capacity is maintained), which is what
+        makes maxUnavailable=0 + maxSurge>0 progress instead of stalling."""
         strat = ris.spec.update_strategy
         if strat.paused:
             return 0.0
@@ -254,11 +308,22 @@ class RoleInstanceSetController:
                 stale.append((i, inst))
         if not stale:
             return 0.0
-        budget = max(0, strat.max_unavailable - not_ready)
+        updatable = [(i, inst) for i, inst in stale if i >= strat.partition]
+        ready_surge = 0
+        if strat.max_surge > 0 and updatable:
+            for s in range(want, want + strat.max_surge):
+                name = instance_ordinal_name(ris.metadata.name, s)
+                surge = instances.get(name)
+                if surge is None:
+                    created = self.store.create(
+                        self._make_instance(ris, s, update_hash))
+                    instances[name] = created
+                elif surge.metadata.deletion_timestamp is None and \
+                        _is_ready(surge):
+                    ready_surge += 1
+        budget = max(0, strat.max_unavailable + ready_surge - not_ready)
         # descending ordinals, only above the partition
-        for i, inst in sorted(stale, key=lambda t: -t[0]):
-            if i < strat.partition:
-                continue
+        for i, inst in sorted(updatable, key=lambda t: -t[0]):
             if budget <= 0:
                 break
             if self._can_update_in_place(ris, inst):
@@ -287,8 +352,27 @@ class RoleInstanceSetController:
                      for c in ris.spec.template.components]
         feasible = old_shape == new_shape
         if not feasible and ris.spec.update_strategy.type == C.UPDATE_IN_PLACE_ONLY:
+            # surface the stall instead of log-and-hold: a Progressing=False
+            # condition (visible in rbgctl status) + a warning Event
+            # (the reference's status-condition discipline)
             log.warning("ris %s: in-place-only update infeasible; holding",
                         ris.metadata.name)
+            msg = (f"in-place-only update to instance {inst.metadata.name} "
+                   "is infeasible (component topology/resources changed); "
+                   "update is held — change updateStrategy to recreate")
+
+            def mark(cur: RoleInstanceSet):
+                set_condition(cur.status.conditions, Condition.new(
+                    C.COND_PROGRESSING, False, "InPlaceInfeasible", msg))
+                return cur
+            try:
+                self.store.apply(C.KIND_ROLE_INSTANCE_SET,
+                                 ris.metadata.name, mark,
+                                 ris.metadata.namespace,
+                                 subresource="status")
+            except KeyError:
+                pass
+            self.recorder.warning(ris, "InPlaceInfeasible", msg)
         return feasible
 
     def _in_place_update(self, ris: RoleInstanceSet, inst: RoleInstance,
