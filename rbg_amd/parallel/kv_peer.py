@@ -53,6 +53,17 @@ def export_meta(cache: PagedKVCache) -> Dict[str, Any]:
     }
 
 
+def export_meta_local(cache: PagedKVCache) -> Dict[str, Any]:
+    """Same-process variant (hipIpc cannot reopen a handle in the exporting
+    process): the raw device pointer is the 'mapping'.  Used when prefill
+    and decode engines share one process (1-GPU P/D bench)."""
+    return {
+        "local_ptr": int(cache.kv.data_ptr()),
+        "num_pages": int(cache.kv.shape[2]),
+        "shape": list(cache.kv.shape),
+    }
+
+
 class PendingPush:
     def __init__(self, event: torch.cuda.Event, nbytes: int):
         self._event = event
@@ -76,6 +87,8 @@ class PeerKVPusher:
         self._lock = threading.Lock()
 
     def _map(self, meta: Dict[str, Any]) -> int:
+        if "local_ptr" in meta:          # same-process pool: no IPC needed
+            return int(meta["local_ptr"])
         key = meta["handle"]
         with self._lock:
             ptr = self._open.get(key)
