@@ -62,10 +62,12 @@ def init_distributed(args, rank: int, world: int, local_rank: int) -> None:
     timeout = datetime.timedelta(seconds=300)
     if args.parallel == "pd":
         # pd: control metadata only — gloo everywhere; the KV dataplane is
-        # hipIpc + xGMI and never touches a collective
+        # hipIpc + xGMI and never touches a collective.  Modulo lets an
+        # oversubscribed test box (2 ranks, 1 GPU) exercise the full
+        # cross-process hipIpc path on one device.
         dist.init_process_group("gloo", timeout=timeout)
         if torch.cuda.is_available():
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         return
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank)

@@ -579,6 +579,10 @@ def load_object(data: Dict[str, Any]):
             data = _legacy.to_v2(data)
         elif data.get("kind") == C.KIND_RBG_SET:
             data = _legacy.set_to_v2(data)
+        elif data.get("kind") == "InstanceSet":
+            data = _legacy.instanceset_to_v2(data)
+        elif data.get("kind") == "Instance":
+            data = _legacy.instance_to_v2(data)
     kind = data.get("kind", "")
     cls = KIND_TO_TYPE.get(kind)
     if cls is None:

@@ -44,6 +44,11 @@ API_VERSION_V1ALPHA1 = "workloads.x-k8s.io/v1alpha1"
 ANNO_WORKLOAD_TYPE = f"{C.PREFIX}/role-workload-type"
 ANNO_POD_GROUP_POLICY = f"{C.PREFIX}/v1alpha1-pod-group-policy"
 ANNO_COORDINATION = f"{C.PREFIX}/v1alpha1-coordination"
+# original per-role rolloutStrategy + LWS patch templates, preserved so the
+# documented v1 -> v2 -> v1 round-trip is lossless for them (the reference
+# keeps these via annotations too; round-1 advisor finding)
+ANNO_ROLLOUT = f"{C.PREFIX}/v1alpha1-rollout-strategy"
+ANNO_LWS = f"{C.PREFIX}/v1alpha1-lws"
 
 _LEGACY_RESTART = {
     "RecreateRoleInstanceOnPodRestart": C.RESTART_POLICY_RECREATE_INSTANCE,
@@ -141,6 +146,10 @@ def _convert_role_to_v2(role: Dict[str, Any],
             if k in ru}}
         if ru.get("type"):
             out["updateStrategyType"] = ru["type"]
+        # keep the original doc for a lossless round-trip
+        saved = json.loads(annos.get(ANNO_ROLLOUT, "{}"))
+        saved[role.get("name", "")] = role["rolloutStrategy"]
+        annos[ANNO_ROLLOUT] = json.dumps(saved, sort_keys=True)
     restart = _LEGACY_RESTART.get(role.get("restartPolicy", ""),
                                   C.RESTART_POLICY_RECREATE_INSTANCE)
     out["restartPolicy"] = restart
@@ -161,6 +170,11 @@ def _convert_role_to_v2(role: Dict[str, Any],
                                                  lws["patchWorkerTemplate"])
         out["pattern"] = C.PATTERN_LEADER_WORKER
         out["leaderWorkerPattern"] = lwp
+        if lws:
+            # preserve the original patches (merge output is not invertible)
+            saved = json.loads(annos.get(ANNO_LWS, "{}"))
+            saved[out["name"]] = lws
+            annos[ANNO_LWS] = json.dumps(saved, sort_keys=True)
     elif role.get("components"):
         out["pattern"] = C.PATTERN_CUSTOM_COMPONENTS
         out["customComponentsPattern"] = {"components": role["components"]}
@@ -206,12 +220,23 @@ def to_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
 
 
 def _convert_role_from_v2(role: Dict[str, Any],
-                          workload_types: Dict[str, str]) -> Dict[str, Any]:
+                          workload_types: Dict[str, str],
+                          rollouts: Dict[str, Any],
+                          lws_saved: Dict[str, Any]) -> Dict[str, Any]:
     out: Dict[str, Any] = {"name": role.get("name", ""),
                            "replicas": role.get("replicas", 1)}
     for k in ("dependencies", "servicePorts", "minReadySeconds", "template"):
         if k in role:
             out[k] = role[k]
+    # restore rolloutStrategy: the preserved original if present, else
+    # synthesize from the v2 fields
+    if out["name"] in rollouts:
+        out["rolloutStrategy"] = rollouts[out["name"]]
+    elif role.get("rolloutStrategy"):
+        ru = dict((role["rolloutStrategy"] or {}).get("rollingUpdate") or {})
+        if role.get("updateStrategyType"):
+            ru["type"] = role["updateStrategyType"]
+        out["rolloutStrategy"] = {"rollingUpdate": ru}
     if role.get("templateRef"):
         out["templateRef"] = {"name": role["templateRef"].get("name", "")}
         if role["templateRef"].get("patch"):
@@ -230,7 +255,13 @@ def _convert_role_from_v2(role: Dict[str, Any],
         lwp = role.get("leaderWorkerPattern") or {}
         out["workload"] = {"apiVersion": "leaderworkerset.x-k8s.io/v1",
                            "kind": "LeaderWorkerSet"}
-        out["leaderWorkerSet"] = {"size": lwp.get("size", 1)}
+        if out["name"] in lws_saved:
+            # restore the original patch templates (preserved: the merged
+            # leader/worker templates are not invertible)
+            out["leaderWorkerSet"] = lws_saved[out["name"]]
+            out["leaderWorkerSet"].setdefault("size", lwp.get("size", 1))
+        else:
+            out["leaderWorkerSet"] = {"size": lwp.get("size", 1)}
     elif pattern == C.PATTERN_CUSTOM_COMPONENTS:
         out["workload"] = {"apiVersion": C.API_VERSION, "kind": "InstanceSet"}
         out["components"] = (role.get("customComponentsPattern") or {}).get(
@@ -249,8 +280,11 @@ def from_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
     meta = dict(doc.get("metadata") or {})
     annos = dict(meta.get("annotations") or {})
     workload_types = json.loads(annos.pop(ANNO_WORKLOAD_TYPE, "{}"))
+    rollouts = json.loads(annos.pop(ANNO_ROLLOUT, "{}"))
+    lws_saved = json.loads(annos.pop(ANNO_LWS, "{}"))
     out_spec: Dict[str, Any] = {
-        "roles": [_convert_role_from_v2(r, workload_types)
+        "roles": [_convert_role_from_v2(r, workload_types, rollouts,
+                                        lws_saved)
                   for r in spec.get("roles", [])]}
     if spec.get("roleTemplates"):
         out_spec["roleTemplates"] = [
@@ -267,5 +301,204 @@ def from_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
     if not annos:
         meta.pop("annotations", None)
     return {"apiVersion": API_VERSION_V1ALPHA1, "kind": C.KIND_RBG,
+            "metadata": meta, "spec": out_spec,
+            **({"status": doc["status"]} if "status" in doc else {})}
+
+
+# ---------------------------------------------------------------------------
+# Legacy InstanceSet / Instance kinds (reference api/workloads/v1alpha1/
+# instanceset_types.go, instance_types.go).  The reference serves these as
+# standalone v1alpha1 CRDs — the predecessors of RoleInstanceSet /
+# RoleInstance — gated by the deprecated-workload toggle.  Here a legacy doc
+# converts structurally on the way in (load_object) and back out for legacy
+# clients.
+# ---------------------------------------------------------------------------
+
+ANNO_IS_LIFECYCLE = f"{C.PREFIX}/v1alpha1-lifecycle"
+ANNO_IS_SCALE_MAXUNAVAILABLE = f"{C.PREFIX}/v1alpha1-scale-max-unavailable"
+ANNO_READY_POLICY = f"{C.PREFIX}/v1alpha1-ready-policy"
+
+_LEGACY_UPDATE_TYPE = {
+    "InPlaceIfPossible": C.UPDATE_IN_PLACE_IF_POSSIBLE,
+    "InPlaceOnly": C.UPDATE_IN_PLACE_ONLY,
+    "RecreatePod": C.UPDATE_RECREATE,
+    "RecreateInstance": C.UPDATE_RECREATE,
+}
+_LEGACY_UPDATE_TYPE_BACK = {
+    C.UPDATE_IN_PLACE_IF_POSSIBLE: "InPlaceIfPossible",
+    C.UPDATE_IN_PLACE_ONLY: "InPlaceOnly",
+    C.UPDATE_RECREATE: "RecreatePod",
+}
+
+
+def _int_or_pct(v: Any, replicas: int, default: int) -> int:
+    """IntOrString: plain int passes through; "N%" rounds UP against
+    replicas (apimachinery intstr semantics for maxUnavailable)."""
+    if v is None:
+        return default
+    if isinstance(v, int):
+        return v
+    s = str(v).strip()
+    if s.endswith("%"):
+        pct = float(s[:-1])
+        return max(0, int(-(-replicas * pct // 100)))
+    return int(s)
+
+
+def instanceset_to_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
+    """v1alpha1 InstanceSet doc -> v1alpha2 RoleInstanceSet doc."""
+    spec = doc.get("spec") or {}
+    meta = dict(doc.get("metadata") or {})
+    annos = dict(meta.get("annotations") or {})
+    replicas = int(spec.get("replicas", 1))
+    out_spec: Dict[str, Any] = {"replicas": replicas}
+    sel = (spec.get("selector") or {}).get("matchLabels")
+    if sel:
+        out_spec["selector"] = dict(sel)
+    tmpl = spec.get("instanceTemplate") or {}
+    out_spec["template"] = {
+        "metadata": tmpl.get("metadata") or {},
+        "components": tmpl.get("components") or
+                      (tmpl.get("spec") or {}).get("components") or [],
+    }
+    rp = (tmpl.get("spec") or {}).get("restartPolicy") or \
+        tmpl.get("restartPolicy")
+    if rp:
+        out_spec["template"]["restartPolicy"] = _LEGACY_RESTART.get(
+            rp, C.RESTART_POLICY_RECREATE_INSTANCE)
+    scale = spec.get("scaleStrategy") or {}
+    if scale.get("instanceToDelete"):
+        annos[C.ANNO_ROLE_INSTANCE_TO_DELETE] = ",".join(
+            scale["instanceToDelete"])
+    if scale.get("maxUnavailable") is not None:
+        annos[ANNO_IS_SCALE_MAXUNAVAILABLE] = str(scale["maxUnavailable"])
+    upd = spec.get("updateStrategy") or {}
+    out_spec["updateStrategy"] = {
+        "type": _LEGACY_UPDATE_TYPE.get(upd.get("type", ""),
+                                        C.UPDATE_IN_PLACE_IF_POSSIBLE),
+        "partition": _int_or_pct(upd.get("partition"), replicas, 0),
+        "maxUnavailable": _int_or_pct(upd.get("maxUnavailable"),
+                                      replicas, 1),
+        "maxSurge": _int_or_pct(upd.get("maxSurge"), replicas, 0),
+        "paused": bool(upd.get("paused", False)),
+        "gracePeriodSeconds": int(
+            (upd.get("inPlaceUpdateStrategy") or {}).get(
+                "gracePeriodSeconds", 0)),
+    }
+    if spec.get("revisionHistoryLimit") is not None:
+        out_spec["revisionHistoryLimit"] = spec["revisionHistoryLimit"]
+    if spec.get("minReadySeconds") is not None:
+        out_spec["minReadySeconds"] = spec["minReadySeconds"]
+    if spec.get("lifecycle"):
+        annos[ANNO_IS_LIFECYCLE] = json.dumps(spec["lifecycle"],
+                                              sort_keys=True)
+    if annos:
+        meta["annotations"] = annos
+    return {"apiVersion": C.API_VERSION, "kind": C.KIND_ROLE_INSTANCE_SET,
+            "metadata": meta, "spec": out_spec,
+            **({"status": doc["status"]} if "status" in doc else {})}
+
+
+def instanceset_from_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
+    """v1alpha2 RoleInstanceSet doc -> v1alpha1 InstanceSet doc."""
+    spec = doc.get("spec") or {}
+    meta = dict(doc.get("metadata") or {})
+    annos = dict(meta.get("annotations") or {})
+    out_spec: Dict[str, Any] = {"replicas": spec.get("replicas", 1)}
+    if spec.get("selector"):
+        out_spec["selector"] = {"matchLabels": dict(spec["selector"])}
+    tmpl = spec.get("template") or {}
+    out_tmpl: Dict[str, Any] = {
+        "metadata": tmpl.get("metadata") or {},
+        "components": tmpl.get("components") or [],
+    }
+    if tmpl.get("restartPolicy"):
+        out_tmpl["restartPolicy"] = _LEGACY_RESTART_BACK.get(
+            tmpl["restartPolicy"], "None")
+    out_spec["instanceTemplate"] = out_tmpl
+    scale: Dict[str, Any] = {}
+    if C.ANNO_ROLE_INSTANCE_TO_DELETE in annos:
+        scale["instanceToDelete"] = [
+            n for n in annos.pop(C.ANNO_ROLE_INSTANCE_TO_DELETE).split(",")
+            if n]
+    if ANNO_IS_SCALE_MAXUNAVAILABLE in annos:
+        raw = annos.pop(ANNO_IS_SCALE_MAXUNAVAILABLE)
+        scale["maxUnavailable"] = int(raw) if raw.lstrip("-").isdigit() \
+            else raw
+    if scale:
+        out_spec["scaleStrategy"] = scale
+    upd = spec.get("updateStrategy") or {}
+    out_upd: Dict[str, Any] = {
+        "type": _LEGACY_UPDATE_TYPE_BACK.get(
+            upd.get("type", ""), "InPlaceIfPossible"),
+        "partition": upd.get("partition", 0),
+        "maxUnavailable": upd.get("maxUnavailable", 1),
+        "maxSurge": upd.get("maxSurge", 0),
+        "paused": upd.get("paused", False),
+    }
+    if upd.get("gracePeriodSeconds"):
+        out_upd["inPlaceUpdateStrategy"] = {
+            "gracePeriodSeconds": upd["gracePeriodSeconds"]}
+    out_spec["updateStrategy"] = out_upd
+    for k in ("revisionHistoryLimit", "minReadySeconds"):
+        if k in spec:
+            out_spec[k] = spec[k]
+    if ANNO_IS_LIFECYCLE in annos:
+        out_spec["lifecycle"] = json.loads(annos.pop(ANNO_IS_LIFECYCLE))
+    meta["annotations"] = annos
+    if not annos:
+        meta.pop("annotations", None)
+    return {"apiVersion": API_VERSION_V1ALPHA1, "kind": "InstanceSet",
+            "metadata": meta, "spec": out_spec,
+            **({"status": doc["status"]} if "status" in doc else {})}
+
+
+def instance_to_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
+    """v1alpha1 Instance doc -> v1alpha2 RoleInstance doc."""
+    spec = doc.get("spec") or {}
+    meta = dict(doc.get("metadata") or {})
+    annos = dict(meta.get("annotations") or {})
+    out_spec: Dict[str, Any] = {
+        "components": spec.get("components") or [],
+    }
+    if spec.get("restartPolicy"):
+        out_spec["restartPolicy"] = _LEGACY_RESTART.get(
+            spec["restartPolicy"], C.RESTART_POLICY_RECREATE_INSTANCE)
+    if spec.get("readinessGates"):
+        out_spec["readinessGates"] = spec["readinessGates"]
+    if spec.get("readyPolicy"):
+        annos[ANNO_READY_POLICY] = spec["readyPolicy"]
+    pgp = spec.get("podGroupPolicy")
+    if pgp:
+        annos[C.ANNO_GANG_SCHEDULING] = "true"
+        annos[ANNO_POD_GROUP_POLICY] = json.dumps(pgp, sort_keys=True)
+    if annos:
+        meta["annotations"] = annos
+    return {"apiVersion": C.API_VERSION, "kind": C.KIND_ROLE_INSTANCE,
+            "metadata": meta, "spec": out_spec,
+            **({"status": doc["status"]} if "status" in doc else {})}
+
+
+def instance_from_v2(doc: Dict[str, Any]) -> Dict[str, Any]:
+    """v1alpha2 RoleInstance doc -> v1alpha1 Instance doc."""
+    spec = doc.get("spec") or {}
+    meta = dict(doc.get("metadata") or {})
+    annos = dict(meta.get("annotations") or {})
+    out_spec: Dict[str, Any] = {"components": spec.get("components") or []}
+    if spec.get("restartPolicy"):
+        out_spec["restartPolicy"] = _LEGACY_RESTART_BACK.get(
+            spec["restartPolicy"], "None")
+    if spec.get("readinessGates"):
+        out_spec["readinessGates"] = spec["readinessGates"]
+    if ANNO_READY_POLICY in annos:
+        out_spec["readyPolicy"] = annos.pop(ANNO_READY_POLICY)
+    if ANNO_POD_GROUP_POLICY in annos:
+        out_spec["podGroupPolicy"] = json.loads(
+            annos.pop(ANNO_POD_GROUP_POLICY))
+        annos.pop(C.ANNO_GANG_SCHEDULING, None)
+    meta["annotations"] = annos
+    if not annos:
+        meta.pop("annotations", None)
+    return {"apiVersion": API_VERSION_V1ALPHA1, "kind": "Instance",
             "metadata": meta, "spec": out_spec,
             **({"status": doc["status"]} if "status" in doc else {})}
