@@ -127,7 +127,7 @@ def pick_decode_splits(num_seqs: int, num_kv_heads: int,
     half the dot2 kernel's at equal splits."""
     if variant < 0:
         variant = DECODE_VARIANT
-    if variant == 4:
+    if variant >= 4:
         base = num_kv_heads * ((num_seqs + 1) // 2)
         by_ctx = max(1, max_context // 32)      # >= one page pair per split
     else:

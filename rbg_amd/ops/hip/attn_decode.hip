@@ -301,8 +301,20 @@ DEV_INLINE int v_img_off(int key, int dim) {
 //  * 2-wave workgroups, ~18 KB LDS per wave -> 4 WGs (8 waves) per CU;
 //    MFMA depth inside the wave covers the low waves/SIMD (the same
 //    budget shape as the 8-wave prefill kernel).
-template <int QPG>
-__global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
+// DIET (variant 5): the round-2 register diet targeting 4 waves/SIMD
+// (docs/ROUND2_DESIGNS.md §1a+1b).  Two changes vs variant 4:
+//  (a) q fragments are RELOADED from global per page pair instead of held
+//      in 16 persistent VGPRs — the rows are the same cache lines every
+//      pair, so after the first touch they come from L1.  Issued BEFORE
+//      the V stage so the compiler's dependency waitcnt for the first S
+//      MFMA is vmcnt(8) (V still in flight), not a pipeline drain.
+//  (b) V stages through 4 of its own registers + 4 of K's (ka0-3 are dead
+//      between DM_WRITE_K and DM_ISSUE_K(pg+2)) — 16 fewer VGPRs.  The
+//      round-67 FULL merge (one 8-set) hit in-loop spills and lost 27%;
+//      this partial merge keeps issue order WAR-safe without aliasing
+//      pressure.  Verified ScratchSize==0 at compile time.
+template <int QPG, int DIET>
+__global__ __launch_bounds__(128, DIET ? 4 : 3) void decode_attn_mfma_kernel(
     float* __restrict__ partial_o,        // [splits, seqs, QH, D]
     float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
     __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
@@ -346,12 +358,13 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
   const int pg_begin = key_begin >> 4;
   const int pg_end = (key_end + 15) >> 4;
 
-  // q fragments: row = head (clamped), 4 k-steps of 32 dims
+  // q fragments: row = head (clamped), 4 k-steps of 32 dims.
+  // DIET: only the row pointer persists; fragments reload per pair (L1).
+  const __hip_bfloat16* qrow =
+      q + (size_t)seq * q_stride +
+      (size_t)(kvh * QPG + min(gl, QPG - 1)) * HEAD_DIM;
   dm_bf8 qa[4];
-  {
-    const int h = min(gl, QPG - 1);
-    const __hip_bfloat16* qrow =
-        q + (size_t)seq * q_stride + (size_t)(kvh * QPG + h) * HEAD_DIM;
+  if constexpr (!DIET) {
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
       union { uint4 u; dm_bf8 v; } cvt;
@@ -415,6 +428,33 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
     *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 48)]) = va6;  \
     *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 56)]) = va7;  \
   } while (0)
+  // DIET: V stages through vb0-7, declared in the loop with liveness
+  // DISJOINT from the q fragments (qd dies at the last S MFMA; vb issues
+  // right after), so the allocator colors them into the same registers —
+  // V latency hides under the softmax shuffle chain instead of under S.
+#define DM_ISSUE_V_D(pg)                                                    \
+  do {                                                                      \
+    const size_t vb_ = page_base(pg, myloc);                                \
+    vb0 = *reinterpret_cast<const uint4*>(val_cache + vb_);                 \
+    vb1 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 8);             \
+    vb2 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 16);            \
+    vb3 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 24);            \
+    vb4 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 32);            \
+    vb5 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 40);            \
+    vb6 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 48);            \
+    vb7 = *reinterpret_cast<const uint4*>(val_cache + vb_ + 56);            \
+  } while (0)
+#define DM_WRITE_V_D()                                                      \
+  do {                                                                      \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk)]) = vb0;       \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 8)]) = vb1;   \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 16)]) = vb2;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 24)]) = vb3;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 32)]) = vb4;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 40)]) = vb5;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 48)]) = vb6;  \
+    *reinterpret_cast<uint4*>(&v_img[v_img_off(srow, schunk + 56)]) = vb7;  \
+  } while (0)
 
   float m[4], l[4];
   f32x4 acc_o[8];
@@ -426,8 +466,20 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
   DM_ISSUE_K(pg_begin);
 
   for (int pg = pg_begin; pg < pg_end; pg += 2) {
+    // DIET: reload q fragments BEFORE the V issue, so the dependency wait
+    // in front of the first S MFMA is vmcnt(8) — V keeps streaming
+    dm_bf8 qd[4];
+    if constexpr (DIET) {
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        union { uint4 u; dm_bf8 v; } cvt;
+        cvt.u = *reinterpret_cast<const uint4*>(qrow + ks * 32 + gslice * 8);
+        qd[ks] = cvt.v;
+      }
+    }
     DM_WRITE_K();
-    DM_ISSUE_V(pg);          // V latency hides under S + softmax
+    uint4 vb0, vb1, vb2, vb3, vb4, vb5, vb6, vb7;
+    if constexpr (!DIET) DM_ISSUE_V(pg);  // V hides under S + softmax
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // S = q K^T over the pair's 32 keys (B-frag: key row = gl / gl+16)
@@ -439,9 +491,13 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
           &k_lds[gl * KP + ks * 32 + gslice * 8]);
       kfB.u = *reinterpret_cast<const uint4*>(
           &k_lds[(16 + gl) * KP + ks * 32 + gslice * 8]);
-      sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[ks], kfA.v, sA, 0, 0, 0);
-      sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[ks], kfB.v, sB, 0, 0, 0);
+      const dm_bf8 qk = DIET ? qd[ks] : qa[ks];
+      sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qk, kfA.v, sA, 0, 0, 0);
+      sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qk, kfB.v, sB, 0, 0, 0);
     }
+    // DIET: q fragments are dead now — V stages into their registers and
+    // its latency hides under the softmax shuffle chain below
+    if constexpr (DIET) DM_ISSUE_V_D(pg);
 
     // online softmax: lane holds rows 4*gslice+r for key cols
     // (pg*16+gl, pg*16+16+gl); ONE group16 round covers both pages
@@ -473,12 +529,13 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
     }
     // V overwrites the K panel: same-wave DS ordering protects the S
     // fragment reads issued above
-    DM_WRITE_V();
+    if constexpr (DIET) DM_WRITE_V_D();
+    else DM_WRITE_V();
     DM_ISSUE_K(pg + 2);      // next pair's K hides under PV
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // O += P V via the 32-k fragment + tr-read V image (attn_prefill PV)
-    {
+    if constexpr (!DIET) {
       union { uint4 u; dm_bf8 v; } paf;
       const int pk = gslice * 8;
       paf.u.x = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk]);
@@ -520,6 +577,62 @@ __global__ __launch_bounds__(128, 3) void decode_attn_mfma_kernel(
         vf.u.hi = vhi[dt];
         acc_o[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             paf.v, vf.vf2, acc_o[dt], 0, 0, 0);
+      }
+    } else {
+      // DIET PV: two 4-dt halves so only 16 tr-read VGPRs are live at a
+      // time (the full 32-reg batch overlapped the 32 in-flight K staging
+      // registers and was the register-pressure peak)
+      union { uint4 u; dm_bf8 v; } paf;
+      const int pk = gslice * 8;
+      paf.u.x = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk]);
+      paf.u.y = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 2]);
+      paf.u.z = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 4]);
+      paf.u.w = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 6]);
+      const unsigned vaddr = (unsigned)(unsigned long long)(
+          &v_img[gslice * 64 + gl * 4]);
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        unsigned long long vlo[4], vhi[4];
+        if (half == 0) {
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %8\n\t"
+              "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+              "ds_read_b64_tr_b16 %2, %8 offset:1024\n\t"
+              "ds_read_b64_tr_b16 %3, %8 offset:1536\n\t"
+              "ds_read_b64_tr_b16 %4, %8 offset:2048\n\t"
+              "ds_read_b64_tr_b16 %5, %8 offset:2560\n\t"
+              "ds_read_b64_tr_b16 %6, %8 offset:3072\n\t"
+              "ds_read_b64_tr_b16 %7, %8 offset:3584\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(vlo[0]), "=&v"(vhi[0]), "=&v"(vlo[1]), "=&v"(vhi[1]),
+                "=&v"(vlo[2]), "=&v"(vhi[2]), "=&v"(vlo[3]), "=&v"(vhi[3])
+              : "v"(vaddr)
+              : "memory");
+        } else {
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %8 offset:4096\n\t"
+              "ds_read_b64_tr_b16 %1, %8 offset:4608\n\t"
+              "ds_read_b64_tr_b16 %2, %8 offset:5120\n\t"
+              "ds_read_b64_tr_b16 %3, %8 offset:5632\n\t"
+              "ds_read_b64_tr_b16 %4, %8 offset:6144\n\t"
+              "ds_read_b64_tr_b16 %5, %8 offset:6656\n\t"
+              "ds_read_b64_tr_b16 %6, %8 offset:7168\n\t"
+              "ds_read_b64_tr_b16 %7, %8 offset:7680\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(vlo[0]), "=&v"(vhi[0]), "=&v"(vlo[1]), "=&v"(vhi[1]),
+                "=&v"(vlo[2]), "=&v"(vhi[2]), "=&v"(vlo[3]), "=&v"(vhi[3])
+              : "v"(vaddr)
+              : "memory");
+        }
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          union { struct { unsigned long long lo, hi; } u; dm_bf8 vf2; } vf;
+          vf.u.lo = vlo[dt];
+          vf.u.hi = vhi[dt];
+          acc_o[half * 4 + dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              paf.v, vf.vf2, acc_o[half * 4 + dt], 0, 0, 0);
+        }
       }
     }
   }
@@ -598,10 +711,11 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits,  \
                      q_stride)
-  if (variant == 4 && page_size == 16) {
+  if ((variant == 4 || variant == 5) && page_size == 16) {
     dim3 mgrid(num_kv_heads, (num_seqs + 1) / 2, num_splits), mblock(128);
-#define LAUNCH_MFMA(QPG)                                                     \
-    hipLaunchKernelGGL((decode_attn_mfma_kernel<QPG>), mgrid, mblock, 0,     \
+#define LAUNCH_MFMA(QPG, DIET)                                               \
+    hipLaunchKernelGGL((decode_attn_mfma_kernel<QPG, DIET>), mgrid, mblock,  \
+                       0,                                                    \
                        stream, (float*)partial_o, (float*)partial_ml,        \
                        (__hip_bfloat16*)out, (const __hip_bfloat16*)q,       \
                        (const __hip_bfloat16*)key_cache,                     \
@@ -609,11 +723,19 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                        (const int*)block_tables, (const int*)context_lens,   \
                        scale, num_kv_heads, max_pages, num_splits,         \
                        num_seqs, q_stride)
-    if (qpg == 1) LAUNCH_MFMA(1);
-    else if (qpg == 2) LAUNCH_MFMA(2);
-    else if (qpg == 4) LAUNCH_MFMA(4);
-    else if (qpg == 8) LAUNCH_MFMA(8);
-    else return;
+    if (variant == 5) {
+      if (qpg == 1) LAUNCH_MFMA(1, 1);
+      else if (qpg == 2) LAUNCH_MFMA(2, 1);
+      else if (qpg == 4) LAUNCH_MFMA(4, 1);
+      else if (qpg == 8) LAUNCH_MFMA(8, 1);
+      else return;
+    } else {
+      if (qpg == 1) LAUNCH_MFMA(1, 0);
+      else if (qpg == 2) LAUNCH_MFMA(2, 0);
+      else if (qpg == 4) LAUNCH_MFMA(4, 0);
+      else if (qpg == 8) LAUNCH_MFMA(8, 0);
+      else return;
+    }
 #undef LAUNCH_MFMA
     if (num_splits > 1) {
       dim3 cgrid(num_seqs, num_q_heads), cblock(HEAD_DIM);

@@ -204,9 +204,11 @@ def test_skinny_gemm_matches_fp32():
 
 
 @pytest.mark.gpu
-def test_decode_attention_mfma_variant4():
-    """MFMA-tiled decode (variant 4) vs fp32 reference, including ragged
-    contexts, multiple splits, and odd page tails."""
+@pytest.mark.parametrize("variant", [4, 5])
+def test_decode_attention_mfma_variant4(variant):
+    """MFMA-tiled decode (variant 4 + register-diet variant 5) vs fp32
+    reference, including ragged contexts, multiple splits, odd page
+    tails."""
     import torch
     from rbg_amd import ops
     from rbg_amd.ops import reference
@@ -229,7 +231,8 @@ def test_decode_attention_mfma_variant4():
     ref = reference.decode_attention(
         q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(), ctx.cpu(), 0.088).to(dev)
     for splits in (1, 2, 4):
-        got = ops._hip.decode_attention(q, kc, vc, bt, ctx, 0.088, splits, 4)
+        got = ops._hip.decode_attention(q, kc, vc, bt, ctx, 0.088, splits,
+                                        variant)
         assert torch.allclose(got.float(), ref.float(), atol=2e-2,
                               rtol=2e-2), \
             (splits, (got.float() - ref.float()).abs().max().item())

@@ -36,7 +36,7 @@ def run(variant):
     torch.cuda.synchronize()
     return (time.monotonic() - t0) / a.iters
 
-VARIANTS = [1, 4]
+VARIANTS = [int(x) for x in os.environ.get("RBG_AB_VARIANTS", "4,5").split(",")]
 for v in VARIANTS:
     run(v)   # warmup
 results = {v: [] for v in VARIANTS}
