@@ -255,6 +255,10 @@ class RoleInstanceController:
         if self.gang.holding(rt.gang_id) is not None:
             return
         rbg_uid = self._owner_uid(inst)
+        # exclusive topology (reference pod_reconciler.go:160-241): the
+        # group's engines pack onto one GPU set that no other group shares
+        exclusive = bool(inst.metadata.annotations.get(
+            C.ANNO_EXCLUSIVE_TOPOLOGY))
         claims = []
         for comp, j, wname in desired:
             gpus = 0
@@ -266,7 +270,8 @@ class RoleInstanceController:
                     hbm = res.hbm_bytes
             prefer = self.bindings.lookup(rbg_uid, GpuBindingStore.key(
                 inst.metadata.name, f"{comp.name}-{j}"))
-            claims.append(GpuClaim(gpus=gpus, hbm_bytes=hbm, prefer=prefer)
+            claims.append(GpuClaim(gpus=gpus, hbm_bytes=hbm, prefer=prefer,
+                                   group=rbg_uid, exclusive=exclusive)
                           if gpus or hbm else GpuClaim(gpus=0))
         if not any(cl.gpus or cl.hbm_bytes for cl in claims):
             self.gang.reserve(rt.gang_id, [])

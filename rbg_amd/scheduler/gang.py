@@ -31,6 +31,11 @@ class GpuClaim:
     gpus: int = 1
     hbm_bytes: int = 0          # 0 => exclusive whole-GPU claim
     prefer: Tuple[int, ...] = ()  # sticky hint: previous device ids
+    # exclusive-topology (reference pod_reconciler.go:160-241, consumed via
+    # the rbg.workloads.x-k8s.io/exclusive-topology annotation): the claim's
+    # GROUP packs onto a shared GPU set that no OTHER group may share
+    group: str = ""
+    exclusive: bool = False
 
 
 @dataclass
@@ -49,6 +54,7 @@ class GangAllocator:
         self._exclusive: Dict[int, str] = {}     # gpu -> gang holding exclusively
         self._shared: Dict[int, Dict[str, int]] = {g: {} for g in self._free_hbm}
         self._gangs: Dict[str, Reservation] = {}
+        self._gang_group: Dict[str, str] = {}    # gang -> exclusive group
 
     # -- queries ------------------------------------------------------------
 
@@ -90,18 +96,42 @@ class GangAllocator:
         snap_shared = {g: dict(v) for g, v in self._shared.items()}
         res = Reservation(gang_id=gang_id)
         ok = True
+        snap_group = dict(self._gang_group)
         for claim in claims:
             if claim.hbm_bytes and claim.gpus == 1:
-                # shared slice of one GPU
-                cand = [g for g in claim.prefer
-                        if g in self._free_hbm and g not in self._exclusive
-                        and self._free_hbm[g] >= claim.hbm_bytes]
+                # shared slice of one GPU.  With exclusive topology a GPU
+                # qualifies only if every current sharer belongs to the
+                # same group (1 group per device — the "Disaggregated
+                # Inference" same-domain use case), and group GPUs are
+                # preferred so the group's roles PACK together (affinity
+                # half of the reference semantics).
+                def allowed(g: int) -> bool:
+                    if g in self._exclusive or \
+                            self._free_hbm[g] < claim.hbm_bytes:
+                        return False
+                    if not claim.exclusive:
+                        # a non-exclusive claim must not intrude on a GPU
+                        # claimed exclusively-by-topology by another group
+                        return all(
+                            self._gang_group.get(o, "") == "" or
+                            self._gang_group.get(o) == claim.group
+                            for o in self._shared[g])
+                    return all(self._gang_group.get(o) == claim.group
+                               for o in self._shared[g])
+
+                def group_share(g: int) -> int:
+                    return sum(v for o, v in self._shared[g].items()
+                               if claim.group and
+                               self._gang_group.get(o) == claim.group)
+
+                cand = [g for g in claim.prefer if g in self._free_hbm
+                        and allowed(g)]
                 if not cand:
                     cand = sorted(
-                        (g for g in self._free_hbm
-                         if g not in self._exclusive
-                         and self._free_hbm[g] >= claim.hbm_bytes),
-                        key=lambda g: -self._free_hbm[g])
+                        (g for g in self._free_hbm if allowed(g)),
+                        key=lambda g: (-group_share(g)
+                                       if claim.exclusive else 0,
+                                       -self._free_hbm[g]))
                 if not cand:
                     ok = False
                     break
@@ -123,7 +153,11 @@ class GangAllocator:
             self._exclusive = snap_excl
             self._free_hbm = snap_hbm
             self._shared = snap_shared
+            self._gang_group = snap_group
             return None
+        grp = next((c.group for c in claims if c.exclusive and c.group), "")
+        if grp:
+            self._gang_group[gang_id] = grp
         self._gangs[gang_id] = res
         return res
 
@@ -160,6 +194,7 @@ class GangAllocator:
                 taken = owners.pop(gang_id, 0)
                 if taken:
                     self._free_hbm[g] += taken
+            self._gang_group.pop(gang_id, None)
             self._lock.notify_all()
 
     def holding(self, gang_id: str) -> Optional[Reservation]:

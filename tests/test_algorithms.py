@@ -187,3 +187,39 @@ def test_restart_tracker_reset_after_stability():
     # healthy past the stability window resets
     t.observe_healthy(now + 10 + stability_window(300) + 1)
     assert t.restart_count == 0
+
+
+# ---- exclusive topology (gang allocator consumer of the annotation) ------
+
+def test_exclusive_topology_packs_group_and_excludes_others():
+    from rbg_amd.scheduler.gang import GangAllocator, GangUnschedulable, GpuClaim
+    from rbg_amd.scheduler.topology import fully_connected
+    GB = 1 << 30
+    alloc = GangAllocator(fully_connected(2))
+    # group A: two shared-slice roles with exclusive topology -> both pack
+    # onto the SAME GPU (affinity half)
+    r1 = alloc.reserve("a-prefill", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                              group="A", exclusive=True)])
+    r2 = alloc.reserve("a-decode", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                             group="A", exclusive=True)])
+    assert r1.assignments[0] == r2.assignments[0]
+    g_a = r1.assignments[0][0]
+    # group B exclusive -> must land on the OTHER GPU (exclusivity half)
+    r3 = alloc.reserve("b-prefill", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                              group="B", exclusive=True)])
+    assert r3.assignments[0][0] != g_a
+    # a third exclusive group cannot fit anywhere: both GPUs are claimed
+    import pytest as _pytest
+    with _pytest.raises(GangUnschedulable):
+        alloc.reserve("c-x", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                       group="C", exclusive=True)],
+                      timeout=0.0)
+    # non-exclusive claims also stay off exclusively-claimed GPUs
+    with _pytest.raises(GangUnschedulable):
+        alloc.reserve("d-x", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                       group="D")], timeout=0.0)
+    # release group B -> group C fits
+    alloc.release("b-prefill")
+    r5 = alloc.reserve("c-x", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                        group="C", exclusive=True)])
+    assert r5.assignments[0][0] != g_a
