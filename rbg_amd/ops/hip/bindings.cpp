@@ -29,6 +29,9 @@ void launch_skinny_gemm(void*, void*, const void*, const void*, int, int,
 void launch_reduce_splits(void*, const void*, int, long, hipStream_t);
 void launch_kv_peer_copy(void*, const void*, const void*, const void*, int,
                          int, int, long, long, long, long, hipStream_t);
+void launch_xgmi_allreduce(void*, const void*, void**, void**, int, int,
+                           long, hipStream_t);
+long xgmi_allreduce_signal_bytes();
 }
 
 namespace {
@@ -320,6 +323,63 @@ void kv_peer_copy(int64_t dst_base, torch::Tensor src_kv,
                       current_stream());
 }
 
+// ---- one-shot xGMI all-reduce (parallel/xgmi_allreduce.py) ---------------
+// Data buffer: ordinary hipMalloc (IPC-exportable base).  Signal buffer:
+// UNCACHED device memory so remote flag stores are immediately visible to
+// the local spin loops.
+
+int64_t ar_alloc_signals() {
+  void* ptr = nullptr;
+  hipError_t err = hipExtMallocWithFlags(
+      &ptr, (size_t)xgmi_allreduce_signal_bytes(), hipDeviceMallocUncached);
+  TORCH_CHECK(err == hipSuccess,
+              "uncached signal alloc failed: ", hipGetErrorString(err));
+  (void)hipMemset(ptr, 0, (size_t)xgmi_allreduce_signal_bytes());
+  (void)hipDeviceSynchronize();
+  return reinterpret_cast<int64_t>(ptr);
+}
+
+py::bytes ar_export_ptr(int64_t ptr) {
+  hipIpcMemHandle_t handle;
+  hipError_t err =
+      hipIpcGetMemHandle(&handle, reinterpret_cast<void*>(ptr));
+  TORCH_CHECK(err == hipSuccess,
+              "hipIpcGetMemHandle failed: ", hipGetErrorString(err));
+  return py::bytes(reinterpret_cast<const char*>(&handle), sizeof(handle));
+}
+
+unsigned ar_error_flag(int64_t sig_ptr) {
+  unsigned err_word = 0;
+  // error word is the last unsigned of Signals
+  char* base = reinterpret_cast<char*>(sig_ptr);
+  (void)hipMemcpy(&err_word,
+                  base + xgmi_allreduce_signal_bytes() - sizeof(unsigned),
+                  sizeof(unsigned), hipMemcpyDeviceToHost);
+  return err_word;
+}
+
+torch::Tensor xgmi_allreduce(torch::Tensor inp,
+                             std::vector<int64_t> sig_ptrs,
+                             std::vector<int64_t> data_ptrs, int64_t rank) {
+  check_bf16_contig(inp, "inp");
+  const int world = (int)sig_ptrs.size();
+  TORCH_CHECK(world >= 1 && world <= 8 &&
+                  data_ptrs.size() == (size_t)world && rank >= 0 &&
+                  rank < world,
+              "bad world/rank");
+  TORCH_CHECK(inp.numel() % 8 == 0, "numel must be a multiple of 8");
+  auto out = torch::empty_like(inp);
+  void* sp[8];
+  void* dp[8];
+  for (int i = 0; i < world; ++i) {
+    sp[i] = reinterpret_cast<void*>(sig_ptrs[i]);
+    dp[i] = reinterpret_cast<void*>(data_ptrs[i]);
+  }
+  launch_xgmi_allreduce(out.data_ptr(), inp.data_ptr(), sp, dp, world,
+                        (int)rank, (long)inp.numel(), current_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "x,residual += ; rmsnorm");
@@ -338,4 +398,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_ipc_close", &kv_ipc_close, "unmap a peer KV pool");
   m.def("kv_peer_copy", &kv_peer_copy,
         "gather local pages -> scatter into peer pool over xGMI");
+  m.def("ar_alloc_signals", &ar_alloc_signals,
+        "uncached signal buffer for the xGMI all-reduce");
+  m.def("ar_export_ptr", &ar_export_ptr, "hipIpc handle of a raw pointer");
+  m.def("ar_error_flag", &ar_error_flag, "read the all-reduce error word");
+  m.def("xgmi_allreduce", &xgmi_allreduce,
+        "one-shot all-reduce over peer-mapped HBM (graph-capturable)");
 }

@@ -65,9 +65,10 @@ def export_meta_local(cache: PagedKVCache) -> Dict[str, Any]:
 
 
 class PendingPush:
-    def __init__(self, event: torch.cuda.Event, nbytes: int):
+    def __init__(self, event: torch.cuda.Event, nbytes: int, keep=()):
         self._event = event
         self.nbytes = nbytes
+        self._keep = keep      # tensors the in-flight kernel reads
 
     def done(self) -> bool:
         return self._event.query()
@@ -108,19 +109,25 @@ class PeerKVPusher:
         assert list(shape)[3:] == list(dst_meta["shape"])[3:], \
             "page layout mismatch between src and dst pools"
         dst_base = self._map(dst_meta)
-        src_t = torch.tensor(src_pages, dtype=torch.int32,
-                             device=cache.kv.device)
-        dst_t = torch.tensor(dst_pages, dtype=torch.int32,
-                             device=cache.kv.device)
         chunk_bytes = shape[3] * shape[4] * shape[5] * 2
         nbytes = len(src_pages) * 2 * shape[0] * chunk_bytes
         ev = torch.cuda.Event()
         with torch.cuda.stream(self._stream):
+            # page-index tensors are allocated AND consumed on the
+            # transfer stream: allocating them on the caller's stream and
+            # letting them die before the kernel ran let the caching
+            # allocator reuse their memory mid-copy (GPU memory fault on
+            # the first hardware run); PendingPush also keeps them alive
+            # until the event resolves
+            src_t = torch.tensor(src_pages, dtype=torch.int32,
+                                 device=cache.kv.device)
+            dst_t = torch.tensor(dst_pages, dtype=torch.int32,
+                                 device=cache.kv.device)
             ops._require_hip().kv_peer_copy(
                 dst_base, cache.kv, src_t, dst_t,
                 int(dst_meta["num_pages"]))
             ev.record(self._stream)
-        return PendingPush(ev, nbytes)
+        return PendingPush(ev, nbytes, keep=(src_t, dst_t))
 
     def close(self) -> None:
         with self._lock:

@@ -28,6 +28,7 @@ HIP_SOURCES = [
     "silu_mul.hip",
     "skinny_gemm.hip",
     "kv_peer_copy.hip",
+    "allreduce.hip",
     "bindings.cpp",
 ]
 sources = [os.path.join(HIP_DIR, f) for f in HIP_SOURCES]

@@ -64,7 +64,7 @@ for _ in range(a.rounds):
 med = statistics.median(times)
 print(f"push {nbytes / 1e6:.1f} MB median {med * 1e3:.3f} ms "
       f"= {nbytes / med / 1e9:.1f} GB/s "
-      f"(read+write {2 * nbytes / med / 1e9:.1f} GB/s HBM)")
+      f"(read+write {2 * nbytes / med / 1e9:.1f} GB/s HBM)", flush=True)
 
 # -- decode overlap ---------------------------------------------------------
 eng = LLMEngine(cfg)
@@ -120,4 +120,4 @@ print(json.dumps({
     "itl_overhead_pct": round((itl_busy / itl_idle - 1) * 100, 1),
     "batch": a.batch, "seq_len": a.seq_len,
     "dst_device": str(dst_dev),
-}))
+}), flush=True)
