@@ -153,6 +153,11 @@ def main() -> int:
         tp_rank=rank if tp_mode else 0,
     )
     tp = TPContext(size=world, rank=rank) if tp_mode else None
+    if tp is not None and device == "cuda" and \
+            os.environ.get("RBG_XGMI_AR"):
+        # one-shot xGMI all-reduce -> TP decode becomes graph-capturable
+        if tp.attach_xgmi():
+            cfg.enforce_eager = args.eager
     pp = PPContext(size=world, stage=rank,
                    instance_ranks=list(range(world)),
                    tp_size=1) if pp_mode else None

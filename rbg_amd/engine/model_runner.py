@@ -158,6 +158,7 @@ class ModelRunner:
         use_graph = (not self.cfg.enforce_eager and
                      self.device.type == "cuda" and
                      self.pp.size == 1 and           # p2p hops: eager path
+                     self.model.tp.graph_safe and    # xGMI AR or no TP
                      n <= GRAPH_BATCH_SIZES[-1])
         if use_graph:
             logits = self._decode_graph(state, slots, n)

@@ -144,6 +144,14 @@ class ServeWorker:
                 self.cfg.tp_size = self.gcomm.tp_size
                 self.cfg.tp_rank = self.gcomm.tp_rank
                 self.model_tp = self.tp
+            if self.cfg.device == "cuda" and (
+                    ctx.args.get("xgmi_allreduce") or
+                    os.environ.get("RBG_XGMI_AR")):
+                # collective across the TP group: every rank constructs
+                # ServeWorker in the same wave (gang start)
+                if self.model_tp.attach_xgmi():
+                    self.cfg.enforce_eager = bool(
+                        ctx.args.get("enforce_eager", False))
         self.my_instance = os.environ.get(C.ENV_ROLE_INSTANCE_NAME, "")
         self.engine = LLMEngine(self.cfg, self.model_tp, self.model_pp)
         self._xfer_lock = threading.Lock()

@@ -60,15 +60,36 @@ def _linear_weight(out_f: int, in_f: int, device, dtype, gen,
 
 
 class TPContext:
-    """Holds the tensor-parallel process group (RCCL over xGMI)."""
+    """Holds the tensor-parallel process group (RCCL over xGMI), plus the
+    optional one-shot xGMI all-reduce (parallel/xgmi_allreduce.py) that
+    replaces RCCL for decode-sized messages — a plain kernel launch, so
+    it is hipGraph-capturable (RBG_XGMI_AR=1 or attach_xgmi())."""
 
     def __init__(self, size: int = 1, rank: int = 0, group=None):
         self.size = size
         self.rank = rank
         self.group = group
+        self.xgmi = None
+
+    def attach_xgmi(self, max_bytes: int = 32 << 20) -> bool:
+        """Initialize the peer-mapped all-reduce (collective: every rank
+        of the group must call this together).  Returns False on CPU."""
+        if self.size <= 1 or not torch.cuda.is_available():
+            return False
+        from ..parallel.xgmi_allreduce import XgmiAllReduce
+        self.xgmi = XgmiAllReduce(self.size, self.rank,
+                                  max_bytes=max_bytes, group=self.group)
+        return True
+
+    @property
+    def graph_safe(self) -> bool:
+        """Collectives in this context can be captured into hipGraphs."""
+        return self.size <= 1 or self.xgmi is not None
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.size > 1:
+            if self.xgmi is not None and self.xgmi.usable(t):
+                return self.xgmi.all_reduce(t)
             torch.distributed.all_reduce(t, group=self.group)
         return t
 
