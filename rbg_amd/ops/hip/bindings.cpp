@@ -24,6 +24,8 @@ void launch_prefill_attention(void*, const void*, const void*, const void*,
                               const void*, const void*, float, int, int, int,
                               int, int, int, hipStream_t);
 void launch_mfma_probe(void*, const void*, const void*, int, int, hipStream_t);
+void launch_mfma_probe32(void*, const void*, const void*, int, int, int,
+                         hipStream_t);
 void launch_skinny_gemm(void*, void*, const void*, const void*, int, int,
                         int, int, hipStream_t);
 void launch_reduce_splits(void*, const void*, int, long, hipStream_t);
@@ -198,6 +200,14 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b, int64_t a_split,
   auto d = torch::zeros({16, 16}, a.options().dtype(torch::kFloat32));
   launch_mfma_probe(d.data_ptr(), a.data_ptr(), b.data_ptr(), (int)a_split,
                     (int)b_split, current_stream());
+  return d;
+}
+
+torch::Tensor mfma_probe32(torch::Tensor a, torch::Tensor b, int64_t a_split,
+                           int64_t b_split, int64_t dmap) {
+  auto d = torch::zeros({32, 32}, a.options().dtype(torch::kFloat32));
+  launch_mfma_probe32(d.data_ptr(), a.data_ptr(), b.data_ptr(), (int)a_split,
+                      (int)b_split, (int)dmap, current_stream());
   return d;
 }
 
@@ -388,6 +398,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attention", &decode_attention, "paged flash-decode");
   m.def("prefill_attention", &prefill_attention, "varlen causal flash prefill");
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe");
+  m.def("mfma_probe32", &mfma_probe32, "32x32x16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "decode-shape GEMM (M<=128)",
         py::arg("a"), py::arg("w"), py::arg("force_splits") = 0);
   m.def("ipc_alloc_bf16", &ipc_alloc_bf16,
