@@ -128,7 +128,9 @@ def pick_decode_splits(num_seqs: int, num_kv_heads: int,
     if variant < 0:
         variant = DECODE_VARIANT
     if variant >= 4:
-        base = num_kv_heads * ((num_seqs + 1) // 2)
+        # v4/v5 pack 2 seqs per WG; v6 runs 1 seq per (2-wave) WG
+        per_wg = 2 if variant in (4, 5) else 1
+        base = num_kv_heads * ((num_seqs + per_wg - 1) // per_wg)
         by_ctx = max(1, max_context // 32)      # >= one page pair per split
     else:
         base = num_seqs * num_kv_heads

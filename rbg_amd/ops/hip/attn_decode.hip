@@ -664,6 +664,264 @@ __global__ __launch_bounds__(128, DIET ? 4 : 3) void decode_attn_mfma_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Producer/consumer MFMA decode (variant 6, ROUND2 §1c): TWO waves per
+// sequence — wave 1 (producer) owns all global->LDS staging, wave 0
+// (consumer) owns all math (S, softmax, PV) and never issues a global
+// load, so the math pipeline never parks on vmcnt.  Unlike a double
+// buffer, K and V keep ONE LDS panel each with a fine-grained flag
+// interlock (monotonic pair counters in LDS):
+//   producer pair i: wait cons_s >= i-1  -> write K_i, flag prod_k = i,
+//                    issue K_{i+1};
+//                    wait cons_pv >= i-1 -> write V_i, flag prod_v = i,
+//                    issue V_{i+1}
+//   consumer pair i: wait prod_k >= i -> S + softmax, flag cons_s = i;
+//                    wait prod_v >= i -> PV, flag cons_pv = i
+// LDS stays ~17 KB/WG (like variant 4's shared panel) and the producer
+// runs a full pair ahead.  Cross-wave LDS ordering: panel ds_writes are
+// drained (lgkmcnt(0)) before each flag write.
+template <int QPG>
+__global__ __launch_bounds__(128, 4) void decode_attn_pc_kernel(
+    float* __restrict__ partial_o,        // [splits, seqs, QH, D]
+    float* __restrict__ partial_ml,       // [splits, seqs, QH, 2]
+    __hip_bfloat16* __restrict__ out,     // [seqs, QH, D] (splits==1 path)
+    const __hip_bfloat16* __restrict__ q, // [seqs, QH, D]
+    const __hip_bfloat16* __restrict__ key_cache,  // [pages, KVH, 16, D]
+    const __hip_bfloat16* __restrict__ val_cache,
+    const int* __restrict__ block_tables, // [seqs, max_pages]
+    const int* __restrict__ context_lens, // [seqs]
+    const float scale, const int num_kv_heads, const int max_pages,
+    const int num_splits, const int num_seqs, const int q_stride) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 dm_bf8;
+  const int kvh = blockIdx.x;
+  const int seq = blockIdx.y;
+  const int split = blockIdx.z;
+  const int wave = threadIdx.x >> 6;     // 0 = consumer, 1 = producer
+  const int lane = threadIdx.x & 63;
+  const int gl = lane & 15;
+  const int gslice = lane >> 4;
+  const int num_q_heads = num_kv_heads * QPG;
+
+  constexpr int KP = 132;
+  __shared__ __hip_bfloat16 k_lds[32 * KP];
+  __shared__ __hip_bfloat16 v_img[32 * 128];
+  __shared__ __hip_bfloat16 p_lds[16 * 40];
+  __shared__ int flags[4];   // prod_k, prod_v, cons_s, cons_pv (pair idx)
+  if (threadIdx.x == 0) {
+    flags[0] = -1; flags[1] = -1; flags[2] = -1; flags[3] = -1;
+  }
+  __syncthreads();
+  volatile int* f_prod_k = &flags[0];
+  volatile int* f_prod_v = &flags[1];
+  volatile int* f_cons_s = &flags[2];
+  volatile int* f_cons_pv = &flags[3];
+
+  const int ctx = context_lens[seq];
+  const int chunk = 16;
+  const int nchunks = (ctx + chunk - 1) / chunk;
+  const int per_split = (nchunks + num_splits - 1) / num_splits;
+  const int key_begin = split * per_split * chunk;
+  const int key_end = min(ctx, (split + 1) * per_split * chunk);
+  const int pg_begin = key_begin >> 4;
+  const int pg_end = (key_end + 15) >> 4;
+  const int npairs = max(0, (pg_end - pg_begin + 1) >> 1);
+
+  const int* btab = block_tables + (size_t)seq * max_pages;
+  // ctx >= 1 so pg_end >= 1; for empty splits the clamp keeps block-table
+  // reads at the last REAL page (same convention as variant 4)
+  const int pg_last = pg_end - 1;
+
+#define PC_SPIN(fl, want)                                                    \
+  while (*(fl) < (want)) __builtin_amdgcn_s_sleep(2)
+
+  if (wave == 1) {
+    // ---- producer -------------------------------------------------------
+    const int srow = lane >> 1, schunk = (lane & 1) * 64;
+    const int myloc = srow >> 4;
+    auto page_base = [&](int pg, int local) -> size_t {
+      const int p = btab[min(pg + local, pg_last)];
+      return (((size_t)p * num_kv_heads + kvh) * 16 + (srow & 15)) *
+                 HEAD_DIM + schunk;
+    };
+    uint4 ka0, ka1, ka2, ka3, ka4, ka5, ka6, ka7;
+    uint4 va0, va1, va2, va3, va4, va5, va6, va7;
+    if (npairs > 0) {
+      DM_LOAD8(key_cache, ka0, ka1, ka2, ka3, ka4, ka5, ka6, ka7,
+               page_base(pg_begin, myloc));
+      DM_LOAD8(val_cache, va0, va1, va2, va3, va4, va5, va6, va7,
+               page_base(pg_begin, myloc));
+    }
+    for (int i = 0; i < npairs; ++i) {
+      const int pg_next = pg_begin + 2 * (i + 1);
+      PC_SPIN(f_cons_s, i - 1);          // K panel free
+      DM_WRITE_K();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (lane == 0) *f_prod_k = i;
+      if (i + 1 < npairs)
+        DM_LOAD8(key_cache, ka0, ka1, ka2, ka3, ka4, ka5, ka6, ka7,
+                 page_base(pg_next, myloc));
+      PC_SPIN(f_cons_pv, i - 1);         // V image free
+      DM_WRITE_V();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (lane == 0) *f_prod_v = i;
+      if (i + 1 < npairs)
+        DM_LOAD8(val_cache, va0, va1, va2, va3, va4, va5, va6, va7,
+                 page_base(pg_next, myloc));
+    }
+    return;
+  }
+
+  // ---- consumer ---------------------------------------------------------
+  dm_bf8 qa[4];
+  {
+    const __hip_bfloat16* qrow =
+        q + (size_t)seq * q_stride +
+        (size_t)(kvh * QPG + min(gl, QPG - 1)) * HEAD_DIM;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      union { uint4 u; dm_bf8 v; } cvt;
+      cvt.u = *reinterpret_cast<const uint4*>(qrow + ks * 32 + gslice * 8);
+      qa[ks] = cvt.v;
+    }
+  }
+  float m[4], l[4];
+  f32x4 acc_o[8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = NEG_INF; l[r] = 0.f; }
+#pragma unroll
+  for (int dt = 0; dt < 8; ++dt) acc_o[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int i = 0; i < npairs; ++i) {
+    const int pg = pg_begin + 2 * i;
+    PC_SPIN(f_prod_k, i);
+    f32x4 sA = {0.f, 0.f, 0.f, 0.f}, sB = sA;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      union { uint4 u; dm_bf8 v; } kfA, kfB;
+      kfA.u = *reinterpret_cast<const uint4*>(
+          &k_lds[gl * KP + ks * 32 + gslice * 8]);
+      kfB.u = *reinterpret_cast<const uint4*>(
+          &k_lds[(16 + gl) * KP + ks * 32 + gslice * 8]);
+      sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[ks], kfA.v, sA, 0, 0, 0);
+      sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[ks], kfB.v, sB, 0, 0, 0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // frag reads done
+    if (lane == 0) *f_cons_s = i;        // K panel may be overwritten
+
+    const int keyA = pg * 16 + gl;
+    const int keyB = keyA + 16;
+    const bool vA = keyA >= key_begin && keyA < key_end;
+    const bool vB = keyB >= key_begin && keyB < key_end;
+    float pA[4], pB[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float a = vA ? sA[r] * scale : NEG_INF;
+      const float b = vB ? sB[r] * scale : NEG_INF;
+      const float tm = group16_max(fmaxf(a, b));
+      const float m_new = fmaxf(m[r], tm);
+      const float alpha = __expf(m[r] - m_new);
+      pA[r] = vA ? __expf(a - m_new) : 0.f;
+      pB[r] = vB ? __expf(b - m_new) : 0.f;
+      l[r] = l[r] * alpha + group16_sum(pA[r] + pB[r]);
+      m[r] = m_new;
+      acc_o[0][r] *= alpha; acc_o[1][r] *= alpha;
+      acc_o[2][r] *= alpha; acc_o[3][r] *= alpha;
+      acc_o[4][r] *= alpha; acc_o[5][r] *= alpha;
+      acc_o[6][r] *= alpha; acc_o[7][r] *= alpha;
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      p_lds[(4 * gslice + r) * 40 + gl] = f2bf(pA[r]);
+      p_lds[(4 * gslice + r) * 40 + 16 + gl] = f2bf(pB[r]);
+    }
+
+    PC_SPIN(f_prod_v, i);
+    {
+      union { uint4 u; dm_bf8 v; } paf;
+      const int pk = gslice * 8;
+      paf.u.x = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk]);
+      paf.u.y = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 2]);
+      paf.u.z = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 4]);
+      paf.u.w = *reinterpret_cast<const uint*>(&p_lds[gl * 40 + pk + 6]);
+      const unsigned vaddr = (unsigned)(unsigned long long)(
+          &v_img[gslice * 64 + gl * 4]);
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        unsigned long long vlo[4], vhi[4];
+        if (half == 0) {
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %8\n\t"
+              "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+              "ds_read_b64_tr_b16 %2, %8 offset:1024\n\t"
+              "ds_read_b64_tr_b16 %3, %8 offset:1536\n\t"
+              "ds_read_b64_tr_b16 %4, %8 offset:2048\n\t"
+              "ds_read_b64_tr_b16 %5, %8 offset:2560\n\t"
+              "ds_read_b64_tr_b16 %6, %8 offset:3072\n\t"
+              "ds_read_b64_tr_b16 %7, %8 offset:3584\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(vlo[0]), "=&v"(vhi[0]), "=&v"(vlo[1]), "=&v"(vhi[1]),
+                "=&v"(vlo[2]), "=&v"(vhi[2]), "=&v"(vlo[3]), "=&v"(vhi[3])
+              : "v"(vaddr)
+              : "memory");
+        } else {
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %8 offset:4096\n\t"
+              "ds_read_b64_tr_b16 %1, %8 offset:4608\n\t"
+              "ds_read_b64_tr_b16 %2, %8 offset:5120\n\t"
+              "ds_read_b64_tr_b16 %3, %8 offset:5632\n\t"
+              "ds_read_b64_tr_b16 %4, %8 offset:6144\n\t"
+              "ds_read_b64_tr_b16 %5, %8 offset:6656\n\t"
+              "ds_read_b64_tr_b16 %6, %8 offset:7168\n\t"
+              "ds_read_b64_tr_b16 %7, %8 offset:7680\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(vlo[0]), "=&v"(vhi[0]), "=&v"(vlo[1]), "=&v"(vhi[1]),
+                "=&v"(vlo[2]), "=&v"(vhi[2]), "=&v"(vlo[3]), "=&v"(vhi[3])
+              : "v"(vaddr)
+              : "memory");
+        }
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          union { struct { unsigned long long lo, hi; } u; dm_bf8 vf2; } vf;
+          vf.u.lo = vlo[dt];
+          vf.u.hi = vhi[dt];
+          acc_o[half * 4 + dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              paf.v, vf.vf2, acc_o[half * 4 + dt], 0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    if (lane == 0) *f_cons_pv = i;       // V image may be overwritten
+  }
+#undef PC_SPIN
+
+  // epilogue identical to variant 4
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = 4 * gslice + r;
+    if (row >= QPG) continue;
+    const int qh = kvh * QPG + row;
+    if (num_splits == 1) {
+      const float inv_l = l[r] > 0.f ? 1.f / l[r] : 0.f;
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt)
+        out[((size_t)seq * num_q_heads + qh) * HEAD_DIM + dt * 16 + gl] =
+            f2bf(acc_o[dt][r] * inv_l);
+    } else {
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt)
+        partial_o[(((size_t)split * num_seqs + seq) * num_q_heads + qh) *
+                      HEAD_DIM + dt * 16 + gl] = acc_o[dt][r];
+      if (gl == 0) {
+        float* ml = partial_ml +
+            (((size_t)split * num_seqs + seq) * num_q_heads + qh) * 2;
+        ml[0] = m[r];
+        ml[1] = l[r];
+      }
+    }
+  }
+}
+
 // Merge split partials: one block per (seq, q_head).
 __global__ void decode_combine_kernel(
     __hip_bfloat16* __restrict__ out,       // [seqs, QH, D]
@@ -711,6 +969,31 @@ void launch_decode_attention(void* out, void* partial_o, void* partial_ml,
                      (const int*)block_tables, (const int*)context_lens,      \
                      scale, num_kv_heads, page_size, max_pages, num_splits,  \
                      q_stride)
+  if (variant == 6 && page_size == 16) {
+    dim3 pgrid(num_kv_heads, num_seqs, num_splits), pblock(128);
+#define LAUNCH_PC(QPG)                                                       \
+    hipLaunchKernelGGL((decode_attn_pc_kernel<QPG>), pgrid, pblock, 0,       \
+                       stream, (float*)partial_o, (float*)partial_ml,        \
+                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,       \
+                       (const __hip_bfloat16*)key_cache,                     \
+                       (const __hip_bfloat16*)val_cache,                     \
+                       (const int*)block_tables, (const int*)context_lens,   \
+                       scale, num_kv_heads, max_pages, num_splits,           \
+                       num_seqs, q_stride)
+    if (qpg == 1) LAUNCH_PC(1);
+    else if (qpg == 2) LAUNCH_PC(2);
+    else if (qpg == 4) LAUNCH_PC(4);
+    else if (qpg == 8) LAUNCH_PC(8);
+    else return;
+#undef LAUNCH_PC
+    if (num_splits > 1) {
+      dim3 cgrid(num_seqs, num_q_heads), cblock(HEAD_DIM);
+      hipLaunchKernelGGL(decode_combine_kernel, cgrid, cblock, 0, stream,
+                         (__hip_bfloat16*)out, (const float*)partial_o,
+                         (const float*)partial_ml, num_splits, num_q_heads);
+    }
+    return;
+  }
   if ((variant == 4 || variant == 5) && page_size == 16) {
     dim3 mgrid(num_kv_heads, (num_seqs + 1) / 2, num_splits), mblock(128);
 #define LAUNCH_MFMA(QPG, DIET)                                               \

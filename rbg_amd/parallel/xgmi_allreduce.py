@@ -74,7 +74,7 @@ class XgmiAllReduce:
         hip = ops._require_hip()
         for obj, rank in ((a, 0), (b, 1)):
             obj.world, obj.rank = 2, rank
-            obj.max_bytes = 1 << 20
+            obj.max_bytes = 8 << 20
             obj.device = device
             obj.data = hip.ipc_alloc_bf16([obj.max_bytes // 2])
             obj.sig_ptr = hip.ar_alloc_signals()
@@ -91,6 +91,11 @@ class XgmiAllReduce:
                 t.numel() * 2 <= self.max_bytes)
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if not self.usable(t):
+            raise ValueError(
+                f"tensor ({t.numel() * 2} B, {t.dtype}) does not fit the "
+                f"xGMI all-reduce buffer ({self.max_bytes} B) — callers "
+                "must check usable() and fall back to RCCL")
         return ops._require_hip().xgmi_allreduce(
             t, self.sig_ptrs, self.data_ptrs, self.rank)
 

@@ -90,34 +90,53 @@ def timed_decode(steps):
 
 
 itl_idle = timed_decode(a.decode_steps)
-# saturating push stream on the transfer stream during decode
-stop = False
-pending = []
 
 
-def push_loop():
-    while not stop:
-        pending.append(pusher.push(src, src_pages, meta, dst_pages))
-        while len(pending) > 2:
-            pending.pop(0).wait()
+def measure_during_pushes(interval_s):
+    """Decode ITL while pushes run on the transfer stream: interval None =
+    saturating back-to-back stream (HBM-contention ceiling on 1 GPU — the
+    realistic cross-GPU case only writes ~150 GB/s into this device);
+    otherwise one full-sequence migration per interval (a realistic P/D
+    admission rate)."""
+    import threading
+    stop = threading.Event()
+    pending = []
+
+    def push_loop():
+        nxt = time.monotonic()
+        while not stop.is_set():
+            pending.append(pusher.push(src, src_pages, meta, dst_pages))
+            while len(pending) > 2:
+                pending.pop(0).wait()
+            if interval_s:
+                nxt += interval_s
+                dt = nxt - time.monotonic()
+                if dt > 0:
+                    time.sleep(dt)
+    th = threading.Thread(target=push_loop)
+    th.start()
+    time.sleep(0.05)
+    itl = timed_decode(a.decode_steps)
+    stop.set()
+    th.join()
+    for p in pending:
+        p.wait()
+    return itl
 
 
-import threading
-th = threading.Thread(target=push_loop)
-th.start()
-time.sleep(0.05)
-itl_busy = timed_decode(a.decode_steps)
-stop = True
-th.join()
-for p in pending:
-    p.wait()
+itl_saturated = measure_during_pushes(None)
+# 20 migrations/s = a 2048-token sequence admitted to decode every 50 ms
+itl_rate20 = measure_during_pushes(0.05)
 print(json.dumps({
     "push_mb": round(nbytes / 1e6, 1),
     "push_ms_median": round(med * 1e3, 3),
     "push_gb_s": round(nbytes / med / 1e9, 1),
     "itl_idle_ms": round(itl_idle * 1e3, 3),
-    "itl_during_continuous_push_ms": round(itl_busy * 1e3, 3),
-    "itl_overhead_pct": round((itl_busy / itl_idle - 1) * 100, 1),
+    "itl_saturated_push_ms": round(itl_saturated * 1e3, 3),
+    "itl_saturated_overhead_pct": round(
+        (itl_saturated / itl_idle - 1) * 100, 1),
+    "itl_rate20_push_ms": round(itl_rate20 * 1e3, 3),
+    "itl_rate20_overhead_pct": round((itl_rate20 / itl_idle - 1) * 100, 1),
     "batch": a.batch, "seq_len": a.seq_len,
     "dst_device": str(dst_dev),
 }), flush=True)

@@ -10,6 +10,9 @@ QPS as JSON.
 
 Usage: python tools/stress.py [--groups 10] [--qps 5] [--roles 2]
                               [--replicas 2] [--out stress.json]
+       python tools/stress.py --matrix   # scenario sweep (the reference
+                              harness's multi-scenario mode) with per-phase
+                              CPU / RSS sampling (the pprof-scrape analog)
 """
 from __future__ import annotations
 
@@ -94,6 +97,114 @@ def run_phase(mgr: Manager, names: List[str], qps: float, submit, settled,
     }
 
 
+class ResourceSampler:
+    """Per-phase controller resource profile — the stress harness's pprof
+    scrape analog (reference test/stress/pprof.go): samples process CPU
+    time and RSS at 100 ms while a phase runs."""
+
+    def __init__(self):
+        import threading
+        self._stop = threading.Event()
+        self._thread = None
+        self.samples = []
+
+    def __enter__(self):
+        import threading
+        self._t0 = time.monotonic()
+        self._cpu0 = self._cpu()
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+        return self
+
+    @staticmethod
+    def _cpu() -> float:
+        import resource
+        ru = resource.getrusage(resource.RUSAGE_SELF)
+        return ru.ru_utime + ru.ru_stime
+
+    @staticmethod
+    def _rss() -> int:
+        import resource
+        return resource.getrusage(resource.RUSAGE_SELF).ru_maxrss * 1024
+
+    def _loop(self):
+        while not self._stop.wait(0.1):
+            self.samples.append(self._rss())
+
+    def __exit__(self, *exc):
+        self._stop.set()
+        self._thread.join(timeout=1)
+        wall = max(1e-9, time.monotonic() - self._t0)
+        self.report = {
+            "cpu_s": round(self._cpu() - self._cpu0, 3),
+            "cpu_pct": round(100 * (self._cpu() - self._cpu0) / wall, 1),
+            "rss_peak_mb": round(max(self.samples or [self._rss()]) / 1e6, 1),
+        }
+        return False
+
+
+def run_scenario(groups: int, qps: float, roles: int, replicas: int,
+                 timeout: float) -> Dict:
+    run_root = tempfile.mkdtemp(prefix="rbg-stress-")
+    mgr = Manager(ManagerOptions(run_root=run_root, num_gpus=8,
+                                 resync_period=0.1))
+    mgr.start()
+    names = [f"stress-{i}" for i in range(groups)]
+    report = {"groups": groups, "qps": qps, "roles": roles,
+              "replicas": replicas}
+    try:
+        with ResourceSampler() as rs:
+            report["create"] = run_phase(
+                mgr, names, qps,
+                submit=lambda n: mgr.store.create(
+                    echo_rbg(n, roles, replicas)),
+                settled=lambda n: ready(mgr, n),
+                timeout=timeout)
+        report["create"]["controller"] = rs.report
+
+        def scale_up(n):
+            def mutate(cur):
+                cur.spec.roles[-1].replicas += 1
+                return cur
+            mgr.store.apply(C.KIND_RBG, n, mutate)
+
+        def scaled(n):
+            rbg = mgr.store.try_get(C.KIND_RBG, n)
+            if rbg is None or not ready(mgr, n):
+                return False
+            want = rbg.spec.roles[-1].replicas
+            for st in rbg.status.role_statuses:
+                if st.name == rbg.spec.roles[-1].name:
+                    return st.ready_replicas >= want
+            return False
+        with ResourceSampler() as rs:
+            report["update"] = run_phase(mgr, names, qps, scale_up, scaled,
+                                         timeout)
+        report["update"]["controller"] = rs.report
+
+        def delete(n):
+            def mark(cur):
+                cur.metadata.deletion_timestamp = time.time()
+                return cur
+            mgr.store.apply(C.KIND_RBG, n, mark)
+        with ResourceSampler() as rs:
+            report["delete"] = run_phase(
+                mgr, names, qps, delete,
+                settled=lambda n: mgr.store.try_get(C.KIND_RBG, n) is None,
+                timeout=timeout)
+        report["delete"]["controller"] = rs.report
+    finally:
+        mgr.stop()
+    return report
+
+
+MATRIX = [  # (groups, qps) — reference stress README's scenario table shape
+    (10, 5.0),
+    (25, 10.0),
+    (50, 20.0),
+]
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--groups", type=int, default=10)
@@ -101,8 +212,25 @@ def main() -> int:
     ap.add_argument("--roles", type=int, default=2)
     ap.add_argument("--replicas", type=int, default=2)
     ap.add_argument("--timeout", type=float, default=120.0)
+    ap.add_argument("--matrix", action="store_true",
+                    help="run the scenario sweep instead of one scenario")
     ap.add_argument("--out", default="")
     args = ap.parse_args()
+
+    if args.matrix:
+        scenarios = []
+        for groups, qps in MATRIX:
+            scenarios.append(run_scenario(groups, qps, args.roles,
+                                          args.replicas, args.timeout))
+            print(f"scenario {groups}g@{qps}qps done", file=sys.stderr)
+        out = json.dumps({"matrix": scenarios}, indent=1)
+        print(out)
+        if args.out:
+            with open(args.out, "w") as f:
+                f.write(out)
+        ok = all(sc[ph]["timed_out"] == 0 for sc in scenarios
+                 for ph in ("create", "update", "delete"))
+        return 0 if ok else 1
 
     run_root = tempfile.mkdtemp(prefix="rbg-stress-")
     mgr = Manager(ManagerOptions(run_root=run_root, num_gpus=8,
