@@ -89,6 +89,24 @@ def main():
             row(["phase", "submitted", "settled", "p50 s", "p90 s",
                  "p99 s"], "th") + "".join(rows) + "</table>")
 
+    # stress matrix (round 2: scenario sweep with controller CPU/RSS)
+    d = load_json(os.path.join(args.profiles, "stress_matrix_r2.json"))
+    if d and d.get("matrix"):
+        rows = []
+        for sc in d["matrix"]:
+            for phase in ("create", "update", "delete"):
+                s = sc.get(phase) or {}
+                c = s.get("controller") or {}
+                rows.append(row([f"{sc['groups']}g@{sc['qps']}qps", phase,
+                                 s.get("p50_s"), s.get("p99_s"),
+                                 s.get("timed_out"), c.get("cpu_pct"),
+                                 c.get("rss_peak_mb")]))
+        sections.append(
+            "<h2>Controller stress matrix</h2><table>" +
+            row(["scenario", "phase", "p50 s", "p99 s", "timed out",
+                 "controller cpu %", "rss MB"], "th") +
+            "".join(rows) + "</table>")
+
     # recovery
     d = load_json(os.path.join(args.profiles, "recovery_llama8b_gpu.json"))
     if d:
