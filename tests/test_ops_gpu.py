@@ -328,8 +328,10 @@ def test_xgmi_allreduce_pair_protocol():
 
 @pytest.mark.gpu
 def test_xgmi_allreduce_graph_capture():
-    """The all-reduce captures into a hipGraph and replays correctly —
-    the property that keeps TP decode off the eager path."""
+    """Both ranks capture their all-reduce into hipGraphs (the production
+    TP shape: every rank captures its own decode graph) and replays with
+    changing inputs produce the elementwise sum — the property that keeps
+    TP decode off the eager path."""
     import torch
 
     from rbg_amd.parallel.xgmi_allreduce import XgmiAllReduce
@@ -347,24 +349,25 @@ def test_xgmi_allreduce_graph_capture():
         ob = b.all_reduce(xb)
     torch.cuda.synchronize()
 
-    # capture rank-a's call; rank-b runs eagerly to partner the replays
     ga = torch.cuda.CUDAGraph()
-    with torch.cuda.stream(sa):
-        with torch.cuda.graph(ga, stream=sa):
-            oa = a.all_reduce(xa)
-    # the capture itself does not EXECUTE rank-a's kernel, so rank-b's
-    # eager partner call below pairs with the first replay
+    gb = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(ga, stream=sa):
+        oa = a.all_reduce(xa)
+    with torch.cuda.graph(gb, stream=sb):
+        ob = b.all_reduce(xb)
     for round_ in range(3):
         xa.fill_(float(round_ + 1))
         xb.fill_(float(10 * (round_ + 1)))
         torch.cuda.synchronize()
-        with torch.cuda.stream(sb):
-            ob = b.all_reduce(xb)
         with torch.cuda.stream(sa):
             ga.replay()
+        with torch.cuda.stream(sb):
+            gb.replay()
         torch.cuda.synchronize()
         a.check()
         b.check()
         want = float(round_ + 1) + 10 * (round_ + 1)
-        assert torch.all(oa.float() == want), (round_, oa.float().unique())
-        assert torch.all(ob.float() == want), round_
+        assert torch.all(oa.float() == want), \
+            (round_, "oa", oa.float().unique()[:8].tolist())
+        assert torch.all(ob.float() == want), \
+            (round_, "ob", ob.float().unique()[:8].tolist())

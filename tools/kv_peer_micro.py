@@ -72,7 +72,9 @@ torch.manual_seed(0)
 prompts = [torch.randint(0, model.vocab_size, (512,)).tolist()
            for _ in range(a.batch)]
 for p in prompts:
-    eng.add_request(p, SamplingParams(max_new_tokens=a.decode_steps + 64,
+    # 3 timed phases + warmup must all stay in pure decode: size max_new
+    # so no sequence retires mid-experiment
+    eng.add_request(p, SamplingParams(max_new_tokens=4 * a.decode_steps + 64,
                                       ignore_eos=True))
 while eng.scheduler.waiting:
     eng.step()
@@ -84,7 +86,8 @@ torch.cuda.synchronize()
 def timed_decode(steps):
     t0 = time.monotonic()
     for _ in range(steps):
-        eng.step()
+        mode = eng.step()
+        assert mode == "decode", mode    # retirement would fake the ITL
     torch.cuda.synchronize()
     return (time.monotonic() - t0) / steps
 
