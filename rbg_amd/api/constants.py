@@ -45,6 +45,11 @@ LABEL_SCALING_ADAPTER = f"{PREFIX}/scaling-adapter"
 ANNO_GANG_SCHEDULING = f"{PREFIX}/gang-scheduling"            # "true"/"false"
 ANNO_GANG_TIMEOUT = f"{PREFIX}/gang-scheduling-timeout"       # seconds
 ANNO_EXCLUSIVE_TOPOLOGY = f"{PREFIX}/exclusive-topology"      # topo key, e.g. "xgmi-hive"
+# KV migration dataplane: "peer" (default) = hipIpc + xGMI push, P/D roles
+# need no collective world so replica scale-out never rebuilds comms
+# (ROUND2 design 4 resolved architecturally); "collective" forces the
+# RCCL/gloo send-recv world (CPU engines always use it)
+ANNO_KV_TRANSFER = f"{PREFIX}/kv-transfer"
 ANNO_INPLACE_SCHEDULING = f"{PREFIX}/in-place-scheduling"     # "preferred"|"required"
 ANNO_INPLACE_GRANULARITY = f"{PREFIX}/in-place-scheduling-granularity"  # "instance"|"component"
 ANNO_RESTART_TRIGGER_POLICY = f"{PREFIX}/restart-trigger-policy"  # "Restart"|"Ignore"

@@ -108,10 +108,23 @@ class RoleBasedGroupController:
             engines = [(c, e) for c in comps
                        for e in (c.template.engines if c.template else [])
                        if e.runner == "llm-engine"]
-            communicates = (role.pattern == C.PATTERN_LEADER_WORKER or any(
-                e.args.get("mode") in ("prefill", "decode")
-                for _, e in engines))
-            if not engines or not communicates:
+            is_lw = role.pattern == C.PATTERN_LEADER_WORKER
+            is_pd = any(e.args.get("mode") in ("prefill", "decode")
+                        for _, e in engines)
+            if not engines or not (is_lw or is_pd):
+                continue
+            # peer KV mode (the default): GPU-resident P/D roles migrate
+            # pages over hipIpc/xGMI and need NO collective world — so
+            # scaling a replica pool never tears down communicators
+            # (ROUND2 design 4).  CPU engines (tests) and an explicit
+            # kv-transfer=collective keep the send/recv world.
+            kv_mode = rbg.metadata.annotations.get(C.ANNO_KV_TRANSFER,
+                                                   "peer")
+            all_gpu = all(
+                e.resources is not None and not e.resources.cpu_only and
+                e.resources.gpus + e.resources.hbm_bytes > 0
+                for _, e in engines)
+            if is_pd and not is_lw and kv_mode == "peer" and all_gpu:
                 continue
             for i in range(role.replicas):
                 inst = instance_name(rbg.metadata.name, role.name, i)

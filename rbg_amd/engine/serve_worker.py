@@ -532,21 +532,29 @@ class ServeWorker:
             return {"decode_seq_id": res.get("seq_id"),
                     "ticket": res.get("ticket"),
                     "first_token": first_token, "ttft_s": seq.ttft()}
-        # legacy standalone transfer group (explicit transfer_port args)
-        plan = self.plan()
-        dinst = next(i for i in plan.decode_instances
-                     if i["name"] == decode_instance)
-        client = RpcClient("127.0.0.1", int(dinst["ports"][0]))
+        # No group comm world (peer KV mode, ROUND2 design 4): resolve the
+        # decode RPC endpoint from discovery and push over xGMI.  The
+        # standalone gloo/RCCL transfer group (explicit transfer_port
+        # args) is only formed if the peer handshake is declined.
+        client = RpcClient("127.0.0.1", self._decode_port(decode_instance))
         try:
+            has_group = bool(self.ctx.args.get("transfer_port"))
+            src_rank = (self.plan().rank_of(self.my_instance)
+                        if has_group else -1)
             res = client.call(
-                "import_seq", peer_ok=peer_ok,
-                src_rank=plan.rank_of(self.my_instance), **meta)
+                "import_seq", peer_ok=peer_ok, src_rank=src_rank, **meta)
             if res.get("peer") is not None:
                 self._peer_push(pages, res["peer"], res["dst_pages"])
                 client.call("import_commit", seq_id=res["seq_id"])
+            elif has_group:
+                self.transfer.send_pages(
+                    self.engine.runner.cache, pages,
+                    self.plan().rank_of(decode_instance))
             else:
-                self.transfer.send_pages(self.engine.runner.cache, pages,
-                                         plan.rank_of(decode_instance))
+                raise RuntimeError(
+                    "decode peer declined the xGMI push and no transfer "
+                    "group is configured (set transfer_port or annotate "
+                    "kv-transfer=collective)")
         finally:
             self._release_parked(parked)
             client.close()
@@ -632,6 +640,11 @@ class ServeWorker:
             return {"ticket": ticket}
         recv_from = (self._peer_rank(src_instance)
                      if self.gcomm is not None else src_rank)
+        if recv_from < 0:
+            raise RuntimeError(
+                "import_seq without a peer push needs a transfer group "
+                "(no comm world and src_rank unset — peer KV mode requires "
+                "GPU engines on both sides)")
         seq = self._import_alloc(tokens, num_pages, max_new_tokens,
                                  temperature)
 

@@ -163,4 +163,9 @@ void launch_xgmi_allreduce(void* out, const void* in, void** sig_ptrs,
 
 long xgmi_allreduce_signal_bytes() { return (long)sizeof(Signals); }
 
+long xgmi_allreduce_error_offset() { return (long)offsetof(Signals, error); }
+long xgmi_allreduce_counter_offset() {
+  return (long)offsetof(Signals, counter);
+}
+
 }  // extern "C"

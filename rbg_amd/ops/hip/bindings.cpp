@@ -34,6 +34,8 @@ void launch_kv_peer_copy(void*, const void*, const void*, const void*, int,
 void launch_xgmi_allreduce(void*, const void*, void**, void**, int, int,
                            long, hipStream_t);
 long xgmi_allreduce_signal_bytes();
+long xgmi_allreduce_error_offset();
+long xgmi_allreduce_counter_offset();
 }
 
 namespace {
@@ -360,12 +362,21 @@ py::bytes ar_export_ptr(int64_t ptr) {
 
 unsigned ar_error_flag(int64_t sig_ptr) {
   unsigned err_word = 0;
-  // error word is the last unsigned of Signals
+  // offsetof, NOT sizeof-4: the 128-byte struct alignment pads the tail,
+  // so sizeof-4 read the padding and timeouts were invisible (r2 GPU run)
   char* base = reinterpret_cast<char*>(sig_ptr);
-  (void)hipMemcpy(&err_word,
-                  base + xgmi_allreduce_signal_bytes() - sizeof(unsigned),
+  (void)hipMemcpy(&err_word, base + xgmi_allreduce_error_offset(),
                   sizeof(unsigned), hipMemcpyDeviceToHost);
   return err_word;
+}
+
+torch::Tensor ar_dump_signals(int64_t sig_ptr) {
+  // debug: the whole Signals block as int32 on CPU
+  long n = xgmi_allreduce_signal_bytes() / 4;
+  auto t = torch::empty({n}, torch::TensorOptions().dtype(torch::kInt32));
+  (void)hipMemcpy(t.data_ptr(), reinterpret_cast<void*>(sig_ptr), n * 4,
+                  hipMemcpyDeviceToHost);
+  return t;
 }
 
 torch::Tensor xgmi_allreduce(torch::Tensor inp,
@@ -413,6 +424,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "uncached signal buffer for the xGMI all-reduce");
   m.def("ar_export_ptr", &ar_export_ptr, "hipIpc handle of a raw pointer");
   m.def("ar_error_flag", &ar_error_flag, "read the all-reduce error word");
+  m.def("ar_dump_signals", &ar_dump_signals, "debug: dump the signal block");
+  m.def("ar_counter_offset", &xgmi_allreduce_counter_offset);
   m.def("xgmi_allreduce", &xgmi_allreduce,
         "one-shot all-reduce over peer-mapped HBM (graph-capturable)");
 }

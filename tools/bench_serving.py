@@ -44,14 +44,22 @@ def build_rbg(args, engine_args):
     if args.mode == "pd":
         shared = dict(engine_args,
                       prefill_roles=["prefill"], decode_roles=["decode"])
+        # per-role KV pools: the prefill engine never holds output tokens
+        # (pages migrate at first token), the decode pool carries the full
+        # in+out trace — sized separately so both engines + weights +
+        # decode hipGraph pools fit 288 GB on a 1-GPU colocated run
+        prefill_pool = args.prompts * (args.in_len + 64) + 8192
+        decode_pool = args.prompts * (args.in_len + args.out_len + 64) + 8192
         roles = [
             role("router", "router",
                  {"dispatch": "pd", "prefill_roles": ["prefill"],
                   "decode_roles": ["decode"],
                   "vocab_size": args.vocab}, 0),
-            role("prefill", "llm-engine", dict(shared, mode="prefill"),
+            role("prefill", "llm-engine",
+                 dict(shared, mode="prefill", kv_pool_tokens=prefill_pool),
                  args.gpus_per_engine, deps=("router",)),
-            role("decode", "llm-engine", dict(shared, mode="decode"),
+            role("decode", "llm-engine",
+                 dict(shared, mode="decode", kv_pool_tokens=decode_pool),
                  args.gpus_per_engine, deps=("router",)),
         ]
     else:
