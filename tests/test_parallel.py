@@ -257,3 +257,47 @@ def test_pp2_chunked_prefill_matches(tmp_path):
     ref = s.output_tokens
     assert torch.load(str(tmp_path / "ppc_rank0.pt")) == ref
     assert torch.load(str(tmp_path / "ppc_rank1.pt")) == ref
+
+
+def test_comm_plan_excludes_peer_kv_pd_roles():
+    """GPU-resident P/D-only roles carry NO collective world under the
+    default kv-transfer=peer dataplane (scale-out never rebuilds comms);
+    CPU engines, leaderWorker roles, and kv-transfer=collective keep it."""
+    from rbg_amd.api import constants as C
+    from rbg_amd.api.types import (EngineResources, EngineSpec,
+                                   EngineTemplate, LeaderWorkerPattern,
+                                   ObjectMeta, RoleBasedGroup,
+                                   RoleBasedGroupSpec, RoleSpec)
+    from rbg_amd.controller.rbg_controller import RoleBasedGroupController
+    from rbg_amd.store.store import Store
+
+    def pd_rbg(cpu=False, annotations=None, lw=False):
+        res = (EngineResources(cpu_only=True) if cpu
+               else EngineResources(gpus=1))
+        def tmpl(mode):
+            return EngineTemplate(engines=[EngineSpec(
+                name="engine", runner="llm-engine",
+                args={"mode": mode}, resources=res)])
+        roles = [
+            RoleSpec(name="prefill", replicas=1, template=tmpl("prefill")),
+            RoleSpec(name="decode", replicas=2, template=tmpl("decode")),
+        ]
+        if lw:
+            roles[0].pattern = C.PATTERN_LEADER_WORKER
+            roles[0].leader_worker_pattern = LeaderWorkerPattern(
+                size=2, leader_template=tmpl("prefill"),
+                worker_template=tmpl("prefill"))
+        return RoleBasedGroup(
+            metadata=ObjectMeta(name="g", annotations=annotations or {}),
+            spec=RoleBasedGroupSpec(roles=roles))
+
+    ctrl = RoleBasedGroupController(Store(), None)
+    # GPU P/D roles, default peer mode -> no comm world
+    assert ctrl._comm_plan(pd_rbg()) == {}
+    # CPU engines -> send/recv world kept
+    assert "rbg.comm-world" in ctrl._comm_plan(pd_rbg(cpu=True))
+    # explicit collective mode -> world kept
+    assert "rbg.comm-world" in ctrl._comm_plan(pd_rbg(
+        annotations={C.ANNO_KV_TRANSFER: "collective"}))
+    # leaderWorker (TP) roles always get the world
+    assert "rbg.comm-world" in ctrl._comm_plan(pd_rbg(lw=True))
