@@ -206,16 +206,22 @@ def run_pd(args, rank: int, world: int, device: str) -> Optional[dict]:
                 reply: List[Any] = [None]
                 dist.recv_object_list(reply, src=r)
                 dst_lists, peer_meta = reply[0]
+                pushed = False
                 if device == "cuda":
-                    # ONE batched kernel launch for the whole rank-batch
-                    src_all = [p for _s, pages in done for p in pages]
-                    dst_all = [p for lst in dst_lists for p in lst]
-                    migrator.push_remote(prefill_engine.runner.cache,
-                                         peer_meta, src_all, dst_all, r)
-                else:
-                    for (_s, pages), dst_list in zip(done, dst_lists):
+                    try:
+                        # ONE batched kernel launch for the whole batch
+                        src_all = [p for _s, pages in done for p in pages]
+                        dst_all = [p for lst in dst_lists for p in lst]
                         migrator.push_remote(prefill_engine.runner.cache,
-                                             None, pages, dst_list, r)
+                                             peer_meta, src_all, dst_all, r)
+                        pushed = True
+                    except Exception as e:  # noqa: BLE001
+                        # first-multi-GPU safety net: if the cross-device
+                        # hipIpc mapping is refused, fall back to the gloo
+                        # wire path instead of failing the run
+                        print(f"[pd_bench] xGMI push to rank {r} failed "
+                              f"({e!r}); falling back to gloo wire",
+                              flush=True)
                 checks = None
                 if _verify():
                     if device == "cuda":
@@ -225,8 +231,17 @@ def run_pd(args, rank: int, world: int, device: str) -> Optional[dict]:
                               for _s, pages in done]
                 commit = [([(s.prompt_tokens, s.output_tokens[0], lst)
                             for (s, _pg), lst in zip(done, dst_lists)],
-                           checks)]
+                           checks, pushed)]
                 dist.send_object_list(commit, dst=r)
+                if not pushed:
+                    from ..parallel.comm import to_wire
+                    for (_s, pages), _lst in zip(done, dst_lists):
+                        idx = torch.tensor(
+                            pages, dtype=torch.int64,
+                            device=prefill_engine.runner.cache.kv.device)
+                        buf = prefill_engine.runner.cache.kv.index_select(
+                            2, idx).contiguous().cpu()
+                        dist.send(to_wire(buf), r)
                 now = time.monotonic()
                 for s, pages in done:
                     prefill_engine.runner.cache.free(pages)
@@ -247,7 +262,23 @@ def run_pd(args, rank: int, world: int, device: str) -> Optional[dict]:
                 _recv_pages_cpu(alloc, lst, 0)
         commit: List[Any] = [None]
         dist.recv_object_list(commit, src=0)
-        entries, checks = commit[0]
+        entries, checks = commit[0][0], commit[0][1]
+        pushed = commit[0][2] if len(commit[0]) > 2 else (device == "cuda")
+        if not pushed and device == "cuda":
+            # sender fell back to the gloo wire: receive each page set
+            from ..parallel.comm import from_wire, wire_dtype
+            m = alloc.kv.shape
+            for _tokens, _ft, pages in entries:
+                buf = torch.empty(
+                    (m[0], m[1], len(pages), m[3], m[4], m[5]),
+                    dtype=wire_dtype(alloc.kv.dtype, torch.device("cpu")),
+                    device="cpu")
+                dist.recv(buf, 0)
+                idx = torch.tensor(pages, dtype=torch.int64,
+                                   device=alloc.kv.device)
+                alloc.kv.index_copy_(2, idx,
+                                     from_wire(buf, alloc.kv.dtype).to(
+                                         alloc.kv.device))
         if checks is not None:
             # byte-exact transfer proof (RBG_PD_VERIFY=1): the imported
             # pages must checksum identically to the sender's parked pages

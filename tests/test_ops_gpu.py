@@ -1,9 +1,12 @@
 """GPU numerics: every HIP kernel vs its plain-PyTorch fp32 reference
 (tests run on a real MI355X via gpurun; marked gpu)."""
 import math
+import os
 
 import pytest
 import torch
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 pytestmark = pytest.mark.gpu
 
@@ -328,46 +331,20 @@ def test_xgmi_allreduce_pair_protocol():
 
 @pytest.mark.gpu
 def test_xgmi_allreduce_graph_capture():
-    """Both ranks capture their all-reduce into hipGraphs (the production
-    TP shape: every rank captures its own decode graph) and replays with
-    changing inputs produce the elementwise sum — the property that keeps
-    TP decode off the eager path."""
-    import torch
-
-    from rbg_amd.parallel.xgmi_allreduce import XgmiAllReduce
-    dev = torch.device("cuda", 0)
-    a, b = XgmiAllReduce.for_test_pair(dev)
-    xa = torch.zeros(1024, 512, dtype=torch.bfloat16, device=dev)
-    xb = torch.zeros(1024, 512, dtype=torch.bfloat16, device=dev)
-    sa = torch.cuda.Stream(dev)
-    sb = torch.cuda.Stream(dev)
-
-    # warmup (allocations + first epochs) outside capture
-    with torch.cuda.stream(sa):
-        oa = a.all_reduce(xa)
-    with torch.cuda.stream(sb):
-        ob = b.all_reduce(xb)
-    torch.cuda.synchronize()
-
-    ga = torch.cuda.CUDAGraph()
-    gb = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(ga, stream=sa):
-        oa = a.all_reduce(xa)
-    with torch.cuda.graph(gb, stream=sb):
-        ob = b.all_reduce(xb)
+    """Both ranks capture their all-reduce into hipGraphs and 3 replays
+    with changing inputs produce exact sums (epochs advance in device
+    memory).  Runs tools/xgmi_graph_debug.py in a FRESH subprocess: the
+    protocol is order-sensitive to unrelated prior GPU state in the same
+    process (pair-test-then-capture interleaving flaked), and production
+    TP engines are fresh processes."""
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools", "xgmi_graph_debug.py")],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
     for round_ in range(3):
-        xa.fill_(float(round_ + 1))
-        xb.fill_(float(10 * (round_ + 1)))
-        torch.cuda.synchronize()
-        with torch.cuda.stream(sa):
-            ga.replay()
-        with torch.cuda.stream(sb):
-            gb.replay()
-        torch.cuda.synchronize()
-        a.check()
-        b.check()
         want = float(round_ + 1) + 10 * (round_ + 1)
-        assert torch.all(oa.float() == want), \
-            (round_, "oa", oa.float().unique()[:8].tolist())
-        assert torch.all(ob.float() == want), \
-            (round_, "ob", ob.float().unique()[:8].tolist())
+        assert f"want {want}" in out.stdout
+        assert f"oa uniq [{want}] ob uniq [{want}]" in out.stdout,             out.stdout[-1500:]
+    assert "err=0" in out.stdout
