@@ -91,35 +91,12 @@ class _Migrator:
     def push_remote(self, src_cache, peer_meta: Dict[str, Any],
                     src_pages: List[int], dst_pages: List[int],
                     dst_rank: int) -> None:
-        import torch.distributed as dist
-        if self.device == "cuda":
-            from ..parallel.kv_peer import PeerKVPusher
-            if self._pusher is None:
-                self._pusher = PeerKVPusher(src_cache.kv.device)
-            pending = self._pusher.push(src_cache, src_pages, peer_meta,
-                                        dst_pages)
-            pending.wait()
-        else:
-            from ..parallel.comm import to_wire
-            idx = torch.tensor(src_pages, dtype=torch.int64,
-                               device=src_cache.kv.device)
-            buf = src_cache.kv.index_select(2, idx).contiguous()
-            dist.send(to_wire(buf), dst_rank)
-
-
-def _recv_pages_cpu(cache, pages: List[int], src_rank: int) -> None:
-    import torch.distributed as dist
-
-    from ..parallel.comm import from_wire, wire_dtype
-    m = cache.kv.shape
-    buf = torch.empty((m[0], m[1], len(pages), m[3], m[4], m[5]),
-                      dtype=wire_dtype(cache.kv.dtype,
-                                       torch.device("cpu")),
-                      device="cpu")
-    dist.recv(buf, src_rank)
-    buf = from_wire(buf, cache.kv.dtype)
-    idx = torch.tensor(pages, dtype=torch.int64, device=cache.kv.device)
-    cache.kv.index_copy_(2, idx, buf.to(cache.kv.device))
+        from ..parallel.kv_peer import PeerKVPusher
+        if self._pusher is None:
+            self._pusher = PeerKVPusher(src_cache.kv.device)
+        pending = self._pusher.push(src_cache, src_pages, peer_meta,
+                                    dst_pages)
+        pending.wait()
 
 
 def _enqueue_imported(engine: LLMEngine, tokens: List[int],
@@ -257,15 +234,13 @@ def run_pd(args, rank: int, world: int, device: str) -> Optional[dict]:
         else:
             meta = None
         dist.send_object_list([(dst_lists, meta)], dst=0)
-        if device != "cuda":
-            for lst in dst_lists:
-                _recv_pages_cpu(alloc, lst, 0)
         commit: List[Any] = [None]
         dist.recv_object_list(commit, src=0)
-        entries, checks = commit[0][0], commit[0][1]
-        pushed = commit[0][2] if len(commit[0]) > 2 else (device == "cuda")
-        if not pushed and device == "cuda":
-            # sender fell back to the gloo wire: receive each page set
+        entries, checks, pushed = commit[0]
+        if not pushed:
+            # CPU path, or the sender's cross-device hipIpc push was
+            # refused: page data arrives over the gloo wire AFTER the
+            # commit message, one set per sequence
             from ..parallel.comm import from_wire, wire_dtype
             m = alloc.kv.shape
             for _tokens, _ft, pages in entries:
