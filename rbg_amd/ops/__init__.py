@@ -165,7 +165,8 @@ def decode_attention(q, key_cache, value_cache, block_tables, context_lens,
 def prefill_attention(q, k, v, cu_seqlens, scale: float,
                       cu_seqlens_k=None) -> torch.Tensor:
     if _on_gpu(q):
-        qtile = 128 if (PREFILL_SWZ & 4) else 64   # bit 2 = 8-wave blocks
+        # bit 2 = 8-wave 16x16 blocks; bit 3 = 4-wave 32x32 blocks
+        qtile = 128 if (PREFILL_SWZ & 12) else 64
         block_info, seq_lens = reference.prefill_block_info(
             cu_seqlens.cpu(), qtile=qtile,
             cu_seqlens_k=None if cu_seqlens_k is None else cu_seqlens_k.cpu())

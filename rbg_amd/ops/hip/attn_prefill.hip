@@ -327,6 +327,236 @@ __global__ __launch_bounds__(NW * 64) void prefill_attn_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 32x32x16 MFMA prefill (swz bit 3, ROUND2 §2a): 4 waves x 32 q-rows =
+// the same 128-row block as the 8-wave 16x16 kernel, but each MFMA does
+// 2x the flops, halving the MFMA instruction count and the per-flop
+// fragment-read traffic.  Operand layouts verified empirically
+// (tools/mfma_probe32.py: contiguous-8 k per half-wave for A and B;
+// D row = (i%4) + 4*(l>>5) + 8*(i/4), col = l%32).
+//
+// The PV B-fragment (col = l%32, k = 8*(l>>5)+j) needs its own tr-read
+// image: 4x16 row-major blocks indexed q = 2*(k-half) + (col>=16) — which
+// equals the hardware's (l>>4) group index — so ds_read_b64_tr_b16 with
+// per-lane addr base + (l>>4)*64 + (l&15)*4 lands each lane's column
+// exactly (guide LDS §tr-read).  Staging writes stay 16-byte vectors.
+DEV_INLINE int v_img32_off(int key, int dim) {
+  const int ks = key >> 4, kk = key & 15;
+  const int h = kk >> 3, j7 = kk & 7;
+  return ks * 2048 + (dim >> 5) * 512 + ((j7 >= 4) ? 256 : 0) +
+         (2 * h + ((dim >> 4) & 1)) * 64 + (j7 & 3) * 16 + (dim & 15);
+}
+
+typedef __attribute__((ext_vector_type(16))) float mfma_f16;
+
+template <int SWZ, int DB>
+__global__ __launch_bounds__(256, 2) void prefill_attn32_kernel(
+    __hip_bfloat16* __restrict__ out,        // [T, QH, D]
+    const __hip_bfloat16* __restrict__ q,    // [T, QH, D]
+    const __hip_bfloat16* __restrict__ k,    // [T, KVH, D]
+    const __hip_bfloat16* __restrict__ v,    // [T, KVH, D]
+    const int* __restrict__ block_info,      // [nblocks, 4]
+    const int* __restrict__ seq_lens,        // [nblocks]
+    const float scale, const int num_q_heads, const int num_kv_heads,
+    const int q_stride, const int kv_stride, const int nblocks) {
+  const int qpg_n = num_q_heads / num_kv_heads;
+  int blk, qh;
+  if (SWZ && qpg_n > 1) {
+    const int lin = blockIdx.x;
+    const int G = nblocks * num_kv_heads;
+    const int chunk = 8 * qpg_n;
+    const int full = (G / 8) * chunk;
+    int g, qpg_idx;
+    if (lin < full) {
+      const int c = lin / chunk, r = lin % chunk;
+      qpg_idx = r >> 3;
+      g = c * 8 + (r & 7);
+    } else {
+      const int idx = lin - full;
+      g = (G / 8) * 8 + idx / qpg_n;
+      qpg_idx = idx % qpg_n;
+    }
+    blk = g / num_kv_heads;
+    qh = (g % num_kv_heads) * qpg_n + qpg_idx;
+  } else {
+    blk = blockIdx.x / num_q_heads;
+    qh = blockIdx.x % num_q_heads;
+  }
+  const int kvh = qh / qpg_n;
+  const int q_start = block_info[blk * 4];
+  const int qblock = block_info[blk * 4 + 1];
+  const int kv_start = block_info[blk * 4 + 2];
+  const int q_offset = block_info[blk * 4 + 3];
+  const int seq_len = seq_lens[blk];
+
+  constexpr int QTILE = 128;        // 4 waves x 32 rows
+  constexpr int NTHREADS = 256;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int col = lane & 31;        // fragment column (key / out dim)
+  const int half = lane >> 5;       // k-half of the fragment
+
+  __shared__ __hip_bfloat16 k_lds[KTILE][HEAD_DIM + PAD];
+  __shared__ __hip_bfloat16 v_img[KTILE * HEAD_DIM];   // v_img32 layout
+  __shared__ __hip_bfloat16 p_lds[4][32][KTILE + 4];
+
+  // ---- Q fragments: wave's 32 rows, 8 k-steps of 16 dims ---------------
+  const int num_q_rows = seq_len - q_offset;
+  const int q_row_in_chunk = qblock * QTILE + wave * 32 + col;
+  const int q_row = q_start + min(q_row_in_chunk, num_q_rows - 1);
+  mfma_bf8 q_frag[8];
+#pragma unroll
+  for (int ks = 0; ks < 8; ++ks)
+    q_frag[ks] = load_bf8(q + (size_t)q_row * q_stride +
+                          (size_t)qh * HEAD_DIM + ks * 16 + half * 8);
+
+  // softmax state: the lane's 16 rows r(i) = (i%4) + 4*half + 8*(i/4)
+  float m_run[16], l_run[16];
+  mfma_f16 acc_o[4];                // O[32 x 128]: 4 col-tiles of 32 dims
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { m_run[i] = NEG_INF; l_run[i] = 0.f; }
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int i = 0; i < 16; ++i) acc_o[dt][i] = 0.f;
+
+  const int block_q_max = min(q_offset + qblock * QTILE + QTILE - 1,
+                              seq_len - 1);
+  const int wave_q_max = min(q_offset + qblock * QTILE + wave * 32 + 31,
+                             seq_len - 1);
+  const int nktiles = block_q_max / KTILE + 1;
+
+  constexpr int PIECES = KTILE * (HEAD_DIM / 8) / NTHREADS;   // = 4
+  uint4 kr0 = {}, kr1 = {}, kr2 = {}, kr3 = {};
+  uint4 vr0 = {}, vr1 = {}, vr2 = {}, vr3 = {};
+  auto issue_loads32 = [&](int kt2) {
+    PF_LD(0, kt2, kr0, vr0);
+    PF_LD(1, kt2, kr1, vr1);
+    PF_LD(2, kt2, kr2, vr2);
+    PF_LD(3, kt2, kr3, vr3);
+  };
+#define PF_ST32(s, kr, vr)                                                  \
+  do {                                                                      \
+    const int key = (tid + (s) * NTHREADS) >> 4;                            \
+    const int chunk = ((tid + (s) * NTHREADS) & 15) * 8;                    \
+    *reinterpret_cast<uint4*>(&k_lds[key][chunk]) = kr;                     \
+    *reinterpret_cast<uint4*>(&v_img[v_img32_off(key, chunk)]) = vr;        \
+  } while (0)
+  auto write_tile32 = [&]() {
+    PF_ST32(0, kr0, vr0);
+    PF_ST32(1, kr1, vr1);
+    PF_ST32(2, kr2, vr2);
+    PF_ST32(3, kr3, vr3);
+  };
+
+  if (DB) {
+    issue_loads32(0);
+    write_tile32();
+  }
+
+  for (int kt = 0; kt < nktiles; ++kt) {
+    const int k_base = kt * KTILE;
+    if (DB) {
+      __syncthreads();
+      if (kt + 1 < nktiles)
+        issue_loads32(kt + 1);
+    } else {
+      __syncthreads();
+      issue_loads32(kt);
+      write_tile32();
+      __syncthreads();
+    }
+
+    const bool compute_this = k_base <= wave_q_max;
+    if (compute_this) {
+
+    // ---- S = Q K^T : 2 col-tiles (32 keys) x 8 k-steps ------------------
+    mfma_f16 s[2];
+#pragma unroll
+    for (int ct = 0; ct < 2; ++ct) {
+#pragma unroll
+      for (int i = 0; i < 16; ++i) s[ct][i] = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < 8; ++ks) {
+        mfma_bf8 kf = lds_bf8(
+            &k_lds[ct * 32 + col][ks * 16 + half * 8]);
+        s[ct] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            q_frag[ks], kf, s[ct], 0, 0, 0);
+      }
+    }
+
+    // ---- online softmax: 16 rows per lane, 32-lane row groups -----------
+#pragma unroll
+    for (int i = 0; i < 16; ++i) {
+      const int r = (i % 4) + 4 * half + 8 * (i / 4);
+      const int q_pos = q_offset + qblock * QTILE + wave * 32 + r;
+      float p0 = s[0][i] * scale, p1 = s[1][i] * scale;
+      const int k0 = k_base + col, k1 = k_base + 32 + col;
+      if (k0 > q_pos || k0 >= seq_len || q_pos >= seq_len) p0 = NEG_INF;
+      if (k1 > q_pos || k1 >= seq_len || q_pos >= seq_len) p1 = NEG_INF;
+      float row_max = half_wave_max(fmaxf(p0, p1));
+      const float m_new = fmaxf(m_run[i], row_max);
+      const float alpha = __expf(m_run[i] - m_new);
+      p0 = (p0 <= NEG_INF) ? 0.f : __expf(p0 - m_new);
+      p1 = (p1 <= NEG_INF) ? 0.f : __expf(p1 - m_new);
+      l_run[i] = l_run[i] * alpha + p0 + p1;     // summed in epilogue
+      m_run[i] = m_new;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) acc_o[dt][i] *= alpha;
+      p_lds[wave][r][col] = f2bf(p0);
+      p_lds[wave][r][32 + col] = f2bf(p1);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- O += P V : 4 col-tiles x 4 key-steps ---------------------------
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      mfma_bf8 pa = lds_bf8(&p_lds[wave][col][ks * 16 + half * 8]);
+      const unsigned vbase = (unsigned)(unsigned long long)(
+          &v_img[ks * 2048 + (lane >> 4) * 64 + (lane & 15) * 4]);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        unsigned long long vlo, vhi;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %2 offset:%3\n\t"
+            "ds_read_b64_tr_b16 %1, %2 offset:%4\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(vlo), "=&v"(vhi)
+            : "v"(vbase), "i"(dt * 1024), "i"(dt * 1024 + 512)
+            : "memory");
+        union { struct { unsigned long long lo, hi; } u; mfma_bf8 vf; } vv;
+        vv.u.lo = vlo;
+        vv.u.hi = vhi;
+        acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pa, vv.vf, acc_o[dt], 0, 0, 0);
+      }
+    }
+    }  // compute_this
+
+    if (DB && kt + 1 < nktiles) {
+      __syncthreads();
+      write_tile32();
+    }
+  }
+
+  // ---- epilogue ---------------------------------------------------------
+#pragma unroll
+  for (int i = 0; i < 16; ++i) l_run[i] = half_wave_sum(l_run[i]);
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const int r = (i % 4) + 4 * half + 8 * (i / 4);
+    const int q_row_out = qblock * QTILE + wave * 32 + r;
+    if (q_row_out >= num_q_rows) continue;
+    const float inv_l = l_run[i] > 0.f ? 1.f / l_run[i] : 0.f;
+    __hip_bfloat16* orow =
+        out + ((size_t)(q_start + q_row_out) * num_q_heads + qh) * HEAD_DIM;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      orow[dt * 32 + col] = f2bf(acc_o[dt][i] * inv_l);
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -339,7 +569,26 @@ void launch_prefill_attention(void* out, const void* q, const void* k,
                               hipStream_t stream) {
   dim3 grid(nblocks * num_q_heads);
   // swz bit 0: XCD-affine GQA swizzle; bit 1: T14 double-buffer staging;
-  // bit 2: 8-wave blocks (host must build block_info with qtile=128)
+  // bit 2: 8-wave blocks (host must build block_info with qtile=128);
+  // bit 3: 32x32x16 MFMA kernel (4 waves, qtile=128)
+  if (swz & 8) {
+#define PF32_LAUNCH(S, D)                                                   \
+    hipLaunchKernelGGL((prefill_attn32_kernel<S, D>), grid, dim3(256), 0,   \
+                       stream,                                              \
+                       (__hip_bfloat16*)out, (const __hip_bfloat16*)q,      \
+                       (const __hip_bfloat16*)k, (const __hip_bfloat16*)v,  \
+                       (const int*)block_info, (const int*)seq_lens, scale, \
+                       num_q_heads, num_kv_heads, q_stride, kv_stride,      \
+                       nblocks)
+    switch (swz & 3) {
+      case 0: PF32_LAUNCH(0, 0); break;
+      case 1: PF32_LAUNCH(1, 0); break;
+      case 2: PF32_LAUNCH(0, 1); break;
+      case 3: PF32_LAUNCH(1, 1); break;
+    }
+#undef PF32_LAUNCH
+    return;
+  }
 #define PF_LAUNCH(S, D, NW)                                                 \
   hipLaunchKernelGGL((prefill_attn_kernel<S, D, NW>), grid, dim3(NW * 64),  \
                      0, stream,                                             \

@@ -57,6 +57,13 @@ DEV_INLINE float half_wave_sum(float x) {
   return x;
 }
 
+DEV_INLINE float half_wave_max(float x) {
+#pragma unroll
+  for (int off = 16; off >= 1; off >>= 1)
+    x = fmaxf(x, __shfl_xor(x, off, 64));
+  return x;
+}
+
 // Reduction across 16-lane groups (same l>>4).
 DEV_INLINE float group16_sum(float x) {
 #pragma unroll

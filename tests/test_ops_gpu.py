@@ -348,3 +348,46 @@ def test_xgmi_allreduce_graph_capture():
         assert f"want {want}" in out.stdout
         assert f"oa uniq [{want}] ob uniq [{want}]" in out.stdout,             out.stdout[-1500:]
     assert "err=0" in out.stdout
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("swz", [8, 9, 10])
+@pytest.mark.parametrize("lens", [[128], [64, 200, 1], [1024], [33, 129]])
+def test_prefill_attention_32x32(dev, lens, swz, monkeypatch):
+    """32x32x16 MFMA prefill (swz bit 3; 9 = +XCD swizzle, 10 = +DB) vs
+    fp32 reference across ragged/causal shapes."""
+    monkeypatch.setattr(ops, "PREFILL_SWZ", swz)
+    QH, KVH, D = 32, 8, 128
+    T = sum(lens)
+    torch.manual_seed(5)
+    q = torch.randn(T, QH, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=dev)
+    scale = 1.0 / math.sqrt(D)
+    got = ops.prefill_attention(q, k, v, cu, scale)
+    want = ref.prefill_attention(q, k, v, cu, scale)
+    err = (got.float() - want.float()).abs().max().item()
+    assert err < 3e-2, f"swz={swz} lens={lens} err={err}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("swz", [8])
+def test_prefill_attention_32x32_with_prefix(dev, swz, monkeypatch):
+    """Chunked-prefill path (q_offset > 0) on the 32x32 kernel."""
+    monkeypatch.setattr(ops, "PREFILL_SWZ", swz)
+    QH, KVH, D = 32, 8, 128
+    torch.manual_seed(6)
+    past, new = 192, 160
+    kv_total = past + new
+    q = torch.randn(new, QH, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(kv_total, KVH, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(kv_total, KVH, D, dtype=torch.bfloat16, device=dev)
+    cu_q = torch.tensor([0, new], dtype=torch.int32, device=dev)
+    cu_k = torch.tensor([0, kv_total], dtype=torch.int32, device=dev)
+    scale = 1.0 / math.sqrt(D)
+    got = ops.prefill_attention(q, k, v, cu_q, scale, cu_k)
+    want = ref.prefill_attention(q, k, v, cu_q, scale, cu_k)
+    err = (got.float() - want.float()).abs().max().item()
+    assert err < 3e-2, err

@@ -33,7 +33,7 @@ for qt in (64, 128):
     _binfo[qt] = (bi.to(dev), sl.to(dev))
 
 def run(swz):
-    bi, sl = _binfo[128 if swz & 4 else 64]
+    bi, sl = _binfo[128 if swz & 12 else 64]
     torch.cuda.synchronize()
     t0 = time.monotonic()
     for _ in range(a.iters):
@@ -43,7 +43,9 @@ def run(swz):
 
 flops = a.seqs * 2 * 2 * (a.slen * a.slen / 2) * D * a.qh
 import statistics
-variants = (0, 4, 6)
+import os
+variants = tuple(int(x) for x in os.environ.get(
+    "RBG_PF_VARIANTS", "6,8,9,10").split(","))
 res = {v: [] for v in variants}
 for swz in variants:
     run(swz)
