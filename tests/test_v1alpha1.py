@@ -305,3 +305,61 @@ def test_legacy_instance_round_trip_and_load():
     assert back["spec"]["readyPolicy"] == "AllComponentsReady"
     assert back["spec"]["restartPolicy"] == \
         "RecreateRoleInstanceOnPodRestart"
+
+
+def test_instanceset_round_trip_property():
+    """Property: v1->v2->v1 preserves every semantically-carried
+    InstanceSet field across randomized docs (hypothesis)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from rbg_amd.api.v1alpha1 import instanceset_from_v2, instanceset_to_v2
+
+    upd_types = st.sampled_from(
+        ["InPlaceIfPossible", "InPlaceOnly", "RecreatePod"])
+
+    @settings(max_examples=60, deadline=None)
+    @given(replicas=st.integers(1, 16),
+           partition=st.integers(0, 8),
+           mu=st.integers(0, 4),
+           surge=st.integers(0, 4),
+           paused=st.booleans(),
+           grace=st.integers(0, 30),
+           utype=upd_types,
+           to_delete=st.lists(st.sampled_from(["a-0", "a-1", "b-2"]),
+                              max_size=2, unique=True),
+           min_ready=st.integers(0, 10))
+    def check(replicas, partition, mu, surge, paused, grace, utype,
+              to_delete, min_ready):
+        doc = {"apiVersion": "workloads.x-k8s.io/v1alpha1",
+               "kind": "InstanceSet", "metadata": {"name": "p"},
+               "spec": {
+                   "replicas": replicas,
+                   "instanceTemplate": {
+                       "components": [{"name": "engine", "size": 1}],
+                       "restartPolicy":
+                           "RecreateRoleInstanceOnPodRestart"},
+                   "scaleStrategy": (
+                       {"instanceToDelete": to_delete} if to_delete else {}),
+                   "updateStrategy": {
+                       "type": utype, "partition": partition,
+                       "maxUnavailable": mu, "maxSurge": surge,
+                       "paused": paused,
+                       "inPlaceUpdateStrategy": {
+                           "gracePeriodSeconds": grace}},
+                   "minReadySeconds": min_ready}}
+        back = instanceset_from_v2(instanceset_to_v2(copy.deepcopy(doc)))
+        assert back["spec"]["replicas"] == replicas
+        u = back["spec"]["updateStrategy"]
+        assert (u["type"], u["partition"], u["maxUnavailable"],
+                u["maxSurge"], u["paused"]) == \
+            (utype, partition, mu, surge, paused)
+        if grace:
+            assert u["inPlaceUpdateStrategy"]["gracePeriodSeconds"] == grace
+        if to_delete:
+            assert back["spec"]["scaleStrategy"]["instanceToDelete"] == \
+                to_delete
+        assert back["spec"]["minReadySeconds"] == min_ready
+        assert back["spec"]["instanceTemplate"]["restartPolicy"] == \
+            "RecreateRoleInstanceOnPodRestart"
+    check()
