@@ -122,7 +122,8 @@ class RoleBasedGroupController:
                                                    "peer")
             all_gpu = all(
                 e.resources is not None and not e.resources.cpu_only and
-                e.resources.gpus + e.resources.hbm_bytes > 0
+                e.resources.gpus + e.resources.hbm_bytes > 0 and
+                str(e.args.get("device", "cuda")) != "cpu"
                 for _, e in engines)
             if is_pd and not is_lw and kv_mode == "peer" and all_gpu:
                 continue

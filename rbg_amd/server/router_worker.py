@@ -47,6 +47,8 @@ class _InstanceWatcher:
         self._results: Dict[int, Dict[str, Any]] = {}
         self._thread: threading.Thread = None
 
+    FAIL_WINDOW_S = 10.0    # continuous RPC failure -> declare seqs lost
+
     def wait_for(self, seq_id: int, timeout: float) -> Dict[str, Any]:
         ev = threading.Event()
         with self._lock:
@@ -59,21 +61,48 @@ class _InstanceWatcher:
                 self._waiting.pop(seq_id, None)
             raise TimeoutError(f"seq {seq_id} did not finish in {timeout}s")
         with self._lock:
-            return self._results.pop(seq_id)
+            res = self._results.pop(seq_id)
+        if res.get("lost"):
+            # instance died / was recreated: its sequences are gone — the
+            # router re-dispatches the whole request (InstanceLost)
+            raise InstanceLost(res.get("reason", "instance lost"))
+        return res
+
+    def _fail_all(self, reason: str) -> None:
+        with self._lock:
+            for sid, ev in list(self._waiting.items()):
+                self._results[sid] = {"lost": True, "reason": reason}
+                ev.set()
+            self._waiting.clear()
 
     def _loop(self):
+        fail_since = None
         while True:
             with self._lock:
                 ids = list(self._waiting.keys())
             if not ids:
                 time.sleep(self.tick)
-                with self._lock:
-                    if not self._waiting:
-                        continue
                 continue
             try:
                 results = self.client.call("poll_many", seq_ids=ids)
-            except Exception:  # noqa: BLE001 — instance restarting; retry
+                fail_since = None
+            except Exception as e:  # noqa: BLE001
+                msg = repr(e)
+                if "unknown seq" in msg:
+                    # the engine restarted (gang recreate): pre-restart
+                    # sequences are gone — fail them NOW so the router
+                    # retries elsewhere instead of waiting out the timeout
+                    self._fail_all("engine restarted (unknown seq)")
+                    continue
+                # connection-level failure: the instance is down or
+                # recreating; give it FAIL_WINDOW_S before declaring loss
+                now = time.monotonic()
+                if fail_since is None:
+                    fail_since = now
+                elif now - fail_since > self.FAIL_WINDOW_S:
+                    self._fail_all(f"instance unreachable > "
+                                   f"{self.FAIL_WINDOW_S}s: {msg}")
+                    fail_since = None
                 time.sleep(0.2)
                 continue
             with self._lock:
@@ -82,6 +111,11 @@ class _InstanceWatcher:
                         self._results[sid] = res
                         self._waiting.pop(sid).set()
             time.sleep(self.tick)
+
+
+class InstanceLost(RuntimeError):
+    """A dispatched instance died before finishing the sequence; the
+    request is safe to re-dispatch (full re-prefill) elsewhere."""
 
 
 class Router:
@@ -111,7 +145,9 @@ class Router:
         return out
 
     def _client(self, inst: Dict[str, Any]) -> RpcClient:
-        key = inst["name"]
+        # keyed by (name, ports): a recreated instance may publish a new
+        # RPC port, and a stale cached socket must not shadow it
+        key = (inst["name"], tuple(inst.get("ports", ())))
         with self._lock:
             c = self._clients.get(key)
             if c is None:
@@ -121,7 +157,7 @@ class Router:
             return c
 
     def _watcher(self, inst: Dict[str, Any]) -> _InstanceWatcher:
-        key = inst["name"]
+        key = (inst["name"], tuple(inst.get("ports", ())))
         with self._lock:
             w = self._watchers.get(key)
             if w is None:
@@ -149,9 +185,29 @@ class Router:
 
     def generate(self, tokens: List[int], max_new_tokens: int,
                  temperature: float = 0.0) -> Dict[str, Any]:
+        """Linked-failover continuity: if the dispatched instance dies
+        mid-request (InstanceLost from the watcher, or a submit-time
+        connection error), the whole request re-dispatches — round-robin
+        naturally lands on a surviving replica while the gang recreates
+        the dead one."""
         t0 = time.monotonic()
-        if self.mode == "pd":
-            return self._generate_pd(tokens, max_new_tokens, temperature, t0)
+        last: Exception = None
+        for attempt in range(3):
+            try:
+                if self.mode == "pd":
+                    return self._generate_pd(tokens, max_new_tokens,
+                                             temperature, t0)
+                return self._generate_colocated(tokens, max_new_tokens,
+                                                temperature, t0)
+            except (InstanceLost, ConnectionError, OSError) as e:
+                last = e
+                log.warning("request re-dispatch (attempt %d): %r",
+                            attempt + 1, e)
+                time.sleep(0.1)
+        raise last
+
+    def _generate_colocated(self, tokens, max_new_tokens, temperature,
+                            t0) -> Dict[str, Any]:
         inst = self._pick_role(self.worker_roles)
         sid = self._client(inst).call("submit", tokens=tokens,
                                       max_new_tokens=max_new_tokens,

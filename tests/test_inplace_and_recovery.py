@@ -1,5 +1,6 @@
 """In-place live update + group-recovery metric (reference analog:
 pkg/inplace tests, failure-handling doc semantics, restart_policy envtest)."""
+import os
 import time
 
 import pytest
@@ -149,3 +150,26 @@ def test_readiness_gates_hold_ready(mgr):
         return cur
     mgr.store.apply(C.KIND_ROLE_INSTANCE, name, satisfy, subresource="status")
     assert mgr.wait_for(inst_ready, timeout=30)
+
+
+def test_decode_pool_failover_continuity(tmp_path):
+    """BASELINE config 5 semantics: with a decode POOL (replicas=2), a
+    SIGKILLed decode replica mid-stream loses no requests — the router
+    fails the dead instance's sequences fast (InstanceLost) and
+    re-dispatches them to the survivor while the gang recreates."""
+    import json
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "tools/bench_serving.py", "--mode", "pd",
+         "--model", "tiny", "--device", "cpu", "--prompts", "16",
+         "--rate", "5", "--in-len", "64", "--out-len", "12",
+         "--decode-replicas", "2", "--kill-decode-after", "1.5",
+         "--timeout", "120"],
+        capture_output=True, text=True, timeout=420,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout)
+    assert d["failover"]["killed"].startswith("bench-decode")
+    assert d["errors"] == 0, d
+    assert d["completed"] == 16, d

@@ -47,9 +47,13 @@ def build_rbg(args, engine_args):
         # per-role KV pools: the prefill engine never holds output tokens
         # (pages migrate at first token), the decode pool carries the full
         # in+out trace — sized separately so both engines + weights +
-        # decode hipGraph pools fit 288 GB on a 1-GPU colocated run
+        # decode hipGraph pools fit 288 GB on a 1-GPU colocated run.
+        # A decode POOL (replicas > 1, BASELINE config 5) splits the trace
+        # across replicas (x1.6 slack for round-robin imbalance+failover).
         prefill_pool = args.prompts * (args.in_len + 64) + 8192
-        decode_pool = args.prompts * (args.in_len + args.out_len + 64) + 8192
+        per_decode = max(1, args.prompts // args.decode_replicas)
+        decode_pool = int(per_decode * 1.6) * \
+            (args.in_len + args.out_len + 64) + 8192
         roles = [
             role("router", "router",
                  {"dispatch": "pd", "prefill_roles": ["prefill"],
@@ -62,6 +66,7 @@ def build_rbg(args, engine_args):
                  dict(shared, mode="decode", kv_pool_tokens=decode_pool),
                  args.gpus_per_engine, deps=("router",)),
         ]
+        roles[2].replicas = args.decode_replicas
     else:
         roles = [
             role("router", "router",
@@ -84,6 +89,11 @@ def main() -> int:
     ap.add_argument("--in-len", type=int, default=2048)
     ap.add_argument("--out-len", type=int, default=128)
     ap.add_argument("--gpus-per-engine", type=int, default=1)
+    ap.add_argument("--decode-replicas", type=int, default=1,
+                    help="decode pool size (BASELINE config 5)")
+    ap.add_argument("--kill-decode-after", type=float, default=0.0,
+                    help="SIGKILL one decode engine N seconds into the "
+                         "request stream (linked-failover continuity)")
     ap.add_argument("--hbm-gb", type=int, default=100)
     ap.add_argument("--timeout", type=float, default=900.0)
     args = ap.parse_args()
@@ -158,8 +168,38 @@ def main() -> int:
                 with lock:
                     results.append({"error": repr(e)})
 
+        kill_report = {}
+
+        def kill_one_decode():
+            import os as _os
+            import signal as _signal
+            time.sleep(args.kill_decode_after)
+            for inst in mgr.store.list(KC.KIND_ROLE_INSTANCE, selector={
+                    KC.LABEL_GROUP_NAME: "bench",
+                    KC.LABEL_ROLE_NAME: "decode"}):
+                for w in inst.status.workers:
+                    if w.pid:
+                        kill_report["killed"] = w.name
+                        kill_report["t_kill_s"] = round(
+                            time.monotonic() - bench_start, 2)
+                        _os.kill(w.pid, _signal.SIGKILL)
+                        t0 = time.monotonic()
+                        while ready() and time.monotonic() - t0 < 60:
+                            time.sleep(0.05)
+                        kill_report["detect_s"] = round(
+                            time.monotonic() - t0, 2)
+                        while not ready() and \
+                                time.monotonic() - t0 < 180:
+                            time.sleep(0.1)
+                        kill_report["recovery_s"] = (
+                            round(time.monotonic() - t0, 2)
+                            if ready() else None)
+                        return
+
         threads = []
         bench_start = time.monotonic()
+        if args.kill_decode_after > 0:
+            threading.Thread(target=kill_one_decode, daemon=True).start()
         for i, tokens in enumerate(prompts):
             target = bench_start + i / args.rate
             now = time.monotonic()
@@ -197,6 +237,7 @@ def main() -> int:
                    "prompts": args.prompts, "rate": args.rate,
                    "in_len": args.in_len, "out_len": args.out_len},
         "completed": len(ok), "errors": len(errs),
+        **({"failover": kill_report} if kill_report else {}),
         "total_token_throughput_tok_s": round(total_tokens / wall, 1),
         "output_tok_s": round(out_tokens / wall, 1),
         "mean_ttft_ms": round(1000 * statistics.mean(ttfts), 1) if ok else 0,
