@@ -225,6 +225,20 @@ def main() -> int:
                     except Exception as e:  # noqa: BLE001
                         engine_stats[w.name] = {"error": repr(e)}
     finally:
+        # on failures, surface the engine logs (they die with the box)
+        try:
+            errs_now = sum(1 for r in results if "error" in r)
+            if errs_now:
+                import glob as _glob
+                for lp in sorted(_glob.glob(
+                        os.path.join(run_root, "**", "*.log"),
+                        recursive=True)):
+                    with open(lp, "rb") as f:
+                        f.seek(max(0, os.path.getsize(lp) - 4000))
+                        tail = f.read().decode(errors="replace")
+                    print(f"==== {lp} ====\n{tail}", file=sys.stderr)
+        except Exception:  # noqa: BLE001
+            pass
         mgr.stop()
 
     ok = [r for r in results if "error" not in r]
