@@ -191,20 +191,34 @@ class Router:
         naturally lands on a surviving replica while the gang recreates
         the dead one."""
         t0 = time.monotonic()
+        window = float(self.ctx.args.get("failover_window_s", 45.0))
+        deadline = t0 + window
         last: Exception = None
-        for attempt in range(3):
+        attempt = 0
+        while True:
             try:
                 if self.mode == "pd":
                     return self._generate_pd(tokens, max_new_tokens,
                                              temperature, t0)
                 return self._generate_colocated(tokens, max_new_tokens,
                                                 temperature, t0)
-            except (InstanceLost, ConnectionError, OSError) as e:
+            except (InstanceLost, ConnectionError, OSError,
+                    RuntimeError) as e:
+                # retryable: instance death, connection refusal during a
+                # gang recreate, or a transiently-empty ready set.  The
+                # gang recreates in seconds; requests WAIT (bounded by
+                # failover_window_s) instead of failing the client.
+                retryable = isinstance(
+                    e, (InstanceLost, ConnectionError, OSError)) or \
+                    "no ready instances" in str(e)
+                if not retryable or time.monotonic() >= deadline:
+                    raise
                 last = e
-                log.warning("request re-dispatch (attempt %d): %r",
-                            attempt + 1, e)
-                time.sleep(0.1)
-        raise last
+                attempt += 1
+                if attempt <= 3 or attempt % 20 == 0:
+                    log.warning("request re-dispatch (attempt %d): %r",
+                                attempt, e)
+                time.sleep(0.25)
 
     def _generate_colocated(self, tokens, max_new_tokens, temperature,
                             t0) -> Dict[str, Any]:
