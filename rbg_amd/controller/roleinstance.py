@@ -144,8 +144,15 @@ class RoleInstanceController:
                 self._set_restarting(inst, True,
                                      f"backoff until {tracker.next_allowed_at():.0f}")
                 return max(0.05, tracker.next_allowed_at() - now)
-            log.warning("instance %s: workers %s failed; gang recreate",
-                        inst.metadata.name, failed_fatal)
+            diag = {}
+            for w in failed_fatal:
+                h = rt.handles.get(w)
+                if h is not None:
+                    diag[w] = {"exit": h.exit_code(),
+                               "hb_age_s": round(h.heartbeat_age(), 1),
+                               "status": h.read_status().get("phase")}
+            log.warning("instance %s: workers %s failed (%s); gang recreate",
+                        inst.metadata.name, failed_fatal, diag)
             self._record_bindings(inst, rt)
             self._stop_all(inst, rt)
             tracker.record_restart(now)

@@ -48,6 +48,13 @@ class PagedKVCache:
                 self.kv = ops._require_hip().ipc_alloc_bf16(list(shape))
                 self.kv.zero_()
                 self.ipc_exportable = True
+                # incarnation token: a RECREATED engine's pool can receive
+                # identical hipIpc handle bytes (deterministic allocator),
+                # so peers must key their mappings by (uid, handle), not
+                # handle alone — a stale mapping into a dead process's
+                # freed pool is an uncatchable GPU fault (GPU run r2/13)
+                import uuid as _uuid
+                self.ipc_uid = _uuid.uuid4().hex
             else:
                 self.kv = torch.zeros(shape, dtype=torch.bfloat16,
                                       device=device)

@@ -48,6 +48,7 @@ def export_meta(cache: PagedKVCache) -> Dict[str, Any]:
     handle = ops._require_hip().kv_ipc_export(cache.kv)
     return {
         "handle": base64.b64encode(handle).decode("ascii"),
+        "uid": getattr(cache, "ipc_uid", ""),
         "num_pages": int(cache.kv.shape[2]),
         "shape": list(cache.kv.shape),
     }
@@ -84,17 +85,21 @@ class PeerKVPusher:
     def __init__(self, device: torch.device):
         self.device = device
         self._stream = torch.cuda.Stream(device)
-        self._open: Dict[str, int] = {}     # handle b64 -> mapped base ptr
+        self._open: Dict[tuple, int] = {}   # (uid, handle) -> mapped ptr
         self._lock = threading.Lock()
 
     def _map(self, meta: Dict[str, Any]) -> int:
         if "local_ptr" in meta:          # same-process pool: no IPC needed
             return int(meta["local_ptr"])
-        key = meta["handle"]
+        # (uid, handle): handle BYTES can repeat across an exporting
+        # engine's restarts (deterministic allocator) — the uid is the
+        # pool incarnation, and a cache hit on handle alone would reuse a
+        # mapping into freed memory (fatal GPU fault)
+        key = (meta.get("uid", ""), meta["handle"])
         with self._lock:
             ptr = self._open.get(key)
             if ptr is None:
-                raw = base64.b64decode(key)
+                raw = base64.b64decode(meta["handle"])
                 ptr = ops._require_hip().kv_ipc_open(raw)
                 self._open[key] = ptr
             return ptr
