@@ -290,8 +290,9 @@ py::bytes kv_ipc_export(torch::Tensor pool) {
   return py::bytes(reinterpret_cast<const char*>(&handle), sizeof(handle));
 }
 
-int64_t kv_ipc_open(py::bytes handle_bytes) {
-  std::string raw = handle_bytes;
+int64_t kv_ipc_open(std::string raw) {
+  // std::string parameter: converted from py::bytes during argument
+  // processing (GIL held) so the gil_scoped_release guard is safe
   TORCH_CHECK(raw.size() == sizeof(hipIpcMemHandle_t), "bad handle size");
   hipIpcMemHandle_t handle;
   memcpy(&handle, raw.data(), sizeof(handle));
@@ -412,14 +413,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe32", &mfma_probe32, "32x32x16 MFMA layout probe");
   m.def("skinny_gemm", &skinny_gemm, "decode-shape GEMM (M<=128)",
         py::arg("a"), py::arg("w"), py::arg("force_splits") = 0);
+  // gil_scoped_release on the IPC/copy entry points: mapping or launching
+  // against a dying peer can block inside the HIP runtime, and holding the
+  // GIL there freezes heartbeats until the controller kills the process
   m.def("ipc_alloc_bf16", &ipc_alloc_bf16,
-        "hipMalloc-backed bf16 tensor (IPC-exportable base allocation)");
+        "hipMalloc-backed bf16 tensor (IPC-exportable base allocation)",
+        py::call_guard<py::gil_scoped_release>());
   m.def("kv_ipc_export", &kv_ipc_export, "hipIpcGetMemHandle of a KV pool");
   m.def("kv_ipc_open", &kv_ipc_open,
-        "map a peer KV pool (lazy xGMI peer access)");
-  m.def("kv_ipc_close", &kv_ipc_close, "unmap a peer KV pool");
+        "map a peer KV pool (lazy xGMI peer access)",
+        py::call_guard<py::gil_scoped_release>());
+  m.def("kv_ipc_close", &kv_ipc_close, "unmap a peer KV pool",
+        py::call_guard<py::gil_scoped_release>());
   m.def("kv_peer_copy", &kv_peer_copy,
-        "gather local pages -> scatter into peer pool over xGMI");
+        "gather local pages -> scatter into peer pool over xGMI",
+        py::call_guard<py::gil_scoped_release>());
   m.def("ar_alloc_signals", &ar_alloc_signals,
         "uncached signal buffer for the xGMI all-reduce");
   m.def("ar_export_ptr", &ar_export_ptr, "hipIpc handle of a raw pointer");

@@ -27,6 +27,7 @@ from __future__ import annotations
 
 import base64
 import threading
+import time
 from typing import Any, Dict, List
 
 import torch
@@ -65,17 +66,39 @@ def export_meta_local(cache: PagedKVCache) -> Dict[str, Any]:
     }
 
 
+class PeerDead(RuntimeError):
+    """The push did not complete in time — the exporting engine is gone
+    (its dmabuf is revoked and the copy wavefronts stall).  The peer uid
+    is quarantined; callers re-dispatch the request."""
+
+
 class PendingPush:
-    def __init__(self, event: torch.cuda.Event, nbytes: int, keep=()):
+    def __init__(self, event: torch.cuda.Event, nbytes: int, keep=(),
+                 on_timeout=None):
         self._event = event
         self.nbytes = nbytes
         self._keep = keep      # tensors the in-flight kernel reads
+        self._on_timeout = on_timeout
 
     def done(self) -> bool:
         return self._event.query()
 
-    def wait(self) -> None:
-        self._event.synchronize()
+    def wait(self, timeout_s: float = 10.0) -> None:
+        """Poll the completion event (GIL-friendly) instead of a blocking
+        synchronize: a copy into a DEAD peer's revoked pool never
+        completes, and piling launches onto that wedged stream eventually
+        blocks the launching thread inside the HIP runtime WITH the GIL —
+        freezing the whole process until the controller's hung-HIP
+        heuristic kills it (GPU run r2/15).  A 268 MB push takes 0.15 ms;
+        10 s of non-completion means the peer died mid-transfer."""
+        deadline = time.monotonic() + timeout_s
+        while not self._event.query():
+            if time.monotonic() >= deadline:
+                if self._on_timeout is not None:
+                    self._on_timeout()
+                raise PeerDead(
+                    f"peer push did not complete within {timeout_s}s")
+            time.sleep(0.002)
 
 
 class PeerKVPusher:
@@ -84,9 +107,20 @@ class PeerKVPusher:
 
     def __init__(self, device: torch.device):
         self.device = device
-        self._stream = torch.cuda.Stream(device)
+        # one stream per peer POOL: a wedged copy to a dead peer must not
+        # stall pushes to the survivors
+        self._streams: Dict[str, torch.cuda.Stream] = {}
         self._open: Dict[tuple, int] = {}   # (uid, handle) -> mapped ptr
+        self._bad: set = set()              # quarantined pool uids
         self._lock = threading.Lock()
+
+    def _stream_for(self, uid: str) -> torch.cuda.Stream:
+        with self._lock:
+            st = self._streams.get(uid)
+            if st is None:
+                st = torch.cuda.Stream(self.device)
+                self._streams[uid] = st
+            return st
 
     def _map(self, meta: Dict[str, Any]) -> int:
         if "local_ptr" in meta:          # same-process pool: no IPC needed
@@ -110,6 +144,10 @@ class PeerKVPusher:
         The copy runs on this pusher's dedicated stream so it overlaps any
         compute the caller's engine keeps issuing on the default stream."""
         assert len(src_pages) == len(dst_pages)
+        uid = str(dst_meta.get("uid", dst_meta.get("local_ptr", "")))
+        with self._lock:
+            if uid in self._bad:
+                raise PeerDead(f"pool {uid} is quarantined (dead peer)")
         shape = cache.kv.shape
         assert list(shape)[3:] == list(dst_meta["shape"])[3:], \
             "page layout mismatch between src and dst pools"
@@ -117,7 +155,12 @@ class PeerKVPusher:
         chunk_bytes = shape[3] * shape[4] * shape[5] * 2
         nbytes = len(src_pages) * 2 * shape[0] * chunk_bytes
         ev = torch.cuda.Event()
-        with torch.cuda.stream(self._stream):
+        stream = self._stream_for(uid)
+
+        def quarantine():
+            with self._lock:
+                self._bad.add(uid)
+        with torch.cuda.stream(stream):
             # page-index tensors are allocated AND consumed on the
             # transfer stream: allocating them on the caller's stream and
             # letting them die before the kernel ran let the caching
@@ -131,8 +174,9 @@ class PeerKVPusher:
             ops._require_hip().kv_peer_copy(
                 dst_base, cache.kv, src_t, dst_t,
                 int(dst_meta["num_pages"]))
-            ev.record(self._stream)
-        return PendingPush(ev, nbytes, keep=(src_t, dst_t))
+            ev.record(stream)
+        return PendingPush(ev, nbytes, keep=(src_t, dst_t),
+                           on_timeout=quarantine)
 
     def close(self) -> None:
         with self._lock:
