@@ -210,7 +210,8 @@ class Router:
                 # failover_window_s) instead of failing the client.
                 retryable = isinstance(
                     e, (InstanceLost, ConnectionError, OSError)) or \
-                    "no ready instances" in str(e)
+                    "no ready instances" in str(e) or \
+                    "PeerDead" in str(e)
                 if not retryable or time.monotonic() >= deadline:
                     raise
                 last = e
