@@ -90,15 +90,22 @@ class PendingPush:
         blocks the launching thread inside the HIP runtime WITH the GIL —
         freezing the whole process until the controller's hung-HIP
         heuristic kills it (GPU run r2/15).  A 268 MB push takes 0.15 ms;
-        10 s of non-completion means the peer died mid-transfer."""
-        deadline = time.monotonic() + timeout_s
+        10 s of non-completion means the peer died mid-transfer.
+        Busy-poll for the first few ms (the common case completes in
+        ~0.15 ms; a plain sleep loop quantized every wait to ~2 ms),
+        then back off to sleeps."""
+        start = time.monotonic()
+        deadline = start + timeout_s
+        busy_until = start + 0.004
         while not self._event.query():
-            if time.monotonic() >= deadline:
+            now = time.monotonic()
+            if now >= deadline:
                 if self._on_timeout is not None:
                     self._on_timeout()
                 raise PeerDead(
                     f"peer push did not complete within {timeout_s}s")
-            time.sleep(0.002)
+            if now >= busy_until:
+                time.sleep(0.0005)
 
 
 class PeerKVPusher:
@@ -122,6 +129,8 @@ class PeerKVPusher:
                 self._streams[uid] = st
             return st
 
+    OPEN_TIMEOUT_S = 5.0
+
     def _map(self, meta: Dict[str, Any]) -> int:
         if "local_ptr" in meta:          # same-process pool: no IPC needed
             return int(meta["local_ptr"])
@@ -132,11 +141,36 @@ class PeerKVPusher:
         key = (meta.get("uid", ""), meta["handle"])
         with self._lock:
             ptr = self._open.get(key)
-            if ptr is None:
-                raw = base64.b64decode(meta["handle"])
-                ptr = ops._require_hip().kv_ipc_open(raw)
-                self._open[key] = ptr
+        if ptr is not None:
             return ptr
+        # hipIpcOpenMemHandle against an exporter that died after sending
+        # its handle BLOCKS INDEFINITELY in the driver (GPU runs r2/15-16:
+        # with the GIL it froze heartbeats; without it it wedged the
+        # serialized prefill pipeline).  Run the open under a watchdog
+        # thread; on timeout quarantine the pool and fail the request —
+        # the stuck thread is leaked (daemon, bounded by peer deaths).
+        raw = base64.b64decode(meta["handle"])
+        box: Dict[str, Any] = {}
+
+        def do_open():
+            try:
+                box["ptr"] = ops._require_hip().kv_ipc_open(raw)
+            except Exception as e:  # noqa: BLE001
+                box["err"] = e
+        th = threading.Thread(target=do_open, daemon=True)
+        th.start()
+        th.join(self.OPEN_TIMEOUT_S)
+        if th.is_alive():
+            with self._lock:
+                self._bad.add(str(meta.get("uid", "")))
+            raise PeerDead(
+                f"hipIpcOpenMemHandle blocked > {self.OPEN_TIMEOUT_S}s "
+                "(exporter died mid-handshake); pool quarantined")
+        if "err" in box:
+            raise box["err"]
+        with self._lock:
+            self._open[key] = box["ptr"]
+        return box["ptr"]
 
     def push(self, cache: PagedKVCache, src_pages: List[int],
              dst_meta: Dict[str, Any], dst_pages: List[int]) -> PendingPush:
