@@ -26,6 +26,7 @@ exchange), negotiated per sequence via the import_seq handshake.
 from __future__ import annotations
 
 import base64
+import logging
 import threading
 import time
 from typing import Any, Dict, List
@@ -34,6 +35,8 @@ import torch
 
 from .. import ops
 from ..engine.kv_cache import PagedKVCache
+
+log = logging.getLogger(__name__)
 
 
 def peer_capable(cache: PagedKVCache) -> bool:
@@ -129,7 +132,7 @@ class PeerKVPusher:
                 self._streams[uid] = st
             return st
 
-    OPEN_TIMEOUT_S = 5.0
+    OPEN_TIMEOUT_S = 15.0
 
     def _map(self, meta: Dict[str, Any]) -> int:
         if "local_ptr" in meta:          # same-process pool: no IPC needed
@@ -157,17 +160,22 @@ class PeerKVPusher:
                 box["ptr"] = ops._require_hip().kv_ipc_open(raw)
             except Exception as e:  # noqa: BLE001
                 box["err"] = e
+        t0 = time.monotonic()
         th = threading.Thread(target=do_open, daemon=True)
         th.start()
         th.join(self.OPEN_TIMEOUT_S)
         if th.is_alive():
             with self._lock:
                 self._bad.add(str(meta.get("uid", "")))
+            log.warning("ipc open of pool %s blocked > %.0fs: quarantined",
+                        meta.get("uid"), self.OPEN_TIMEOUT_S)
             raise PeerDead(
                 f"hipIpcOpenMemHandle blocked > {self.OPEN_TIMEOUT_S}s "
                 "(exporter died mid-handshake); pool quarantined")
         if "err" in box:
             raise box["err"]
+        log.info("ipc open of pool %s took %.3fs",
+                 meta.get("uid"), time.monotonic() - t0)
         with self._lock:
             self._open[key] = box["ptr"]
         return box["ptr"]
@@ -192,6 +200,7 @@ class PeerKVPusher:
         stream = self._stream_for(uid)
 
         def quarantine():
+            log.warning("push to pool %s timed out: quarantined", uid)
             with self._lock:
                 self._bad.add(uid)
         with torch.cuda.stream(stream):

@@ -234,7 +234,7 @@ def main() -> int:
                         os.path.join(run_root, "**", "*.log"),
                         recursive=True)):
                     with open(lp, "rb") as f:
-                        f.seek(max(0, os.path.getsize(lp) - 4000))
+                        f.seek(max(0, os.path.getsize(lp) - 20000))
                         tail = f.read().decode(errors="replace")
                     print(f"==== {lp} ====\n{tail}", file=sys.stderr)
         except Exception:  # noqa: BLE001
@@ -243,6 +243,8 @@ def main() -> int:
 
     ok = [r for r in results if "error" not in r]
     errs = [r for r in results if "error" in r]
+    from collections import Counter
+    err_kinds = Counter(r["error"][:120] for r in errs)
     ttfts = sorted(r["ttft"] for r in ok)
     total_tokens = sum(r["out"] for r in ok) + args.in_len * len(ok)
     out_tokens = sum(r["out"] for r in ok)
@@ -251,6 +253,7 @@ def main() -> int:
                    "prompts": args.prompts, "rate": args.rate,
                    "in_len": args.in_len, "out_len": args.out_len},
         "completed": len(ok), "errors": len(errs),
+        **({"error_kinds": dict(err_kinds.most_common(6))} if errs else {}),
         **({"failover": kill_report} if kill_report else {}),
         "total_token_throughput_tok_s": round(total_tokens / wall, 1),
         "output_tok_s": round(out_tokens / wall, 1),
