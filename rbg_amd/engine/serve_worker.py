@@ -508,25 +508,34 @@ class ServeWorker:
             client = RpcClient("127.0.0.1",
                                self._decode_port(decode_instance))
             try:
-                res = client.call("import_seq", peer_ok=peer_ok,
-                                  src_instance=self.my_instance, **meta)
-                if res.get("peer") is not None:
-                    # xGMI fast path (non-TP only — _peer_possible): push
-                    # straight into the decode pool, then commit; no
-                    # collective in the path
-                    self._peer_push(pages, res["peer"], res["dst_pages"])
-                    client.call("import_commit", seq_id=res["seq_id"])
-                    self._release_parked(parked)
-                elif self.tp is not None:
+                if self.tp is not None:
+                    res = client.call("import_seq", peer_ok=peer_ok,
+                                      src_instance=self.my_instance, **meta)
                     # every rank sends its KV shard to its counterpart;
                     # followers park identical page ids (lockstep)
                     self._tp_submit_op({"kind": "migrate",
                                         "seq_id": seq.seq_id,
                                         "dst_instance": decode_instance})
                 else:
-                    self._send_pages(pages,
-                                     self._peer_rank(decode_instance))
-                    self._release_parked(parked)
+                    # non-TP: parked pages must return to the pool on EVERY
+                    # path — a failed push (dead peer) otherwise leaks them
+                    # and repeated failover retries drain the prefill pool
+                    try:
+                        res = client.call("import_seq", peer_ok=peer_ok,
+                                          src_instance=self.my_instance,
+                                          **meta)
+                        if res.get("peer") is not None:
+                            # xGMI fast path: push straight into the
+                            # decode pool, then commit; no collective
+                            self._peer_push(pages, res["peer"],
+                                            res["dst_pages"])
+                            client.call("import_commit",
+                                        seq_id=res["seq_id"])
+                        else:
+                            self._send_pages(
+                                pages, self._peer_rank(decode_instance))
+                    finally:
+                        self._release_parked(parked)
             finally:
                 client.close()
             return {"decode_seq_id": res.get("seq_id"),
