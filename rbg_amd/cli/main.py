@@ -101,6 +101,15 @@ def cmd_status(client: BaseClient, args) -> int:
                      rs.updated_replicas])
     print()
     _print_table(rows, ["ROLE", "READY", "UPDATED"])
+    # surface stalled role workloads (e.g. Progressing=False when an
+    # in-place-only update is infeasible) right in `rbgctl status`
+    for rs in rbg.status.role_statuses:
+        ris = client.get(C.KIND_ROLE_INSTANCE_SET,
+                         f"{args.name}-{rs.name}", args.namespace)
+        for cond in getattr(getattr(ris, "status", None), "conditions", []):
+            if cond.type != C.COND_READY and cond.status != "True":
+                print(f"  role {rs.name}: {cond.type}={cond.status} "
+                      f"({cond.reason}) {cond.message}")
     insts = client.list(C.KIND_ROLE_INSTANCE, args.namespace,
                         selector={C.LABEL_GROUP_NAME: args.name})
     rows = []
