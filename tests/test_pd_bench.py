@@ -119,3 +119,54 @@ def test_local_peer_push_matches_index_copy():
     got = dst.kv.index_select(
         2, torch.tensor(dst_pages, dtype=torch.int64, device=dev))
     assert torch.equal(want, got)
+
+
+def test_pusher_watchdog_quarantines_blocked_open(monkeypatch):
+    """hipIpcOpenMemHandle that never returns (dead exporter / the ROCm
+    dmabuf hazard) must become PeerDead within OPEN_TIMEOUT_S and
+    quarantine the pool uid — never block the caller indefinitely."""
+    import threading
+    import time as _time
+
+    import rbg_amd.ops as ops_mod
+    from rbg_amd.parallel import kv_peer
+
+    class StubHip:
+        def kv_ipc_open(self, raw):
+            _time.sleep(3600)
+
+    monkeypatch.setattr(ops_mod, "_hip", StubHip())
+    monkeypatch.setattr(ops_mod, "HAVE_HIP", True)
+
+    pusher = kv_peer.PeerKVPusher.__new__(kv_peer.PeerKVPusher)
+    pusher._open = {}
+    pusher._bad = set()
+    pusher._streams = {}
+    pusher._lock = threading.Lock()
+    pusher.OPEN_TIMEOUT_S = 0.3
+    meta = {"uid": "deadbeef", "handle": "QUJDRA==",
+            "num_pages": 4, "shape": [2, 2, 4, 1, 16, 128]}
+    t0 = _time.monotonic()
+    with pytest.raises(kv_peer.PeerDead):
+        pusher._map(meta)
+    assert _time.monotonic() - t0 < 2.0        # bounded, not blocked
+    assert "deadbeef" in pusher._bad
+
+
+def test_pending_push_timeout_raises_peerdead():
+    import time as _time
+
+    from rbg_amd.parallel.kv_peer import PeerDead, PendingPush
+
+    class NeverEvent:
+        def query(self):
+            return False
+
+    hit = []
+    p = PendingPush(NeverEvent(), nbytes=1,
+                    on_timeout=lambda: hit.append(1))
+    t0 = _time.monotonic()
+    with pytest.raises(PeerDead):
+        p.wait(timeout_s=0.2)
+    assert hit == [1]
+    assert _time.monotonic() - t0 < 1.5
