@@ -89,6 +89,28 @@ def main():
             row(["phase", "submitted", "settled", "p50 s", "p90 s",
                  "p99 s"], "th") + "".join(rows) + "</table>")
 
+    # P/D serving + failover (round 2)
+    rows = []
+    for name, label in (
+            ("bench_serving_pd_llama8b_rate10b.json", "P/D disagg 300@10"),
+            ("bench_serving_llama8b_rate10_r2.json", "colocated 300@10"),
+            ("bench_serving_qwen3_32b_rate10_r2.json", "qwen3-32b 300@10"),
+            ("bench_decode_pool_failover.json",
+             "decode pool 4x + mid-stream kill")):
+        d = load_json(os.path.join(args.profiles, name))
+        if d:
+            fo = d.get("failover") or {}
+            rows.append(row([label, d.get("completed"), d.get("errors"),
+                             d.get("total_token_throughput_tok_s"),
+                             d.get("mean_ttft_ms"), d.get("mean_itl_ms"),
+                             fo.get("recovery_s", "-")]))
+    if rows:
+        sections.append(
+            "<h2>Serving (round 2)</h2><table>" +
+            row(["config", "completed", "errors", "tok/s", "mean TTFT ms",
+                 "mean ITL ms", "recover s"], "th") +
+            "".join(rows) + "</table>")
+
     # stress matrix (round 2: scenario sweep with controller CPU/RSS)
     d = load_json(os.path.join(args.profiles, "stress_matrix_r2.json"))
     if d and d.get("matrix"):
