@@ -49,6 +49,10 @@ __global__ void kv_peer_copy_kernel(
                (long)dst_pages[page_i] * dst_page_stride_vec;
   for (int i = threadIdx.x; i < chunk_vec; i += blockDim.x)
     dst[i] = src[i];
+  // system-scope release: the completion event is observed by the HOST
+  // and relayed to the PEER process (import_commit), so the remote
+  // stores must be visible across agents, not just this one
+  __threadfence_system();
 }
 
 }  // namespace
