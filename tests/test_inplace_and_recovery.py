@@ -173,3 +173,40 @@ def test_decode_pool_failover_continuity(tmp_path):
     assert d["failover"]["killed"].startswith("bench-decode")
     assert d["errors"] == 0, d
     assert d["completed"] == 16, d
+
+
+def test_watcher_fast_fails_on_engine_restart():
+    """A recreated engine answers poll_many with 'unknown seq' for
+    pre-restart sequences: the watcher must fail those waiters as
+    InstanceLost immediately (the router re-dispatches) instead of
+    letting them run out the request timeout."""
+    from rbg_amd.server.router_worker import InstanceLost, _InstanceWatcher
+
+    class RestartedClient:
+        def call(self, method, **kw):
+            raise RuntimeError("rpc poll_many failed: "
+                               "KeyError('unknown seq 7')")
+
+    w = _InstanceWatcher(RestartedClient(), tick=0.005)
+    t0 = time.time()
+    with pytest.raises(InstanceLost):
+        w.wait_for(7, timeout=30.0)
+    assert time.time() - t0 < 5.0     # fast-fail, not the 30 s timeout
+
+
+def test_watcher_declares_loss_after_connection_window(monkeypatch):
+    """Sustained connection failure (instance down, gang recreating)
+    converts to InstanceLost after FAIL_WINDOW_S."""
+    from rbg_amd.server import router_worker as rw
+
+    class DeadClient:
+        def call(self, method, **kw):
+            raise ConnectionRefusedError(111, "refused")
+
+    monkeypatch.setattr(rw._InstanceWatcher, "FAIL_WINDOW_S", 0.5)
+    w = rw._InstanceWatcher(DeadClient(), tick=0.005)
+    t0 = time.time()
+    with pytest.raises(rw.InstanceLost):
+        w.wait_for(1, timeout=30.0)
+    dt = time.time() - t0
+    assert 0.4 < dt < 5.0, dt
