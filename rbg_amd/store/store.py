@@ -63,14 +63,21 @@ class Watch:
         self._closed = False
 
     def _offer(self, ev: Event) -> None:
+        # called under the store lock: ev.obj is the STORED object (a
+        # replace-only reference, never mutated in place), so enqueueing
+        # the reference is ordering-safe and costs nothing; the deep copy
+        # each consumer needs happens on DEQUEUE, outside the store lock
+        # (the clone-under-lock was the controller's scaling bound at
+        # 50 groups/20 qps — profiles/stress_matrix_r2.json)
         if not self._closed and (self._kinds is None or ev.kind in self._kinds):
             self._q.put(ev)
 
     def get(self, timeout: Optional[float] = None) -> Optional[Event]:
         try:
-            return self._q.get(timeout=timeout)
+            ev = self._q.get(timeout=timeout)
         except queue.Empty:
             return None
+        return Event(ev.type, ev.kind, clone(ev.obj))
 
     def __iter__(self) -> Iterator[Event]:
         while not self._closed:
@@ -104,20 +111,24 @@ class Store:
             m.creation_timestamp = m.creation_timestamp or time.time()
             stored = clone(obj)
             self._objects[key] = stored
-            out = clone(stored)
             # enqueue while still holding the lock: watchers must observe
             # events in resourceVersion order (advisor round-1 finding —
             # an out-of-order MODIFIED could otherwise be the version the
-            # write-behind persister durably records)
-            self._notify(Event("ADDED", obj.kind, clone(stored)))
-        return out
+            # write-behind persister durably records).  The event carries
+            # the stored REFERENCE; consumers clone on dequeue.
+            self._notify(Event("ADDED", obj.kind, stored))
+        return clone(stored)
 
     def get(self, kind: str, name: str, namespace: str = "default") -> Any:
+        # stored objects are replace-only: the reference stays a
+        # consistent snapshot after the lock drops, so the deep copy
+        # happens OUTSIDE the lock (reader-side clones were the dominant
+        # store-lock hold time under controller load)
         with self._lock:
             obj = self._objects.get((kind, namespace, name))
-            if obj is None:
-                raise NotFound(f"{kind} {namespace}/{name}")
-            return clone(obj)
+        if obj is None:
+            raise NotFound(f"{kind} {namespace}/{name}")
+        return clone(obj)
 
     def try_get(self, kind: str, name: str, namespace: str = "default") -> Optional[Any]:
         try:
@@ -160,9 +171,8 @@ class Store:
             else:
                 m.generation = cur.metadata.generation
             self._objects[key] = stored
-            out = clone(stored)
-            self._notify(Event("MODIFIED", obj.kind, clone(stored)))
-        return out
+            self._notify(Event("MODIFIED", obj.kind, stored))
+        return clone(stored)
 
     def apply(self, kind: str, name: str, mutate: Callable[[Any], Any],
               namespace: str = "default", subresource: str = "") -> Any:
@@ -183,7 +193,7 @@ class Store:
             obj = self._objects.pop(key, None)
             if obj is None:
                 raise NotFound(f"{kind} {namespace}/{name}")
-            self._notify(Event("DELETED", kind, clone(obj)))
+            self._notify(Event("DELETED", kind, obj))
         return obj
 
     def try_delete(self, kind: str, name: str, namespace: str = "default") -> bool:
@@ -197,17 +207,13 @@ class Store:
 
     def list(self, kind: str, namespace: Optional[str] = "default",
              selector: Optional[Dict[str, str]] = None) -> List[Any]:
+        # select references under the lock, deep-copy outside it
         with self._lock:
-            out = []
-            for (k, ns, _), obj in self._objects.items():
-                if k != kind:
-                    continue
-                if namespace is not None and ns != namespace:
-                    continue
-                if selector and not match_labels(obj.metadata.labels, selector):
-                    continue
-                out.append(clone(obj))
-            return out
+            refs = [obj for (k, ns, _), obj in self._objects.items()
+                    if k == kind and (namespace is None or ns == namespace)
+                    and (not selector or
+                         match_labels(obj.metadata.labels, selector))]
+        return [clone(obj) for obj in refs]
 
     def list_owned(self, kind: str, owner_uid: str,
                    namespace: str = "default") -> List[Any]:
@@ -229,7 +235,7 @@ class Store:
             self._watches.append(w)
             if replay:
                 for obj in self._objects.values():
-                    w._offer(Event("ADDED", obj.kind, clone(obj)))
+                    w._offer(Event("ADDED", obj.kind, obj))
         return w
 
     def _drop_watch(self, w: Watch) -> None:
