@@ -472,7 +472,20 @@ class RoleBasedGroupController:
                 ready = c is not None and c.status == "True"
                 gpu_ids = sorted({g for w in inst.status.workers
                                   for g in w.gpu_ids})
-                ports = [p for w in inst.status.workers for p in w.ports]
+                workers = list(inst.status.workers)
+                # sharedServiceSelection=LeaderOnly (reference
+                # rolebasedgroup_types.go:355-402 / KEP-260): the role's
+                # discovery endpoints expose only the LEADER component —
+                # engines with an internal rank-0 entrypoint (the usual
+                # TP serving shape) are addressed through it
+                lwp = role.leader_worker_pattern
+                if role.pattern == C.PATTERN_LEADER_WORKER and \
+                        lwp is not None and \
+                        lwp.shared_service_selection == "LeaderOnly":
+                    prefix = f"{inst.metadata.name}-leader-"
+                    workers = [w for w in workers
+                               if w.name.startswith(prefix)]
+                ports = [p for w in workers for p in w.ports]
                 items.append({"name": inst.metadata.name,
                               "address": "127.0.0.1",
                               "ports": ports or role.service_ports,

@@ -355,3 +355,56 @@ def test_rolling_recreate_respects_max_unavailable(mgr):
     assert all(i.metadata.uid not in old_uids for i in insts), \
         "rollout did not complete"
     assert min_ready >= 2, f"availability dipped to {min_ready}"
+
+
+def test_leader_only_shared_service_selection():
+    """sharedServiceSelection=LeaderOnly (reference KEP-260 /
+    rolebasedgroup_types.go:355-402): the role's discovery endpoints
+    expose only the LEADER component's ports."""
+    from rbg_amd.api.types import (ComponentSpec, EngineSpec, EngineTemplate,
+                                   LeaderWorkerPattern, RoleBasedGroup,
+                                   RoleBasedGroupSpec, RoleSpec, WorkerStatus)
+    from rbg_amd.controller.rbg_controller import RoleBasedGroupController
+    from rbg_amd.store.store import Store, set_owner
+    from rbg_amd.api.types import ObjectMeta, RoleInstance, RoleInstanceSpec
+
+    store = Store()
+
+    class CapturingRegistry:
+        def __init__(self):
+            self.last = None
+
+        def publish(self, rbg, instances):
+            self.last = instances
+
+        def path_for(self, ns, name):
+            return f"/tmp/{ns}-{name}.yaml"
+
+    reg = CapturingRegistry()
+    ctrl = RoleBasedGroupController(store, reg)
+    tmpl = EngineTemplate(engines=[EngineSpec(name="engine",
+                                              runner="llm-engine")])
+    rbg = store.create(RoleBasedGroup(
+        metadata=ObjectMeta(name="lw"),
+        spec=RoleBasedGroupSpec(roles=[RoleSpec(
+            name="tp", replicas=1, pattern=C.PATTERN_LEADER_WORKER,
+            leader_worker_pattern=LeaderWorkerPattern(
+                size=3, leader_template=tmpl, worker_template=tmpl,
+                shared_service_selection="LeaderOnly"))])))
+    inst = RoleInstance(
+        metadata=ObjectMeta(name="lw-tp-0",
+                            labels={C.LABEL_GROUP_NAME: "lw",
+                                    C.LABEL_ROLE_NAME: "tp"}),
+        spec=RoleInstanceSpec(components=[
+            ComponentSpec(name="leader", size=1, template=tmpl),
+            ComponentSpec(name="worker", size=2, template=tmpl)]))
+    inst.status.workers = [
+        WorkerStatus(name="lw-tp-0-leader-0", phase="Ready", ports=[1111]),
+        WorkerStatus(name="lw-tp-0-worker-0", phase="Ready", ports=[2222]),
+        WorkerStatus(name="lw-tp-0-worker-1", phase="Ready", ports=[3333]),
+    ]
+    set_owner(inst, rbg)
+    store.create(inst)
+    ctrl._publish_discovery(store.get(C.KIND_RBG, "lw"))
+    entry = reg.last["tp"][0]
+    assert entry["ports"] == [1111], entry
