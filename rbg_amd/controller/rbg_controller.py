@@ -457,6 +457,29 @@ class RoleBasedGroupController:
 
     # ------------------------------------------------------------------
 
+    def _ensure_discovery_mode(self, rbg: RoleBasedGroup) -> str:
+        """Sticky discovery-config mode (reference
+        ensureDiscoveryConfigMode, KEP-133): once chosen it never changes
+        for a group.  Pre-existing reconciled groups keep the LEGACY
+        per-role config files; new groups get the refined single config."""
+        mode = rbg.metadata.annotations.get(C.ANNO_DISCOVERY_MODE, "")
+        if mode in ("legacy", "refined"):
+            return mode
+        legacy = (rbg.status.observed_generation > 0 or
+                  bool(rbg.status.role_statuses))
+        mode = "legacy" if legacy else "refined"
+
+        def mark(cur: RoleBasedGroup):
+            cur.metadata.annotations.setdefault(C.ANNO_DISCOVERY_MODE, mode)
+            return cur
+        try:
+            self.store.apply(C.KIND_RBG, rbg.metadata.name, mark,
+                             rbg.metadata.namespace)
+        except KeyError:
+            pass
+        rbg.metadata.annotations[C.ANNO_DISCOVERY_MODE] = mode
+        return mode
+
     def _publish_discovery(self, rbg: RoleBasedGroup) -> None:
         if self.registry is None:
             return
@@ -492,7 +515,8 @@ class RoleBasedGroupController:
                               "gpu_ids": gpu_ids, "ready": ready})
             items.sort(key=lambda i: i["name"])
             instances[role.name] = items
-        self.registry.publish(rbg, instances)
+        self.registry.publish(rbg, instances,
+                              mode=self._ensure_discovery_mode(rbg))
 
     # ------------------------------------------------------------------
 

@@ -91,15 +91,39 @@ class TopologyRegistry:
     def path_for(self, namespace: str, rbg_name: str) -> str:
         return os.path.join(self.root_dir, namespace, rbg_name, "config.yaml")
 
-    def publish(self, rbg, instances: Dict[str, List[Dict[str, Any]]]) -> str:
+    def role_path_for(self, namespace: str, rbg_name: str,
+                      role: str) -> str:
+        """Legacy discovery mode: one config per role (the reference's
+        pre-KEP-133 per-role ConfigMaps `{rbg}-{role}`)."""
+        return os.path.join(self.root_dir, namespace, rbg_name,
+                            f"config-{role}.yaml")
+
+    def publish(self, rbg, instances: Dict[str, List[Dict[str, Any]]],
+                mode: str = "refined") -> str:
         doc = build_config(rbg, instances)
         path = self.path_for(rbg.metadata.namespace, rbg.metadata.name)
         write_config(doc, path)
+        if mode == "legacy":
+            # legacy groups ALSO get per-role files so engines written
+            # against the old per-role schema keep working (reference
+            # ensureDiscoveryConfigMode keeps pre-existing groups on the
+            # legacy ConfigMaps; KEP-133)
+            for role_doc in doc["group"]["roles"]:
+                write_config(
+                    {"group": {"name": rbg.metadata.name,
+                               "size": len(role_doc["instances"]),
+                               "roles": [role_doc]}},
+                    self.role_path_for(rbg.metadata.namespace,
+                                       rbg.metadata.name,
+                                       role_doc["name"]))
         return path
 
     def remove(self, namespace: str, rbg_name: str) -> None:
+        import glob
         path = self.path_for(namespace, rbg_name)
-        try:
-            os.remove(path)
-        except FileNotFoundError:
-            pass
+        for p in [path] + glob.glob(os.path.join(
+                os.path.dirname(path), "config-*.yaml")):
+            try:
+                os.remove(p)
+            except FileNotFoundError:
+                pass

@@ -374,7 +374,7 @@ def test_leader_only_shared_service_selection():
         def __init__(self):
             self.last = None
 
-        def publish(self, rbg, instances):
+        def publish(self, rbg, instances, mode="refined"):
             self.last = instances
 
         def path_for(self, ns, name):
@@ -408,3 +408,26 @@ def test_leader_only_shared_service_selection():
     ctrl._publish_discovery(store.get(C.KIND_RBG, "lw"))
     entry = reg.last["tp"][0]
     assert entry["ports"] == [1111], entry
+
+
+def test_discovery_mode_sticky_and_legacy_per_role(mgr, tmp_path):
+    """KEP-133 analog: a NEW group gets the refined single config; a
+    group annotated legacy ALSO gets per-role config files, and the mode
+    is sticky once set."""
+    import yaml as _yaml
+    mgr.store.create(router_worker_rbg(name="dm"))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "dm"), timeout=30)
+    rbg = mgr.store.get(C.KIND_RBG, "dm")
+    assert rbg.metadata.annotations.get(C.ANNO_DISCOVERY_MODE) == "refined"
+
+    # explicit legacy group: per-role files appear next to the group one
+    leg = router_worker_rbg(name="dml")
+    leg.metadata.annotations[C.ANNO_DISCOVERY_MODE] = "legacy"
+    mgr.store.create(leg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "dml"), timeout=30)
+    role_path = mgr.registry.role_path_for("default", "dml", "worker")
+    assert mgr.wait_for(lambda: os.path.exists(role_path), timeout=10)
+    with open(role_path) as f:
+        doc = _yaml.safe_load(f)
+    assert doc["group"]["roles"][0]["name"] == "worker"
+    assert len(doc["group"]["roles"]) == 1
