@@ -266,6 +266,14 @@ class RoleInstanceController:
         # group's engines pack onto one GPU set that no other group shares
         exclusive = bool(inst.metadata.annotations.get(
             C.ANNO_EXCLUSIVE_TOPOLOGY))
+        # in-place scheduling knobs (reference node_binding.go:276-423):
+        # mode required makes the sticky GPUs a hard constraint;
+        # granularity instance lets any engine reclaim any of the
+        # instance's previous devices
+        ips_mode = inst.metadata.annotations.get(
+            C.ANNO_INPLACE_SCHEDULING, "preferred")
+        ips_gran = inst.metadata.annotations.get(
+            C.ANNO_INPLACE_GRANULARITY, "component")
         claims = []
         for comp, j, wname in desired:
             gpus = 0
@@ -275,9 +283,14 @@ class RoleInstanceController:
                 if not res.cpu_only:
                     gpus = res.gpus
                     hbm = res.hbm_bytes
-            prefer = self.bindings.lookup(rbg_uid, GpuBindingStore.key(
-                inst.metadata.name, f"{comp.name}-{j}"))
+            if ips_gran == "instance":
+                prefer = self.bindings.lookup_instance(
+                    rbg_uid, inst.metadata.name)
+            else:
+                prefer = self.bindings.lookup(rbg_uid, GpuBindingStore.key(
+                    inst.metadata.name, f"{comp.name}-{j}"))
             claims.append(GpuClaim(gpus=gpus, hbm_bytes=hbm, prefer=prefer,
+                                   require_prefer=(ips_mode == "required"),
                                    group=rbg_uid, exclusive=exclusive)
                           if gpus or hbm else GpuClaim(gpus=0))
         if not any(cl.gpus or cl.hbm_bytes for cl in claims):

@@ -31,6 +31,10 @@ class GpuClaim:
     gpus: int = 1
     hbm_bytes: int = 0          # 0 => exclusive whole-GPU claim
     prefer: Tuple[int, ...] = ()  # sticky hint: previous device ids
+    # in-place-scheduling mode (reference node_binding.go Required
+    # affinity): require_prefer=True makes the sticky devices a HARD
+    # constraint — the reservation waits for exactly those GPUs
+    require_prefer: bool = False
     # exclusive-topology (reference pod_reconciler.go:160-241, consumed via
     # the rbg.workloads.x-k8s.io/exclusive-topology annotation): the claim's
     # GROUP packs onto a shared GPU set that no OTHER group may share
@@ -126,7 +130,7 @@ class GangAllocator:
 
                 cand = [g for g in claim.prefer if g in self._free_hbm
                         and allowed(g)]
-                if not cand:
+                if not cand and not (claim.require_prefer and claim.prefer):
                     cand = sorted(
                         (g for g in self._free_hbm if allowed(g)),
                         key=lambda g: (-group_share(g)
@@ -142,6 +146,10 @@ class GangAllocator:
                 res.assignments.append([g])
             else:
                 gpus = self._pick_gpus(claim.gpus, claim.prefer)
+                if gpus is not None and claim.require_prefer and \
+                        claim.prefer and \
+                        not set(gpus) <= set(claim.prefer):
+                    gpus = None      # required stickiness: wait for them
                 if gpus is None:
                     ok = False
                     break

@@ -47,3 +47,16 @@ class GpuBindingStore:
     def snapshot(self, rbg_uid: str) -> Dict[str, Tuple[int, ...]]:
         with self._lock:
             return dict(self._bindings.get(rbg_uid, {}))
+
+    def lookup_instance(self, rbg_uid: str,
+                        instance_name: str) -> Tuple[int, ...]:
+        """Granularity=instance: the union of every component binding of
+        the instance — any engine may reclaim any of the instance's
+        previous GPUs."""
+        prefix = instance_name + "/"
+        with self._lock:
+            out: List[int] = []
+            for k, gpus in self._bindings.get(rbg_uid, {}).items():
+                if k == instance_name or k.startswith(prefix):
+                    out.extend(g for g in gpus if g not in out)
+            return tuple(out)

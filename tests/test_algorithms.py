@@ -223,3 +223,40 @@ def test_exclusive_topology_packs_group_and_excludes_others():
     r5 = alloc.reserve("c-x", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
                                         group="C", exclusive=True)])
     assert r5.assignments[0][0] != g_a
+
+
+def test_required_stickiness_waits_for_previous_gpus():
+    """in-place-scheduling=required: a claim with sticky devices only
+    accepts those devices — the reservation waits (GangUnschedulable on
+    timeout) while they are busy, instead of landing elsewhere."""
+    from rbg_amd.scheduler.gang import GangAllocator, GangUnschedulable, GpuClaim
+    from rbg_amd.scheduler.topology import fully_connected
+    GB = 1 << 30
+    alloc = GangAllocator(fully_connected(2))
+    # someone exclusively holds GPU 1 (the sticky target)
+    alloc.reserve("other", [GpuClaim(gpus=1, hbm_bytes=0, prefer=(1,))])
+    with pytest.raises(GangUnschedulable):
+        alloc.reserve("sticky", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                          prefer=(1,),
+                                          require_prefer=True)],
+                      timeout=0.0)
+    # preferred mode lands on the free GPU instead
+    res = alloc.reserve("flex", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                          prefer=(1,))])
+    assert res.assignments[0] == [0]
+    # release the holder: required stickiness now succeeds on GPU 1
+    alloc.release("other")
+    res2 = alloc.reserve("sticky", [GpuClaim(gpus=1, hbm_bytes=10 * GB,
+                                             prefer=(1,),
+                                             require_prefer=True)])
+    assert res2.assignments[0] == [1]
+
+
+def test_instance_granularity_binding_union():
+    from rbg_amd.scheduler.placement import GpuBindingStore
+    st = GpuBindingStore()
+    st.record("uid1", GpuBindingStore.key("g-r-0", "leader-0"), [2])
+    st.record("uid1", GpuBindingStore.key("g-r-0", "worker-0"), [5])
+    st.record("uid1", GpuBindingStore.key("g-r-1", "leader-0"), [7])
+    assert set(st.lookup_instance("uid1", "g-r-0")) == {2, 5}
+    assert st.lookup_instance("uid1", "g-r-9") == ()
