@@ -34,6 +34,10 @@ LABEL_GROUP_NAME = f"{PREFIX}/name"                  # owning RBG name
 LABEL_ROLE_NAME = f"{PREFIX}/role"                   # role within the group
 LABEL_ROLE_INDEX = f"{PREFIX}/role-index"            # ordinal of the instance
 LABEL_INSTANCE_NAME = f"{PREFIX}/instance-name"      # owning RoleInstance
+# component identity: on one node the pod-label plane is realized as the
+# RBG_COMPONENT_NAME / RBG_COMPONENT_INDEX env every engine process gets
+# (discovery/env_builder.identity_env); these labels are the wire
+# vocabulary for clients selecting by component
 LABEL_COMPONENT_NAME = f"{PREFIX}/component-name"    # component within instance
 LABEL_COMPONENT_INDEX = f"{PREFIX}/component-index"
 LABEL_REVISION_HASH = f"{PREFIX}/revision-hash"      # controller-revision hash

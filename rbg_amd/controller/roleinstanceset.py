@@ -95,6 +95,8 @@ class RoleInstanceSetController:
         labels = dict(ris.metadata.labels)
         labels.update(tmpl.metadata.labels)
         labels[C.LABEL_ROLE_INDEX] = str(ordinal)
+        labels[C.LABEL_INSTANCE_NAME] = instance_ordinal_name(
+            ris.metadata.name, ordinal)
         labels[C.LABEL_REVISION_HASH] = revision
         annotations = dict(ris.metadata.annotations)
         annotations.update(tmpl.metadata.annotations)
@@ -257,6 +259,7 @@ class RoleInstanceSetController:
         inst = self._make_instance(ris, 0, revision)
         inst.metadata.name = name
         inst.metadata.labels[C.LABEL_ROLE_INDEX] = "0"
+        inst.metadata.labels[C.LABEL_INSTANCE_NAME] = name
         return inst
 
     def _ordinal(self, inst: RoleInstance) -> Optional[int]:

@@ -38,6 +38,17 @@ class ScalingAdapterController:
                     r.replicas = ad.spec.replicas
                 return cur
             self.store.apply(C.KIND_RBG, rbg.metadata.name, mutate, namespace)
+        # bind labels (reference rolebasedgroupscalingadapter_controller:
+        # the adapter is selectable by its target group/role)
+        want_labels = {C.LABEL_GROUP_NAME: rbg.metadata.name,
+                       C.LABEL_SCALING_ADAPTER: role.name}
+        if any(ad.metadata.labels.get(k) != v
+               for k, v in want_labels.items()):
+            def label(cur):
+                cur.metadata.labels.update(want_labels)
+                return cur
+            self.store.apply(C.KIND_SCALING_ADAPTER, ad.metadata.name,
+                             label, namespace)
         ris = self.store.try_get(
             C.KIND_ROLE_INSTANCE_SET,
             f"{rbg.metadata.name}-{role.name}", namespace)
