@@ -172,6 +172,8 @@ class ServeWorker:
         # the non-TP P/D pair; negotiated per sequence in import_seq.
         self._peer_pusher = None
         self._pending_imports: Dict[int, tuple] = {}
+        # migration gauges (the xGMI-bandwidth metric of SURVEY §5)
+        self._xfer_stats = {"pushes": 0, "bytes": 0, "seconds": 0.0}
         self._register_rpc()
 
     # -- xGMI peer push ------------------------------------------------------
@@ -190,9 +192,13 @@ class ServeWorker:
         if self._peer_pusher is None:
             self._peer_pusher = PeerKVPusher(
                 self.engine.runner.cache.kv.device)
+        t0 = time.monotonic()
         pending = self._peer_pusher.push(
             self.engine.runner.cache, pages, peer_meta, dst_pages)
         pending.wait()
+        self._xfer_stats["pushes"] += 1
+        self._xfer_stats["bytes"] += pending.nbytes
+        self._xfer_stats["seconds"] += time.monotonic() - t0
 
     # lazy: the discovery config lists peer instances only once the
     # controller has created them (same dependency wave) — re-read until
@@ -379,7 +385,16 @@ class ServeWorker:
         if self.rpc is None:
             return     # TP follower ranks take orders via broadcast only
         self.rpc.register("ping", lambda: "pong")
-        self.rpc.register("stats", lambda: self.engine.stats.snapshot())
+
+        def stats():
+            out = self.engine.stats.snapshot()
+            if self._xfer_stats["pushes"]:
+                xs = dict(self._xfer_stats)
+                xs["gb_per_s"] = round(
+                    xs["bytes"] / max(xs["seconds"], 1e-9) / 1e9, 2)
+                out["kv_migration"] = xs
+            return out
+        self.rpc.register("stats", stats)
         self.rpc.register("generate", self._rpc_generate)
         self.rpc.register("submit", self._rpc_submit)
         self.rpc.register("poll", self._rpc_poll)
