@@ -124,7 +124,8 @@ class WarmupController:
                     except Exception as e:  # noqa: BLE001
                         st.retries = attempt + 1
                         st.message = repr(e)
-                        time.sleep(min(2.0 ** attempt, 10.0))
+                        if attempt < pol.backoff_limit_per_gpu:
+                            time.sleep(min(2.0 ** attempt, 10.0))
                 st.phase = "Failed"
 
         threads = [threading.Thread(target=warm_one, args=(g,)) for g in gpu_ids]

@@ -42,6 +42,117 @@ def test_warmup_job_lifecycle(mgr):
         lambda: mgr.store.try_get(C.KIND_WARMUP, "warm") is None, timeout=30)
 
 
+def test_warmup_retry_backoff_then_success(mgr):
+    """A flaky action is retried up to backoffLimitPerGpu times and the
+    per-GPU retry count is recorded (reference warmup controller's
+    backoffLimitPerNode semantics)."""
+    from rbg_amd.controller import warmup as wmod
+    calls = {"n": 0}
+
+    def flaky(gpu_id):
+        calls["n"] += 1
+        if calls["n"] == 1:
+            raise RuntimeError("transient")
+
+    wmod.ACTIONS["test-flaky"] = flaky
+    try:
+        wu = RoleBasedGroupWarmup(
+            metadata=ObjectMeta(name="flaky"),
+            spec=WarmupSpec(gpu_ids=[0], actions=["test-flaky"]))
+        wu.spec.policies.backoff_limit_per_gpu = 2
+        mgr.store.create(wu)
+
+        def done():
+            cur = mgr.store.try_get(C.KIND_WARMUP, "flaky")
+            return cur is not None and cur.status.phase == "Succeeded"
+        assert mgr.wait_for(done, timeout=60)
+        cur = mgr.store.get(C.KIND_WARMUP, "flaky")
+        assert cur.status.gpus[0].retries == 1
+        assert cur.status.gpus[0].phase == "Succeeded"
+    finally:
+        wmod.ACTIONS.pop("test-flaky", None)
+
+
+@pytest.mark.parametrize("budget,want_phase", [(1, "Succeeded"), (0, "Failed")])
+def test_warmup_max_failed_gpus_budget(mgr, budget, want_phase):
+    """maxFailedGpus is the failure budget: one bad GPU out of two is
+    tolerated at budget 1, fatal at budget 0 (reference maxFailedNodes)."""
+    from rbg_amd.controller import warmup as wmod
+
+    def bad_gpu1(gpu_id):
+        if gpu_id == 1:
+            raise RuntimeError("ECC storm")
+
+    wmod.ACTIONS["test-bad1"] = bad_gpu1
+    try:
+        name = f"budget{budget}"
+        wu = RoleBasedGroupWarmup(
+            metadata=ObjectMeta(name=name),
+            spec=WarmupSpec(gpu_ids=[0, 1], actions=["test-bad1"]))
+        wu.spec.policies.backoff_limit_per_gpu = 0
+        wu.spec.policies.max_failed_gpus = budget
+        mgr.store.create(wu)
+
+        def done():
+            cur = mgr.store.try_get(C.KIND_WARMUP, name)
+            return cur is not None and \
+                cur.status.phase in ("Succeeded", "Failed")
+        assert mgr.wait_for(done, timeout=60)
+        cur = mgr.store.get(C.KIND_WARMUP, name)
+        assert cur.status.phase == want_phase
+        by_gpu = {g.gpu_id: g for g in cur.status.gpus}
+        assert by_gpu[0].phase == "Succeeded"
+        assert by_gpu[1].phase == "Failed"
+        assert "ECC storm" in by_gpu[1].message
+    finally:
+        wmod.ACTIONS.pop("test-bad1", None)
+
+
+def test_warmup_unknown_action_fails(mgr):
+    """An action name outside the registry fails the GPU (and the job) with
+    a diagnostic message rather than silently succeeding."""
+    wu = RoleBasedGroupWarmup(
+        metadata=ObjectMeta(name="unknown"),
+        spec=WarmupSpec(gpu_ids=[0], actions=["no-such-action"]))
+    wu.spec.policies.backoff_limit_per_gpu = 0
+    wu.spec.policies.max_failed_gpus = 0
+    mgr.store.create(wu)
+
+    def done():
+        cur = mgr.store.try_get(C.KIND_WARMUP, "unknown")
+        return cur is not None and cur.status.phase == "Failed"
+    assert mgr.wait_for(done, timeout=60)
+    cur = mgr.store.get(C.KIND_WARMUP, "unknown")
+    assert "no-such-action" in cur.status.gpus[0].message
+
+
+def test_warmup_gpu_set_derived_from_target_rbg(tmp_run_dir):
+    """spec.targetRbg derives the GPU set from the group's scheduled
+    instances (reference: warmup pods derived from an RBG's scheduled pods)."""
+    from rbg_amd.api.types import (RoleInstance, RoleInstanceStatus,
+                                   WorkerStatus)
+    from rbg_amd.controller.warmup import WarmupController
+    from rbg_amd.store.store import Store
+    store = Store()
+    for i, gpus in enumerate([[0, 1], [2, 3]]):
+        store.create(RoleInstance(
+            metadata=ObjectMeta(name=f"g-prefill-{i}",
+                                labels={C.LABEL_GROUP_NAME: "g"}),
+            status=RoleInstanceStatus(
+                workers=[WorkerStatus(name="w", gpu_ids=gpus)])))
+    # an instance of another group must not contribute
+    store.create(RoleInstance(
+        metadata=ObjectMeta(name="other-x-0",
+                            labels={C.LABEL_GROUP_NAME: "other"}),
+        status=RoleInstanceStatus(
+            workers=[WorkerStatus(name="w", gpu_ids=[7])])))
+    ctl = WarmupController(store)
+    wu = RoleBasedGroupWarmup(
+        metadata=ObjectMeta(name="derive"),
+        spec=WarmupSpec(target_rbg="g"))
+    assert ctl._gpu_ids(wu) == [0, 1, 2, 3]
+
+
 def test_scaling_adapter_drives_role_replicas(mgr):
     """/scale on the adapter overrides the role's declared replicas
     (reference rolebasedgroup_controller.go:853-901 + adapter controller)."""
