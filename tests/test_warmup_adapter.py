@@ -126,7 +126,7 @@ def test_warmup_unknown_action_fails(mgr):
     assert "no-such-action" in cur.status.gpus[0].message
 
 
-def test_warmup_gpu_set_derived_from_target_rbg(tmp_run_dir):
+def test_warmup_gpu_set_derived_from_target_rbg():
     """spec.targetRbg derives the GPU set from the group's scheduled
     instances (reference: warmup pods derived from an RBG's scheduled pods)."""
     from rbg_amd.api.types import (RoleInstance, RoleInstanceStatus,
