@@ -3,6 +3,9 @@
 Model presets are random-init shapes (no network for checkpoints —
 BASELINE.json: synthetic data / random weights); llama-3-8b is the flagship
 benchmark config, llama-3-70b the TP=4+ config, tiny the CPU test config.
+The engine-arg surface mirrors what the reference passes to SGLang via
+role YAML (reference examples/inference/pd-disagg-standalone.yaml:96-135:
+model, tp-size, disaggregation-mode, mem-fraction).
 """
 from __future__ import annotations
 

@@ -1,7 +1,9 @@
 """LLM serving engine — scheduler + model runner + metrics.
 
 The GPU-resident realization of one "role instance" engine process
-(SURVEY §2.3): add_request() enqueues, step() runs one prefill or decode
+(the reference externalizes this to SGLang — examples/
+inference/pd-disagg-standalone.yaml:96-135 runs sglang.launch_server per
+role): add_request() enqueues, step() runs one prefill or decode
 iteration, and the stats feed the Prometheus-style gauges the controller
 scrapes (tok/s, TTFT, pool occupancy).
 """

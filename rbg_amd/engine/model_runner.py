@@ -4,7 +4,9 @@ Owns the device, model, KV pool and (optionally) the hipGraph-captured
 decode path: decode steps at a given batch size are launch-bound (dozens of
 small kernels), so the runner captures one graph per padded batch-size
 bucket and replays it with fresh inputs copied into static buffers
-(guide: "capture launch-bound inner loops in hipGraphs").
+(guide: "capture launch-bound inner loops in hipGraphs").  This is the
+MI355X-native replacement for the model-execution half the reference
+delegates to SGLang (SURVEY §2.3 prefill/decode engine rows).
 """
 from __future__ import annotations
 

@@ -1,4 +1,10 @@
-"""Request/sequence state for continuous batching."""
+"""Request/sequence state for continuous batching.
+
+The in-engine analog of one routed request in the reference's P/D flow
+(reference examples/inference/pd-disagg-standalone.yaml router role:
+a request is prefills on one engine, decodes on another; here Sequence
+carries the KV page list that migrates between them).
+"""
 from __future__ import annotations
 
 import itertools

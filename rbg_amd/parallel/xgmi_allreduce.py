@@ -1,5 +1,9 @@
 """One-shot xGMI all-reduce — the TP decode collective, graph-capturable.
 
+Realizes the per-layer TP all-reduce the reference's leaderWorkerPattern
+exists to bootstrap (reference env_builder.go:50-74 rank env; SURVEY §2.4
+sizes the decode collective at [batch, hidden] bf16 per token).
+
 Python side of ops/hip/allreduce.hip: each rank allocates an IPC-exportable
 data buffer (hipMalloc base) + an UNCACHED signal buffer, the handles are
 exchanged once over the existing torch.distributed control group (gloo or
