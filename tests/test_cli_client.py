@@ -212,3 +212,46 @@ def test_remote_watch_generator(daemon):
     daemon.manager.store.create(make_rbg("watch-gen"))
     assert done.wait(timeout=10), got
     assert got[-1]["kind"] == "RoleBasedGroup"
+
+
+def test_update_with_retry_survives_concurrent_bumps():
+    """CLI get-modify-update retries on optimistic-concurrency Conflict
+    (a controller bumping resourceVersion between the CLI's read and
+    write — the soak-exposed race in rollout undo; the daemon RPC path
+    enforces resourceVersion strictly, unlike InProcessClient)."""
+    from rbg_amd.cli.main import _update_with_retry
+    from rbg_amd.store.store import Conflict
+
+    class FlakyClient:
+        def __init__(self):
+            self.obj = make_rbg(name="racy")
+            self.conflicts_left = 2
+            self.attempts = 0
+
+        def get(self, kind, name, namespace):
+            return self.obj
+
+        def update(self, obj):
+            self.attempts += 1
+            if self.conflicts_left:
+                self.conflicts_left -= 1
+                # remote clients surface the daemon's wrapped error text
+                raise RuntimeError(
+                    "rpc store_update failed: Conflict('rv 372 != 399')")
+            self.obj = obj
+            return obj
+
+    c = FlakyClient()
+    def mutate(cur):
+        cur.metadata.labels["rolled"] = "yes"
+    _update_with_retry(c, C.KIND_RBG, "racy", "default", mutate)
+    assert c.attempts == 3
+    assert c.obj.metadata.labels["rolled"] == "yes"
+
+    # a non-Conflict RuntimeError must NOT be swallowed
+    class BrokenClient(FlakyClient):
+        def update(self, obj):
+            raise RuntimeError("rpc store_update failed: socket closed")
+    with pytest.raises(RuntimeError, match="socket closed"):
+        _update_with_retry(BrokenClient(), C.KIND_RBG, "racy", "default",
+                           mutate)
