@@ -17,7 +17,6 @@ from __future__ import annotations
 import argparse
 import difflib
 import sys
-import time
 from typing import List, Optional
 
 import yaml
@@ -25,7 +24,8 @@ import yaml
 from ..api import constants as C
 from ..api.serde import asdict, fromdict
 from ..api.types import RoleBasedGroupSpec, load_object
-from ..client.client import BaseClient, RemoteClient
+from ..client.client import (BaseClient, RemoteClient,
+                             update_with_retry)
 
 
 def _print_table(rows: List[List[str]], header: List[str]) -> None:
@@ -35,31 +35,6 @@ def _print_table(rows: List[List[str]], header: List[str]) -> None:
     print(fmt.format(*header))
     for r in rows:
         print(fmt.format(*[str(c) for c in r]))
-
-
-def _update_with_retry(client: BaseClient, kind: str, name: str,
-                       namespace: str, mutate) -> None:
-    """Get-modify-update with optimistic-concurrency retry: controllers
-    bump resourceVersion concurrently (status, annotations), so a bare
-    update can hit a Conflict — re-read and re-apply, like kubectl."""
-    from ..store.store import Conflict
-    last = None
-    for _ in range(8):
-        cur = client.get(kind, name, namespace)
-        if cur is None:
-            raise SystemExit(f"{kind} {name} vanished during update")
-        mutate(cur)
-        try:
-            client.update(cur)
-            return
-        except Conflict as e:           # in-process client
-            last = e
-        except RuntimeError as e:       # remote client wraps the error text
-            if "Conflict" not in str(e):
-                raise
-            last = e
-        time.sleep(0.05)
-    raise RuntimeError(f"update of {kind}/{name} kept conflicting: {last}")
 
 
 def cmd_apply(client: BaseClient, args) -> int:
@@ -77,8 +52,8 @@ def cmd_apply(client: BaseClient, args) -> int:
                 cur.spec = obj.spec
                 cur.metadata.labels = obj.metadata.labels
                 cur.metadata.annotations = obj.metadata.annotations
-            _update_with_retry(client, obj.kind, obj.metadata.name,
-                               obj.metadata.namespace, configure)
+            update_with_retry(client, obj.kind, obj.metadata.name,
+                              obj.metadata.namespace, configure)
             print(f"{obj.kind}/{obj.metadata.name} configured")
     return 0
 
@@ -230,7 +205,7 @@ def cmd_rollout(client: BaseClient, args) -> int:
     if args.action == "undo":
         def roll_back(cur, data=target.data):
             cur.spec = fromdict(RoleBasedGroupSpec, data)
-        _update_with_retry(client, C.KIND_RBG, args.name, args.namespace,
+        update_with_retry(client, C.KIND_RBG, args.name, args.namespace,
                            roll_back)
         print(f"rolebasedgroup/{args.name} rolled back to revision "
               f"{target.revision}")

@@ -8,13 +8,38 @@ Objects cross the wire as the same camelCase dicts the YAML uses.
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+import time
+from typing import Any, Callable, Dict, List, Optional
 
 from ..api import constants as C
 from ..api.serde import asdict
 from ..api.types import load_object
 from ..api.validation import validate_rbg, validate_rbg_update
-from ..store.store import Store
+from ..store.store import Conflict, Store
+
+
+def update_with_retry(client: "BaseClient", kind: str, name: str,
+                      namespace: str, mutate: Callable[[Any], None]) -> Any:
+    """Get-modify-update with optimistic-concurrency retry: controllers
+    bump resourceVersion concurrently (status, labels, annotations), so a
+    bare update over the daemon RPC can hit a Conflict — re-read and
+    re-apply, like kubectl / client-go's RetryOnConflict."""
+    last = None
+    for _ in range(8):
+        cur = client.get(kind, name, namespace)
+        if cur is None:
+            raise KeyError(f"{kind} {name} vanished during update")
+        mutate(cur)
+        try:
+            return client.update(cur)
+        except Conflict as e:           # in-process transport
+            last = e
+        except RuntimeError as e:       # remote transport wraps the error text
+            if "Conflict" not in str(e):
+                raise
+            last = e
+        time.sleep(0.05)
+    raise RuntimeError(f"update of {kind}/{name} kept conflicting: {last}")
 
 
 class BaseClient:
@@ -64,12 +89,14 @@ class BaseClient:
 
     def scale(self, adapter_name: str, replicas: int,
               namespace: str = "default") -> None:
-        """The /scale subresource verb."""
-        ad = self.get(C.KIND_SCALING_ADAPTER, adapter_name, namespace)
-        if ad is None:
+        """The /scale subresource verb (conflict-retried: the adapter
+        controller bumps the object's rv via status/label writes)."""
+        if self.get(C.KIND_SCALING_ADAPTER, adapter_name, namespace) is None:
             raise KeyError(f"scaling adapter {adapter_name} not found")
-        ad.spec.replicas = replicas
-        self.update(ad)
+        def set_replicas(ad):
+            ad.spec.replicas = replicas
+        update_with_retry(self, C.KIND_SCALING_ADAPTER, adapter_name,
+                          namespace, set_replicas)
 
     def revisions(self, rbg_name: str, namespace: str = "default"):
         revs = [r for r in self.list(C.KIND_CONTROLLER_REVISION, namespace)

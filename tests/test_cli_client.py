@@ -219,7 +219,7 @@ def test_update_with_retry_survives_concurrent_bumps():
     (a controller bumping resourceVersion between the CLI's read and
     write — the soak-exposed race in rollout undo; the daemon RPC path
     enforces resourceVersion strictly, unlike InProcessClient)."""
-    from rbg_amd.cli.main import _update_with_retry
+    from rbg_amd.client.client import update_with_retry
     from rbg_amd.store.store import Conflict
 
     class FlakyClient:
@@ -244,7 +244,7 @@ def test_update_with_retry_survives_concurrent_bumps():
     c = FlakyClient()
     def mutate(cur):
         cur.metadata.labels["rolled"] = "yes"
-    _update_with_retry(c, C.KIND_RBG, "racy", "default", mutate)
+    update_with_retry(c, C.KIND_RBG, "racy", "default", mutate)
     assert c.attempts == 3
     assert c.obj.metadata.labels["rolled"] == "yes"
 
@@ -253,5 +253,5 @@ def test_update_with_retry_survives_concurrent_bumps():
         def update(self, obj):
             raise RuntimeError("rpc store_update failed: socket closed")
     with pytest.raises(RuntimeError, match="socket closed"):
-        _update_with_retry(BrokenClient(), C.KIND_RBG, "racy", "default",
+        update_with_retry(BrokenClient(), C.KIND_RBG, "racy", "default",
                            mutate)
