@@ -9,10 +9,11 @@ scrapes (tok/s, TTFT, pool occupancy).
 """
 from __future__ import annotations
 
+import collections
 import logging
 import time
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Deque, Dict, List, Optional
 
 from ..models.llama import TPContext
 from .config import EngineConfig
@@ -34,7 +35,10 @@ class EngineStats:
     idle_wall_s: float = 0.0
     sched_wall_s: float = 0.0
     finished: int = 0
-    ttfts: List[float] = field(default_factory=list)
+    # recent-window TTFTs: bounded so a long-running server neither leaks
+    # nor pays an ever-growing sort on every stats scrape
+    ttfts: Deque[float] = field(
+        default_factory=lambda: collections.deque(maxlen=65536))
     started: float = field(default_factory=time.monotonic)
 
     def snapshot(self) -> Dict[str, float]:
