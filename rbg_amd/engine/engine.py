@@ -67,7 +67,6 @@ class LLMEngine:
         self.runner = ModelRunner(cfg, tp, pp)
         self.scheduler = Scheduler(cfg, self.runner.cache)
         self.stats = EngineStats()
-        self._finished: List[Sequence] = []
 
     # ------------------------------------------------------------------
 
@@ -115,13 +114,7 @@ class LLMEngine:
             self.stats.decode_wall_s += time.monotonic() - t0
         else:
             self.stats.idle_wall_s += time.monotonic() - t0
-        self._collect_finished()
         return mode
-
-    def _collect_finished(self) -> None:
-        # sequences leave scheduler.running when finished; callers keep their
-        # own Sequence handles (status/output live on the object)
-        pass
 
     def generate(self, prompts: List[List[int]],
                  sampling: Optional[SamplingParams] = None,
