@@ -644,7 +644,7 @@ class ServeWorker:
                                      temperature)
             self._pending_imports[seq.seq_id] = (
                 seq, int(first_token), int(max_new_tokens),
-                float(arrival_time))
+                float(arrival_time), time.monotonic())
             return {"seq_id": seq.seq_id,
                     "peer": kv_peer.export_meta(self.engine.runner.cache),
                     "dst_pages": list(seq.block_table.pages)}
@@ -686,10 +686,33 @@ class ServeWorker:
         """Sender finished its xGMI push (copy event resolved on its side):
         enqueue the sequence.  The pages were written remotely; no local
         copy or CU work happened here."""
-        seq, first_token, max_new_tokens, arrival_time = \
+        seq, first_token, max_new_tokens, arrival_time, _t0 = \
             self._pending_imports.pop(int(seq_id))
         self._import_enqueue(seq, first_token, max_new_tokens, arrival_time)
         return {"seq_id": seq.seq_id}
+
+    # a push over xGMI completes in milliseconds and the sender's own
+    # wait() gives up after ~10 s, so a pending import with no commit after
+    # IMPORT_TTL_S means the prefill died mid-migration: reclaim the pages
+    # (a commit arriving later hits a KeyError -> the sender fails the
+    # migration and the router re-dispatches, so nothing corrupt enqueues)
+    IMPORT_TTL_S = 30.0
+
+    def _sweep_pending_imports(self) -> None:
+        now = time.monotonic()
+        for sid in list(self._pending_imports):
+            entry = self._pending_imports.get(sid)
+            if entry is None or now - entry[4] < self.IMPORT_TTL_S:
+                continue
+            seq = entry[0]
+            self._pending_imports.pop(sid, None)
+            if seq.block_table is not None and seq.block_table.pages:
+                self.engine.runner.cache.free(seq.block_table.pages)
+                seq.block_table.pages = []
+            self.results.pop(sid, None)
+            log.warning("pending KV import %d expired after %.0fs "
+                        "(prefill died mid-migration?); pages reclaimed",
+                        sid, self.IMPORT_TTL_S)
 
     def _rpc_resolve_ticket(self, ticket: str) -> Dict[str, Any]:
         sid = self._tickets.get(ticket)
@@ -730,6 +753,7 @@ class ServeWorker:
                     continue
                 with self.lock:
                     mode = self.engine.step()
+                self._sweep_pending_imports()
                 if mode == "idle":
                     time.sleep(0.002)
         finally:

@@ -170,3 +170,38 @@ def test_pending_push_timeout_raises_peerdead():
         p.wait(timeout_s=0.2)
     assert hit == [1]
     assert _time.monotonic() - t0 < 1.5
+
+
+def test_pending_import_ttl_sweep_reclaims_pages():
+    """Decode-side leak guard: an import whose sender died before
+    import_commit is expired after IMPORT_TTL_S and its KV pages return
+    to the pool (otherwise repeated prefill crashes mid-migration drain
+    the decode pool)."""
+    import time as _time
+    from types import SimpleNamespace as NS
+    from rbg_amd.engine.serve_worker import ServeWorker
+
+    class FakeCache:
+        def __init__(self):
+            self.freed = []
+
+        def free(self, pages):
+            self.freed.extend(pages)
+
+    w = object.__new__(ServeWorker)       # unit: no full worker bring-up
+    cache = FakeCache()
+    w.engine = NS(runner=NS(cache=cache))
+    w.results = {7: "seq"}
+    stale = NS(block_table=NS(pages=[1, 2, 3]))
+    fresh = NS(block_table=NS(pages=[9]))
+    w._pending_imports = {
+        7: (stale, 0, 8, 0.0, _time.monotonic() - ServeWorker.IMPORT_TTL_S - 1),
+        8: (fresh, 0, 8, 0.0, _time.monotonic()),
+    }
+    w._sweep_pending_imports()
+    assert 7 not in w._pending_imports          # expired entry reclaimed
+    assert cache.freed == [1, 2, 3]
+    assert stale.block_table.pages == []
+    assert 7 not in w.results
+    assert 8 in w._pending_imports              # fresh entry untouched
+    assert fresh.block_table.pages == [9]
