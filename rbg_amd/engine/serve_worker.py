@@ -172,6 +172,9 @@ class ServeWorker:
         # the non-TP P/D pair; negotiated per sequence in import_seq.
         self._peer_pusher = None
         self._pending_imports: Dict[int, tuple] = {}
+        # finished-result retention: when each sid was first seen FINISHED
+        self._finished_at: Dict[int, float] = {}
+        self._last_sweep = 0.0
         # migration gauges (the xGMI-bandwidth metric of SURVEY §5)
         self._xfer_stats = {"pushes": 0, "bytes": 0, "seconds": 0.0}
         self._register_rpc()
@@ -373,6 +376,7 @@ class ServeWorker:
                     for op in wire:
                         self._tp_apply_op(op)
                 mode = self.engine.step()
+                self._sweep()
                 if mode == "idle" and not wire:
                     time.sleep(0.002)
         finally:
@@ -698,6 +702,26 @@ class ServeWorker:
     # migration and the router re-dispatches, so nothing corrupt enqueues)
     IMPORT_TTL_S = 30.0
 
+    # finished sequences stay readable for RESULT_TTL_S after completion
+    # (a poller collects within one tick — milliseconds — so this is a
+    # ~10^5x grace), then drop: retaining every Sequence forever leaks
+    # ~70 KB of token lists per request in a long-running server
+    RESULT_TTL_S = 120.0
+
+    def _sweep(self) -> None:
+        now = time.monotonic()
+        if now - self._last_sweep < 1.0:
+            return
+        self._last_sweep = now
+        self._sweep_pending_imports()
+        for sid, seq in list(self.results.items()):
+            if getattr(seq, "status", None) != FINISHED:
+                continue
+            t0 = self._finished_at.setdefault(sid, now)
+            if now - t0 > self.RESULT_TTL_S:
+                self.results.pop(sid, None)
+                self._finished_at.pop(sid, None)
+
     def _sweep_pending_imports(self) -> None:
         now = time.monotonic()
         for sid in list(self._pending_imports):
@@ -753,7 +777,7 @@ class ServeWorker:
                     continue
                 with self.lock:
                     mode = self.engine.step()
-                self._sweep_pending_imports()
+                self._sweep()
                 if mode == "idle":
                     time.sleep(0.002)
         finally:

@@ -205,3 +205,31 @@ def test_pending_import_ttl_sweep_reclaims_pages():
     assert 7 not in w.results
     assert 8 in w._pending_imports              # fresh entry untouched
     assert fresh.block_table.pages == [9]
+
+
+def test_finished_result_ttl_sweep():
+    """Finished sequences are dropped RESULT_TTL_S after completion;
+    running ones and recently-finished ones are retained (pollers read
+    within one tick)."""
+    import time as _time
+    from types import SimpleNamespace as NS
+    from rbg_amd.engine.serve_worker import ServeWorker
+    from rbg_amd.engine.sequence import FINISHED
+
+    w = object.__new__(ServeWorker)
+    w.engine = NS(runner=NS(cache=NS(free=lambda p: None)))
+    w._pending_imports = {}
+    w._finished_at = {}
+    w._last_sweep = 0.0
+    old = NS(status=FINISHED, block_table=None)
+    recent = NS(status=FINISHED, block_table=None)
+    running = NS(status="running", block_table=None)
+    w.results = {1: old, 2: recent, 3: running}
+    w._finished_at[1] = _time.monotonic() - ServeWorker.RESULT_TTL_S - 1
+    w._sweep()
+    assert 1 not in w.results
+    assert 2 in w.results and 3 in w.results
+    # a second sweep within the rate gate is a no-op (cheap in the loop)
+    w._finished_at[2] = _time.monotonic() - ServeWorker.RESULT_TTL_S - 1
+    w._sweep()
+    assert 2 in w.results
