@@ -46,6 +46,18 @@ class _InstanceWatcher:
         self._waiting: Dict[int, threading.Event] = {}
         self._results: Dict[int, Dict[str, Any]] = {}
         self._thread: threading.Thread = None
+        self._stopped = False
+
+    def stop(self) -> None:
+        """Retire the watcher (its instance left the topology): fail any
+        still-waiting requests so they re-dispatch, stop the poll loop,
+        release the socket."""
+        self._stopped = True
+        self._fail_all("instance left the topology")
+        try:
+            self.client.close()
+        except Exception:  # noqa: BLE001
+            pass
 
     FAIL_WINDOW_S = 10.0    # continuous RPC failure -> declare seqs lost
 
@@ -77,7 +89,7 @@ class _InstanceWatcher:
 
     def _loop(self):
         fail_since = None
-        while True:
+        while not self._stopped:
             with self._lock:
                 ids = list(self._waiting.keys())
             if not ids:
@@ -131,6 +143,44 @@ class Router:
         self._watchers: Dict[str, _InstanceWatcher] = {}
         self._rr = itertools.count()
         self._lock = threading.Lock()
+        # stale client/watcher GC: a recreated instance publishes new ports,
+        # so (name, ports) keys accumulate across gang recreates — retire
+        # entries absent from the topology for STALE_GRACE_S (grace covers
+        # transient unready flaps during rolling updates)
+        self._stale_since: Dict[Any, float] = {}
+        self._last_gc = 0.0
+
+    STALE_GRACE_S = 30.0
+
+    def _gc_stale(self) -> None:
+        now = time.monotonic()
+        if now - self._last_gc < 5.0:
+            return
+        self._last_gc = now
+        live = set()
+        topo = self.ctx.load_topology().get("group", {})
+        for role in topo.get("roles", []):
+            for inst in role.get("instances", []):
+                if inst.get("ports"):
+                    live.add((inst["name"], tuple(inst["ports"])))
+        with self._lock:
+            for key in set(self._clients) | set(self._watchers):
+                if key in live:
+                    self._stale_since.pop(key, None)
+                    continue
+                t0 = self._stale_since.setdefault(key, now)
+                if now - t0 < self.STALE_GRACE_S:
+                    continue
+                self._stale_since.pop(key, None)
+                c = self._clients.pop(key, None)
+                if c is not None:
+                    try:
+                        c.close()
+                    except Exception:  # noqa: BLE001
+                        pass
+                w = self._watchers.pop(key, None)
+                if w is not None:
+                    w.stop()
 
     # -- topology ------------------------------------------------------------
 
@@ -191,6 +241,7 @@ class Router:
         naturally lands on a surviving replica while the gang recreates
         the dead one."""
         t0 = time.monotonic()
+        self._gc_stale()
         window = float(self.ctx.args.get("failover_window_s", 45.0))
         deadline = t0 + window
         last: Exception = None

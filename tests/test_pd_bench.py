@@ -233,3 +233,46 @@ def test_finished_result_ttl_sweep():
     w._finished_at[2] = _time.monotonic() - ServeWorker.RESULT_TTL_S - 1
     w._sweep()
     assert 2 in w.results
+
+
+def test_router_gc_retires_stale_watchers_and_clients():
+    """A recreated instance publishes new ports; the router must retire
+    the old (name, ports) client + watcher after the grace period and
+    fail their waiters so requests re-dispatch."""
+    import time as _time
+    from types import SimpleNamespace as NS
+    from rbg_amd.server.router_worker import Router, _InstanceWatcher
+
+    topo = {"group": {"roles": [{"name": "decode", "instances": [
+        {"name": "d-0", "ready": True, "ports": [2001]}]}]}}
+    ctx = NS(args={}, load_topology=lambda: topo)
+    r = Router(ctx)
+
+    class FakeClient:
+        def __init__(self):
+            self.closed = False
+
+        def close(self):
+            self.closed = True
+
+    live_c, stale_c = FakeClient(), FakeClient()
+    stale_w = _InstanceWatcher(FakeClient())
+    r._clients = {("d-0", (2001,)): live_c, ("d-0", (1999,)): stale_c}
+    r._watchers = {("d-0", (1999,)): stale_w}
+
+    r._gc_stale()                       # first sighting: starts the clock
+    assert ("d-0", (1999,)) in r._clients
+    # age the stale entry past the grace and re-run (bypass the rate gate)
+    r._stale_since[("d-0", (1999,))] -= Router.STALE_GRACE_S + 1
+    r._last_gc = 0.0
+    r._gc_stale()
+    assert ("d-0", (1999,)) not in r._clients and stale_c.closed
+    assert ("d-0", (1999,)) not in r._watchers and stale_w._stopped
+    assert ("d-0", (2001,)) in r._clients and not live_c.closed
+    # a retired watcher fails (not hangs) any later waiter
+    import pytest as _pytest
+    from rbg_amd.server.router_worker import InstanceLost
+    ev_res = stale_w._results
+    stale_w._waiting[5] = __import__("threading").Event()
+    stale_w._fail_all("instance left the topology")
+    assert ev_res[5]["lost"]
