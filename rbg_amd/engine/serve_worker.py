@@ -329,6 +329,10 @@ class ServeWorker:
                                 float(op.get("arrival_time", 0.0)),
                                 self._peer_rank(op["src_instance"]))
             self._tickets[op["ticket"]] = seq.seq_id
+            # bounded FIFO (dicts are insertion-ordered): tickets resolve
+            # within milliseconds; keep a deep tail for RPC retries only
+            while len(self._tickets) > 4096:
+                self._tickets.pop(next(iter(self._tickets)))
         ev = op.get("_ev")
         if ev is not None:
             ev.set()
