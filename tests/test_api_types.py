@@ -138,3 +138,38 @@ def test_scaling_adapter_replica_immutability():
     with pytest.raises(ValidationError) as e:
         validate_rbg_update(old, new)
     assert "scalingAdapter" in str(e.value)
+
+
+def test_role_template_ref_merge():
+    """roleTemplates + templateRef patch (reference rolebasedgroup_types.go
+    RoleTemplates/TemplateRef strategic merge): the role's expanded
+    template is the named base with per-engine arg/runner overrides, and
+    the base itself is never mutated."""
+    from rbg_amd.api.types import (EngineSpec, EngineTemplate,
+                                   RoleBasedGroup, RoleBasedGroupSpec,
+                                   RoleSpec, TemplateRef)
+    from rbg_amd.controller.rbg_controller import expand_pattern
+
+    base = EngineTemplate(engines=[
+        EngineSpec(name="engine", runner="llm-engine",
+                   args={"model": "llama-3-8b", "mode": "colocated"})])
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="t"),
+        spec=RoleBasedGroupSpec(
+            role_templates={"llm": base},
+            roles=[RoleSpec(
+                name="prefill", replicas=1,
+                template_ref=TemplateRef(
+                    name="llm",
+                    patch={"engines": [{"name": "engine",
+                                        "args": {"mode": "prefill"}}]}))]))
+    comps = expand_pattern(rbg, rbg.spec.roles[0])
+    eng = comps[0].template.engines[0]
+    assert eng.args == {"model": "llama-3-8b", "mode": "prefill"}
+    assert eng.runner == "llm-engine"
+    # base untouched (merge clones)
+    assert base.engines[0].args["mode"] == "colocated"
+    # unknown ref name falls back to the role's own template (None here)
+    rbg.spec.roles[0].template_ref = TemplateRef(name="nope")
+    comps = expand_pattern(rbg, rbg.spec.roles[0])
+    assert comps[0].template is None
