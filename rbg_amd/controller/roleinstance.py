@@ -117,8 +117,19 @@ class RoleInstanceController:
             if h is None:
                 continue
             phase = h.phase()
-            if phase in ("Failed",) and not self._restart_ignored(comp):
-                failed_fatal.append(wname)
+            if phase in ("Failed",):
+                if self._restart_ignored(comp):
+                    # auxiliary component (RestartTriggerPolicy=Ignore):
+                    # respawn IT alone — reap the handle so the missing-
+                    # worker path below recreates it; never condemn the
+                    # gang (reference annotation.go:150-176)
+                    self.runner.stop(h)
+                    rt.handles.pop(wname, None)
+                    self.recorder.normal(
+                        inst, "AuxComponentRestart",
+                        f"{wname} failed; restarting alone (Ignore policy)")
+                else:
+                    failed_fatal.append(wname)
             elif phase == "Ready":
                 rt.gpu_by_worker[wname] = h.gpu_ids
         healthy = not failed_fatal and all(

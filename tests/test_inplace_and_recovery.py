@@ -256,3 +256,59 @@ def test_required_sticky_gpus_across_gang_recreate(tmp_run_dir):
         assert second == first, (first, second)
     finally:
         m.stop()
+
+
+def test_ignore_policy_component_restarts_alone(mgr):
+    """A component annotated restart-trigger-policy=Ignore that crashes is
+    respawned by itself: the main component's process survives (no gang
+    recreate), the instance returns Ready, and restart_count stays 0
+    (reference annotation.go:150-176 RestartTriggerPolicy semantics)."""
+    from rbg_amd.api.types import (ComponentSpec, CustomComponentsPattern,
+                                   EngineResources, EngineSpec,
+                                   EngineTemplate, ObjectMeta,
+                                   RoleBasedGroup, RoleBasedGroupSpec,
+                                   RoleSpec)
+
+    def tmpl(args=None):
+        return EngineTemplate(engines=[EngineSpec(
+            name="engine", runner="echo", args=args or {},
+            resources=EngineResources(gpus=0, cpu_only=True))])
+
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="aux"),
+        spec=RoleBasedGroupSpec(roles=[RoleSpec(
+            name="main", replicas=1, pattern=C.PATTERN_CUSTOM_COMPONENTS,
+            custom_components_pattern=CustomComponentsPattern(components=[
+                ComponentSpec(name="engine", size=1, template=tmpl()),
+                ComponentSpec(
+                    name="sidecar", size=1,
+                    template=tmpl(args={"crash_after": 1.0}),
+                    annotations={C.ANNO_RESTART_TRIGGER_POLICY: "Ignore"}),
+            ]))]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "aux"), timeout=30)
+
+    insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                           selector={C.LABEL_GROUP_NAME: "aux"})
+    assert len(insts) == 1
+    main_pid = next(w.pid for w in insts[0].status.workers
+                    if w.component == "engine")
+
+    # the sidecar crashes after ~1s; wait for it to be respawned (new pid)
+    def sidecar_respawned():
+        cur = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                             selector={C.LABEL_GROUP_NAME: "aux"})
+        if not cur:
+            return False
+        sc = [w for w in cur[0].status.workers if w.component == "sidecar"]
+        return bool(sc) and sc[0].phase == "Ready" and sc[0].restart_count >= 0 \
+            and cur[0].status.restart_count == 0
+    time.sleep(2.0)        # let the crash happen
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "aux"), timeout=30)
+    cur = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                         selector={C.LABEL_GROUP_NAME: "aux"})[0]
+    # gang NOT condemned: main engine kept its process
+    main_now = next(w.pid for w in cur.status.workers
+                    if w.component == "engine")
+    assert main_now == main_pid, "gang was recreated despite Ignore policy"
+    assert cur.status.restart_count == 0
