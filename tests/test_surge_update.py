@@ -172,3 +172,65 @@ def test_stateless_surge_keeps_capacity():
     assert len(live) == 3
     assert all(i.metadata.labels.get(C.LABEL_REVISION_HASH) == new_hash
                for i in live)
+
+
+def test_ordered_ready_gates_each_ordinal():
+    """podManagementPolicy=OrderedReady creates ordinal i+1 only after
+    ordinal i is Ready (reference stateful OrderedReady semantics);
+    Parallel (the default elsewhere in this file) creates all at once."""
+    store = Store()
+    ctrl = RoleInstanceSetController(store)
+    ris = make_ris(replicas=3, max_surge=0, max_unavailable=1)
+    ris.spec.pod_management_policy = C.POD_MANAGEMENT_ORDERED_READY
+    ris = store.create(ris)
+
+    ctrl.reconcile("web")
+    names = sorted(i.metadata.name for i in live_instances(store, ris))
+    assert names == ["web-0"], names          # gated on web-0 readiness
+
+    ctrl.reconcile("web")                     # still not ready -> no web-1
+    assert sorted(i.metadata.name
+                  for i in live_instances(store, ris)) == ["web-0"]
+
+    set_ready(store, "web-0")
+    ctrl.reconcile("web")
+    assert sorted(i.metadata.name
+                  for i in live_instances(store, ris)) == ["web-0", "web-1"]
+    set_ready(store, "web-1")
+    ctrl.reconcile("web")
+    assert sorted(i.metadata.name for i in live_instances(store, ris)) == \
+        ["web-0", "web-1", "web-2"]
+
+
+@pytest.mark.parametrize("pattern", ["Stateful", "Stateless"])
+def test_paused_update_holds_until_unpaused(pattern):
+    """updateStrategy.paused freezes the rollout (old revision stays) and
+    clearing it resumes to completion."""
+    store = Store()
+    ctrl = RoleInstanceSetController(store)
+    ris = store.create(make_ris(pattern=pattern, max_unavailable=1,
+                                max_surge=0))
+    converge(store, ctrl, "web")
+    old_hash = template_hash(store.get(C.KIND_ROLE_INSTANCE_SET, "web"))
+
+    def bump_paused(cur):
+        cur.spec.template.components[0].template.engines[0].args["rev"] = "v2"
+        cur.spec.update_strategy.paused = True
+        return cur
+    ris = store.apply(C.KIND_ROLE_INSTANCE_SET, "web", bump_paused)
+    converge(store, ctrl, "web", rounds=10)
+    live = live_instances(store, ris)
+    assert len(live) == 3
+    assert all(i.metadata.labels.get(C.LABEL_REVISION_HASH) == old_hash
+               for i in live), "paused rollout must not replace instances"
+
+    def unpause(cur):
+        cur.spec.update_strategy.paused = False
+        return cur
+    ris = store.apply(C.KIND_ROLE_INSTANCE_SET, "web", unpause)
+    new_hash = template_hash(ris)
+    converge(store, ctrl, "web", rounds=60)
+    live = live_instances(store, ris)
+    assert len(live) == 3
+    assert all(i.metadata.labels.get(C.LABEL_REVISION_HASH) == new_hash
+               for i in live)
