@@ -117,10 +117,38 @@ class RoleInstanceSetController:
                 inst.metadata.owner_references.append(ref)
         return inst
 
-    def _delete_instance(self, inst: RoleInstance) -> None:
+    ANNO_DELETE_AFTER = f"{C.PREFIX}/delete-after"
+
+    def _delete_instance(self, inst: RoleInstance, grace: float = 0.0) -> None:
+        """Mark an instance deleted.  With ``grace`` > 0 (updateStrategy
+        gracePeriodSeconds / v1alpha1 preDelete markPodNotReady): first
+        flip Ready=False and stamp a delete-after time so traffic drains,
+        then set deletion_timestamp once the grace elapses (callers
+        re-invoke every reconcile until then)."""
+        import time as _t
+        now = _t.time()
+        if grace > 0 and inst.metadata.deletion_timestamp is None:
+            after = inst.metadata.annotations.get(self.ANNO_DELETE_AFTER)
+            if after is None:
+                def drain(cur: RoleInstance):
+                    cur.metadata.annotations[self.ANNO_DELETE_AFTER] = \
+                        str(now + grace)
+                    set_condition(cur.status.conditions, Condition.new(
+                        C.COND_READY, False, "Draining",
+                        f"pre-delete grace {grace:.0f}s"))
+                    return cur
+                try:
+                    self.store.apply(C.KIND_ROLE_INSTANCE,
+                                     inst.metadata.name, drain,
+                                     inst.metadata.namespace)
+                except KeyError:
+                    pass
+                return
+            if now < float(after):
+                return                      # still draining
+
         def mark(cur: RoleInstance):
-            import time as _t
-            cur.metadata.deletion_timestamp = _t.time()
+            cur.metadata.deletion_timestamp = now
             return cur
         try:
             self.store.apply(C.KIND_ROLE_INSTANCE, inst.metadata.name, mark,
@@ -251,7 +279,8 @@ class RoleInstanceSetController:
             if self._can_update_in_place(ris, inst):
                 self._in_place_update(ris, inst, update_hash)
             else:
-                self._delete_instance(inst)
+                self._delete_instance(
+                    inst, grace=strat.grace_period_seconds)
         return 0.2
 
     def _make_instance_named(self, ris: RoleInstanceSet, name: str,
@@ -333,7 +362,8 @@ class RoleInstanceSetController:
                 self._in_place_update(ris, inst, update_hash)
             else:
                 # recreate: delete; _scale recreates at the new revision
-                self._delete_instance(inst)
+                self._delete_instance(
+                    inst, grace=strat.grace_period_seconds)
             budget -= 1
         return 0.2
 

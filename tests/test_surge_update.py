@@ -234,3 +234,63 @@ def test_paused_update_holds_until_unpaused(pattern):
     assert len(live) == 3
     assert all(i.metadata.labels.get(C.LABEL_REVISION_HASH) == new_hash
                for i in live)
+
+
+@pytest.mark.parametrize("pattern", ["Stateful", "Stateless"])
+def test_grace_period_drains_before_recreate(pattern):
+    """updateStrategy.gracePeriodSeconds: a to-be-recreated instance is
+    first flipped NotReady (drain window, the v1alpha1 preDelete
+    markPodNotReady semantics) and only deleted after the grace elapses."""
+    import time as _time
+    store = Store()
+    ctrl = RoleInstanceSetController(store)
+    ris = make_ris(pattern=pattern, replicas=2, max_unavailable=1,
+                   max_surge=0)
+    ris.spec.update_strategy.grace_period_seconds = 1
+    ris = store.create(ris)
+    converge(store, ctrl, "web")
+    assert len(live_instances(store, ris)) == 2
+
+    def bump(cur):
+        cur.spec.template.components[0].template.engines[0].args["rev"] = "v2"
+        return cur
+    ris = store.apply(C.KIND_ROLE_INSTANCE_SET, "web", bump)
+    new_hash = template_hash(ris)
+
+    # first reconciles: draining, nothing deleted yet
+    ctrl.reconcile("web")
+    ctrl.reconcile("web")
+    insts = store.list_owned(C.KIND_ROLE_INSTANCE, ris.metadata.uid)
+    draining = [i for i in insts if
+                RoleInstanceSetController.ANNO_DELETE_AFTER
+                in i.metadata.annotations]
+    assert draining, "grace must stamp a delete-after drain marker"
+    assert all(i.metadata.deletion_timestamp is None for i in insts)
+    assert any(c.type == C.COND_READY and c.status == "False" and
+               c.reason == "Draining"
+               for i in draining for c in i.status.conditions)
+
+    # within the grace window the instance survives reconciles
+    ctrl.reconcile("web")
+    assert all(i.metadata.deletion_timestamp is None
+               for i in store.list_owned(C.KIND_ROLE_INSTANCE,
+                                         ris.metadata.uid))
+
+    _time.sleep(1.1)
+    for _ in range(80):
+        ctrl.reconcile("web")
+        for inst in store.list_owned(C.KIND_ROLE_INSTANCE, ris.metadata.uid):
+            if inst.metadata.deletion_timestamp is not None:
+                store.try_delete(C.KIND_ROLE_INSTANCE, inst.metadata.name)
+            else:
+                set_ready(store, inst.metadata.name)
+        live = live_instances(store, ris)
+        if len(live) == 2 and all(
+                i.metadata.labels.get(C.LABEL_REVISION_HASH) == new_hash
+                for i in live):
+            break
+        _time.sleep(0.05)    # each remaining stale instance drains 1s
+    live = live_instances(store, ris)
+    assert len(live) == 2
+    assert all(i.metadata.labels.get(C.LABEL_REVISION_HASH) == new_hash
+               for i in live)
