@@ -175,6 +175,7 @@ class RoleBasedGroupController:
         if rbg.metadata.deletion_timestamp is not None:
             self._teardown(rbg)
             return 0.0
+        self._ensure_scaling_adapters(rbg)
         rbg = self._apply_scaling_adapter_override(rbg)
         try:
             validate_rbg(rbg)
@@ -246,6 +247,37 @@ class RoleBasedGroupController:
             spec=CoordinatedPolicySpec(rules=rules))
         set_owner(policy, rbg)
         self.store.create(policy)
+
+    def _ensure_scaling_adapters(self, rbg: RoleBasedGroup) -> None:
+        """Auto-provision one RoleBasedGroupScalingAdapter per role with
+        `scalingAdapter.enable: true`, named `{rbg}-{role}` and owned by
+        the group; delete owned adapters whose role disabled or vanished
+        (reference rolebasedgroup_controller.go scale-adapter management)."""
+        from ..api.types import (RoleBasedGroupScalingAdapter,
+                                 ScaleTargetRef, ScalingAdapterSpecFull)
+        want = {f"{rbg.metadata.name}-{r.name}": r.name
+                for r in rbg.spec.roles
+                if r.scaling_adapter is not None and r.scaling_adapter.enable}
+        for name, role_name in want.items():
+            if self.store.try_get(C.KIND_SCALING_ADAPTER, name,
+                                  rbg.metadata.namespace) is not None:
+                continue
+            ad = RoleBasedGroupScalingAdapter(
+                metadata=ObjectMeta(name=name,
+                                    namespace=rbg.metadata.namespace),
+                spec=ScalingAdapterSpecFull(scale_target_ref=ScaleTargetRef(
+                    name=rbg.metadata.name, role=role_name)))
+            set_owner(ad, rbg)
+            self.store.create(ad)
+            self.recorder.normal(rbg, "ScalingAdapterCreated",
+                                 f"auto-provisioned adapter {name}")
+        for ad in self.store.list_owned(C.KIND_SCALING_ADAPTER,
+                                        rbg.metadata.uid,
+                                        rbg.metadata.namespace):
+            if ad.metadata.name not in want:
+                self.store.try_delete(C.KIND_SCALING_ADAPTER,
+                                      ad.metadata.name,
+                                      ad.metadata.namespace)
 
     def _apply_scaling_adapter_override(self, rbg: RoleBasedGroup) -> RoleBasedGroup:
         """Adapter-driven replicas win over spec (reference
@@ -593,6 +625,11 @@ class RoleBasedGroupController:
             for rev in self.revisions.list_for(rbg):
                 self.store.try_delete(C.KIND_CONTROLLER_REVISION,
                                       rev.metadata.name, rbg.metadata.namespace)
+            for ad in self.store.list_owned(C.KIND_SCALING_ADAPTER,
+                                            rbg.metadata.uid,
+                                            rbg.metadata.namespace):
+                self.store.try_delete(C.KIND_SCALING_ADAPTER,
+                                      ad.metadata.name, ad.metadata.namespace)
             self.store.try_delete(C.KIND_RBG, rbg.metadata.name,
                                   rbg.metadata.namespace)
 

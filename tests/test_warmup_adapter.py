@@ -182,3 +182,39 @@ def test_scaling_adapter_drives_role_replicas(mgr):
         cur = mgr.store.get(C.KIND_SCALING_ADAPTER, "auto-worker")
         return cur.status.replicas == 3 and cur.status.last_scale_time > 0
     assert mgr.wait_for(status_converged, timeout=30)
+
+
+def test_scaling_adapter_auto_provisioned_from_role_spec(mgr):
+    """roles[].scalingAdapter.enable=true auto-creates an owned
+    `{rbg}-{role}` adapter that binds and scales the role; disabling it
+    deletes the adapter (reference RBG-controller adapter management)."""
+    from rbg_amd.api.types import ScalingAdapterSpec
+    rbg = router_worker_rbg(name="autoprov", worker_replicas=1)
+    rbg.spec.role("worker").scaling_adapter = ScalingAdapterSpec(enable=True)
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "autoprov"), timeout=60)
+
+    def adapter_bound():
+        ad = mgr.store.try_get(C.KIND_SCALING_ADAPTER, "autoprov-worker")
+        return ad is not None and ad.status.phase == C.SCALING_ADAPTER_BOUND \
+            and ad.spec.scale_target_ref.role == "worker"
+    assert mgr.wait_for(adapter_bound, timeout=30)
+
+    # driving the auto-provisioned adapter scales the role
+    scale_adapter(mgr.store, "autoprov-worker", 2)
+
+    def scaled():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                               selector={C.LABEL_GROUP_NAME: "autoprov",
+                                         C.LABEL_ROLE_NAME: "worker"})
+        return len(insts) == 2
+    assert mgr.wait_for(scaled, timeout=60)
+
+    # disabling removes the owned adapter
+    def disable(cur):
+        cur.spec.role("worker").scaling_adapter = None
+        return cur
+    mgr.store.apply(C.KIND_RBG, "autoprov", disable)
+    assert mgr.wait_for(
+        lambda: mgr.store.try_get(C.KIND_SCALING_ADAPTER,
+                                  "autoprov-worker") is None, timeout=30)
