@@ -97,8 +97,14 @@ def echo_runner(ctx: WorkerContext) -> None:
     ready_delay = float(ctx.args.get("ready_delay", 0) or 0)
     if ready_delay:
         ctx.wait(ready_delay)
+    extra = {}
+    if os.environ.get("PORT_HTTP"):
+        # surface the allocator-injected port like a real engine would
+        # (WorkerStatus.ports picks up http_port/rpc_port)
+        extra["http_port"] = int(os.environ["PORT_HTTP"].split(",")[0])
     ctx.set_ready(topology_roles=[
-        r.get("name") for r in ctx.load_topology().get("group", {}).get("roles", [])])
+        r.get("name") for r in ctx.load_topology().get("group", {}).get("roles", [])],
+        **extra)
     start = time.time()
     while not ctx.should_stop():
         if crash_after and time.time() - start > crash_after:

@@ -203,3 +203,47 @@ def test_teardown_stops_in_reverse_dependency_order():
     ctrl.teardown(inst)
     assert stopped == ["g-r-0-frontend-0", "g-r-0-engine-0",
                        "g-r-0-cache-0"]
+
+
+@pytest.mark.parametrize("scope,same", [("RoleScoped", True),
+                                        ("PodScoped", False)])
+def test_port_allocation_annotation_scopes(mgr, scope, same):
+    """The port-allocation annotation contract (reference
+    pkg/port-allocator parser.go): RoleScoped allocates ONE port shared by
+    every replica of the role; PodScoped gives each engine its own.  The
+    echo engine surfaces PORT_HTTP in its worker status."""
+    import json as _json
+    from rbg_amd.api.types import (ComponentSpec, CustomComponentsPattern,
+                                   EngineResources, EngineSpec,
+                                   EngineTemplate, ObjectMeta,
+                                   RoleBasedGroup, RoleBasedGroupSpec,
+                                   RoleSpec)
+    name = f"ports-{scope.lower()}"
+    tmpl = EngineTemplate(engines=[EngineSpec(
+        name="engine", runner="echo",
+        resources=EngineResources(gpus=0, cpu_only=True))])
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name=name),
+        spec=RoleBasedGroupSpec(roles=[RoleSpec(
+            name="web", replicas=2, pattern=C.PATTERN_CUSTOM_COMPONENTS,
+            custom_components_pattern=CustomComponentsPattern(components=[
+                ComponentSpec(
+                    name="engine", size=1, template=tmpl,
+                    annotations={C.ANNO_PORT_ALLOCATION: _json.dumps(
+                        [{"name": "http", "scope": scope, "count": 1}])}),
+            ]))]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, name), timeout=60)
+
+    def collected():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                               selector={C.LABEL_GROUP_NAME: name})
+        ports = [p for i in insts for w in i.status.workers for p in w.ports]
+        return ports if len(ports) == 2 else None
+    assert mgr.wait_for(lambda: collected() is not None, timeout=30)
+    ports = collected()
+    assert all(30000 <= p < 40000 for p in ports), ports
+    if same:
+        assert ports[0] == ports[1], f"RoleScoped must share: {ports}"
+    else:
+        assert ports[0] != ports[1], f"PodScoped must differ: {ports}"
