@@ -36,6 +36,19 @@ class RoleBasedGroupSetController:
             return 0.2
         want = rbgset.spec.replicas
         owned = {r.metadata.name: r for r in self._owned(rbgset)}
+
+        def normalize(spec_dict):
+            # adapter-owned roles: the member's ScalingAdapter drives
+            # `replicas`, so template propagation must not revert it
+            # (otherwise the set and the adapter fight every reconcile)
+            import copy as _copy
+            d = _copy.deepcopy(spec_dict)
+            for role in d.get("roles") or []:
+                sa = role.get("scalingAdapter")
+                if sa and sa.get("enable"):
+                    role["replicas"] = None
+            return d
+
         for i in range(want):
             mname = member_name(name, i)
             cur = owned.get(mname)
@@ -49,9 +62,15 @@ class RoleBasedGroupSetController:
                     spec=fromdict(RoleBasedGroupSpec, desired_spec))
                 set_owner(rbg, rbgset)
                 self.store.create(rbg)
-            elif asdict(cur.spec) != desired_spec:
+            elif normalize(asdict(cur.spec)) != normalize(desired_spec):
                 def mutate(obj, spec=desired_spec):
+                    keep = {r.name: r.replicas for r in obj.spec.roles
+                            if r.scaling_adapter is not None and
+                            r.scaling_adapter.enable}
                     obj.spec = fromdict(RoleBasedGroupSpec, spec)
+                    for r in obj.spec.roles:
+                        if r.name in keep:
+                            r.replicas = keep[r.name]
                     return obj
                 self.store.apply(C.KIND_RBG, mname, mutate, namespace)
         for mname, rbg in owned.items():

@@ -431,3 +431,51 @@ def test_discovery_mode_sticky_and_legacy_per_role(mgr, tmp_path):
         doc = _yaml.safe_load(f)
     assert doc["group"]["roles"][0]["name"] == "worker"
     assert len(doc["group"]["roles"]) == 1
+
+
+def test_rbgset_update_propagates_but_respects_adapter_scaling(mgr):
+    """RBGSet template updates reach existing members, but an
+    adapter-scaled role's replicas are adapter-owned: the set must not
+    revert them on its propagation loop (set-vs-adapter fight)."""
+    from rbg_amd.api.types import (RoleBasedGroupSet, RoleBasedGroupSetSpec,
+                                   ScalingAdapterSpec)
+    from rbg_amd.controller.scalingadapter import scale_adapter
+    template = router_worker_rbg(worker_replicas=1).spec
+    template.role("worker").scaling_adapter = ScalingAdapterSpec(enable=True)
+    rbgset = RoleBasedGroupSet(
+        metadata=ObjectMeta(name="fl2"),
+        spec=RoleBasedGroupSetSpec(replicas=2, template=template))
+    mgr.store.create(rbgset)
+    assert mgr.wait_for(
+        lambda: rbg_ready(mgr, "fl2-0") and rbg_ready(mgr, "fl2-1"),
+        timeout=60)
+    # the members auto-provision adapters; scale member 0's worker pool
+    assert mgr.wait_for(
+        lambda: mgr.store.try_get(C.KIND_SCALING_ADAPTER,
+                                  "fl2-0-worker") is not None, timeout=30)
+    scale_adapter(mgr.store, "fl2-0-worker", 2)
+
+    def member0_scaled():
+        insts = mgr.store.list(C.KIND_ROLE_INSTANCE,
+                               selector={C.LABEL_GROUP_NAME: "fl2-0",
+                                         C.LABEL_ROLE_NAME: "worker"})
+        return len(insts) == 2
+    assert mgr.wait_for(member0_scaled, timeout=60)
+
+    # a template update (new arg) propagates to both members...
+    def bump(cur):
+        cur.spec.template.role("worker").template.engines[0].args["rev"] = "v2"
+        return cur
+    mgr.store.apply(C.KIND_RBG_SET, "fl2", bump)
+
+    def propagated():
+        return all(
+            mgr.store.get(C.KIND_RBG, m).spec.role("worker")
+            .template.engines[0].args.get("rev") == "v2"
+            for m in ("fl2-0", "fl2-1"))
+    assert mgr.wait_for(propagated, timeout=30)
+
+    # ...while member 0 keeps its adapter-driven scale (several resyncs)
+    time.sleep(1.0)
+    assert member0_scaled(), "set propagation reverted adapter scaling"
+    assert mgr.store.get(C.KIND_RBG, "fl2-0").spec.role("worker").replicas == 2
