@@ -16,6 +16,9 @@ import threading
 from typing import Any, Callable, Dict, Optional
 
 _LEN = struct.Struct("!I")
+# control messages are small (KV bytes never cross these sockets); a frame
+# beyond this is a corrupt/hostile peer, not a real request
+MAX_FRAME = 1 << 28     # 256 MB
 
 
 def _send_msg(sock: socket.socket, obj: Any) -> None:
@@ -26,7 +29,13 @@ def _send_msg(sock: socket.socket, obj: Any) -> None:
 def _recv_msg(sock: socket.socket) -> Any:
     hdr = _recv_exact(sock, _LEN.size)
     (n,) = _LEN.unpack(hdr)
-    return json.loads(_recv_exact(sock, n).decode())
+    if n > MAX_FRAME:
+        raise ConnectionError(f"oversized rpc frame ({n} bytes)")
+    try:
+        return json.loads(_recv_exact(sock, n).decode())
+    except (ValueError, UnicodeDecodeError) as e:
+        # corrupt frame: the stream is unrecoverable — drop the connection
+        raise ConnectionError(f"malformed rpc frame: {e}") from e
 
 
 def _recv_exact(sock: socket.socket, n: int) -> bytes:

@@ -255,3 +255,26 @@ def test_update_with_retry_survives_concurrent_bumps():
     with pytest.raises(RuntimeError, match="socket closed"):
         update_with_retry(BrokenClient(), C.KIND_RBG, "racy", "default",
                            mutate)
+
+
+def test_rpc_server_survives_malformed_frames(daemon):
+    """A corrupt or oversized frame drops THAT connection; the server keeps
+    serving well-formed clients (control-plane robustness)."""
+    import socket
+    import struct
+
+    port = daemon.rpc.port if hasattr(daemon, "rpc") else daemon.port
+    # oversized length prefix
+    s = socket.create_connection(("127.0.0.1", daemon.port), timeout=5)
+    s.sendall(struct.pack("!I", (1 << 31)))
+    s.close()
+    # garbage that is not JSON
+    s = socket.create_connection(("127.0.0.1", daemon.port), timeout=5)
+    s.sendall(struct.pack("!I", 4) + b"\xff\xfe\x00\x01")
+    s.close()
+    # a real client still works
+    c = RemoteClient("127.0.0.1", daemon.port)
+    try:
+        assert c.list_raw("RoleBasedGroup", "default", None) == []
+    finally:
+        c.rpc.close()
