@@ -166,6 +166,7 @@ class RoleInstanceController:
                         inst.metadata.name, failed_fatal, diag)
             self._record_bindings(inst, rt)
             self._stop_all(inst, rt)
+            self._bounce_comm_siblings(inst)
             tracker.record_restart(now)
             inst.status.restart_count = tracker.restart_count
             inst.status.last_restart_time = now
@@ -404,6 +405,44 @@ class RoleInstanceController:
                 suffix = wname[len(inst.metadata.name) + 1:]
                 self.bindings.record(rbg_uid, GpuBindingStore.key(
                     inst.metadata.name, suffix), gpus)
+
+    def _bounce_comm_siblings(self, inst: RoleInstance) -> None:
+        """Linked failover across a communicator world (abort
+        comm -> gang-recreate the engine GROUP): when this instance is
+        gang-recreated, every other instance sharing its RBG-wide
+        collective world still holds the dead world — its ranks would
+        block the new rendezvous forever.  Stop their workers too (handles
+        cleared -> they respawn as 'missing', with NO restart count and no
+        recursive bounce) so the whole world re-forms together.  Peer-KV
+        P/D roles carry no world and are never bounced."""
+        import json as _json
+        raw = inst.metadata.annotations.get("rbg.comm-members", "")
+        if not raw:
+            return
+        try:
+            members = _json.loads(raw)
+        except ValueError:
+            return
+        if not isinstance(members, dict) or len(members) <= 1:
+            return
+        for sib_name in members:
+            if sib_name == inst.metadata.name:
+                continue
+            sib = self.store.try_get(C.KIND_ROLE_INSTANCE, sib_name,
+                                     inst.metadata.namespace)
+            if sib is None:
+                continue
+            rt = self._runtimes.get(sib.metadata.uid)
+            if rt is None or not rt.handles:
+                continue
+            log.warning("linked failover: bouncing %s (shares comm world "
+                        "with restarted %s)", sib_name, inst.metadata.name)
+            self.recorder.normal(
+                sib, "LinkedRestart",
+                f"comm world member {inst.metadata.name} restarted; "
+                "rebuilding the collective world")
+            self._record_bindings(sib, rt)
+            self._stop_ordered(sib, rt)   # clears handles: respawn follows
 
     def _stop_ordered(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
         """Stop workers in reverse dependency order (reference

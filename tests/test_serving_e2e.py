@@ -311,3 +311,76 @@ def test_tp2_pp2_leader_worker_serving(mgr):
     assert len(res["tokens"]) == 5
     assert res["tokens"] == _local_reference_tokens(prompt, 5,
                                                     model="tiny-tp")
+
+
+@pytest.mark.timeout(600)
+def test_tp_world_linked_failover(mgr):
+    """Kill ONE rank of the prefill TP pair: every instance sharing the
+    collective world must bounce (survivors hold a dead world and would
+    block the new rendezvous), the world re-forms, and serving resumes
+    with bit-exact output — the reference's linked failover (abort comm →
+    recreate group) on the gloo world."""
+    import os
+    import signal
+    from rbg_amd.api.types import LeaderWorkerPattern
+    args = dict(ENGINE_ARGS, model="tiny-tp", cpu_model="tiny-tp",
+                comm_backend="gloo")
+
+    def tp_role(name, mode):
+        return RoleSpec(
+            name=name, replicas=1, dependencies=["router"],
+            pattern=C.PATTERN_LEADER_WORKER,
+            leader_worker_pattern=LeaderWorkerPattern(size=2),
+            template=EngineTemplate(engines=[EngineSpec(
+                name="engine", runner="llm-engine",
+                args=dict(args, mode=mode),
+                resources=EngineResources(cpu_only=True))]))
+
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="tplf"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("pd", {"prefill_roles": ["prefill"],
+                               "decode_roles": ["decode"],
+                               "vocab_size": 500}),
+            tp_role("prefill", "prefill"),
+            tp_role("decode", "decode"),
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "tplf"), timeout=240)
+
+    def pids_by_instance():
+        out = {}
+        for inst in mgr.store.list(C.KIND_ROLE_INSTANCE,
+                                   selector={C.LABEL_GROUP_NAME: "tplf"}):
+            if "router" in inst.metadata.name:
+                continue
+            out[inst.metadata.name] = sorted(
+                w.pid for w in inst.status.workers if w.pid)
+        return out
+
+    before = pids_by_instance()
+    assert len(before) == 2 and all(len(v) == 2 for v in before.values())
+    victim = before["tplf-prefill-0"][0]
+    os.kill(victim, signal.SIGKILL)
+
+    # linked recovery: BOTH instances' worlds bounce -> all four pids new
+    def fully_bounced():
+        cur = pids_by_instance()
+        if set(cur) != set(before):
+            return False
+        old = {p for v in before.values() for p in v}
+        new = {p for v in cur.values() for p in v}
+        return len(new) == 4 and not (old & new) and rbg_ready(mgr, "tplf")
+    assert mgr.wait_for(fully_bounced, timeout=240), pids_by_instance()
+
+    # the re-formed world serves, and output is still bit-exact
+    assert mgr.wait_for(
+        lambda: (_router_http_port(mgr, "tplf") is not None), timeout=30)
+    port = _router_http_port(mgr, "tplf")
+    torch.manual_seed(77)
+    prompt = torch.randint(0, 500, (17,)).tolist()
+    res = _http_post(port, "/generate",
+                     {"prompt_tokens": prompt, "max_new_tokens": 5},
+                     timeout=240)
+    assert res["tokens"] == _local_reference_tokens(prompt, 5,
+                                                    model="tiny-tp")
