@@ -384,3 +384,71 @@ def test_tp_world_linked_failover(mgr):
                      timeout=240)
     assert res["tokens"] == _local_reference_tokens(prompt, 5,
                                                     model="tiny-tp")
+
+
+@pytest.mark.timeout(600)
+def test_tp_request_survives_world_bounce(mgr):
+    """A request streaming from a TP decode group when one rank is
+    SIGKILLed: the router detects the loss, waits out the linked world
+    re-formation, re-dispatches, and the client still gets the full
+    bit-exact answer (failover_window_s bounds the wait)."""
+    import os
+    import signal
+    import threading
+    from rbg_amd.api.types import LeaderWorkerPattern
+    args = dict(ENGINE_ARGS, model="tiny-tp", cpu_model="tiny-tp",
+                comm_backend="gloo")
+
+    def tp_role(name, mode):
+        return RoleSpec(
+            name=name, replicas=1, dependencies=["router"],
+            pattern=C.PATTERN_LEADER_WORKER,
+            leader_worker_pattern=LeaderWorkerPattern(size=2),
+            template=EngineTemplate(engines=[EngineSpec(
+                name="engine", runner="llm-engine",
+                args=dict(args, mode=mode),
+                resources=EngineResources(cpu_only=True))]))
+
+    rbg = RoleBasedGroup(
+        metadata=ObjectMeta(name="tpcx"),
+        spec=RoleBasedGroupSpec(roles=[
+            router_role("pd", {"prefill_roles": ["prefill"],
+                               "decode_roles": ["decode"],
+                               "vocab_size": 500,
+                               "failover_window_s": 240}),
+            tp_role("prefill", "prefill"),
+            tp_role("decode", "decode"),
+        ]))
+    mgr.store.create(rbg)
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "tpcx"), timeout=240)
+    assert mgr.wait_for(
+        lambda: (_router_http_port(mgr, "tpcx") is not None), timeout=30)
+    port = _router_http_port(mgr, "tpcx")
+
+    torch.manual_seed(99)
+    prompt = torch.randint(0, 500, (19,)).tolist()
+    box = {}
+
+    def ask():
+        try:
+            box["res"] = _http_post(
+                port, "/generate",
+                {"prompt_tokens": prompt, "max_new_tokens": 48},
+                timeout=300)
+        except Exception as e:  # noqa: BLE001
+            box["err"] = e
+    t = threading.Thread(target=ask, daemon=True)
+    t.start()
+    time.sleep(0.8)          # request in flight (decode streaming)
+
+    decode = next(i for i in mgr.store.list(
+        C.KIND_ROLE_INSTANCE, selector={C.LABEL_GROUP_NAME: "tpcx"})
+        if "decode" in i.metadata.name)
+    victim = next(w.pid for w in decode.status.workers if w.pid)
+    os.kill(victim, signal.SIGKILL)
+
+    t.join(timeout=400)
+    assert not t.is_alive(), "request never completed"
+    assert "res" in box, box.get("err")
+    assert box["res"]["tokens"] == _local_reference_tokens(
+        prompt, 48, model="tiny-tp")
