@@ -566,6 +566,13 @@ class RoleBasedGroupController:
             self.recorder.warning(rbg, "GroupNotReady",
                                   "one or more roles lost readiness")
 
+        # UpdateInProgress (reference rolebasedgroup_types.go:541-553):
+        # true while any role still has replicas off the current revision
+        updating = [r.name for r in rbg.spec.roles
+                    if r.name in statuses and
+                    statuses[r.name].updated_replicas <
+                    statuses[r.name].replicas]
+
         def mutate(cur: RoleBasedGroup):
             cur.status.observed_generation = cur.metadata.generation
             cur.status.role_statuses = [statuses[r.name] for r in cur.spec.roles
@@ -577,6 +584,11 @@ class RoleBasedGroupController:
                           for r, s in ((r, statuses[r.name])
                                        for r in cur.spec.roles
                                        if r.name in statuses))))
+            set_condition(cur.status.conditions, Condition.new(
+                C.COND_UPDATE_IN_PROGRESS, bool(updating),
+                "RollingUpdate" if updating else "UpToDate",
+                f"roles updating: {', '.join(updating)}" if updating
+                else "all replicas on the current revision"))
             return cur
         try:
             self.store.apply(C.KIND_RBG, rbg.metadata.name, mutate,

@@ -479,3 +479,32 @@ def test_rbgset_update_propagates_but_respects_adapter_scaling(mgr):
     time.sleep(1.0)
     assert member0_scaled(), "set propagation reverted adapter scaling"
     assert mgr.store.get(C.KIND_RBG, "fl2-0").spec.role("worker").replicas == 2
+
+
+def test_update_in_progress_condition(mgr):
+    """status condition UpdateInProgress (reference
+    rolebasedgroup_types.go:541-553): true while any role has replicas
+    off the current revision, false once the rollout completes."""
+    from rbg_amd.api.types import get_condition as _gc
+    mgr.store.create(router_worker_rbg(name="uip", worker_replicas=2))
+    assert mgr.wait_for(lambda: rbg_ready(mgr, "uip"), timeout=60)
+
+    def cond():
+        cur = mgr.store.get(C.KIND_RBG, "uip")
+        return _gc(cur.status.conditions, C.COND_UPDATE_IN_PROGRESS)
+    assert mgr.wait_for(
+        lambda: cond() is not None and cond().status == "False", timeout=30)
+
+    # slow-ready new revision keeps the in-progress window observable
+    def bump(cur):
+        role = cur.spec.role("worker")
+        role.update_strategy_type = C.UPDATE_RECREATE
+        role.template.engines[0].args["ready_delay"] = 1.0
+        role.template.engines[0].args["rev"] = "v2"
+        return cur
+    mgr.store.apply(C.KIND_RBG, "uip", bump)
+    assert mgr.wait_for(
+        lambda: cond() is not None and cond().status == "True", timeout=30)
+    assert mgr.wait_for(
+        lambda: cond().status == "False" and rbg_ready(mgr, "uip"),
+        timeout=90)
