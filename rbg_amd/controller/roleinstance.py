@@ -49,6 +49,10 @@ class InstanceRuntime:
     recovery_started: float = 0.0        # failure detected, not Ready yet
     applied_args: Dict[str, str] = field(default_factory=dict)  # in-place
     unschedulable: str = ""              # last gang-reserve failure message
+    # linked failover: set by a comm-world sibling's reconcile; consumed by
+    # THIS instance's own reconcile (per-key serialization avoids touching
+    # another runtime's handles concurrently)
+    bounce_requested: str = ""
 
 
 class RoleInstanceController:
@@ -109,6 +113,24 @@ class RoleInstanceController:
     def _sync(self, inst: RoleInstance, rt: InstanceRuntime) -> float:
         desired = self._desired_workers(inst)
         tracker = self.restarts.for_key(inst.metadata.uid)
+
+        if rt.bounce_requested and rt.handles:
+            # a comm-world sibling was gang-recreated: our ranks hold the
+            # dead world and would block its new rendezvous — stop them so
+            # the whole world re-forms (handles cleared -> respawn below,
+            # no restart count, no recursive bounce)
+            src_name = rt.bounce_requested
+            rt.bounce_requested = ""
+            log.warning("linked failover: bouncing %s (shares comm world "
+                        "with restarted %s)", inst.metadata.name, src_name)
+            self.recorder.normal(
+                inst, "LinkedRestart",
+                f"comm world member {src_name} restarted; rebuilding the "
+                "collective world")
+            self._record_bindings(inst, rt)
+            self._stop_ordered(inst, rt)
+            return 0.05
+        rt.bounce_requested = ""
 
         # 1. observe failures
         failed_fatal = []
@@ -411,10 +433,11 @@ class RoleInstanceController:
         comm -> gang-recreate the engine GROUP): when this instance is
         gang-recreated, every other instance sharing its RBG-wide
         collective world still holds the dead world — its ranks would
-        block the new rendezvous forever.  Stop their workers too (handles
-        cleared -> they respawn as 'missing', with NO restart count and no
-        recursive bounce) so the whole world re-forms together.  Peer-KV
-        P/D roles carry no world and are never bounced."""
+        block the new rendezvous forever.  Flag each sibling runtime; the
+        sibling's own reconcile stops its workers (handles cleared -> they
+        respawn as 'missing', with NO restart count and no recursive
+        bounce) so the whole world re-forms together.  Peer-KV P/D roles
+        carry no world and are never bounced."""
         import json as _json
         raw = inst.metadata.annotations.get("rbg.comm-members", "")
         if not raw:
@@ -435,14 +458,10 @@ class RoleInstanceController:
             rt = self._runtimes.get(sib.metadata.uid)
             if rt is None or not rt.handles:
                 continue
-            log.warning("linked failover: bouncing %s (shares comm world "
-                        "with restarted %s)", sib_name, inst.metadata.name)
-            self.recorder.normal(
-                sib, "LinkedRestart",
-                f"comm world member {inst.metadata.name} restarted; "
-                "rebuilding the collective world")
-            self._record_bindings(sib, rt)
-            self._stop_ordered(sib, rt)   # clears handles: respawn follows
+            # flag only: the sibling stops ITSELF on its next reconcile —
+            # reconciles are serialized per key, so two workers never
+            # mutate the same runtime's handles concurrently
+            rt.bounce_requested = inst.metadata.name
 
     def _stop_ordered(self, inst: RoleInstance, rt: InstanceRuntime) -> None:
         """Stop workers in reverse dependency order (reference
