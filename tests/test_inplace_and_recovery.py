@@ -165,7 +165,7 @@ def test_decode_pool_failover_continuity(tmp_path):
          "--model", "tiny", "--device", "cpu", "--prompts", "16",
          "--rate", "5", "--in-len", "64", "--out-len", "12",
          "--decode-replicas", "2", "--kill-decode-after", "1.5",
-         "--timeout", "120"],
+         "--timeout", "200", "--failover-window", "180"],
         capture_output=True, text=True, timeout=420,
         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert out.returncode == 0, out.stderr[-2000:]

@@ -58,6 +58,7 @@ def build_rbg(args, engine_args):
             role("router", "router",
                  {"dispatch": "pd", "prefill_roles": ["prefill"],
                   "decode_roles": ["decode"],
+                  "failover_window_s": args.failover_window,
                   "vocab_size": args.vocab}, 0),
             role("prefill", "llm-engine",
                  dict(shared, mode="prefill", kv_pool_tokens=prefill_pool),
@@ -96,6 +97,10 @@ def main() -> int:
                          "request stream (linked-failover continuity)")
     ap.add_argument("--hbm-gb", type=int, default=100)
     ap.add_argument("--timeout", type=float, default=900.0)
+    ap.add_argument("--failover-window", type=float, default=45.0,
+                    help="router re-dispatch window (s); raise on "
+                         "oversubscribed CPU test boxes where a collective "
+                         "world bounce restarts the whole pool")
     args = ap.parse_args()
     from rbg_amd.engine.config import ModelConfig
     model_cfg = ModelConfig.preset(args.model)
