@@ -156,7 +156,13 @@ def test_decode_pool_failover_continuity(tmp_path):
     """BASELINE config 5 semantics: with a decode POOL (replicas=2), a
     SIGKILLed decode replica mid-stream loses no requests — the router
     fails the dead instance's sequences fast (InstanceLost) and
-    re-dispatches them to the survivor while the gang recreates."""
+    re-dispatches them while the gang recreates.  On CPU the P/D roles
+    share a collective world (gloo wire), so the recreate bounces the
+    whole world (linked failover) and requests ride the re-dispatch
+    window across the blip.  Calibrated for a non-oversubscribed box
+    (the driver runs the suite serially): under multi-suite CPU
+    starvation the recovery can exceed any finite failover window and
+    requests are then SHED BY DESIGN rather than hung."""
     import json
     import subprocess
     import sys
