@@ -82,6 +82,45 @@ def test_watch_events():
     w.stop()
 
 
+def test_watch_order_under_concurrent_writers():
+    """Events fan out while the store mutation lock is held, so a watcher
+    must see MODIFIED events in strictly increasing resourceVersion order
+    per object, ending on the durable final version — the property the
+    write-behind StorePersister depends on (advisor finding r1: a notify
+    outside the lock let a stale version be the one persisted)."""
+    s = Store()
+    w = s.watch(kinds=(C.KIND_RBG,))
+    s.create(make_rbg())
+    assert w.get(timeout=1.0).type == "ADDED"
+
+    def bump():
+        for _ in range(25):
+            s.apply(C.KIND_RBG, "demo",
+                    lambda cur: (setattr(cur.spec.roles[0], "replicas",
+                                         cur.spec.roles[0].replicas + 1),
+                                 cur)[1])
+
+    threads = [threading.Thread(target=bump) for _ in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+
+    versions = []
+    while True:
+        ev = w.get(timeout=0.5)
+        if ev is None:
+            break
+        assert ev.type == "MODIFIED"
+        versions.append(ev.obj.metadata.resource_version)
+    w.stop()
+    assert len(versions) == 100
+    assert versions == sorted(versions)
+    assert len(set(versions)) == 100
+    # the last event delivered IS the live stored version (persist-safe)
+    assert versions[-1] == s.get(C.KIND_RBG, "demo").metadata.resource_version
+
+
 def test_list_selector_and_owned():
     s = Store()
     parent = s.create(make_rbg("parent"))
