@@ -65,7 +65,21 @@ def cmd_get(client: BaseClient, args) -> int:
         if obj is None:
             print(f"{kind}/{args.name} not found", file=sys.stderr)
             return 1
-        print(yaml.safe_dump(asdict(obj), sort_keys=False))
+        doc = asdict(obj)
+        if getattr(args, "output", "") == "v1alpha1":
+            # serve the deprecated API version on the way out (the
+            # conversion-webhook read path; conversion annotations written
+            # by apply make the round-trip lossless)
+            from ..api import v1alpha1 as _legacy
+            if kind == C.KIND_RBG:
+                doc = _legacy.from_v2(doc)
+            elif kind == C.KIND_RBG_SET:
+                doc = _legacy.set_from_v2(doc)
+            else:
+                print(f"-o v1alpha1 supports {C.KIND_RBG} and "
+                      f"{C.KIND_RBG_SET} only", file=sys.stderr)
+                return 1
+        print(yaml.safe_dump(doc, sort_keys=False))
         return 0
     if kind == C.KIND_EVENT:
         import time as _t
@@ -248,6 +262,8 @@ def build_parser() -> argparse.ArgumentParser:
     p = sub.add_parser("get")
     p.add_argument("kind")
     p.add_argument("name", nargs="?")
+    p.add_argument("-o", "--output", default="",
+                   help="'v1alpha1' emits the deprecated API version")
     p = sub.add_parser("status")
     p.add_argument("name")
     p = sub.add_parser("describe")

@@ -278,3 +278,48 @@ def test_rpc_server_survives_malformed_frames(daemon):
         assert c.list_raw("RoleBasedGroup", "default", None) == []
     finally:
         c.rpc.close()
+
+
+def test_get_output_v1alpha1_round_trip(daemon, tmp_path):
+    """`rbgctl get -o v1alpha1` serves the deprecated API version on the
+    read path (conversion-webhook analog): a legacy doc applied as-is
+    comes back with its workload kind, restartPolicy and rolloutStrategy
+    restored from the conversion annotations."""
+    import yaml as _yaml
+    legacy = textwrap.dedent("""\
+        apiVersion: workloads.x-k8s.io/v1alpha1
+        kind: RoleBasedGroup
+        metadata:
+          name: legacy
+        spec:
+          roles:
+          - name: engine
+            replicas: 1
+            workload: {apiVersion: apps/v1, kind: StatefulSet}
+            restartPolicy: RecreateRoleInstanceOnPodRestart
+            rolloutStrategy:
+              type: RollingUpdate
+              rollingUpdate: {maxUnavailable: 1, maxSurge: 0}
+            template:
+              engines:
+              - {name: engine, runner: echo, resources: {cpuOnly: true}}
+    """)
+    f = tmp_path / "legacy.yaml"
+    f.write_text(legacy)
+    rc, out = run_cli(daemon, "apply", "-f", str(f))
+    assert rc == 0 and "created" in out
+    rc, out = run_cli(daemon, "get", "rbg", "legacy", "-o", "v1alpha1")
+    assert rc == 0
+    doc = _yaml.safe_load(out)
+    assert doc["apiVersion"] == "workloads.x-k8s.io/v1alpha1"
+    (role,) = doc["spec"]["roles"]
+    assert role["workload"]["kind"] == "StatefulSet"
+    assert role["restartPolicy"] == "RecreateRoleInstanceOnPodRestart"
+    assert role["rolloutStrategy"]["rollingUpdate"]["maxUnavailable"] == 1
+    # the storage version remains v1alpha2
+    rc, out = run_cli(daemon, "get", "rbg", "legacy")
+    assert rc == 0
+    assert _yaml.safe_load(out)["apiVersion"].endswith("v1alpha2")
+    # unsupported kinds fail cleanly rather than emitting v2 YAML
+    rc, _ = run_cli(daemon, "get", "roleinstance", "nope", "-o", "v1alpha1")
+    assert rc == 1
