@@ -25,7 +25,8 @@ def update_with_retry(client: "BaseClient", kind: str, name: str,
     bare update over the daemon RPC can hit a Conflict — re-read and
     re-apply, like kubectl / client-go's RetryOnConflict."""
     last = None
-    for _ in range(8):
+    delay = 0.02
+    for _ in range(12):
         cur = client.get(kind, name, namespace)
         if cur is None:
             raise KeyError(f"{kind} {name} vanished during update")
@@ -38,7 +39,11 @@ def update_with_retry(client: "BaseClient", kind: str, name: str,
             if "Conflict" not in str(e):
                 raise
             last = e
-        time.sleep(0.05)
+        # exponential backoff (client-go RetryOnConflict shape): a busy
+        # controller bumps status every resync tick, so a fixed short
+        # sleep can lose the race every attempt on a loaded box
+        time.sleep(delay)
+        delay = min(delay * 1.6, 0.5)
     raise RuntimeError(f"update of {kind}/{name} kept conflicting: {last}")
 
 
